@@ -1,0 +1,340 @@
+"""TEST INFRASTRUCTURE ONLY — numpy restatement of flox's grouped-reduce semantics.
+
+Restates (does not copy) the behaviour of the reference ``groupby_reduce`` eager
+path: reference core.py:739 (validation / min_count defaulting core.py:1026-1038)
+-> factorize (factorize.py:42-108, 147-213) -> chunk_reduce (core.py:214-394)
+-> finalize (core.py:410-475), with the aggregation recipes of
+aggregations.py:304-546 and the dtype/fill rules of xrdtypes.py:153-209.
+
+Numerics contract: float32 inputs are accumulated in float64 (the
+``engine="numpy"``/numpy_groupies behaviour pinned by reference
+tests/test_properties.py:146-151), then cast to the final dtype.
+
+This module is the checker for parity tests; it must never be imported by the
+product package ``flox_amd``.
+"""
+
+from __future__ import annotations
+
+import math
+
+import numpy as np
+
+ALL_FUNCS = (
+    "count",
+    "sum",
+    "nansum",
+    "prod",
+    "nanprod",
+    "mean",
+    "nanmean",
+    "var",
+    "nanvar",
+    "std",
+    "nanstd",
+    "min",
+    "nanmin",
+    "max",
+    "nanmax",
+)
+
+_FLOAT_FUNCS = {"mean", "nanmean", "var", "nanvar", "std", "nanstd"}
+_NAN_SKIP = {"nansum", "nanprod", "nanmean", "nanvar", "nanstd", "nanmin", "nanmax", "count"}
+# funcs whose output dtype equals the input dtype (reference: preserves_dtype=True,
+# aggregations.py:529-546)
+_PRESERVES_DTYPE = {"min", "nanmin", "max", "nanmax"}
+
+
+def _isnull(a: np.ndarray) -> np.ndarray:
+    # reference xrutils.isnull, restricted to numeric dtypes
+    if a.dtype.kind in "fc":
+        return np.isnan(a)
+    return np.zeros(a.shape, dtype=bool)
+
+
+def _factorize_single(flat: np.ndarray, expect) -> tuple[np.ndarray, np.ndarray]:
+    """labels -> dense codes in [0, ngroups); -1 for NaN / not-in-expected.
+
+    Restates reference factorize.py:42-99 (_factorize_single): RangeIndex-like
+    fast path (44-53), searchsorted-vs-expected (84-94), unique/factorize (96).
+    Returns (codes int64, found_group_values).
+    """
+    if expect is None:
+        # hash/factorize path (reference line 96: pd.factorize(flat, sort=True))
+        nanmask = _isnull(flat)
+        uniq = np.unique(flat[~nanmask])
+        codes = np.searchsorted(uniq, flat)
+        # values not present can't occur (uniq built from flat); mask NaNs
+        codes = codes.astype(np.int64)
+        codes[nanmask] = -1
+        return codes, uniq
+
+    expect = np.asarray(expect)
+    n = len(expect)
+    if expect.dtype.kind in "iu" and n and expect[0] == 0 and expect[-1] == n - 1 and np.array_equal(expect, np.arange(n)):
+        # RangeIndex path (reference factorize.py:44-53): codes are the labels,
+        # anything above the last expected label -> -1
+        codes = flat.astype(np.int64, copy=True)
+        codes[codes > n - 1] = -1
+        codes[codes < -1] = -1  # out-of-domain negatives are treated as missing
+        if flat.dtype.kind in "fc":
+            codes[_isnull(flat)] = -1
+        return codes, expect
+    # searchsorted path (reference factorize.py:84-94), sort=True so expect sorted
+    sorter = np.argsort(expect)
+    sorted_expect = expect[sorter]
+    idx = np.searchsorted(sorted_expect, flat)
+    idx = idx.astype(np.int64)
+    oob = idx == n
+    idx_clipped = np.where(oob, 0, idx)
+    mask = oob | (sorted_expect[idx_clipped] != flat) | _isnull(flat)
+    idx[mask] = -1
+    return idx, sorted_expect
+
+
+def _ravel_codes(codes_list, grp_shape) -> np.ndarray:
+    """Combine multiple factorized label arrays into one code
+    (reference factorize.py:102-108)."""
+    valid = np.ones(codes_list[0].shape, dtype=bool)
+    for c in codes_list:
+        valid &= c >= 0
+    clipped = [np.where(c < 0, 0, c) for c in codes_list]
+    group_idx = np.ravel_multi_index(clipped, grp_shape, mode="wrap").astype(np.int64)
+    group_idx[~valid] = -1
+    return group_idx
+
+
+def _final_dtype(func: str, array_dtype: np.dtype, dtype=None) -> np.dtype:
+    """Output dtype rules: xrdtypes.py:153-186 (_normalize_dtype/_maybe_promote_int)."""
+    if dtype is not None:
+        return np.dtype(dtype)
+    if func == "count":
+        return np.dtype(np.intp)
+    if func in _FLOAT_FUNCS:
+        # "mean, std, var always result in floating, preserving a floating input
+        # dtype" (xrdtypes.py:161-167)
+        if array_dtype.kind in "fc":
+            return array_dtype
+        return np.dtype("float64")
+    if func in _PRESERVES_DTYPE:
+        return array_dtype
+    # sum/prod: promote small ints to platform int (xrdtypes.py:175-185)
+    if array_dtype.kind == "i":
+        return np.result_type(array_dtype, np.int_)
+    if array_dtype.kind == "u":
+        return np.result_type(array_dtype, np.uint)
+    return array_dtype
+
+
+def _fill_default(func: str, out_dtype: np.dtype):
+    """final_fill_value per aggregation (aggregations.py:304-546):
+    count->0, prod->1, others -> dtype-NA (NaN for floats, iinfo extremes for
+    ints per xrdtypes.py:188-209)."""
+    if func == "count":
+        return 0
+    if func in ("prod",):
+        return 1
+    if out_dtype.kind in "fc":
+        return np.nan
+    # NA for integer output: get_neg_infinity(min_for_int=True) (xrdtypes.py:205-206)
+    return np.iinfo(out_dtype).min
+
+
+def groupby_reduce(
+    array,
+    *by,
+    func: str,
+    expected_groups=None,
+    axis=None,
+    fill_value=None,
+    dtype=None,
+    min_count=None,
+    finalize_kwargs=None,
+):
+    """Eager grouped reduction with flox semantics. Returns (result, *groups).
+
+    Supports: by aligned with the trailing ``by[0].ndim`` dims of ``array``,
+    reduction over all dims of by (axis=None or the full trailing tuple).
+    """
+    array = np.asarray(array)
+    bys = tuple(np.asarray(b) for b in by)
+    nby = len(bys)
+    if nby == 0:
+        raise ValueError("need at least one by")
+    if func not in ALL_FUNCS:
+        raise NotImplementedError(func)
+    by_ndim = bys[0].ndim
+    for b in bys:
+        assert b.shape == bys[0].shape
+    assert array.shape[array.ndim - by_ndim :] == bys[0].shape, (array.shape, bys[0].shape)
+    if axis is not None:
+        ax = axis if isinstance(axis, tuple) else (axis,)
+        ax = tuple(a % array.ndim for a in ax)
+        assert ax == tuple(range(array.ndim - by_ndim, array.ndim)), "oracle reduces over all by dims"
+
+    if expected_groups is not None and not isinstance(expected_groups, tuple):
+        expected_groups = (expected_groups,)
+    if expected_groups is None:
+        expected_groups = (None,) * nby
+    provided_expected = any(e is not None for e in expected_groups)
+
+    if array.dtype.kind == "b":
+        array = array.astype(np.int_)
+
+    # --- factorize (early, like reference core.py:943-949) ---
+    codes_list, found = [], []
+    for b, e in zip(bys, expected_groups):
+        c, f = _factorize_single(b.reshape(-1), e)
+        codes_list.append(c)
+        found.append(f)
+    grp_shape = tuple(len(f) for f in found)
+    ngroups = math.prod(grp_shape)
+    codes = _ravel_codes(codes_list, grp_shape) if nby > 1 else codes_list[0]
+
+    # --- min_count defaulting (reference core.py:1026-1038) ---
+    if min_count is None:
+        min_count_ = 1 if (fill_value is not None and provided_expected) else 0
+    else:
+        min_count_ = min_count
+    if func in ("nanmin", "nanmax") and min_count_ == 0:
+        # reference aggregations.py:997-1003 nanmin/nanmax hack
+        min_count_ = 1
+        if fill_value is None:
+            fill_value = np.nan if array.dtype.kind in "fc" else None
+    if min_count_ > 0 and func in ("nansum", "nanprod") and fill_value is None:
+        fill_value = np.nan  # reference core.py:1035-1038
+
+    out_dtype = _final_dtype(func, array.dtype, dtype)
+
+    # --- chunk reduction over flattened group dims ---
+    lead_shape = array.shape[: array.ndim - by_ndim]
+    vals = array.reshape(lead_shape + (-1,))
+    M = int(np.prod(lead_shape, dtype=np.int64)) if lead_shape else 1
+    vals2d = vals.reshape(M, -1)
+
+    nanmask_v = _isnull(vals2d)
+    valid_code = codes >= 0
+
+    acc_dtype = np.float64 if array.dtype.kind in "fc" else np.int64
+    if array.dtype == np.complex64 or array.dtype == np.complex128:
+        acc_dtype = np.complex128
+
+    def bincount_rows(weights2d, mask2d):
+        """per-row np.bincount with f64 accumulation (npg semantics)."""
+        out = np.zeros((M, ngroups), dtype=acc_dtype)
+        for r in range(M):
+            m = mask2d[r] if mask2d is not None else valid_code
+            if weights2d is None:
+                out[r] = np.bincount(codes[m], minlength=ngroups).astype(acc_dtype)
+            else:
+                out[r] = np.bincount(codes[m], weights=weights2d[r][m].astype(acc_dtype), minlength=ngroups)
+        return out
+
+    skipna = func in _NAN_SKIP
+    m_all = valid_code[None, :] & (~nanmask_v if skipna else np.ones_like(nanmask_v))
+    counts = bincount_rows(None, valid_code[None, :] & ~nanmask_v)  # nanlen, always NaN-skipping
+
+    def grouped_extreme(op_at, init):
+        out = np.full((M, ngroups), init, dtype=array.dtype if array.dtype.kind != "b" else np.int_)
+        seen = np.zeros((M, ngroups), dtype=bool)
+        for r in range(M):
+            m = m_all[r]
+            op_at(out[r], codes[m], vals2d[r][m])
+            np.logical_or.at(seen[r], codes[m], True)
+        return out, seen
+
+    result = None
+    if func == "count":
+        result = counts.astype(out_dtype)
+        empty_mask = counts == 0
+    elif func in ("sum", "nansum"):
+        sums = bincount_rows(vals2d, m_all)
+        result = sums.astype(out_dtype)
+        empty_mask = bincount_rows(None, m_all) == 0
+    elif func in ("prod", "nanprod"):
+        out = np.ones((M, ngroups), dtype=acc_dtype)
+        for r in range(M):
+            m = m_all[r]
+            np.multiply.at(out[r], codes[m], vals2d[r][m].astype(acc_dtype))
+        result = out.astype(out_dtype)
+        empty_mask = bincount_rows(None, m_all) == 0
+    elif func in ("mean", "nanmean"):
+        sums = bincount_rows(vals2d, None if skipna else valid_code[None, :].repeat(M, 0))
+        if not skipna:
+            # non-skip mean: sum propagates NaN; divide by nanlen
+            # (reference aggregate_flox.py:251-257: mean = sum / nanlen)
+            sums = np.zeros((M, ngroups), dtype=acc_dtype)
+            for r in range(M):
+                m = valid_code
+                np.add.at(sums[r], codes[m], vals2d[r][m].astype(acc_dtype))
+        else:
+            sums = bincount_rows(vals2d, m_all)
+        with np.errstate(invalid="ignore", divide="ignore"):
+            result = (sums / counts).astype(out_dtype)
+        empty_mask = counts == 0
+    elif func in ("var", "nanvar", "std", "nanstd"):
+        # restates var_chunk (aggregations.py:348-389): len, sum, then sum of
+        # squared deviations about the per-group mean, all f64-accumulated
+        sums = bincount_rows(vals2d, m_all)
+        with np.errstate(invalid="ignore", divide="ignore"):
+            means = sums / counts
+        ssd_dtype = np.complex128 if acc_dtype == np.complex128 else np.float64
+        ssd = np.zeros((M, ngroups), dtype=ssd_dtype)
+        for r in range(M):
+            m = m_all[r] if skipna else valid_code
+            dev = vals2d[r][m].astype(acc_dtype) - means[r][codes[m]]
+            np.add.at(ssd[r], codes[m], dev * dev)
+        ddof = (finalize_kwargs or {}).get("ddof", 0)
+        den = counts - ddof
+        with np.errstate(invalid="ignore", divide="ignore"):
+            v = ssd / den
+        v[den < 0] = np.nan
+        v[counts == 0] = np.nan
+        if func in ("std", "nanstd"):
+            v = np.sqrt(v)
+        result = v.astype(out_dtype)
+        empty_mask = counts == 0
+    elif func in ("min", "nanmin"):
+        if array.dtype.kind in "fc":
+            init = np.inf
+        else:
+            init = np.iinfo(array.dtype).max
+        out, seen = grouped_extreme(np.minimum.at, init)
+        result = out.astype(out_dtype)
+        empty_mask = ~seen
+    elif func in ("max", "nanmax"):
+        if array.dtype.kind in "fc":
+            init = -np.inf
+        else:
+            init = np.iinfo(array.dtype).min
+        out, seen = grouped_extreme(np.maximum.at, init)
+        result = out.astype(out_dtype)
+        empty_mask = ~seen
+    else:  # pragma: no cover
+        raise NotImplementedError(func)
+
+    # nanmin/nanmax of an all-NaN group is NaN (reference aggregate_flox.py:195-207)
+    # handled below via min_count masking (counts==0 -> fill), with fill=NaN default.
+
+    # --- fill for empty groups / min_count mask (reference core.py:437-459) ---
+    user_fill = fill_value
+    if min_count_ > 0:
+        mask = counts < min_count_
+        fv = user_fill
+        if fv is None and mask.any():
+            raise ValueError("Filling is required but fill_value is None.")
+        if mask.any():
+            if np.asarray(fv).dtype.kind in "fc" and out_dtype.kind not in "fc":
+                result = result.astype(np.result_type(out_dtype, np.asarray(fv).dtype))
+            result = np.where(mask, fv, result)
+    else:
+        fv = user_fill if user_fill is not None else _fill_default(func, out_dtype)
+        if empty_mask.any():
+            result = np.where(empty_mask, fv, result)
+
+    result = np.asarray(result)
+    if result.dtype != out_dtype and (user_fill is None or np.asarray(user_fill).dtype.kind not in "fc" or out_dtype.kind in "fc"):
+        result = result.astype(out_dtype, copy=False)
+
+    result = result.reshape(lead_shape + grp_shape)
+    return (result, *found)

@@ -1,0 +1,215 @@
+"""Generate golden parity fixtures from the reference implementation.
+
+Runs ONLY in the build container, where the reference (xarray-contrib/flox)
+is mounted read-only at /root/reference. The reference cannot be imported
+as-is there (Python 3.10 vs 3.11 syntax; missing numpy_groupies/toolz), so
+this script loads it with three mechanical in-memory shims (verified
+equivalent in SURVEY.md §8c) and module stubs — nothing from the reference is
+ever written into this repository; only the resulting input/output vectors
+are committed, as tests/golden/golden_cases.npz.
+
+Usage:  python tests/golden/generate.py
+"""
+
+from __future__ import annotations
+
+import os
+import re
+import sys
+import tempfile
+import types
+
+import numpy as np
+
+REF = "/root/reference/flox"
+OUT = os.path.join(os.path.dirname(__file__), "golden_cases.npz")
+
+
+def load_reference():
+    """Import the reference flox package with mechanical Python-3.10 shims."""
+    tmp = tempfile.mkdtemp(prefix="floxref_")
+    pkg = os.path.join(tmp, "floxref")
+    os.makedirs(pkg)
+    for fname in os.listdir(REF):
+        if not fname.endswith(".py"):
+            continue
+        with open(os.path.join(REF, fname)) as f:
+            src = f.read()
+        # shim 1: PEP-646 star-subscript (aggregations.py:400,408) -> tuple()
+        src = src.replace("array[*not_last]", "array[tuple(not_last)]")
+        src = src.replace("array[*not_first]", "array[tuple(not_first)]")
+        # shim 2: starred return annotation (core.py:754)
+        src = re.sub(
+            r"-> tuple\[DaskArray, \*tuple\[np\.ndarray \| DaskArray, \.\.\.\]\]:", ":", src
+        )
+        # shim 3: typing.Self (multiarray.py:2)
+        src = src.replace("from typing import Self", "from typing_extensions import Self")
+        # intra-package imports: flox.X -> floxref.X
+        src = re.sub(r"\bfrom flox(\.|\b)", r"from floxref\1", src)
+        src = re.sub(r"\bimport flox\b", "import floxref", src)
+        with open(os.path.join(pkg, fname), "w") as f:
+            f.write(src)
+
+    # stub the absent hard deps; neither is touched by the eager engine="flox" path
+    npg_stub = types.ModuleType("numpy_groupies")
+
+    def _unavailable(*a, **k):  # pragma: no cover
+        raise RuntimeError("numpy_groupies stub: not available in this container")
+
+    npg_stub.aggregate_numpy = types.SimpleNamespace(aggregate=_unavailable)
+    npg_stub.aggregate_numba = types.SimpleNamespace(aggregate=_unavailable)
+    npg_stub.aggregate = _unavailable
+    sys.modules.setdefault("numpy_groupies", npg_stub)
+
+    toolz_stub = types.ModuleType("toolz")
+    toolz_stub.partition_all = _unavailable
+    toolz_stub.unique = _unavailable
+    toolz_stub.memoize = lambda f=None, **k: (f if f is not None else (lambda g: g))
+    sys.modules.setdefault("toolz", toolz_stub)
+
+    sys.path.insert(0, tmp)
+    import floxref  # noqa: F401
+    import floxref.core as core
+
+    return core
+
+
+def gen_cases():
+    """Yield (name, kwargs-dict) cases. Mirrors the shapes of the reference's
+    own golden tests (test_core.py:126-206) plus seeded random sweeps."""
+    rng = np.random.default_rng(42)
+    labels_basic = np.array([0, 0, 2, 2, 2, 1, 1, 2, 2, 1, 1, 0])
+    nan_labels = labels_basic.astype(float).copy()
+    nan_labels[[1, 4, 5]] = np.nan
+
+    funcs = [
+        "count", "sum", "nansum", "prod", "nanprod", "mean", "nanmean",
+        "var", "nanvar", "std", "nanstd", "min", "nanmin", "max", "nanmax",
+    ]
+
+    ones = np.ones((12,))
+    vals_f8 = rng.standard_normal(12) * 10
+    vals_f4 = vals_f8.astype(np.float32)
+    vals_nan = vals_f8.copy()
+    vals_nan[[2, 5, 11]] = np.nan
+    vals_int = rng.integers(-50, 50, 12).astype(np.int64)
+    vals_i32 = vals_int.astype(np.int32)
+
+    for func in funcs:
+        yield f"{func}_ones_basic", dict(array=ones, by=labels_basic, func=func)
+        yield f"{func}_f64_basic", dict(array=vals_f8, by=labels_basic, func=func)
+        yield f"{func}_f32_basic", dict(array=vals_f4, by=labels_basic.copy(), func=func)
+        yield f"{func}_f64_nanvals", dict(array=vals_nan, by=labels_basic, func=func)
+        yield f"{func}_i64_basic", dict(array=vals_int, by=labels_basic, func=func)
+        yield f"{func}_i32_basic", dict(array=vals_i32, by=labels_basic, func=func)
+        # NaN in by -> those rows are dropped (factorize.py:201-210)
+        yield f"{func}_nanby", dict(array=vals_f8, by=nan_labels, func=func)
+        # expected_groups superset with a missing group + fill_value
+        yield (
+            f"{func}_expected_fill",
+            dict(
+                array=vals_f8,
+                by=labels_basic,
+                func=func,
+                expected_groups=np.array([0, 1, 2, 5]),
+                fill_value=123.0,
+            ),
+        )
+        # expected_groups as a range, labels uniform
+        big_by = rng.integers(0, 17, 200)
+        big_vals = rng.standard_normal(200)
+        yield (
+            f"{func}_range17",
+            dict(array=big_vals, by=big_by, func=func, expected_groups=np.arange(17)),
+        )
+        # 2-D array, 1-D by (reduce over trailing axis)
+        arr2 = rng.standard_normal((3, 40))
+        by2 = rng.integers(0, 5, 40)
+        yield f"{func}_2d_lead", dict(array=arr2, by=by2, func=func, expected_groups=np.arange(5))
+        # ddof for var/std
+        if "var" in func or "std" in func:
+            yield (
+                f"{func}_ddof1",
+                dict(array=vals_f8, by=labels_basic, func=func, finalize_kwargs={"ddof": 1}),
+            )
+
+    # multi-by (2-D groupby, like BASELINE config 5)
+    by_a = rng.integers(0, 4, 300)
+    by_b = rng.integers(0, 6, 300)
+    vals = rng.standard_normal(300)
+    vals[::17] = np.nan
+    for func in ["nanmean", "sum", "count", "nanmax", "nanvar"]:
+        yield (
+            f"{func}_multi_by",
+            dict(
+                array=vals,
+                by=(by_a, by_b),
+                func=func,
+                expected_groups=(np.arange(4), np.arange(6)),
+            ),
+        )
+    # empty groups at the tail of the range
+    yield "mean_sparse_groups", dict(
+        array=rng.standard_normal(50),
+        by=rng.integers(0, 3, 50),
+        func="mean",
+        expected_groups=np.arange(10),
+    )
+    yield "sum_sparse_groups_fill", dict(
+        array=rng.standard_normal(50),
+        by=rng.integers(0, 3, 50),
+        func="sum",
+        expected_groups=np.arange(10),
+        fill_value=-1.0,
+    )
+
+
+def main():
+    core = load_reference()
+    out = {}
+    n_done, n_skip = 0, 0
+    for name, kw in gen_cases():
+        by = kw.pop("by")
+        bys = by if isinstance(by, tuple) else (by,)
+        try:
+            result, *groups = core.groupby_reduce(
+                kw.pop("array1") if "array1" in kw else kw.pop("array"),
+                *bys,
+                engine="flox",
+                **kw,
+            )
+        except Exception as e:  # pragma: no cover
+            print(f"SKIP {name}: {type(e).__name__}: {e}")
+            n_skip += 1
+            continue
+        # store inputs + outputs
+        case = dict(kw)
+        arr = None
+        out[f"{name}::result"] = np.asarray(result)
+        for i, g in enumerate(groups):
+            out[f"{name}::groups{i}"] = np.asarray(g)
+        n_done += 1
+    # re-run to also store the inputs (gen_cases is deterministic)
+    for name, kw in gen_cases():
+        if f"{name}::result" not in out:
+            continue
+        by = kw.pop("by")
+        bys = by if isinstance(by, tuple) else (by,)
+        out[f"{name}::array"] = np.asarray(kw.pop("array"))
+        for i, b in enumerate(bys):
+            out[f"{name}::by{i}"] = np.asarray(b)
+        eg = kw.pop("expected_groups", None)
+        if eg is not None:
+            egs = eg if isinstance(eg, tuple) else (eg,)
+            for i, e in enumerate(egs):
+                out[f"{name}::expected{i}"] = np.asarray(e)
+        if kw.get("fill_value") is not None:
+            out[f"{name}::fill_value"] = np.asarray(kw["fill_value"])
+        if kw.get("finalize_kwargs"):
+            out[f"{name}::ddof"] = np.asarray(kw["finalize_kwargs"]["ddof"])
+    np.savez_compressed(OUT, **out)
+    print(f"wrote {OUT}: {n_done} cases ({n_skip} skipped)")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,61 @@
+"""Helpers to iterate the committed golden fixtures (tests/golden/golden_cases.npz)."""
+
+from __future__ import annotations
+
+import os
+
+import numpy as np
+
+GOLDEN = os.path.join(os.path.dirname(__file__), "golden", "golden_cases.npz")
+
+
+def load_golden_cases():
+    """Yield (name, inputs-dict, expected-result, expected-groups-list)."""
+    data = np.load(GOLDEN, allow_pickle=False)
+    names = sorted({k.split("::")[0] for k in data.files})
+    for name in names:
+        def get(suffix, default=None):
+            key = f"{name}::{suffix}"
+            return data[key] if key in data.files else default
+
+        bys = []
+        i = 0
+        while get(f"by{i}") is not None:
+            bys.append(get(f"by{i}"))
+            i += 1
+        expected = []
+        i = 0
+        while get(f"expected{i}") is not None:
+            expected.append(get(f"expected{i}"))
+            i += 1
+        groups = []
+        i = 0
+        while get(f"groups{i}") is not None:
+            groups.append(get(f"groups{i}"))
+            i += 1
+        func = name.split("_")[0]
+        kw = {}
+        if expected:
+            kw["expected_groups"] = tuple(expected) if len(expected) > 1 else expected[0]
+        fv = get("fill_value")
+        if fv is not None:
+            kw["fill_value"] = fv.item()
+        ddof = get("ddof")
+        if ddof is not None:
+            kw["finalize_kwargs"] = {"ddof": int(ddof)}
+        yield name, dict(array=get("array"), by=tuple(bys), func=func, **kw), get("result"), groups
+
+
+def tolerance_for(name, result_dtype):
+    """fp tolerance per case class: engine="flox" (the golden producer)
+    accumulates fp32 in fp32 while our builds accumulate in f64 (npg
+    semantics, reference tests/test_properties.py:146-151) -> loose rtol for
+    f32; tight for f64 (reference tests/__init__.py:96-99 uses
+    rtol=1e-15/atol=1e-18 for f64, var/std rtol 1e-13 test_core.py:259)."""
+    if result_dtype.kind in "iub":
+        return dict(rtol=0, atol=0)
+    if result_dtype.itemsize == 4:
+        return dict(rtol=2e-6, atol=1e-7)
+    if "var" in name or "std" in name:
+        return dict(rtol=1e-12, atol=1e-14)
+    return dict(rtol=1e-13, atol=1e-16)
