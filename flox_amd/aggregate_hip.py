@@ -1,0 +1,352 @@
+"""engine="hip": grouped aggregations on MI355X via the floxhip C ABI.
+
+Two levels:
+
+* ``grouped_partials(...)`` — the fused low-level call used by
+  ``flox_amd.core.groupby_reduce``: one kernel pass produces every per-group
+  partial an aggregation needs (e.g. mean -> {sum(f64), count(i64)}), which
+  is also exactly the per-rank state the RCCL combine reduces.
+
+* reference-shaped per-reduction callables (``sum``, ``nansum``, ``mean``,
+  ``nanlen`` ...) with the signature of the reference engine seam
+  (flox/aggregations.py:60-133 generic_aggregate ->
+  f(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None)),
+  so ``engine="hip"`` registers beside the reference's "flox"/"numpy" engines.
+
+All tensors are torch CUDA tensors; there is NO CPU fallback — if the HIP
+library or a GPU is missing these raise.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import math
+
+import numpy as np
+import torch
+
+from . import _ffi
+from ._ffi import (
+    FLAG_SKIPNAN,
+    SET_COUNT,
+    SET_MAX_COUNT,
+    SET_MAX_FULL,
+    SET_MIN_COUNT,
+    SET_MIN_FULL,
+    SET_PROD,
+    SET_SSD,
+    SET_SUM_COUNT,
+    SET_SUM_COUNT_PRESENT,
+    FhCall,
+)
+
+_TORCH_VDTYPE = {
+    torch.float32: _ffi.F32,
+    torch.float64: _ffi.F64,
+    torch.int64: _ffi.I64,
+    torch.int32: _ffi.I32,
+}
+_TORCH_LDTYPE = {torch.int64: _ffi.L_I64, torch.int32: _ffi.L_I32}
+
+# op-set membership: which output tensors each set produces
+_SET_MEMBERS = {
+    SET_SUM_COUNT: ("sum", "count"),
+    SET_SUM_COUNT_PRESENT: ("sum", "count", "present"),
+    SET_COUNT: ("count",),
+    SET_MIN_FULL: ("min", "count", "present", "nanflag"),
+    SET_MIN_COUNT: ("min", "count"),
+    SET_MAX_FULL: ("max", "count", "present", "nanflag"),
+    SET_MAX_COUNT: ("max", "count"),
+    SET_SSD: ("sum",),
+    SET_PROD: ("sum", "count", "present"),
+}
+
+
+def _acc_dtype(value_dtype: torch.dtype) -> torch.dtype:
+    """sum/ssd accumulator dtype: f64 for floats (npg contract), i64 for ints."""
+    return torch.float64 if value_dtype.is_floating_point else torch.int64
+
+
+def _require_gpu_tensor(t: torch.Tensor, name: str) -> None:
+    if not t.is_cuda:
+        raise RuntimeError(
+            f"flox_amd engine='hip' requires {name} on the GPU (got device={t.device}); "
+            "there is deliberately no CPU fallback."
+        )
+
+
+def grouped_partials(
+    op_set: int,
+    values: torch.Tensor,
+    labels: torch.Tensor,
+    ngroups: int,
+    *,
+    labels2: torch.Tensor | None = None,
+    grp_shape: tuple[int, int] | None = None,
+    means: torch.Tensor | None = None,
+    skipnan: bool = False,
+    force_path: int = 0,
+) -> dict[str, torch.Tensor]:
+    """One fused factorize+reduce pass. Returns per-group partial tensors.
+
+    values, labels: contiguous 1-D CUDA tensors of equal length.
+    labels2/grp_shape: fused 2-D groupby (reference factorize.py:102-108).
+    means: f64[ngroups] for SET_SSD (var pass 2).
+    force_path: 0 auto, 1 LDS-binned, 2 global-atomic (testing).
+    """
+    lib = _ffi.load_library()
+    _require_gpu_tensor(values, "values")
+    _require_gpu_tensor(labels, "labels")
+    if ngroups == 0:
+        dev0 = values.device
+        acc = _acc_dtype(values.dtype)
+        empty = {
+            "sum": torch.empty(0, dtype=acc, device=dev0),
+            "count": torch.empty(0, dtype=torch.int64, device=dev0),
+            "present": torch.empty(0, dtype=torch.int32, device=dev0),
+            "min": torch.empty(0, dtype=values.dtype, device=dev0),
+            "max": torch.empty(0, dtype=values.dtype, device=dev0),
+            "nanflag": torch.empty(0, dtype=torch.int32, device=dev0),
+        }
+        return {k: empty[k] for k in _SET_MEMBERS[op_set]}
+    assert values.ndim == 1 and labels.ndim == 1 and values.numel() == labels.numel()
+    values = values.contiguous()
+    labels = labels.contiguous()
+    if labels.dtype not in _TORCH_LDTYPE:
+        labels = labels.to(torch.int64)
+    if values.dtype not in _TORCH_VDTYPE:
+        raise NotImplementedError(f"engine='hip' does not support values dtype {values.dtype}")
+    dev = values.device
+
+    c = FhCall()
+    c.op_set = op_set
+    c.vdtype = _TORCH_VDTYPE[values.dtype]
+    c.ldtype = _TORCH_LDTYPE[labels.dtype]
+    c.flags = (FLAG_SKIPNAN if skipnan else 0) | (
+        _ffi.FLAG_FORCE_LDS if force_path == 1 else _ffi.FLAG_FORCE_ATOMIC if force_path == 2 else 0
+    )
+    c.n = values.numel()
+    c.ngroups = ngroups
+    c.values = values.data_ptr()
+    c.labels = labels.data_ptr()
+    if labels2 is not None:
+        _require_gpu_tensor(labels2, "labels2")
+        labels2 = labels2.contiguous()
+        if labels2.dtype != labels.dtype:
+            labels2 = labels2.to(labels.dtype)
+        assert grp_shape is not None and grp_shape[0] * grp_shape[1] == ngroups
+        c.labels2 = labels2.data_ptr()
+        c.g0, c.g1 = grp_shape
+    if means is not None:
+        assert means.dtype == torch.float64 and means.is_cuda
+        means = means.contiguous()
+        c.means = means.data_ptr()
+
+    members = _SET_MEMBERS[op_set]
+    out: dict[str, torch.Tensor] = {}
+    if "sum" in members:
+        out["sum"] = torch.empty(ngroups, dtype=_acc_dtype(values.dtype), device=dev)
+        if op_set == SET_SSD:
+            out["sum"] = torch.empty(ngroups, dtype=torch.float64, device=dev)
+        c.out_sum = out["sum"].data_ptr()
+    if "count" in members:
+        out["count"] = torch.empty(ngroups, dtype=torch.int64, device=dev)
+        c.out_count = out["count"].data_ptr()
+    if "present" in members:
+        out["present"] = torch.empty(ngroups, dtype=torch.int32, device=dev)
+        c.out_present = out["present"].data_ptr()
+    if "min" in members:
+        out["min"] = torch.empty(ngroups, dtype=values.dtype, device=dev)
+        c.out_min = out["min"].data_ptr()
+    if "max" in members:
+        out["max"] = torch.empty(ngroups, dtype=values.dtype, device=dev)
+        c.out_max = out["max"].data_ptr()
+    if "nanflag" in members:
+        out["nanflag"] = torch.empty(ngroups, dtype=torch.int32, device=dev)
+        c.out_nanflag = out["nanflag"].data_ptr()
+
+    nscratch = lib.fh_scratch_bytes(ctypes.byref(c))
+    if nscratch < 0:
+        raise RuntimeError("fh_scratch_bytes failed")
+    scratch = None
+    if nscratch:
+        scratch = torch.empty(nscratch, dtype=torch.uint8, device=dev)
+        c.scratch = scratch.data_ptr()
+        c.scratch_bytes = nscratch
+    c.stream = torch.cuda.current_stream(dev).cuda_stream
+
+    _ffi.check(lib.fh_grouped_reduce(ctypes.byref(c)))
+    out["_path"] = c.path_used  # type: ignore[assignment]
+    # keep tensors alive until the stream consumes them (torch caching allocator
+    # ties lifetime to the stream via recorded events only for torch ops; we
+    # record explicitly)
+    for t in (values, labels, labels2, means, scratch):
+        if isinstance(t, torch.Tensor):
+            t.record_stream(torch.cuda.current_stream(dev))
+    return out
+
+
+# ---------------------------------------------------------------------------
+# reference-shaped engine callables (the generic_aggregate seam)
+# ---------------------------------------------------------------------------
+
+
+def _prep(group_idx, array):
+    """Coerce engine-seam inputs (torch CUDA tensors; numpy moves to GPU)."""
+    if not isinstance(array, torch.Tensor):
+        array = torch.as_tensor(np.ascontiguousarray(array))
+    if not isinstance(group_idx, torch.Tensor):
+        group_idx = torch.as_tensor(np.ascontiguousarray(group_idx))
+    if not array.is_cuda:
+        if not torch.cuda.is_available():
+            raise RuntimeError("engine='hip' requires a GPU; none is available")
+        array = array.cuda()
+    group_idx = group_idx.to(array.device)
+    return group_idx.reshape(-1), array.reshape(-1)
+
+
+def _size_of(group_idx, size):
+    if size is not None:
+        return int(size)
+    return int(group_idx.max().item()) + 1 if group_idx.numel() else 0
+
+
+def _fill(result: torch.Tensor, mask: torch.Tensor, fill_value):
+    if fill_value is None:
+        return result
+    if mask.any():
+        fv = torch.as_tensor(fill_value, dtype=result.dtype, device=result.device)
+        result = torch.where(mask, fv, result)
+    return result
+
+
+def _to_dtype(result: torch.Tensor, dtype):
+    if dtype is None:
+        return result
+    import torch.utils.dlpack  # noqa: F401
+
+    td = torch.from_numpy(np.empty(0, dtype=np.dtype(dtype))).dtype
+    return result.to(td)
+
+
+def _sum_like(group_idx, array, *, skipnan, axis=-1, size=None, fill_value=None, dtype=None):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_SUM_COUNT_PRESENT, array, group_idx, ng, skipnan=skipnan)
+    res = p["sum"]
+    if array.dtype.is_floating_point:
+        res = res.to(array.dtype if dtype is None else res.dtype)
+    res = _fill(res, p["present"] == 0, fill_value)
+    return _to_dtype(res, dtype)
+
+
+def sum(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):  # noqa: A001
+    return _sum_like(group_idx, array, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nansum(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _sum_like(group_idx, array, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanlen(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_COUNT, array, group_idx, ng, skipnan=True)
+    return _to_dtype(p["count"], dtype if dtype is not None else np.intp)
+
+
+count = nanlen
+
+
+def _mean_like(group_idx, array, *, skipnan, size=None, fill_value=None, dtype=None):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_SUM_COUNT, array, group_idx, ng, skipnan=skipnan)
+    res = p["sum"] / p["count"]
+    if array.dtype.is_floating_point and dtype is None:
+        res = res.to(array.dtype)
+    res = _fill(res, p["count"] == 0, fill_value)
+    return _to_dtype(res, dtype)
+
+
+def mean(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _mean_like(group_idx, array, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanmean(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _mean_like(group_idx, array, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def _minmax(group_idx, array, *, ismin, skipnan, size=None, fill_value=None, dtype=None):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    op = (SET_MIN_COUNT if skipnan else SET_MIN_FULL) if ismin else (
+        SET_MAX_COUNT if skipnan else SET_MAX_FULL
+    )
+    p = grouped_partials(op, array, group_idx, ng, skipnan=skipnan)
+    res = p["min" if ismin else "max"]
+    if not skipnan and array.dtype.is_floating_point:
+        res = torch.where(p["nanflag"] != 0, torch.full_like(res, float("nan")), res)
+    empty = (p["present"] == 0) if "present" in p else (p["count"] == 0)
+    res = _fill(res, empty, fill_value)
+    return _to_dtype(res, dtype)
+
+
+def min(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):  # noqa: A001
+    return _minmax(group_idx, array, ismin=True, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanmin(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _minmax(group_idx, array, ismin=True, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def max(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):  # noqa: A001
+    return _minmax(group_idx, array, ismin=False, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanmax(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _minmax(group_idx, array, ismin=False, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def prod(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_PROD, array, group_idx, ng, skipnan=False)
+    res = p["sum"]
+    if array.dtype.is_floating_point:
+        res = res.to(array.dtype if dtype is None else res.dtype)
+    res = _fill(res, p["present"] == 0, fill_value)
+    return _to_dtype(res, dtype)
+
+
+def nanprod(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_PROD, array, group_idx, ng, skipnan=True)
+    res = p["sum"]
+    if array.dtype.is_floating_point:
+        res = res.to(array.dtype if dtype is None else res.dtype)
+    res = _fill(res, p["present"] == 0, fill_value)
+    return _to_dtype(res, dtype)
+
+
+def var_partials(group_idx, array, *, skipnan, size, global_counts=None, global_sums=None):
+    """The (ssd, sum, len) triple of the reference's var_chunk
+    (flox/aggregations.py:348-389): pass 1 sum+count -> means, pass 2
+    sum of squared deviations about those means.
+
+    When global_counts/sums are given (multi-GPU), deviations are taken about
+    the GLOBAL mean so the cross-rank combine is a plain sum (algebraically
+    the reference's _var_combine with zero adjustment terms)."""
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    if global_counts is None:
+        p1 = grouped_partials(SET_SUM_COUNT, array, group_idx, ng, skipnan=skipnan)
+        sums, counts = p1["sum"], p1["count"]
+    else:
+        sums, counts = global_sums, global_counts
+    # empty groups give mean = 0/0 = NaN, but no row carries their code, so the
+    # SSD kernel never reads those entries
+    means = sums.to(torch.float64) / counts
+    p2 = grouped_partials(SET_SSD, array, group_idx, ng, skipnan=skipnan, means=means)
+    return p2["sum"], sums, counts

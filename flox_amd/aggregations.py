@@ -1,0 +1,107 @@
+"""Aggregation blueprints and the engine-dispatch seam.
+
+Keeps the reference's operator API intact (flox/aggregations.py:161-301
+``Aggregation``; :60-133 ``generic_aggregate``): every reduction is described
+by the per-block partials it needs (``chunk``), how partials merge across
+blocks/GPUs (``combine``), and how partials become the result (``finalize``).
+Here ``chunk`` maps onto ONE fused HIP op-set per pass (include/floxhip.h),
+and ``combine`` is the RCCL reduction applied to each partial.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+from . import _ffi
+
+
+@dataclass
+class Aggregation:
+    """Blueprint for a grouped reduction (mirrors reference
+    flox/aggregations.py:161-301, re-expressed for the fused HIP engine).
+
+    op_set: the fused kernel pass (fh_opset) producing this aggregation's
+        partials; None for multi-pass aggregations handled specially (var).
+    skipnan: NaN values contribute nothing (the reference's nan* family).
+    combine: per-partial cross-block/cross-GPU reduction: name -> "sum" |
+        "min" | "max" | "or" (reference combine recipes,
+        flox/aggregations.py:304-546).
+    """
+
+    name: str
+    op_set: int | None
+    skipnan: bool = False
+    combine: dict = field(default_factory=dict)
+    preserves_dtype: bool = False
+
+
+_SUMLIKE_COMBINE = {"sum": "sum", "count": "sum", "present": "max"}
+
+REDUCTIONS: dict[str, Aggregation] = {
+    "count": Aggregation("count", _ffi.SET_COUNT, skipnan=True, combine={"count": "sum"}),
+    "sum": Aggregation("sum", _ffi.SET_SUM_COUNT_PRESENT, combine=dict(_SUMLIKE_COMBINE)),
+    "nansum": Aggregation(
+        "nansum", _ffi.SET_SUM_COUNT_PRESENT, skipnan=True, combine=dict(_SUMLIKE_COMBINE)
+    ),
+    "prod": Aggregation("prod", _ffi.SET_PROD, combine={"sum": "prod", "count": "sum", "present": "max"}),
+    "nanprod": Aggregation(
+        "nanprod", _ffi.SET_PROD, skipnan=True, combine={"sum": "prod", "count": "sum", "present": "max"}
+    ),
+    "mean": Aggregation("mean", _ffi.SET_SUM_COUNT, combine={"sum": "sum", "count": "sum"}),
+    "nanmean": Aggregation(
+        "nanmean", _ffi.SET_SUM_COUNT, skipnan=True, combine={"sum": "sum", "count": "sum"}
+    ),
+    "min": Aggregation(
+        "min",
+        _ffi.SET_MIN_FULL,
+        combine={"min": "min", "count": "sum", "present": "max", "nanflag": "max"},
+        preserves_dtype=True,
+    ),
+    "nanmin": Aggregation(
+        "nanmin", _ffi.SET_MIN_COUNT, skipnan=True, combine={"min": "min", "count": "sum"},
+        preserves_dtype=True,
+    ),
+    "max": Aggregation(
+        "max",
+        _ffi.SET_MAX_FULL,
+        combine={"max": "max", "count": "sum", "present": "max", "nanflag": "max"},
+        preserves_dtype=True,
+    ),
+    "nanmax": Aggregation(
+        "nanmax", _ffi.SET_MAX_COUNT, skipnan=True, combine={"max": "max", "count": "sum"},
+        preserves_dtype=True,
+    ),
+    # var family: two fused passes (SUM_COUNT then SSD); see core._reduce_var
+    "var": Aggregation("var", None),
+    "nanvar": Aggregation("nanvar", None, skipnan=True),
+    "std": Aggregation("std", None),
+    "nanstd": Aggregation("nanstd", None, skipnan=True),
+}
+
+
+def generic_aggregate(
+    group_idx,
+    array,
+    *,
+    engine: str,
+    func: str,
+    axis=-1,
+    size=None,
+    fill_value=None,
+    dtype=None,
+    **kwargs,
+):
+    """Engine dispatch, the reference's plugin seam
+    (flox/aggregations.py:60-133). flox_amd registers exactly one engine,
+    "hip"; anything else is an explicit error — there is no CPU fallback."""
+    if engine != "hip":
+        raise ValueError(
+            f"flox_amd implements engine='hip' only (got {engine!r}). "
+            "For CPU semantics use the reference implementation."
+        )
+    from . import aggregate_hip
+
+    method = getattr(aggregate_hip, func, None)
+    if method is None:
+        raise NotImplementedError(f"engine='hip' does not implement {func!r} yet")
+    return method(group_idx, array, axis=axis, size=size, fill_value=fill_value, dtype=dtype, **kwargs)

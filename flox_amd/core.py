@@ -1,0 +1,299 @@
+"""groupby_reduce: the flox-shaped top-level API on MI355X.
+
+Mirrors the reference's eager path (flox/core.py:739-1222 groupby_reduce ->
+_reduce_blockwise -> chunk_reduce -> _finalize_results) with the per-block
+kernel layer replaced by fused HIP passes (aggregate_hip.grouped_partials)
+and the dask combine tree replaced by RCCL all-reduces of the per-group
+partial bins (distributed.py).
+
+Scope (round 1): reduction over ALL dims of ``by`` (axis=None or the full
+trailing tuple), ``array`` fully covered by ``by`` (no leading batch dims on
+the GPU path yet), dtypes f32/f64/i32/i64/bool, 1 or more ``by`` arrays.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Any
+
+import numpy as np
+import torch
+
+from . import _ffi, distributed, xrdtypes
+from .aggregate_hip import grouped_partials, var_partials
+from .aggregations import REDUCTIONS
+
+_TORCH_TO_NP = {
+    torch.float32: np.dtype("float32"),
+    torch.float64: np.dtype("float64"),
+    torch.int64: np.dtype("int64"),
+    torch.int32: np.dtype("int32"),
+    torch.bool: np.dtype(bool),
+}
+
+
+def _np_dtype(t: torch.dtype) -> np.dtype:
+    return _TORCH_TO_NP[t]
+
+
+def _torch_dtype(d: np.dtype) -> torch.dtype:
+    for k, v in _TORCH_TO_NP.items():
+        if v == np.dtype(d):
+            return k
+    raise NotImplementedError(f"unsupported dtype {d}")
+
+
+def _as_device_tensor(x, device) -> torch.Tensor:
+    if isinstance(x, torch.Tensor):
+        t = x
+    else:
+        t = torch.from_numpy(np.ascontiguousarray(x))
+    if t.device != device:
+        t = t.to(device, non_blocking=True)
+    return t
+
+
+class _FactorizedBy:
+    """One by-array turned into dense integer codes on device.
+
+    direct=True: codes ARE the raw labels (expected_groups is 0..n-1), so the
+    kernel's bounds check is the whole factorize step (the GPU analogue of
+    the reference's RangeIndex fast path, factorize.py:44-53)."""
+
+    def __init__(self, codes: torch.Tensor, groups: np.ndarray, direct: bool):
+        self.codes = codes
+        self.groups = groups
+        self.ngroups = len(groups)
+        self.direct = direct
+
+
+def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
+    """labels -> codes in [0, ngroups), invalid/NaN -> out-of-range
+    (restates reference _factorize_single, factorize.py:42-99)."""
+    if expect is None:
+        if not sort:
+            raise NotImplementedError("sort=False without expected_groups")
+        fl = flat
+        if fl.dtype.is_floating_point:
+            fl = fl[~torch.isnan(fl)]
+        uniq = torch.unique(fl)  # sorted
+        codes = torch.searchsorted(uniq, flat)
+        if flat.dtype.is_floating_point:
+            codes = torch.where(torch.isnan(flat), torch.full_like(codes, -1), codes)
+        return _FactorizedBy(codes, uniq.cpu().numpy(), direct=False)
+
+    expect_np = np.asarray(expect)
+    n = len(expect_np)
+    if (
+        expect_np.dtype.kind in "iu"
+        and not flat.dtype.is_floating_point
+        and n > 0
+        and expect_np[0] == 0
+        and expect_np[-1] == n - 1
+        and np.array_equal(expect_np, np.arange(n))
+    ):
+        # direct path: labels are already the codes; the kernel's unsigned
+        # bounds check drops anything outside [0, n)
+        return _FactorizedBy(flat, expect_np, direct=True)
+
+    sorted_expect = np.sort(expect_np) if sort else expect_np
+    if sort is False and not np.all(np.diff(expect_np) >= 0):
+        raise NotImplementedError("unsorted expected_groups with sort=False")
+    exp_t = torch.from_numpy(np.ascontiguousarray(sorted_expect)).to(flat.device)
+    if exp_t.dtype != flat.dtype:
+        common = torch.promote_types(exp_t.dtype, flat.dtype)
+        exp_t = exp_t.to(common)
+        flat = flat.to(common)
+    codes = torch.searchsorted(exp_t, flat)
+    clipped = torch.clamp(codes, max=n - 1)
+    bad = (codes == n) | (exp_t[clipped] != flat)
+    if flat.dtype.is_floating_point:
+        bad |= torch.isnan(flat)
+    codes = torch.where(bad, torch.full_like(codes, -1), codes)
+    return _FactorizedBy(codes, sorted_expect, direct=False)
+
+
+def _combined_codes(facs: list[_FactorizedBy]):
+    """Combine multiple factorized by-arrays into one code stream.
+
+    Returns (labels, labels2, grp_pair): when exactly two direct int code
+    arrays, the ravel is fused into the kernel (labels2 path, mirroring
+    reference _ravel_factorized factorize.py:102-108); otherwise codes are
+    raveled here with invalid propagation."""
+    if len(facs) == 1:
+        return facs[0].codes, None, None
+    if len(facs) == 2 and all(f.direct for f in facs):
+        return facs[0].codes, facs[1].codes, (facs[0].ngroups, facs[1].ngroups)
+    code = facs[0].codes.to(torch.int64)
+    bad = code < 0
+    for f in facs[1:]:
+        c = f.codes.to(torch.int64)
+        bad = bad | (c < 0) | (code < 0)
+        code = code * f.ngroups + c
+    code = torch.where(bad, torch.full_like(code, -1), code)
+    return code, None, None
+
+
+def groupby_reduce(
+    array,
+    *by,
+    func: str,
+    expected_groups=None,
+    sort: bool = True,
+    isbin: bool = False,
+    axis=None,
+    fill_value=None,
+    dtype=None,
+    min_count: int | None = None,
+    engine: str = "hip",
+    finalize_kwargs: dict[str, Any] | None = None,
+    distributed_combine: bool | None = None,
+):
+    """Grouped reduction with flox semantics on MI355X. Returns (result, *groups).
+
+    ``array``/``by``: numpy arrays or torch tensors (numpy is moved to the
+    GPU; note the PCIe cost — keep tensors resident for performance).
+    ``distributed_combine``: all-reduce the per-group partials across the
+    initialized torch.distributed world (default: auto when initialized).
+    Each rank passes its own row shard; results are full-group on every rank.
+    """
+    if engine != "hip":
+        raise ValueError(f"flox_amd implements engine='hip' only (got {engine!r})")
+    if isbin:
+        raise NotImplementedError("isbin=True (bin edges) is not implemented yet")
+    if func not in REDUCTIONS:
+        raise NotImplementedError(f"reduction {func!r}")
+    agg = REDUCTIONS[func]
+
+    if not torch.cuda.is_available():
+        raise RuntimeError("flox_amd.groupby_reduce requires a GPU (engine='hip')")
+    device = torch.device("cuda", torch.cuda.current_device())
+
+    return_numpy = not isinstance(array, torch.Tensor)
+    arr = _as_device_tensor(array, device)
+    if arr.dtype == torch.bool:
+        arr = arr.to(torch.int64)  # reference core.py:916-917
+    in_np_dtype = _np_dtype(arr.dtype)
+    bys = tuple(_as_device_tensor(b, device) for b in by)
+    nby = len(bys)
+    if nby == 0:
+        raise ValueError("need at least one by array")
+    by_shape = bys[0].shape
+    for b in bys:
+        assert b.shape == by_shape
+    if tuple(arr.shape[arr.ndim - len(by_shape) :]) != tuple(by_shape):
+        raise ValueError(f"by {tuple(by_shape)} must align with trailing dims of array {tuple(arr.shape)}")
+    if axis is not None:
+        ax = axis if isinstance(axis, (tuple, list)) else (axis,)
+        ax = tuple(a % arr.ndim for a in ax)
+        if tuple(sorted(ax)) != tuple(range(arr.ndim - len(by_shape), arr.ndim)):
+            raise NotImplementedError("GPU path reduces over all dims of by (axis subset: next row)")
+    lead_shape = tuple(arr.shape[: arr.ndim - len(by_shape)])
+    if math.prod(lead_shape) != 1:
+        raise NotImplementedError("leading (non-grouped) array dims: next row (config-4 kernel)")
+
+    if expected_groups is not None and not isinstance(expected_groups, tuple):
+        expected_groups = (expected_groups,)
+    if expected_groups is None:
+        expected_groups = (None,) * nby
+    provided_expected = any(e is not None for e in expected_groups)
+
+    facs = [_factorize_device(b.reshape(-1), e, sort) for b, e in zip(bys, expected_groups)]
+    grp_shape = tuple(f.ngroups for f in facs)
+    ngroups = math.prod(grp_shape)
+    labels, labels2, grp_pair = _combined_codes(facs)
+    vals = arr.reshape(-1)
+
+    # min_count defaulting (reference core.py:1026-1038 + aggregations.py:997-1003)
+    if min_count is None:
+        min_count_ = 1 if (fill_value is not None and provided_expected) else 0
+    else:
+        min_count_ = min_count
+    out_dtype = xrdtypes.final_dtype(func, in_np_dtype, dtype)
+    if func in ("nanmin", "nanmax") and min_count_ == 0:
+        min_count_ = 1
+        if fill_value is None:
+            fill_value = xrdtypes.fill_default(func, out_dtype)
+    if min_count_ > 0 and func in ("nansum", "nanprod") and fill_value is None:
+        fill_value = float("nan")
+    if fill_value is not None:
+        # user fill promotes the output dtype (reference xrdtypes.py:170-171)
+        out_dtype = np.result_type(out_dtype, fill_value)
+
+    dist_on = distributed.is_active() if distributed_combine is None else distributed_combine
+    ddof = (finalize_kwargs or {}).get("ddof", 0)
+
+    kw = dict(labels2=labels2, grp_shape=grp_pair)
+    if func in ("var", "nanvar", "std", "nanstd"):
+        skip = agg.skipnan
+        if dist_on:
+            p1 = grouped_partials(_ffi.SET_SUM_COUNT, vals, labels, ngroups, skipnan=skip, **kw)
+            distributed.all_reduce_(p1["sum"], "sum")
+            distributed.all_reduce_(p1["count"], "sum")
+            ssd, sums, counts = var_partials(
+                labels, vals, skipnan=skip, size=ngroups,
+                global_counts=p1["count"], global_sums=p1["sum"],
+            )
+            distributed.all_reduce_(ssd, "sum")
+        else:
+            ssd, sums, counts = var_partials(labels, vals, skipnan=skip, size=ngroups)
+        den = counts.to(torch.float64) - ddof
+        result = ssd / den
+        nan_t = torch.full_like(result, float("nan"))
+        result = torch.where((den < 0) | (counts == 0), nan_t, result)
+        if func in ("std", "nanstd"):
+            result = torch.sqrt(result)
+        counts_for_mask = counts
+        empty_mask = counts == 0
+    else:
+        p = grouped_partials(agg.op_set, vals, labels, ngroups, skipnan=agg.skipnan, **kw)
+        if dist_on:
+            distributed.combine_partials(p, agg.combine)
+        if func == "count":
+            result = p["count"]
+            empty_mask = p["count"] == 0
+            counts_for_mask = p["count"]
+        elif func in ("sum", "nansum", "prod", "nanprod"):
+            result = p["sum"]
+            empty_mask = p["present"] == 0
+            counts_for_mask = p["count"]
+        elif func in ("mean", "nanmean"):
+            result = p["sum"].to(torch.float64) / p["count"]
+            empty_mask = p["count"] == 0
+            counts_for_mask = p["count"]
+        elif func in ("min", "nanmin", "max", "nanmax"):
+            result = p["min" if "min" in p else "max"]
+            if "nanflag" in p and arr.dtype.is_floating_point:
+                nan_t = torch.full_like(result, float("nan"))
+                result = torch.where(p["nanflag"] != 0, nan_t, result)
+            empty_mask = (p["present"] == 0) if "present" in p else (p["count"] == 0)
+            counts_for_mask = p["count"]
+        else:  # pragma: no cover
+            raise NotImplementedError(func)
+
+    # --- finalize: masking + fills + final dtype (reference core.py:410-475) ---
+    t_out_dtype = _torch_dtype(out_dtype)
+    if min_count_ > 0:
+        mask = counts_for_mask < min_count_
+        if bool(mask.any().item()):
+            if fill_value is None:
+                raise ValueError("Filling is required but fill_value is None.")
+            if not torch.is_floating_point(torch.empty(0, dtype=t_out_dtype)) and isinstance(
+                fill_value, float
+            ) and math.isnan(fill_value):
+                t_out_dtype = torch.float64
+            result = result.to(t_out_dtype)
+            result = torch.where(mask, torch.tensor(fill_value, dtype=t_out_dtype, device=device), result)
+    else:
+        fv = fill_value if fill_value is not None else xrdtypes.fill_default(func, out_dtype)
+        if bool(empty_mask.any().item()):
+            result = result.to(t_out_dtype)
+            result = torch.where(empty_mask, torch.tensor(fv, dtype=t_out_dtype, device=device), result)
+
+    result = result.to(t_out_dtype)
+    result = result.reshape(lead_shape + grp_shape)
+
+    groups = tuple(f.groups for f in facs)
+    if return_numpy:
+        return (result.cpu().numpy(), *groups)
+    return (result, *groups)

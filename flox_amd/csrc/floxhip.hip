@@ -1,0 +1,599 @@
+/* floxhip — MI355X (gfx950, CDNA4) grouped-reduction kernels.
+ *
+ * Re-implements, from scratch and HBM-first, the hot path of
+ * xarray-contrib/flox: factorize + per-group sum/count/min/max/var
+ * (reference flox/core.py:214-394 chunk_reduce, flox/aggregate_flox.py,
+ * flox/aggregate_npg.py). Design notes in /root/repo/DESIGN.md.
+ *
+ * Two paths, selected by group cardinality (the GPU analogue of the
+ * reference's _choose_engine, flox/core.py:712-736):
+ *
+ *  1. LDS-binned scatter (ngroups small enough that the per-group partial
+ *     bins fit in the CU's 160 KiB LDS): each 1024-thread workgroup keeps
+ *     private bins in LDS, streams its grid-stride share of the rows with
+ *     16-byte vector loads (values) + int64 label loads, updates bins with
+ *     native LDS atomics (ds_add_f64 / ds_add_u32 / ds_min_u32|u64), then
+ *     writes its bins to a per-block slab in HBM with plain coalesced
+ *     stores. A combine kernel folds the slab over blocks (fixed order).
+ *     Nothing in this path uses inter-workgroup communication.
+ *
+ *  2. Global-atomic scatter (large ngroups, e.g. the 1e7-group config):
+ *     bins live in HBM (LLC-cached), rows update them with device-scope
+ *     atomics (global_atomic_add_f64 / _umin / _umax ...). min/max bins are
+ *     kept in an order-preserving unsigned encoding so init is a memset and
+ *     the atomic is an integer min/max; a decode kernel maps back.
+ *
+ * Numerics: float32 sums/counts/means/var accumulate in float64 (the
+ * numpy_groupies contract, reference tests/test_properties.py:146-151).
+ * count/min/max are bit-exact. fp sums use atomics, so the reduction order
+ * is not fixed: results are reproducible only to fp-roundoff tolerance
+ * (stated in DESIGN.md and the parity tests).
+ */
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+#include <cstdio>
+
+#include "../../include/floxhip.h"
+
+#define FH_CHECK(x)                       \
+  do {                                    \
+    hipError_t _e = (x);                  \
+    if (_e != hipSuccess) return (int)_e + 1000; \
+  } while (0)
+
+namespace {
+
+constexpr int BLOCK_LDS = 1024;   /* 16 waves/CU at 1 block/CU */
+constexpr int BLOCK_ATOMIC = 256;
+constexpr int NUM_CU = 256;
+constexpr int64_t LDS_MAX = 160 * 1024 - 512; /* gfx950 LDS per workgroup */
+
+/* ---- op-set membership ------------------------------------------------- */
+enum {
+  B_SUM = 1,
+  B_CNT = 2,
+  B_PRESENT = 4,
+  B_MIN = 8,
+  B_MAX = 16,
+  B_NANFLAG = 32,
+  B_SSD = 64,
+  B_PROD = 128,
+};
+
+__host__ __device__ constexpr int set_bits(int op_set) {
+  switch (op_set) {
+    case FH_SET_SUM_COUNT: return B_SUM | B_CNT;
+    case FH_SET_SUM_COUNT_PRESENT: return B_SUM | B_CNT | B_PRESENT;
+    case FH_SET_COUNT: return B_CNT;
+    case FH_SET_MIN_FULL: return B_MIN | B_CNT | B_PRESENT | B_NANFLAG;
+    case FH_SET_MIN_COUNT: return B_MIN | B_CNT;
+    case FH_SET_MAX_FULL: return B_MAX | B_CNT | B_PRESENT | B_NANFLAG;
+    case FH_SET_MAX_COUNT: return B_MAX | B_CNT;
+    case FH_SET_SSD: return B_SSD;
+    case FH_SET_PROD: return B_PROD | B_CNT | B_PRESENT;
+    default: return 0;
+  }
+}
+
+/* ---- per-dtype traits --------------------------------------------------- */
+template <typename V> struct Traits;
+
+template <> struct Traits<float> {
+  using Acc = double;          /* f64 accumulation (npg contract) */
+  using Enc = uint32_t;        /* order-preserving encoding for min/max */
+  static constexpr int VEC = 4;
+  static __device__ __forceinline__ bool isnan_(float v) { return v != v; }
+  static __device__ __forceinline__ Enc enc(float v) {
+    uint32_t u = __float_as_uint(v);
+    return (u & 0x80000000u) ? ~u : (u | 0x80000000u);
+  }
+  static __device__ __forceinline__ float dec(Enc u) {
+    uint32_t r = (u & 0x80000000u) ? (u ^ 0x80000000u) : ~u;
+    return __uint_as_float(r);
+  }
+  static __device__ __forceinline__ float pos_inf() { return __builtin_inff(); }
+};
+
+template <> struct Traits<double> {
+  using Acc = double;
+  using Enc = uint64_t;
+  static constexpr int VEC = 2;
+  static __device__ __forceinline__ bool isnan_(double v) { return v != v; }
+  static __device__ __forceinline__ Enc enc(double v) {
+    uint64_t u = __double_as_longlong(v);
+    return (u & 0x8000000000000000ull) ? ~u : (u | 0x8000000000000000ull);
+  }
+  static __device__ __forceinline__ double dec(Enc u) {
+    uint64_t r = (u & 0x8000000000000000ull) ? (u ^ 0x8000000000000000ull) : ~u;
+    return __longlong_as_double((long long)r);
+  }
+  static __device__ __forceinline__ double pos_inf() { return __builtin_inf(); }
+};
+
+template <> struct Traits<int32_t> {
+  using Acc = int64_t;         /* numpy promotes i32 sums to platform int */
+  using Enc = uint32_t;
+  static constexpr int VEC = 4;
+  static __device__ __forceinline__ bool isnan_(int32_t) { return false; }
+  static __device__ __forceinline__ Enc enc(int32_t v) {
+    return (uint32_t)v ^ 0x80000000u;   /* order-preserving signed->unsigned */
+  }
+  static __device__ __forceinline__ int32_t dec(Enc u) { return (int32_t)(u ^ 0x80000000u); }
+  static __device__ __forceinline__ int32_t pos_inf() { return INT32_MAX; }
+};
+
+template <> struct Traits<int64_t> {
+  using Acc = int64_t;
+  using Enc = uint64_t;
+  static constexpr int VEC = 2;
+  static __device__ __forceinline__ bool isnan_(int64_t) { return false; }
+  static __device__ __forceinline__ Enc enc(int64_t v) {
+    return (uint64_t)v ^ 0x8000000000000000ull;
+  }
+  static __device__ __forceinline__ int64_t dec(Enc u) {
+    return (int64_t)(u ^ 0x8000000000000000ull);
+  }
+  static __device__ __forceinline__ int64_t pos_inf() { return INT64_MAX; }
+};
+
+/* atomic add on the accumulator type (LDS or global pointer) */
+__device__ __forceinline__ void acc_add(double* p, double v) { atomicAdd(p, v); }
+__device__ __forceinline__ void acc_add(int64_t* p, int64_t v) {
+  atomicAdd(reinterpret_cast<unsigned long long*>(p), (unsigned long long)v);
+}
+__device__ __forceinline__ void enc_min(uint32_t* p, uint32_t v) { atomicMin(p, v); }
+__device__ __forceinline__ void enc_min(uint64_t* p, uint64_t v) {
+  atomicMin(reinterpret_cast<unsigned long long*>(p), (unsigned long long)v);
+}
+__device__ __forceinline__ void enc_max(uint32_t* p, uint32_t v) { atomicMax(p, v); }
+__device__ __forceinline__ void enc_max(uint64_t* p, uint64_t v) {
+  atomicMax(reinterpret_cast<unsigned long long*>(p), (unsigned long long)v);
+}
+/* CAS product (prod is rare; contention-tolerant CAS loop) */
+__device__ __forceinline__ void acc_mul(double* p, double v) {
+  unsigned long long* u = reinterpret_cast<unsigned long long*>(p);
+  unsigned long long old = *u, assumed;
+  do {
+    assumed = old;
+    double cur = __longlong_as_double((long long)assumed);
+    old = atomicCAS(u, assumed, (unsigned long long)__double_as_longlong(cur * v));
+  } while (old != assumed);
+}
+__device__ __forceinline__ void acc_mul(int64_t* p, int64_t v) {
+  unsigned long long* u = reinterpret_cast<unsigned long long*>(p);
+  unsigned long long old = *u, assumed;
+  do {
+    assumed = old;
+    old = atomicCAS(u, assumed, (unsigned long long)((int64_t)assumed * v));
+  } while (old != assumed);
+}
+
+/* 16-byte-aligned vector type for coalesced loads */
+template <typename T, int N> struct alignas(sizeof(T) * N) Vec { T v[N]; };
+
+/* ---- bin layout (shared between LDS carve, slab and combine) ------------ */
+struct BinLayout {
+  int64_t sum_off, cnt_off, present_off, minmax_off, nanflag_off;
+  int64_t bytes;  /* per block-copy of the bins */
+};
+
+template <typename V>
+__host__ __device__ BinLayout bin_layout(int bits, int64_t ngroups, int64_t cnt_elem_size) {
+  BinLayout L{};
+  int64_t off = 0;
+  auto carve = [&](int64_t elem) {
+    int64_t o = off;
+    off += ((ngroups * elem + 255) / 256) * 256; /* 256-B aligned sections */
+    return o;
+  };
+  L.sum_off = (bits & (B_SUM | B_SSD | B_PROD)) ? carve(8) : -1;
+  L.cnt_off = (bits & B_CNT) ? carve(cnt_elem_size) : -1;
+  L.present_off = (bits & B_PRESENT) ? carve(4) : -1;
+  L.minmax_off = (bits & (B_MIN | B_MAX)) ? carve(sizeof(typename Traits<V>::Enc)) : -1;
+  L.nanflag_off = (bits & B_NANFLAG) ? carve(4) : -1;
+  L.bytes = off;
+  return L;
+}
+
+/* ---- kernel 1: LDS-binned grouped reduce -------------------------------- */
+template <typename V, typename L, int OPS>
+__launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
+    const V* __restrict__ values, const L* __restrict__ labels,
+    const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
+    int64_t g1, const double* __restrict__ means, int skipnan,
+    char* __restrict__ slab, BinLayout lay) {
+  using TR = Traits<V>;
+  using Acc = typename TR::Acc;
+  using Enc = typename TR::Enc;
+  constexpr int VEC = TR::VEC;
+  constexpr bool IS_PROD = (OPS & B_PROD) != 0;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  Acc* s_sum = (OPS & (B_SUM | B_SSD | B_PROD)) ? (Acc*)(smem + lay.sum_off) : nullptr;
+  uint32_t* s_cnt = (OPS & B_CNT) ? (uint32_t*)(smem + lay.cnt_off) : nullptr;
+  uint32_t* s_present = (OPS & B_PRESENT) ? (uint32_t*)(smem + lay.present_off) : nullptr;
+  Enc* s_mm = (OPS & (B_MIN | B_MAX)) ? (Enc*)(smem + lay.minmax_off) : nullptr;
+  uint32_t* s_nanflag = (OPS & B_NANFLAG) ? (uint32_t*)(smem + lay.nanflag_off) : nullptr;
+
+  const int tid = threadIdx.x;
+  for (int64_t g = tid; g < ngroups; g += blockDim.x) {
+    if (OPS & (B_SUM | B_SSD)) s_sum[g] = (Acc)0;
+    if (IS_PROD) s_sum[g] = (Acc)1;
+    if (OPS & B_CNT) s_cnt[g] = 0u;
+    if (OPS & B_PRESENT) s_present[g] = 0u;
+    if (OPS & B_MIN) s_mm[g] = (Enc)~(Enc)0;
+    if (OPS & B_MAX) s_mm[g] = (Enc)0;
+    if (OPS & B_NANFLAG) s_nanflag[g] = 0u;
+  }
+  __syncthreads();
+
+  const bool twolab = labels2 != nullptr;
+
+  auto process = [&](V v, int64_t l0raw, int64_t l1raw) {
+    uint64_t code;
+    if (twolab) {
+      if ((uint64_t)l0raw >= (uint64_t)g0 || (uint64_t)l1raw >= (uint64_t)g1) return;
+      code = (uint64_t)l0raw * (uint64_t)g1 + (uint64_t)l1raw;
+    } else {
+      if ((uint64_t)l0raw >= (uint64_t)ngroups) return;
+      code = (uint64_t)l0raw;
+    }
+    const bool vnan = TR::isnan_(v);
+    if (OPS & B_PRESENT) s_present[code] = 1u;  /* benign write race: all store 1 */
+    if (vnan && skipnan) return;
+    if (OPS & B_SUM) acc_add(&s_sum[code], (Acc)v);
+    if (IS_PROD) acc_mul(&s_sum[code], (Acc)v);
+    if (OPS & B_SSD) {
+      double d = (double)v - means[code];
+      atomicAdd((double*)&s_sum[code], d * d);
+    }
+    if (OPS & B_CNT) {
+      if (!vnan) atomicAdd(&s_cnt[code], 1u);
+    }
+    if (OPS & (B_MIN | B_MAX)) {
+      if (vnan) {
+        if (OPS & B_NANFLAG) s_nanflag[code] = 1u;
+      } else {
+        if (OPS & B_MIN) enc_min(&s_mm[code], TR::enc(v));
+        if (OPS & B_MAX) enc_max(&s_mm[code], TR::enc(v));
+      }
+    }
+  };
+
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t gtid = (int64_t)blockIdx.x * blockDim.x + tid;
+  const int64_t nvec = n / VEC;
+  for (int64_t i = gtid; i < nvec; i += stride) {
+    Vec<V, VEC> vv = *reinterpret_cast<const Vec<V, VEC>*>(values + i * VEC);
+    Vec<L, VEC> lv = *reinterpret_cast<const Vec<L, VEC>*>(labels + i * VEC);
+    if (twolab) {
+      Vec<L, VEC> lv2 = *reinterpret_cast<const Vec<L, VEC>*>(labels2 + i * VEC);
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], (int64_t)lv2.v[k]);
+    } else {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], 0);
+    }
+  }
+  /* tail */
+  for (int64_t i = nvec * VEC + gtid; i < n; i += stride) {
+    process(values[i], (int64_t)labels[i], twolab ? (int64_t)labels2[i] : 0);
+  }
+
+  __syncthreads();
+  /* flush bins to this block's slab section with plain coalesced stores */
+  char* my = slab + (int64_t)blockIdx.x * lay.bytes;
+  for (int64_t g = tid; g < ngroups; g += blockDim.x) {
+    if (OPS & (B_SUM | B_SSD | B_PROD)) ((Acc*)(my + lay.sum_off))[g] = s_sum[g];
+    if (OPS & B_CNT) ((uint32_t*)(my + lay.cnt_off))[g] = s_cnt[g];
+    if (OPS & B_PRESENT) ((uint32_t*)(my + lay.present_off))[g] = s_present[g];
+    if (OPS & (B_MIN | B_MAX)) ((Enc*)(my + lay.minmax_off))[g] = s_mm[g];
+    if (OPS & B_NANFLAG) ((uint32_t*)(my + lay.nanflag_off))[g] = s_nanflag[g];
+  }
+}
+
+/* ---- kernel 2: combine per-block slabs (fixed order over blocks) --------- */
+template <typename V, int OPS>
+__global__ void k_combine(const char* __restrict__ slab, int nblocks,
+                          int64_t ngroups, BinLayout lay, void* out_sum,
+                          int64_t* out_count, uint32_t* out_present,
+                          void* out_min, void* out_max, uint32_t* out_nanflag) {
+  using TR = Traits<V>;
+  using Acc = typename TR::Acc;
+  using Enc = typename TR::Enc;
+  constexpr bool IS_PROD = (OPS & B_PROD) != 0;
+  const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= ngroups) return;
+
+  Acc s = IS_PROD ? (Acc)1 : (Acc)0;
+  int64_t c = 0;
+  uint32_t p = 0, nf = 0;
+  Enc mn = (Enc)~(Enc)0, mx = (Enc)0;
+  for (int b = 0; b < nblocks; ++b) {
+    const char* blk = slab + (int64_t)b * lay.bytes;
+    if (OPS & (B_SUM | B_SSD)) s += ((const Acc*)(blk + lay.sum_off))[g];
+    if (IS_PROD) s *= ((const Acc*)(blk + lay.sum_off))[g];
+    if (OPS & B_CNT) c += (int64_t)((const uint32_t*)(blk + lay.cnt_off))[g];
+    if (OPS & B_PRESENT) p |= ((const uint32_t*)(blk + lay.present_off))[g];
+    if (OPS & B_MIN) {
+      Enc e = ((const Enc*)(blk + lay.minmax_off))[g];
+      mn = e < mn ? e : mn;
+    }
+    if (OPS & B_MAX) {
+      Enc e = ((const Enc*)(blk + lay.minmax_off))[g];
+      mx = e > mx ? e : mx;
+    }
+    if (OPS & B_NANFLAG) nf |= ((const uint32_t*)(blk + lay.nanflag_off))[g];
+  }
+  const bool present = (OPS & B_PRESENT) ? (p != 0) : (c != 0);
+  if (OPS & (B_SUM | B_SSD | B_PROD)) ((Acc*)out_sum)[g] = s;
+  if (OPS & B_CNT) out_count[g] = c;
+  if (OPS & B_PRESENT) out_present[g] = p;
+  if (OPS & B_MIN) ((V*)out_min)[g] = present ? TR::dec(mn) : TR::pos_inf();
+  if (OPS & B_MAX) ((V*)out_max)[g] = present ? TR::dec(mx) : (V)-TR::pos_inf();
+  if (OPS & B_NANFLAG) out_nanflag[g] = nf;
+}
+
+/* ---- global-atomic path -------------------------------------------------- */
+/* bins live in HBM/LLC; min/max bins are ENCODED in out_min/out_max during
+ * accumulation (memset-friendly init) and decoded by k_decode afterwards. */
+template <typename V, typename L, int OPS>
+__launch_bounds__(BLOCK_ATOMIC) __global__ void k_reduce_atomic(
+    const V* __restrict__ values, const L* __restrict__ labels,
+    const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
+    int64_t g1, const double* __restrict__ means, int skipnan, void* out_sum,
+    int64_t* out_count, uint32_t* out_present, void* out_min, void* out_max,
+    uint32_t* out_nanflag) {
+  using TR = Traits<V>;
+  using Acc = typename TR::Acc;
+  using Enc = typename TR::Enc;
+  constexpr int VEC = TR::VEC;
+  constexpr bool IS_PROD = (OPS & B_PROD) != 0;
+  const bool twolab = labels2 != nullptr;
+
+  auto process = [&](V v, int64_t l0raw, int64_t l1raw) {
+    uint64_t code;
+    if (twolab) {
+      if ((uint64_t)l0raw >= (uint64_t)g0 || (uint64_t)l1raw >= (uint64_t)g1) return;
+      code = (uint64_t)l0raw * (uint64_t)g1 + (uint64_t)l1raw;
+    } else {
+      if ((uint64_t)l0raw >= (uint64_t)ngroups) return;
+      code = (uint64_t)l0raw;
+    }
+    const bool vnan = TR::isnan_(v);
+    if (OPS & B_PRESENT) out_present[code] = 1u;
+    if (vnan && skipnan) return;
+    if (OPS & B_SUM) acc_add(&((Acc*)out_sum)[code], (Acc)v);
+    if (IS_PROD) acc_mul(&((Acc*)out_sum)[code], (Acc)v);
+    if (OPS & B_SSD) {
+      double d = (double)v - means[code];
+      atomicAdd(&((double*)out_sum)[code], d * d);
+    }
+    if (OPS & B_CNT) {
+      if (!vnan)
+        atomicAdd(reinterpret_cast<unsigned long long*>(&out_count[code]), 1ull);
+    }
+    if (OPS & (B_MIN | B_MAX)) {
+      if (vnan) {
+        if (OPS & B_NANFLAG) out_nanflag[code] = 1u;
+      } else {
+        if (OPS & B_MIN) enc_min(&((Enc*)out_min)[code], TR::enc(v));
+        if (OPS & B_MAX) enc_max(&((Enc*)out_max)[code], TR::enc(v));
+      }
+    }
+  };
+
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const int64_t gtid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t nvec = n / VEC;
+  for (int64_t i = gtid; i < nvec; i += stride) {
+    Vec<V, VEC> vv = *reinterpret_cast<const Vec<V, VEC>*>(values + i * VEC);
+    Vec<L, VEC> lv = *reinterpret_cast<const Vec<L, VEC>*>(labels + i * VEC);
+    if (twolab) {
+      Vec<L, VEC> lv2 = *reinterpret_cast<const Vec<L, VEC>*>(labels2 + i * VEC);
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], (int64_t)lv2.v[k]);
+    } else {
+#pragma unroll
+      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], 0);
+    }
+  }
+  for (int64_t i = nvec * VEC + gtid; i < n; i += stride) {
+    process(values[i], (int64_t)labels[i], twolab ? (int64_t)labels2[i] : 0);
+  }
+}
+
+/* init product bins to 1 (memset cannot) */
+__global__ void k_fill_f64(double* p, int64_t n, double v) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = v;
+}
+__global__ void k_fill_i64(int64_t* p, int64_t n, int64_t v) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = v;
+}
+
+/* decode min/max bins in place (global-atomic path) */
+template <typename V, int OPS>
+__global__ void k_decode(int64_t ngroups, void* out_min, void* out_max,
+                         const int64_t* out_count, const uint32_t* out_present) {
+  using TR = Traits<V>;
+  using Enc = typename TR::Enc;
+  const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= ngroups) return;
+  const bool present =
+      (OPS & B_PRESENT) ? (out_present[g] != 0) : (out_count && out_count[g] != 0);
+  if (OPS & B_MIN) {
+    Enc e = ((Enc*)out_min)[g];
+    ((V*)out_min)[g] = present ? TR::dec(e) : TR::pos_inf();
+  }
+  if (OPS & B_MAX) {
+    Enc e = ((Enc*)out_max)[g];
+    ((V*)out_max)[g] = present ? TR::dec(e) : (V)-TR::pos_inf();
+  }
+}
+
+/* ---- host-side dispatch -------------------------------------------------- */
+
+template <typename V, typename L, int OPS>
+int launch_typed(fh_call* c) {
+  using TR = Traits<V>;
+  hipStream_t stream = (hipStream_t)c->stream;
+  const bool skipnan = (c->flags & FH_SKIPNAN) != 0;
+  BinLayout lay = bin_layout<V>(OPS, c->ngroups, 4);
+
+  bool use_lds = lay.bytes <= LDS_MAX;
+  if (c->flags & FH_FORCE_ATOMIC) use_lds = false;
+  if (c->flags & FH_FORCE_LDS) {
+    if (lay.bytes > LDS_MAX) return 2; /* cannot honor */
+    use_lds = true;
+  }
+
+  if (use_lds) {
+    int blocks_per_cu = lay.bytes * 2 <= LDS_MAX ? 2 : 1;
+    int nblocks = NUM_CU * blocks_per_cu;
+    /* do not launch more blocks than there is work or scratch for */
+    int64_t work_blocks = (c->n + BLOCK_LDS - 1) / BLOCK_LDS;
+    if (work_blocks < nblocks) nblocks = (int)(work_blocks > 0 ? work_blocks : 1);
+    if ((int64_t)nblocks * lay.bytes > c->scratch_bytes) return 3;
+
+    auto kern = k_reduce_lds<V, L, OPS>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)lay.bytes));
+    hipLaunchKernelGGL(kern, dim3(nblocks), dim3(BLOCK_LDS), lay.bytes, stream,
+                       (const V*)c->values, (const L*)c->labels,
+                       (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
+                       c->means, (int)skipnan, (char*)c->scratch, lay);
+    FH_CHECK(hipGetLastError());
+    int cb = (int)((c->ngroups + 255) / 256);
+    hipLaunchKernelGGL((k_combine<V, OPS>), dim3(cb), dim3(256), 0, stream,
+                       (const char*)c->scratch, nblocks, c->ngroups, lay,
+                       c->out_sum, c->out_count, c->out_present, c->out_min,
+                       c->out_max, c->out_nanflag);
+    FH_CHECK(hipGetLastError());
+    c->path_used = 1;
+    return 0;
+  }
+
+  /* global-atomic path: memset-style init of the bins */
+  const int64_t ng = c->ngroups;
+  if (OPS & (B_SUM | B_SSD))
+    FH_CHECK(hipMemsetAsync(c->out_sum, 0, ng * 8, stream));
+  if (OPS & B_PROD) {
+    int fb = (int)((ng + 255) / 256);
+    if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
+      hipLaunchKernelGGL(k_fill_f64, dim3(fb), dim3(256), 0, stream,
+                         (double*)c->out_sum, ng, 1.0);
+    else
+      hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                         (int64_t*)c->out_sum, ng, (int64_t)1);
+    FH_CHECK(hipGetLastError());
+  }
+  if (OPS & B_CNT) FH_CHECK(hipMemsetAsync(c->out_count, 0, ng * 8, stream));
+  if (OPS & B_PRESENT) FH_CHECK(hipMemsetAsync(c->out_present, 0, ng * 4, stream));
+  if (OPS & B_MIN)
+    FH_CHECK(hipMemsetAsync(c->out_min, 0xFF, ng * sizeof(typename TR::Enc), stream));
+  if (OPS & B_MAX)
+    FH_CHECK(hipMemsetAsync(c->out_max, 0x00, ng * sizeof(typename TR::Enc), stream));
+  if (OPS & B_NANFLAG) FH_CHECK(hipMemsetAsync(c->out_nanflag, 0, ng * 4, stream));
+
+  int64_t work_blocks = (c->n / TR::VEC + BLOCK_ATOMIC - 1) / BLOCK_ATOMIC;
+  int nblocks = (int)(work_blocks < 2048 ? (work_blocks > 0 ? work_blocks : 1) : 2048);
+  hipLaunchKernelGGL((k_reduce_atomic<V, L, OPS>), dim3(nblocks),
+                     dim3(BLOCK_ATOMIC), 0, stream, (const V*)c->values,
+                     (const L*)c->labels, (const L*)c->labels2, c->n, c->ngroups,
+                     c->g0, c->g1, c->means, (int)skipnan, c->out_sum,
+                     c->out_count, c->out_present, c->out_min, c->out_max,
+                     c->out_nanflag);
+  FH_CHECK(hipGetLastError());
+  if (OPS & (B_MIN | B_MAX)) {
+    int db = (int)((ng + 255) / 256);
+    hipLaunchKernelGGL((k_decode<V, OPS>), dim3(db), dim3(256), 0, stream, ng,
+                       c->out_min, c->out_max, c->out_count, c->out_present);
+    FH_CHECK(hipGetLastError());
+  }
+  c->path_used = 2;
+  return 0;
+}
+
+template <typename V, typename L>
+int dispatch_ops(fh_call* c) {
+  switch (set_bits(c->op_set)) {
+    case B_SUM | B_CNT: return launch_typed<V, L, B_SUM | B_CNT>(c);
+    case B_SUM | B_CNT | B_PRESENT:
+      return launch_typed<V, L, B_SUM | B_CNT | B_PRESENT>(c);
+    case B_CNT: return launch_typed<V, L, B_CNT>(c);
+    case B_MIN | B_CNT | B_PRESENT | B_NANFLAG:
+      return launch_typed<V, L, B_MIN | B_CNT | B_PRESENT | B_NANFLAG>(c);
+    case B_MIN | B_CNT: return launch_typed<V, L, B_MIN | B_CNT>(c);
+    case B_MAX | B_CNT | B_PRESENT | B_NANFLAG:
+      return launch_typed<V, L, B_MAX | B_CNT | B_PRESENT | B_NANFLAG>(c);
+    case B_MAX | B_CNT: return launch_typed<V, L, B_MAX | B_CNT>(c);
+    case B_SSD: return launch_typed<V, L, B_SSD>(c);
+    case B_PROD | B_CNT | B_PRESENT:
+      return launch_typed<V, L, B_PROD | B_CNT | B_PRESENT>(c);
+    default: return 4;
+  }
+}
+
+template <typename V>
+int dispatch_label(fh_call* c) {
+  switch (c->ldtype) {
+    case FH_L_I64: return dispatch_ops<V, int64_t>(c);
+    case FH_L_I32: return dispatch_ops<V, int32_t>(c);
+    default: return 5;
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+int64_t fh_scratch_bytes(const fh_call* c) {
+  int bits = set_bits(c->op_set);
+  int64_t per_block;
+  switch (c->vdtype) {
+    case FH_F32: per_block = bin_layout<float>(bits, c->ngroups, 4).bytes; break;
+    case FH_F64: per_block = bin_layout<double>(bits, c->ngroups, 4).bytes; break;
+    case FH_I64: per_block = bin_layout<int64_t>(bits, c->ngroups, 4).bytes; break;
+    case FH_I32: per_block = bin_layout<int32_t>(bits, c->ngroups, 4).bytes; break;
+    default: return -1;
+  }
+  if (per_block > LDS_MAX && !(c->flags & FH_FORCE_LDS)) return 0; /* atomic path */
+  int blocks_per_cu = per_block * 2 <= LDS_MAX ? 2 : 1;
+  return (int64_t)NUM_CU * blocks_per_cu * per_block;
+}
+
+int fh_grouped_reduce(fh_call* c) {
+  if (!c || !c->values || !c->labels) return 6;
+  if (c->labels2 && c->g0 * c->g1 != c->ngroups) return 7;
+  if (c->op_set == FH_SET_SSD && !c->means) return 8;
+  switch (c->vdtype) {
+    case FH_F32: return dispatch_label<float>(c);
+    case FH_F64: return dispatch_label<double>(c);
+    case FH_I64: return dispatch_label<int64_t>(c);
+    case FH_I32: return dispatch_label<int32_t>(c);
+    default: return 9;
+  }
+}
+
+const char* fh_error_string(int code) {
+  switch (code) {
+    case 0: return "ok";
+    case 2: return "FH_FORCE_LDS but bins exceed LDS capacity";
+    case 3: return "scratch buffer too small (call fh_scratch_bytes)";
+    case 4: return "unknown op_set";
+    case 5: return "unknown label dtype";
+    case 6: return "null values/labels pointer";
+    case 7: return "labels2 given but g0*g1 != ngroups";
+    case 8: return "FH_SET_SSD requires means";
+    case 9: return "unknown value dtype";
+    default: return code >= 1000 ? hipGetErrorString((hipError_t)(code - 1000)) : "unknown error";
+  }
+}
+
+int fh_version(void) { return 1; }
+
+}  /* extern "C" */

@@ -1,0 +1,107 @@
+/* floxhip — MI355X (gfx950) grouped-reduction engine: C ABI.
+ *
+ * This is the drop-in boundary for the hot path of xarray-contrib/flox
+ * (reference: flox/core.py:214-394 `chunk_reduce` and the per-engine
+ * aggregate modules flox/aggregate_flox.py / flox/aggregate_npg.py).
+ * The Python shim flox_amd/aggregate_hip.py implements the reference's
+ * engine-plugin interface (flox/aggregations.py:60-133 `generic_aggregate`:
+ * one callable per reduction name, signature
+ * f(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None))
+ * on top of exactly these entry points.
+ *
+ * One call = one fused factorize+reduce pass over `n` rows producing
+ * per-group partial bins (the "intermediates" of the reference's
+ * IntermediateDict, flox/types.py:28). Multi-GPU combine = an RCCL
+ * all-reduce of these bins, applying the reference's combine recipes
+ * (flox/aggregations.py:304-546).
+ *
+ * All pointers are DEVICE pointers; the caller owns every buffer and
+ * synchronises via `stream` (a hipStream_t). Calls are stateless and
+ * thread-safe. Errors: non-zero return, message via fh_error_string().
+ */
+#ifndef FLOXHIP_H
+#define FLOXHIP_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* value dtypes */
+enum fh_dtype { FH_F32 = 0, FH_F64 = 1, FH_I64 = 2, FH_I32 = 3 };
+/* label dtypes */
+enum fh_ldtype { FH_L_I64 = 0, FH_L_I32 = 1 };
+
+/* Fused op sets: the per-group partials one pass produces.
+ * (SUM accumulates float inputs in float64 — the numpy_groupies numerics
+ * contract pinned by reference tests/test_properties.py:146-151.) */
+enum fh_opset {
+  FH_SET_SUM_COUNT = 0,         /* sum(f64/i64) + count(i64): mean, var pass 1 */
+  FH_SET_SUM_COUNT_PRESENT = 1, /* + per-group "any row seen" flag: sum/nansum */
+  FH_SET_COUNT = 2,             /* count only */
+  FH_SET_MIN_FULL = 3,          /* min + count + present + nanflag: min */
+  FH_SET_MIN_COUNT = 4,         /* min + count: nanmin */
+  FH_SET_MAX_FULL = 5,          /* max + count + present + nanflag: max */
+  FH_SET_MAX_COUNT = 6,         /* max + count: nanmax */
+  FH_SET_SSD = 7,               /* sum of (x - mean[g])^2, f64: var pass 2 */
+  FH_SET_PROD = 8,              /* product + count + present (CAS loop) */
+};
+
+/* flags */
+enum fh_flags {
+  FH_SKIPNAN = 1 << 0,    /* nan* reductions: NaN values contribute nothing */
+  FH_FORCE_LDS = 1 << 1,  /* testing: force the LDS-binned path */
+  FH_FORCE_ATOMIC = 1 << 2, /* testing: force the global-atomic path */
+};
+
+typedef struct fh_call {
+  int op_set;     /* fh_opset */
+  int vdtype;     /* fh_dtype of `values` */
+  int ldtype;     /* fh_ldtype of `labels` (and `labels2`) */
+  int flags;      /* fh_flags */
+  int64_t n;      /* rows */
+  int64_t ngroups;
+  /* 2-D groupby (reference factorize.py:102-108 _ravel_factorized):
+   * labels2 != NULL means code = labels[i]*g1 + labels2[i], with either
+   * label out of [0,g0)/[0,g1) meaning "not in any group" (the reference's
+   * -1 sentinel). ngroups must equal g0*g1. */
+  const void* values;
+  const void* labels;
+  const void* labels2;
+  int64_t g0, g1;
+  /* per-group means (f64[ngroups]) for FH_SET_SSD, else NULL */
+  const double* means;
+
+  /* outputs: each f64/i64/u32[ngroups]; only the set's members are written.
+   * out_sum: f64 for float inputs (also receives the SSD result), i64 for
+   *          int inputs. out_min/out_max: in VALUE dtype, +inf/-inf
+   *          (INT_MAX/MIN) for empty groups. */
+  void* out_sum;
+  int64_t* out_count;
+  uint32_t* out_present;
+  void* out_min;
+  void* out_max;
+  uint32_t* out_nanflag;
+
+  /* scratch: device buffer of fh_scratch_bytes() bytes (may be NULL when 0) */
+  void* scratch;
+  int64_t scratch_bytes;
+
+  void* stream;   /* hipStream_t */
+  int path_used;  /* out: 1 = LDS-binned, 2 = global-atomic */
+} fh_call;
+
+/* scratch requirement for this call (0 when the global-atomic path is used) */
+int64_t fh_scratch_bytes(const fh_call* c);
+
+/* run the fused grouped reduction; returns 0 on success */
+int fh_grouped_reduce(fh_call* c);
+
+const char* fh_error_string(int code);
+int fh_version(void);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* FLOXHIP_H */

@@ -1,0 +1,99 @@
+"""Multi-process (gloo, world_size=2) tests of the cross-GPU combine logic.
+
+Runs on CPU: the combine recipes (distributed.combine_partials and the
+two-phase var combine) are device-agnostic torch collectives; the kernels
+that produce per-rank partials are exercised by the gpu-marked tests. Here
+each rank builds its shard's partial bins in numpy/torch and the combined
+result must equal the whole-data answer (bit-exact for count/min/max).
+"""
+
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+
+N, NG = 40_000, 97
+SEED = 123
+
+
+def _shard_data(rank, world):
+    rng = np.random.default_rng(SEED)
+    vals = rng.standard_normal(N)
+    labels = rng.integers(0, NG, N)
+    sl = slice(rank * N // world, (rank + 1) * N // world)
+    return vals, labels, vals[sl], labels[sl]
+
+
+def _partials_np(vals, labels):
+    sums = np.bincount(labels, weights=vals, minlength=NG)
+    counts = np.bincount(labels, minlength=NG)
+    mins = np.full(NG, np.inf)
+    np.minimum.at(mins, labels, vals)
+    maxs = np.full(NG, -np.inf)
+    np.maximum.at(maxs, labels, vals)
+    return {
+        "sum": torch.tensor(sums),
+        "count": torch.tensor(counts),
+        "min": torch.tensor(mins),
+        "max": torch.tensor(maxs),
+    }
+
+
+def _worker(rank, world, port, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import flox_amd.distributed as fdist
+
+        vals, labels, sv, sl = _shard_data(rank, world)
+
+        # 1. plain partial combine == whole-data partials
+        p = _partials_np(sv, sl)
+        fdist.combine_partials(p, {"sum": "sum", "count": "sum", "min": "min", "max": "max"})
+        whole = _partials_np(vals, labels)
+        np.testing.assert_array_equal(p["count"].numpy(), whole["count"].numpy())
+        np.testing.assert_array_equal(p["min"].numpy(), whole["min"].numpy())
+        np.testing.assert_array_equal(p["max"].numpy(), whole["max"].numpy())
+        np.testing.assert_allclose(p["sum"].numpy(), whole["sum"].numpy(), rtol=1e-13)
+
+        # 2. two-phase var combine (deviations about the GLOBAL mean; the
+        #    reference's _var_combine reduces to a plain sum of ssd then)
+        sums = torch.tensor(np.bincount(sl, weights=sv, minlength=NG))
+        counts = torch.tensor(np.bincount(sl, minlength=NG))
+        fdist.all_reduce_(sums, "sum")
+        fdist.all_reduce_(counts, "sum")
+        means = (sums / counts).numpy()
+        ssd_local = np.bincount(sl, weights=(sv - means[sl]) ** 2, minlength=NG)
+        ssd = torch.tensor(ssd_local)
+        fdist.all_reduce_(ssd, "sum")
+        var_combined = ssd.numpy() / counts.numpy()
+
+        from oracle import groupby_reduce as oracle_reduce
+
+        want, _ = oracle_reduce(vals, labels, func="var", expected_groups=np.arange(NG))
+        np.testing.assert_allclose(var_combined, want, rtol=1e-10, atol=1e-12)
+
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}")
+        raise
+
+
+def test_gloo_world2_combine():
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.Queue()
+    port = 29517
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, fail_q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+    errs = []
+    while not fail_q.empty():
+        errs.append(fail_q.get())
+    assert not errs, errs
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
