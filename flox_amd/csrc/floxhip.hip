@@ -32,6 +32,7 @@
 
 #include <hip/hip_runtime.h>
 #include <cstdint>
+#include <type_traits>
 #include <cstdio>
 
 #include "../../include/floxhip.h"
@@ -205,12 +206,13 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
     char* __restrict__ slab, BinLayout lay) {
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
+  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
   using Enc = typename TR::Enc;
   constexpr int VEC = TR::VEC;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  Acc* s_sum = (OPS & (B_SUM | B_SSD | B_PROD)) ? (Acc*)(smem + lay.sum_off) : nullptr;
+  SumT* s_sum = (OPS & (B_SUM | B_SSD | B_PROD)) ? (SumT*)(smem + lay.sum_off) : nullptr;
   uint32_t* s_cnt = (OPS & B_CNT) ? (uint32_t*)(smem + lay.cnt_off) : nullptr;
   uint32_t* s_present = (OPS & B_PRESENT) ? (uint32_t*)(smem + lay.present_off) : nullptr;
   Enc* s_mm = (OPS & (B_MIN | B_MAX)) ? (Enc*)(smem + lay.minmax_off) : nullptr;
@@ -218,8 +220,8 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
 
   const int tid = threadIdx.x;
   for (int64_t g = tid; g < ngroups; g += blockDim.x) {
-    if (OPS & (B_SUM | B_SSD)) s_sum[g] = (Acc)0;
-    if (IS_PROD) s_sum[g] = (Acc)1;
+    if (OPS & (B_SUM | B_SSD)) s_sum[g] = (SumT)0;
+    if (IS_PROD) s_sum[g] = (SumT)1;
     if (OPS & B_CNT) s_cnt[g] = 0u;
     if (OPS & B_PRESENT) s_present[g] = 0u;
     if (OPS & B_MIN) s_mm[g] = (Enc)~(Enc)0;
@@ -246,7 +248,7 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
     if (IS_PROD) acc_mul(&s_sum[code], (Acc)v);
     if (OPS & B_SSD) {
       double d = (double)v - means[code];
-      atomicAdd((double*)&s_sum[code], d * d);
+      atomicAdd((double*)s_sum + code, d * d);
     }
     if (OPS & B_CNT) {
       if (!vnan) atomicAdd(&s_cnt[code], 1u);
@@ -285,7 +287,7 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
   /* flush bins to this block's slab section with plain coalesced stores */
   char* my = slab + (int64_t)blockIdx.x * lay.bytes;
   for (int64_t g = tid; g < ngroups; g += blockDim.x) {
-    if (OPS & (B_SUM | B_SSD | B_PROD)) ((Acc*)(my + lay.sum_off))[g] = s_sum[g];
+    if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)(my + lay.sum_off))[g] = s_sum[g];
     if (OPS & B_CNT) ((uint32_t*)(my + lay.cnt_off))[g] = s_cnt[g];
     if (OPS & B_PRESENT) ((uint32_t*)(my + lay.present_off))[g] = s_present[g];
     if (OPS & (B_MIN | B_MAX)) ((Enc*)(my + lay.minmax_off))[g] = s_mm[g];
@@ -301,19 +303,20 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
                           void* out_min, void* out_max, uint32_t* out_nanflag) {
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
+  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
   using Enc = typename TR::Enc;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
   const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (g >= ngroups) return;
 
-  Acc s = IS_PROD ? (Acc)1 : (Acc)0;
+  SumT s = IS_PROD ? (SumT)1 : (SumT)0;
   int64_t c = 0;
   uint32_t p = 0, nf = 0;
   Enc mn = (Enc)~(Enc)0, mx = (Enc)0;
   for (int b = 0; b < nblocks; ++b) {
     const char* blk = slab + (int64_t)b * lay.bytes;
-    if (OPS & (B_SUM | B_SSD)) s += ((const Acc*)(blk + lay.sum_off))[g];
-    if (IS_PROD) s *= ((const Acc*)(blk + lay.sum_off))[g];
+    if (OPS & (B_SUM | B_SSD)) s += ((const SumT*)(blk + lay.sum_off))[g];
+    if (IS_PROD) s *= ((const SumT*)(blk + lay.sum_off))[g];
     if (OPS & B_CNT) c += (int64_t)((const uint32_t*)(blk + lay.cnt_off))[g];
     if (OPS & B_PRESENT) p |= ((const uint32_t*)(blk + lay.present_off))[g];
     if (OPS & B_MIN) {
@@ -327,7 +330,7 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
     if (OPS & B_NANFLAG) nf |= ((const uint32_t*)(blk + lay.nanflag_off))[g];
   }
   const bool present = (OPS & B_PRESENT) ? (p != 0) : (c != 0);
-  if (OPS & (B_SUM | B_SSD | B_PROD)) ((Acc*)out_sum)[g] = s;
+  if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)out_sum)[g] = s;
   if (OPS & B_CNT) out_count[g] = c;
   if (OPS & B_PRESENT) out_present[g] = p;
   if (OPS & B_MIN) ((V*)out_min)[g] = present ? TR::dec(mn) : TR::pos_inf();

@@ -192,19 +192,19 @@ def groupby_reduce(
     codes = _ravel_codes(codes_list, grp_shape) if nby > 1 else codes_list[0]
 
     # --- min_count defaulting (reference core.py:1026-1038) ---
+    out_dtype = _final_dtype(func, array.dtype, dtype)
     if min_count is None:
         min_count_ = 1 if (fill_value is not None and provided_expected) else 0
     else:
         min_count_ = min_count
     if func in ("nanmin", "nanmax") and min_count_ == 0:
-        # reference aggregations.py:997-1003 nanmin/nanmax hack
+        # reference aggregations.py:997-1003 nanmin/nanmax hack: user fill
+        # defaults to the dtype's NA (NaN for float, iinfo.min for int)
         min_count_ = 1
         if fill_value is None:
-            fill_value = np.nan if array.dtype.kind in "fc" else None
+            fill_value = _fill_default(func, out_dtype)
     if min_count_ > 0 and func in ("nansum", "nanprod") and fill_value is None:
         fill_value = np.nan  # reference core.py:1035-1038
-
-    out_dtype = _final_dtype(func, array.dtype, dtype)
 
     # --- chunk reduction over flattened group dims ---
     lead_shape = array.shape[: array.ndim - by_ndim]
@@ -233,6 +233,11 @@ def groupby_reduce(
     skipna = func in _NAN_SKIP
     m_all = valid_code[None, :] & (~nanmask_v if skipna else np.ones_like(nanmask_v))
     counts = bincount_rows(None, valid_code[None, :] & ~nanmask_v)  # nanlen, always NaN-skipping
+    # "group present" = any row carries its code, NaN or not (reference: npg
+    # writes fill_value only for codes with no rows at all; an all-NaN group
+    # still yields nansum=0 / nanprod=1)
+    present = np.bincount(codes[valid_code], minlength=ngroups) > 0
+    absent_mask = np.broadcast_to(~present, (M, ngroups))
 
     def grouped_extreme(op_at, init):
         out = np.full((M, ngroups), init, dtype=array.dtype if array.dtype.kind != "b" else np.int_)
@@ -250,14 +255,14 @@ def groupby_reduce(
     elif func in ("sum", "nansum"):
         sums = bincount_rows(vals2d, m_all)
         result = sums.astype(out_dtype)
-        empty_mask = bincount_rows(None, m_all) == 0
+        empty_mask = absent_mask
     elif func in ("prod", "nanprod"):
         out = np.ones((M, ngroups), dtype=acc_dtype)
         for r in range(M):
             m = m_all[r]
             np.multiply.at(out[r], codes[m], vals2d[r][m].astype(acc_dtype))
         result = out.astype(out_dtype)
-        empty_mask = bincount_rows(None, m_all) == 0
+        empty_mask = absent_mask
     elif func in ("mean", "nanmean"):
         sums = bincount_rows(vals2d, None if skipna else valid_code[None, :].repeat(M, 0))
         if not skipna:
