@@ -65,6 +65,10 @@ class FhCall(ctypes.Structure):
         ("scratch_bytes", ctypes.c_int64),
         ("stream", ctypes.c_void_p),
         ("path_used", ctypes.c_int),
+        # column path (fh_grouped_reduce_cols)
+        ("perm", ctypes.c_void_p),
+        ("m", ctypes.c_int64),
+        ("ldm", ctypes.c_int64),
     ]
 
 
@@ -84,6 +88,8 @@ def load_library():
     lib = ctypes.CDLL(_LIB_PATH)
     lib.fh_grouped_reduce.argtypes = [ctypes.POINTER(FhCall)]
     lib.fh_grouped_reduce.restype = ctypes.c_int
+    lib.fh_grouped_reduce_cols.argtypes = [ctypes.POINTER(FhCall)]
+    lib.fh_grouped_reduce_cols.restype = ctypes.c_int
     lib.fh_scratch_bytes.argtypes = [ctypes.POINTER(FhCall)]
     lib.fh_scratch_bytes.restype = ctypes.c_int64
     lib.fh_error_string.argtypes = [ctypes.c_int]

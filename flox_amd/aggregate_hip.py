@@ -330,7 +330,84 @@ def nanprod(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None
     return _to_dtype(res, dtype)
 
 
-def var_partials(group_idx, array, *, skipnan, size, global_counts=None, global_sums=None):
+def grouped_partials_cols(
+    op_set: int,
+    values2d: torch.Tensor,
+    codes_sorted: torch.Tensor,
+    perm: torch.Tensor,
+    ngroups: int,
+    *,
+    means: torch.Tensor | None = None,
+    skipnan: bool = False,
+) -> dict[str, torch.Tensor]:
+    """Grouped reduce over the LEADING axis of a 2-D array (n_t, m) with
+    column stride 1 (row stride = values2d.stride(0)). codes_sorted/perm:
+    int32 tensors from a stable sort of the per-row group codes. Returns
+    per-group partials of shape (ngroups, m)."""
+    lib = _ffi.load_library()
+    _require_gpu_tensor(values2d, "values")
+    assert values2d.ndim == 2 and values2d.stride(1) == 1
+    assert codes_sorted.dtype == torch.int32 and perm.dtype == torch.int32
+    n_t, m = values2d.shape
+    dev = values2d.device
+    if values2d.dtype not in _TORCH_VDTYPE:
+        raise NotImplementedError(f"engine=hip does not support values dtype {values2d.dtype}")
+
+    c = FhCall()
+    c.op_set = op_set
+    c.vdtype = _TORCH_VDTYPE[values2d.dtype]
+    c.ldtype = _ffi.L_I32
+    c.flags = FLAG_SKIPNAN if skipnan else 0
+    c.n = n_t
+    c.m = m
+    c.ldm = values2d.stride(0)
+    c.ngroups = ngroups
+    c.values = values2d.data_ptr()
+    codes_sorted = codes_sorted.contiguous()
+    perm = perm.contiguous()
+    c.labels = codes_sorted.data_ptr()
+    c.perm = perm.data_ptr()
+    if means is not None:
+        assert means.dtype == torch.float64 and means.is_cuda
+        means = means.contiguous()
+        c.means = means.data_ptr()
+
+    members = _SET_MEMBERS[op_set]
+    out: dict[str, torch.Tensor] = {}
+    shape = (ngroups, m)
+    if "sum" in members:
+        dt = torch.float64 if op_set == SET_SSD else _acc_dtype(values2d.dtype)
+        out["sum"] = torch.empty(shape, dtype=dt, device=dev)
+        c.out_sum = out["sum"].data_ptr()
+    if "count" in members:
+        out["count"] = torch.empty(shape, dtype=torch.int64, device=dev)
+        c.out_count = out["count"].data_ptr()
+    if "present" in members:
+        out["present"] = torch.empty(shape, dtype=torch.int32, device=dev)
+        c.out_present = out["present"].data_ptr()
+    if "min" in members:
+        out["min"] = torch.empty(shape, dtype=values2d.dtype, device=dev)
+        c.out_min = out["min"].data_ptr()
+    if "max" in members:
+        out["max"] = torch.empty(shape, dtype=values2d.dtype, device=dev)
+        c.out_max = out["max"].data_ptr()
+    if "nanflag" in members:
+        out["nanflag"] = torch.empty(shape, dtype=torch.int32, device=dev)
+        c.out_nanflag = out["nanflag"].data_ptr()
+
+    c.stream = torch.cuda.current_stream(dev).cuda_stream
+    _ffi.check(lib.fh_grouped_reduce_cols(ctypes.byref(c)))
+    out["_path"] = c.path_used  # type: ignore[assignment]
+    for t in (values2d, codes_sorted, perm, means):
+        if isinstance(t, torch.Tensor):
+            t.record_stream(torch.cuda.current_stream(dev))
+    return out
+
+
+def var_partials(
+    group_idx, array, *, skipnan, size, labels2=None, grp_shape=None,
+    global_counts=None, global_sums=None,
+):
     """The (ssd, sum, len) triple of the reference's var_chunk
     (flox/aggregations.py:348-389): pass 1 sum+count -> means, pass 2
     sum of squared deviations about those means.
@@ -340,13 +417,14 @@ def var_partials(group_idx, array, *, skipnan, size, global_counts=None, global_
     the reference's _var_combine with zero adjustment terms)."""
     group_idx, array = _prep(group_idx, array)
     ng = _size_of(group_idx, size)
+    kw = dict(labels2=labels2, grp_shape=grp_shape)
     if global_counts is None:
-        p1 = grouped_partials(SET_SUM_COUNT, array, group_idx, ng, skipnan=skipnan)
+        p1 = grouped_partials(SET_SUM_COUNT, array, group_idx, ng, skipnan=skipnan, **kw)
         sums, counts = p1["sum"], p1["count"]
     else:
         sums, counts = global_sums, global_counts
     # empty groups give mean = 0/0 = NaN, but no row carries their code, so the
     # SSD kernel never reads those entries
     means = sums.to(torch.float64) / counts
-    p2 = grouped_partials(SET_SSD, array, group_idx, ng, skipnan=skipnan, means=means)
+    p2 = grouped_partials(SET_SSD, array, group_idx, ng, skipnan=skipnan, means=means, **kw)
     return p2["sum"], sums, counts

@@ -20,7 +20,7 @@ import numpy as np
 import torch
 
 from . import _ffi, distributed, xrdtypes
-from .aggregate_hip import grouped_partials, var_partials
+from .aggregate_hip import grouped_partials, grouped_partials_cols
 from .aggregations import REDUCTIONS
 
 _TORCH_TO_NP = {
@@ -189,8 +189,7 @@ def groupby_reduce(
         if tuple(sorted(ax)) != tuple(range(arr.ndim - len(by_shape), arr.ndim)):
             raise NotImplementedError("GPU path reduces over all dims of by (axis subset: next row)")
     lead_shape = tuple(arr.shape[: arr.ndim - len(by_shape)])
-    if math.prod(lead_shape) != 1:
-        raise NotImplementedError("leading (non-grouped) array dims: next row (config-4 kernel)")
+    lead_M = math.prod(lead_shape) if lead_shape else 1
 
     if expected_groups is not None and not isinstance(expected_groups, tuple):
         expected_groups = (expected_groups,)
@@ -223,20 +222,63 @@ def groupby_reduce(
     dist_on = distributed.is_active() if distributed_combine is None else distributed_combine
     ddof = (finalize_kwargs or {}).get("ddof", 0)
 
-    kw = dict(labels2=labels2, grp_shape=grp_pair)
+    if lead_M == 1:
+        def run_set(op_set, skipnan, means=None):
+            return grouped_partials(
+                op_set, vals, labels, ngroups, skipnan=skipnan,
+                labels2=labels2, grp_shape=grp_pair, means=means,
+            )
+    else:
+        # column path: grouped dims to the front, lead dims flattened as
+        # columns (a zero-copy view when the caller's layout is
+        # grouped-axis-major, e.g. time-major climatology read through a
+        # .permute view; otherwise one transpose copy)
+        nbydims = len(by_shape)
+        N = math.prod(by_shape)
+        arr_t = arr.movedim(
+            tuple(range(arr.ndim - nbydims, arr.ndim)), tuple(range(nbydims))
+        )
+        vt = arr_t.reshape((N, lead_M))
+        if vt.stride(1) != 1:
+            vt = vt.contiguous()
+        if N >= 2**31 or ngroups * lead_M >= 2**62:
+            raise NotImplementedError("column path: axis length must fit int32")
+        # bound codes here (the column kernel trusts codes in [-1, ngroups))
+        if labels2 is not None:
+            g0, g1 = grp_pair
+            c0, c1 = labels.to(torch.int64), labels2.to(torch.int64)
+            bad = (c0 < 0) | (c0 >= g0) | (c1 < 0) | (c1 >= g1)
+            codes_full = torch.where(bad, torch.full_like(c0, -1), c0 * g1 + c1)
+        else:
+            c0 = labels.to(torch.int64)
+            bad = (c0 < 0) | (c0 >= ngroups)
+            codes_full = torch.where(bad, torch.full_like(c0, -1), c0)
+        # stable sort of the (small) per-row code vector: rows are then
+        # walked in group order, one contiguous segment per group (the GPU
+        # analogue of _prepare_for_flox, reference aggregate_flox.py:9-23)
+        scodes64, perm64 = torch.sort(codes_full, stable=True)
+        scodes, perm = scodes64.to(torch.int32), perm64.to(torch.int32)
+
+        def run_set(op_set, skipnan, means=None):
+            return grouped_partials_cols(
+                op_set, vt, scodes, perm, ngroups, skipnan=skipnan, means=means
+            )
+
     if func in ("var", "nanvar", "std", "nanstd"):
         skip = agg.skipnan
+        p1 = run_set(_ffi.SET_SUM_COUNT, skip)
         if dist_on:
-            p1 = grouped_partials(_ffi.SET_SUM_COUNT, vals, labels, ngroups, skipnan=skip, **kw)
             distributed.all_reduce_(p1["sum"], "sum")
             distributed.all_reduce_(p1["count"], "sum")
-            ssd, sums, counts = var_partials(
-                labels, vals, skipnan=skip, size=ngroups,
-                global_counts=p1["count"], global_sums=p1["sum"],
-            )
-            distributed.all_reduce_(ssd, "sum")
-        else:
-            ssd, sums, counts = var_partials(labels, vals, skipnan=skip, size=ngroups)
+        counts = p1["count"]
+        # deviations about the (global) per-group mean: the cross-rank ssd
+        # combine is then a plain sum (the reference's _var_combine,
+        # aggregations.py:392-451, with zero adjustment terms)
+        means = (p1["sum"].to(torch.float64) / counts).contiguous()
+        p2 = run_set(_ffi.SET_SSD, skip, means=means)
+        if dist_on:
+            distributed.all_reduce_(p2["sum"], "sum")
+        ssd = p2["sum"]
         den = counts.to(torch.float64) - ddof
         result = ssd / den
         nan_t = torch.full_like(result, float("nan"))
@@ -246,7 +288,7 @@ def groupby_reduce(
         counts_for_mask = counts
         empty_mask = counts == 0
     else:
-        p = grouped_partials(agg.op_set, vals, labels, ngroups, skipnan=agg.skipnan, **kw)
+        p = run_set(agg.op_set, agg.skipnan)
         if dist_on:
             distributed.combine_partials(p, agg.combine)
         if func == "count":
@@ -291,6 +333,10 @@ def groupby_reduce(
             result = torch.where(empty_mask, torch.tensor(fv, dtype=t_out_dtype, device=device), result)
 
     result = result.to(t_out_dtype)
+    if lead_M > 1:
+        # column partials are (ngroups, M) group-major; the API result puts
+        # the group dims last
+        result = result.reshape(ngroups, lead_M).t().contiguous()
     result = result.reshape(lead_shape + grp_shape)
 
     groups = tuple(f.groups for f in facs)

@@ -407,6 +407,103 @@ __launch_bounds__(BLOCK_ATOMIC) __global__ void k_reduce_atomic(
   }
 }
 
+
+/* ---- column path: grouped reduce over a strided/leading axis ------------- */
+/* For array (..., reduced axis) with leading dims — e.g. the hour-of-day
+ * climatology (reference asv ERA5 shapes, BASELINE config 4). The label
+ * vector lives on the (small) reduced axis only, so the host argsorts it
+ * once (the GPU analogue of the reference's _prepare_for_flox sort,
+ * aggregate_flox.py:9-23 — but over N_t labels, not N_t*M rows) and the
+ * kernel walks rows in GROUP ORDER: each thread owns one column, keeps ONE
+ * running accumulator in registers, and writes it out at each (wave-uniform)
+ * segment boundary. No atomics, no LDS bins, any ngroups.
+ * Element (t, c) is at values[t*ldm + c] (column stride 1 — the natural
+ * layout for time-major climate data; a torch permute view, no transpose).
+ * Outputs are (ngroups, m) group-major so flush stores coalesce. */
+constexpr int COLS_BLOCK = 256;
+constexpr int COLS_TTILE = 2048;
+
+template <typename V, int OPS>
+__launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
+    const V* __restrict__ values, const int* __restrict__ codes_sorted,
+    const int* __restrict__ perm, int64_t n_t, int64_t m, int64_t ldm,
+    int64_t ngroups, const double* __restrict__ means, int skipnan,
+    void* out_sum, int64_t* out_count, uint32_t* out_present, void* out_min,
+    void* out_max, uint32_t* out_nanflag) {
+  using TR = Traits<V>;
+  using Acc = typename TR::Acc;
+  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
+  using Enc = typename TR::Enc;
+  constexpr bool IS_PROD = (OPS & B_PROD) != 0;
+
+  __shared__ int s_code[COLS_TTILE];
+  __shared__ int s_perm[COLS_TTILE];
+
+  const int64_t c = (int64_t)blockIdx.x * COLS_BLOCK + threadIdx.x;
+  const bool active = c < m;
+
+  SumT acc = IS_PROD ? (SumT)1 : (SumT)0;
+  uint32_t cnt = 0, nanflag = 0;
+  Enc mn = (Enc)~(Enc)0, mx = (Enc)0;
+  double mean_g = 0.0;
+  int cur_g = -1;
+
+  auto flush = [&](int g) {
+    if (!active || g < 0) return;
+    const int64_t o = (int64_t)g * m + c;
+    if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)out_sum)[o] = acc;
+    if (OPS & B_CNT) out_count[o] = (int64_t)cnt;
+    if (OPS & B_PRESENT) out_present[o] = 1u;
+    if (OPS & B_MIN) ((Enc*)out_min)[o] = mn;
+    if (OPS & B_MAX) ((Enc*)out_max)[o] = mx;
+    if (OPS & B_NANFLAG) out_nanflag[o] = nanflag;
+  };
+
+  for (int64_t t0 = 0; t0 < n_t; t0 += COLS_TTILE) {
+    const int nt = (int)((n_t - t0 < COLS_TTILE) ? (n_t - t0) : COLS_TTILE);
+    for (int i = threadIdx.x; i < nt; i += COLS_BLOCK) {
+      s_code[i] = codes_sorted[t0 + i];
+      s_perm[i] = perm[t0 + i];
+    }
+    __syncthreads();
+    for (int i = 0; i < nt; ++i) {
+      const int g = s_code[i];
+      if (g != cur_g) {
+        flush(cur_g);
+        acc = IS_PROD ? (SumT)1 : (SumT)0;
+        cnt = 0;
+        nanflag = 0;
+        mn = (Enc)~(Enc)0;
+        mx = (Enc)0;
+        cur_g = g;
+        if ((OPS & B_SSD) && g >= 0 && active) mean_g = means[(int64_t)g * m + c];
+      }
+      if (g < 0 || !active) continue;
+      const V v = values[(int64_t)s_perm[i] * ldm + c];
+      const bool vnan = TR::isnan_(v);
+      if (vnan && skipnan) continue;
+      if (OPS & B_SUM) acc += (SumT)v;
+      if (IS_PROD) acc *= (SumT)v;
+      if (OPS & B_SSD) {
+        const double d = (double)v - mean_g;
+        acc += d * d;
+      }
+      if (OPS & B_CNT) cnt += vnan ? 0u : 1u;
+      if (OPS & (B_MIN | B_MAX)) {
+        if (vnan) {
+          if (OPS & B_NANFLAG) nanflag = 1u;
+        } else {
+          const Enc e = TR::enc(v);
+          if (OPS & B_MIN) mn = e < mn ? e : mn;
+          if (OPS & B_MAX) mx = e > mx ? e : mx;
+        }
+      }
+    }
+    __syncthreads();
+  }
+  flush(cur_g);
+}
+
 /* init product bins to 1 (memset cannot) */
 __global__ void k_fill_f64(double* p, int64_t n, double v) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -550,6 +647,74 @@ int dispatch_label(fh_call* c) {
   }
 }
 
+
+/* init + decode shared with the atomic path, bins sized ngroups*m */
+template <typename V, int OPS>
+int launch_cols(fh_call* c) {
+  using TR = Traits<V>;
+  hipStream_t stream = (hipStream_t)c->stream;
+  const int64_t nbins = c->ngroups * c->m;
+  if (OPS & (B_SUM | B_SSD))
+    FH_CHECK(hipMemsetAsync(c->out_sum, 0, nbins * 8, stream));
+  if (OPS & B_PROD) {
+    int fb = (int)((nbins + 255) / 256);
+    if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
+      hipLaunchKernelGGL(k_fill_f64, dim3(fb), dim3(256), 0, stream,
+                         (double*)c->out_sum, nbins, 1.0);
+    else
+      hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                         (int64_t*)c->out_sum, nbins, (int64_t)1);
+    FH_CHECK(hipGetLastError());
+  }
+  if (OPS & B_CNT) FH_CHECK(hipMemsetAsync(c->out_count, 0, nbins * 8, stream));
+  if (OPS & B_PRESENT) FH_CHECK(hipMemsetAsync(c->out_present, 0, nbins * 4, stream));
+  if (OPS & B_MIN)
+    FH_CHECK(hipMemsetAsync(c->out_min, 0xFF, nbins * sizeof(typename TR::Enc), stream));
+  if (OPS & B_MAX)
+    FH_CHECK(hipMemsetAsync(c->out_max, 0x00, nbins * sizeof(typename TR::Enc), stream));
+  if (OPS & B_NANFLAG) FH_CHECK(hipMemsetAsync(c->out_nanflag, 0, nbins * 4, stream));
+
+  const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
+  int nblocks = (int)((c->m + COLS_BLOCK - 1) / COLS_BLOCK);
+  if (nblocks == 0) nblocks = 1;
+  hipLaunchKernelGGL((k_reduce_cols<V, OPS>), dim3(nblocks), dim3(COLS_BLOCK),
+                     0, stream, (const V*)c->values,
+                     (const int*)c->labels, (const int*)c->perm, c->n,
+                     c->m, c->ldm, c->ngroups, c->means, skipnan, c->out_sum,
+                     c->out_count, c->out_present, c->out_min, c->out_max,
+                     c->out_nanflag);
+  FH_CHECK(hipGetLastError());
+  if (OPS & (B_MIN | B_MAX)) {
+    int db = (int)((nbins + 255) / 256);
+    hipLaunchKernelGGL((k_decode<V, OPS>), dim3(db), dim3(256), 0, stream,
+                       nbins, c->out_min, c->out_max, c->out_count,
+                       c->out_present);
+    FH_CHECK(hipGetLastError());
+  }
+  c->path_used = 3;
+  return 0;
+}
+
+template <typename V>
+int dispatch_cols_ops(fh_call* c) {
+  switch (set_bits(c->op_set)) {
+    case B_SUM | B_CNT: return launch_cols<V, B_SUM | B_CNT>(c);
+    case B_SUM | B_CNT | B_PRESENT:
+      return launch_cols<V, B_SUM | B_CNT | B_PRESENT>(c);
+    case B_CNT: return launch_cols<V, B_CNT>(c);
+    case B_MIN | B_CNT | B_PRESENT | B_NANFLAG:
+      return launch_cols<V, B_MIN | B_CNT | B_PRESENT | B_NANFLAG>(c);
+    case B_MIN | B_CNT: return launch_cols<V, B_MIN | B_CNT>(c);
+    case B_MAX | B_CNT | B_PRESENT | B_NANFLAG:
+      return launch_cols<V, B_MAX | B_CNT | B_PRESENT | B_NANFLAG>(c);
+    case B_MAX | B_CNT: return launch_cols<V, B_MAX | B_CNT>(c);
+    case B_SSD: return launch_cols<V, B_SSD>(c);
+    case B_PROD | B_CNT | B_PRESENT:
+      return launch_cols<V, B_PROD | B_CNT | B_PRESENT>(c);
+    default: return 4;
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -578,6 +743,23 @@ int fh_grouped_reduce(fh_call* c) {
     case FH_F64: return dispatch_label<double>(c);
     case FH_I64: return dispatch_label<int64_t>(c);
     case FH_I32: return dispatch_label<int32_t>(c);
+    default: return 9;
+  }
+}
+
+
+int fh_grouped_reduce_cols(fh_call* c) {
+  /* column path: values (n_t rows x m columns, column stride 1, row stride
+   * ldm); labels = group codes pre-sorted ascending (int32), perm = the
+   * argsort permutation (int32) such that codes_sorted[i] = codes[perm[i]].
+   * Outputs are (ngroups, m) group-major. */
+  if (!c || !c->values || !c->labels || !c->perm) return 6;
+  if (c->op_set == FH_SET_SSD && !c->means) return 8;
+  switch (c->vdtype) {
+    case FH_F32: return dispatch_cols_ops<float>(c);
+    case FH_F64: return dispatch_cols_ops<double>(c);
+    case FH_I64: return dispatch_cols_ops<int64_t>(c);
+    case FH_I32: return dispatch_cols_ops<int32_t>(c);
     default: return 9;
   }
 }

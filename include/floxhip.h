@@ -89,7 +89,14 @@ typedef struct fh_call {
   int64_t scratch_bytes;
 
   void* stream;   /* hipStream_t */
-  int path_used;  /* out: 1 = LDS-binned, 2 = global-atomic */
+  int path_used;  /* out: 1 = LDS-binned, 2 = global-atomic, 3 = column */
+
+  /* column path only (fh_grouped_reduce_cols): array is (n rows x m
+   * columns), element (t,c) at values[t*ldm + c]; labels then holds the
+   * int32 group codes SORTED ascending and perm the int32 argsort
+   * permutation; outputs are (ngroups, m) group-major. */
+  const void* perm;
+  int64_t m, ldm;
 } fh_call;
 
 /* scratch requirement for this call (0 when the global-atomic path is used) */
@@ -97,6 +104,11 @@ int64_t fh_scratch_bytes(const fh_call* c);
 
 /* run the fused grouped reduction; returns 0 on success */
 int fh_grouped_reduce(fh_call* c);
+
+/* grouped reduce over the leading/strided axis of a multi-column array
+ * (the reference's axis-subset case, core.py:272-316 + factorize.py:24-39,
+ * computed without offsetting labels: one segmented pass per column) */
+int fh_grouped_reduce_cols(fh_call* c);
 
 const char* fh_error_string(int code);
 int fh_version(void);

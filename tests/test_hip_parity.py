@@ -11,8 +11,7 @@ from tests.golden_util import load_golden_cases, tolerance_for
 
 pytestmark = pytest.mark.gpu
 
-CASES = [c for c in load_golden_cases() if c[1]["array"].ndim == len(c[1]["by"][0].shape)]
-LEAD_CASES = [c for c in load_golden_cases() if c[1]["array"].ndim > len(c[1]["by"][0].shape)]
+CASES = list(load_golden_cases())
 
 
 @pytest.mark.parametrize("name,inputs,expected,groups", CASES, ids=[c[0] for c in CASES])
@@ -130,13 +129,36 @@ def test_size_independent_properties_large():
     assert abs(res_s.sum(dtype=torch.float64).item() - total) < 1e-4 * abs(total) + 1e-3
 
 
-def test_lead_dims_not_silently_wrong():
-    """Cases with leading array dims are a declared next row: the engine must
-    refuse rather than return a wrong shape."""
-    if not LEAD_CASES:
-        pytest.skip("no lead-dim golden cases")
-    name, inputs, expected, groups = LEAD_CASES[0]
-    kw = dict(inputs)
-    arr, bys = kw.pop("array"), kw.pop("by")
-    with pytest.raises(NotImplementedError):
-        flox_amd.groupby_reduce(arr, *bys, **kw)
+@pytest.mark.parametrize("func", ["sum", "mean", "nanmean", "var", "count", "min", "nanmax"])
+def test_column_path_climatology_shape(func):
+    """Config-4 shape (scaled down): time-major (n_t, y, x) array reduced by
+    hour-of-day along axis 0, fed as a .permute view (no transpose copy)."""
+    rng = np.random.default_rng(5)
+    n_t, y, x = 24 * 60, 36, 72
+    arr = rng.standard_normal((n_t, y, x)).astype(np.float32)
+    arr[rng.random((n_t, y, x)) < 0.02] = np.nan
+    hours = (np.arange(n_t) % 24).astype(np.int64)
+    want, *_ = oracle_reduce(
+        arr.transpose(1, 2, 0), hours, func=func, expected_groups=np.arange(24)
+    )
+    arr_t = torch.tensor(arr, device="cuda").permute(1, 2, 0)
+    got, _ = flox_amd.groupby_reduce(
+        arr_t, torch.tensor(hours, device="cuda"), func=func, expected_groups=np.arange(24)
+    )
+    assert got.shape == (y, x, 24)
+    g = got.cpu().numpy()
+    assert g.dtype == want.dtype
+    tol = dict(rtol=0, atol=0) if g.dtype.kind in "iu" else dict(rtol=3e-6, atol=1e-6)
+    np.testing.assert_allclose(g, want, equal_nan=True, **tol)
+
+
+def test_column_path_large_ngroups():
+    """The column kernel has no group-count limit (register-segment design)."""
+    rng = np.random.default_rng(9)
+    n_t, m, ng = 3000, 500, 366
+    arr = rng.standard_normal((n_t, m))
+    labels = rng.integers(0, ng, n_t)
+    want, *_ = oracle_reduce(arr.T, labels, func="mean", expected_groups=np.arange(ng))
+    arr_t = torch.tensor(arr, device="cuda").permute(1, 0)
+    got, _ = flox_amd.groupby_reduce(arr_t, torch.tensor(labels, device="cuda"), func="mean", expected_groups=np.arange(ng))
+    np.testing.assert_allclose(got.cpu().numpy(), want, equal_nan=True, rtol=1e-12, atol=1e-14)
