@@ -395,10 +395,18 @@ def grouped_partials_cols(
         out["nanflag"] = torch.empty(shape, dtype=torch.int32, device=dev)
         c.out_nanflag = out["nanflag"].data_ptr()
 
+    nscratch = lib.fh_scratch_bytes(ctypes.byref(c))
+    if nscratch < 0:
+        raise RuntimeError("fh_scratch_bytes failed")
+    scratch = None
+    if nscratch:
+        scratch = torch.empty(nscratch, dtype=torch.uint8, device=dev)
+        c.scratch = scratch.data_ptr()
+        c.scratch_bytes = nscratch
     c.stream = torch.cuda.current_stream(dev).cuda_stream
     _ffi.check(lib.fh_grouped_reduce_cols(ctypes.byref(c)))
     out["_path"] = c.path_used  # type: ignore[assignment]
-    for t in (values2d, codes_sorted, perm, means):
+    for t in (values2d, codes_sorted, perm, means, scratch):
         if isinstance(t, torch.Tensor):
             t.record_stream(torch.cuda.current_stream(dev))
     return out

@@ -201,7 +201,9 @@ def groupby_reduce(
     grp_shape = tuple(f.ngroups for f in facs)
     ngroups = math.prod(grp_shape)
     labels, labels2, grp_pair = _combined_codes(facs)
-    vals = arr.reshape(-1)
+    # only the 1-D path flattens the values (a no-op for contiguous input);
+    # the column path reads the caller's strided view directly
+    vals = arr.reshape(-1) if lead_M == 1 else None
 
     # min_count defaulting (reference core.py:1026-1038 + aggregations.py:997-1003)
     if min_count is None:

@@ -423,11 +423,17 @@ __launch_bounds__(BLOCK_ATOMIC) __global__ void k_reduce_atomic(
 constexpr int COLS_BLOCK = 256;
 constexpr int COLS_TTILE = 2048;
 
-template <typename V, int OPS>
+/* VC columns per thread (16-B vector loads per G13); blockIdx.y = row chunk.
+ * With several row chunks each chunk writes PARTIAL bins (cnt as u32) into
+ * its section of the scratch slab, identical in layout to the 1-D path's
+ * per-block slab, and k_combine folds the chunks; with one chunk the final
+ * bins are written directly (cnt as i64). */
+template <typename V, int OPS, int VC, bool SLAB>
 __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
     const V* __restrict__ values, const int* __restrict__ codes_sorted,
     const int* __restrict__ perm, int64_t n_t, int64_t m, int64_t ldm,
     int64_t ngroups, const double* __restrict__ means, int skipnan,
+    int64_t chunk_rows, char* __restrict__ slab, BinLayout lay,
     void* out_sum, int64_t* out_count, uint32_t* out_present, void* out_min,
     void* out_max, uint32_t* out_nanflag) {
   using TR = Traits<V>;
@@ -439,28 +445,81 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
   __shared__ int s_code[COLS_TTILE];
   __shared__ int s_perm[COLS_TTILE];
 
-  const int64_t c = (int64_t)blockIdx.x * COLS_BLOCK + threadIdx.x;
-  const bool active = c < m;
+  const int64_t c0 = ((int64_t)blockIdx.x * COLS_BLOCK + threadIdx.x) * VC;
+  const bool full = c0 + VC <= m;
+  const int64_t t_begin = (int64_t)blockIdx.y * chunk_rows;
+  const int64_t t_end = (t_begin + chunk_rows < n_t) ? t_begin + chunk_rows : n_t;
 
-  SumT acc = IS_PROD ? (SumT)1 : (SumT)0;
-  uint32_t cnt = 0, nanflag = 0;
-  Enc mn = (Enc)~(Enc)0, mx = (Enc)0;
-  double mean_g = 0.0;
+  char* my_slab = SLAB ? slab + (int64_t)blockIdx.y * lay.bytes : nullptr;
+
+  SumT acc[VC];
+  uint32_t cnt[VC];
+  Enc mn[VC], mx[VC];
+  uint32_t nanflag = 0;
+  double mean_g[VC];
   int cur_g = -1;
 
+  auto reset = [&]() {
+#pragma unroll
+    for (int k = 0; k < VC; ++k) {
+      acc[k] = IS_PROD ? (SumT)1 : (SumT)0;
+      cnt[k] = 0;
+      mn[k] = (Enc)~(Enc)0;
+      mx[k] = (Enc)0;
+    }
+    nanflag = 0;
+  };
+  reset();
+
   auto flush = [&](int g) {
-    if (!active || g < 0) return;
-    const int64_t o = (int64_t)g * m + c;
-    if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)out_sum)[o] = acc;
-    if (OPS & B_CNT) out_count[o] = (int64_t)cnt;
-    if (OPS & B_PRESENT) out_present[o] = 1u;
-    if (OPS & B_MIN) ((Enc*)out_min)[o] = mn;
-    if (OPS & B_MAX) ((Enc*)out_max)[o] = mx;
-    if (OPS & B_NANFLAG) out_nanflag[o] = nanflag;
+    if (g < 0) return;
+#pragma unroll
+    for (int k = 0; k < VC; ++k) {
+      if (c0 + k >= m) break;
+      const int64_t o = (int64_t)g * m + c0 + k;
+      if (SLAB) {
+        if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)(my_slab + lay.sum_off))[o] = acc[k];
+        if (OPS & B_CNT) ((uint32_t*)(my_slab + lay.cnt_off))[o] = cnt[k];
+        if (OPS & B_PRESENT) ((uint32_t*)(my_slab + lay.present_off))[o] = 1u;
+        if (OPS & B_MIN) ((Enc*)(my_slab + lay.minmax_off))[o] = mn[k];
+        if (OPS & B_MAX) ((Enc*)(my_slab + lay.minmax_off))[o] = mx[k];
+        if (OPS & B_NANFLAG) ((uint32_t*)(my_slab + lay.nanflag_off))[o] = nanflag;
+      } else {
+        if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)out_sum)[o] = acc[k];
+        if (OPS & B_CNT) out_count[o] = (int64_t)cnt[k];
+        if (OPS & B_PRESENT) out_present[o] = 1u;
+        if (OPS & B_MIN) ((Enc*)out_min)[o] = mn[k];
+        if (OPS & B_MAX) ((Enc*)out_max)[o] = mx[k];
+        if (OPS & B_NANFLAG) out_nanflag[o] = nanflag;
+      }
+    }
   };
 
-  for (int64_t t0 = 0; t0 < n_t; t0 += COLS_TTILE) {
-    const int nt = (int)((n_t - t0 < COLS_TTILE) ? (n_t - t0) : COLS_TTILE);
+  auto consume = [&](V v, int k, bool lanes_ok) {
+    if (!lanes_ok) return;
+    const bool vnan = TR::isnan_(v);
+    if (vnan && skipnan) return;
+    if (OPS & B_SUM) acc[k] += (SumT)v;
+    if (IS_PROD) acc[k] *= (SumT)v;
+    if (OPS & B_SSD) {
+      const double d = (double)v - mean_g[k];
+      acc[k] += d * d;
+    }
+    if (OPS & B_CNT) cnt[k] += vnan ? 0u : 1u;
+    if (OPS & (B_MIN | B_MAX)) {
+      if (vnan) {
+        if (OPS & B_NANFLAG) nanflag = 1u;
+      } else {
+        const Enc e = TR::enc(v);
+        if (OPS & B_MIN) mn[k] = e < mn[k] ? e : mn[k];
+        if (OPS & B_MAX) mx[k] = e > mx[k] ? e : mx[k];
+      }
+    }
+  };
+
+  for (int64_t t0 = t_begin; t0 < t_end; t0 += COLS_TTILE) {
+    const int nt = (int)((t_end - t0 < COLS_TTILE) ? (t_end - t0) : COLS_TTILE);
+    __syncthreads();
     for (int i = threadIdx.x; i < nt; i += COLS_BLOCK) {
       s_code[i] = codes_sorted[t0 + i];
       s_perm[i] = perm[t0 + i];
@@ -470,36 +529,26 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
       const int g = s_code[i];
       if (g != cur_g) {
         flush(cur_g);
-        acc = IS_PROD ? (SumT)1 : (SumT)0;
-        cnt = 0;
-        nanflag = 0;
-        mn = (Enc)~(Enc)0;
-        mx = (Enc)0;
+        reset();
         cur_g = g;
-        if ((OPS & B_SSD) && g >= 0 && active) mean_g = means[(int64_t)g * m + c];
-      }
-      if (g < 0 || !active) continue;
-      const V v = values[(int64_t)s_perm[i] * ldm + c];
-      const bool vnan = TR::isnan_(v);
-      if (vnan && skipnan) continue;
-      if (OPS & B_SUM) acc += (SumT)v;
-      if (IS_PROD) acc *= (SumT)v;
-      if (OPS & B_SSD) {
-        const double d = (double)v - mean_g;
-        acc += d * d;
-      }
-      if (OPS & B_CNT) cnt += vnan ? 0u : 1u;
-      if (OPS & (B_MIN | B_MAX)) {
-        if (vnan) {
-          if (OPS & B_NANFLAG) nanflag = 1u;
-        } else {
-          const Enc e = TR::enc(v);
-          if (OPS & B_MIN) mn = e < mn ? e : mn;
-          if (OPS & B_MAX) mx = e > mx ? e : mx;
+        if ((OPS & B_SSD) && g >= 0) {
+#pragma unroll
+          for (int k = 0; k < VC; ++k)
+            if (c0 + k < m) mean_g[k] = means[(int64_t)g * m + c0 + k];
         }
       }
+      if (g < 0 || c0 >= m) continue;
+      const int64_t row = (int64_t)s_perm[i] * ldm;
+      if (VC > 1 && full) {
+        Vec<V, VC> vv = *reinterpret_cast<const Vec<V, VC>*>(values + row + c0);
+#pragma unroll
+        for (int k = 0; k < VC; ++k) consume(vv.v[k], k, true);
+      } else {
+#pragma unroll
+        for (int k = 0; k < VC; ++k)
+          consume((c0 + k < m) ? values[row + c0 + k] : (V)0, k, c0 + k < m);
+      }
     }
-    __syncthreads();
   }
   flush(cur_g);
 }
@@ -648,43 +697,126 @@ int dispatch_label(fh_call* c) {
 }
 
 
+/* launch geometry for the column path: vector width per thread, column
+ * blocks, and how many row chunks to split into so the chip is filled
+ * (~2048 workgroups) when the column count alone is too small */
+struct ColsPlan {
+  int vc;
+  int64_t ncolblk;
+  int nchunks;
+  int64_t chunk_rows;
+  BinLayout lay;  /* per-chunk slab layout over ngroups*m bins */
+};
+
+template <typename V>
+ColsPlan cols_plan(const fh_call* c) {
+  ColsPlan p{};
+  p.vc = Traits<V>::VEC;
+  if (c->ldm % p.vc != 0 || ((uintptr_t)c->values % 16) != 0) p.vc = 1;
+  p.ncolblk = (c->m + (int64_t)COLS_BLOCK * p.vc - 1) / ((int64_t)COLS_BLOCK * p.vc);
+  if (p.ncolblk == 0) p.ncolblk = 1;
+  int want = (int)((2048 + p.ncolblk - 1) / p.ncolblk);
+  if (want < 1) want = 1;
+  if (want > 64) want = 64;
+  p.lay = bin_layout<V>(set_bits(c->op_set), c->ngroups * c->m, 4);
+  /* cap slab at 2 GiB */
+  while (want > 1 && (int64_t)want * p.lay.bytes > (int64_t)2 << 30) want--;
+  if ((int64_t)want * COLS_TTILE > c->n) want = (int)((c->n + COLS_TTILE - 1) / COLS_TTILE);
+  if (want < 1) want = 1;
+  p.nchunks = want;
+  p.chunk_rows = (c->n + p.nchunks - 1) / p.nchunks;
+  return p;
+}
+
 /* init + decode shared with the atomic path, bins sized ngroups*m */
 template <typename V, int OPS>
 int launch_cols(fh_call* c) {
   using TR = Traits<V>;
   hipStream_t stream = (hipStream_t)c->stream;
   const int64_t nbins = c->ngroups * c->m;
-  if (OPS & (B_SUM | B_SSD))
-    FH_CHECK(hipMemsetAsync(c->out_sum, 0, nbins * 8, stream));
-  if (OPS & B_PROD) {
-    int fb = (int)((nbins + 255) / 256);
-    if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
-      hipLaunchKernelGGL(k_fill_f64, dim3(fb), dim3(256), 0, stream,
-                         (double*)c->out_sum, nbins, 1.0);
-    else
-      hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
-                         (int64_t*)c->out_sum, nbins, (int64_t)1);
-    FH_CHECK(hipGetLastError());
-  }
-  if (OPS & B_CNT) FH_CHECK(hipMemsetAsync(c->out_count, 0, nbins * 8, stream));
-  if (OPS & B_PRESENT) FH_CHECK(hipMemsetAsync(c->out_present, 0, nbins * 4, stream));
-  if (OPS & B_MIN)
-    FH_CHECK(hipMemsetAsync(c->out_min, 0xFF, nbins * sizeof(typename TR::Enc), stream));
-  if (OPS & B_MAX)
-    FH_CHECK(hipMemsetAsync(c->out_max, 0x00, nbins * sizeof(typename TR::Enc), stream));
-  if (OPS & B_NANFLAG) FH_CHECK(hipMemsetAsync(c->out_nanflag, 0, nbins * 4, stream));
-
+  const ColsPlan plan = cols_plan<V>(c);
+  const bool slab_mode = plan.nchunks > 1;
   const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
-  int nblocks = (int)((c->m + COLS_BLOCK - 1) / COLS_BLOCK);
-  if (nblocks == 0) nblocks = 1;
-  hipLaunchKernelGGL((k_reduce_cols<V, OPS>), dim3(nblocks), dim3(COLS_BLOCK),
-                     0, stream, (const V*)c->values,
-                     (const int*)c->labels, (const int*)c->perm, c->n,
-                     c->m, c->ldm, c->ngroups, c->means, skipnan, c->out_sum,
-                     c->out_count, c->out_present, c->out_min, c->out_max,
-                     c->out_nanflag);
-  FH_CHECK(hipGetLastError());
-  if (OPS & (B_MIN | B_MAX)) {
+
+  if (slab_mode) {
+    if ((int64_t)plan.nchunks * plan.lay.bytes > c->scratch_bytes) return 3;
+    /* identity-fill every chunk's slab sections (bins no segment touches) */
+    for (int ch = 0; ch < plan.nchunks; ++ch) {
+      char* s = (char*)c->scratch + (int64_t)ch * plan.lay.bytes;
+      if (OPS & (B_SUM | B_SSD))
+        FH_CHECK(hipMemsetAsync(s + plan.lay.sum_off, 0, nbins * 8, stream));
+      if (OPS & B_PROD) {
+        int fb = (int)((nbins + 255) / 256);
+        if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
+          hipLaunchKernelGGL(k_fill_f64, dim3(fb), dim3(256), 0, stream,
+                             (double*)(s + plan.lay.sum_off), nbins, 1.0);
+        else
+          hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                             (int64_t*)(s + plan.lay.sum_off), nbins, (int64_t)1);
+        FH_CHECK(hipGetLastError());
+      }
+      if (OPS & B_CNT) FH_CHECK(hipMemsetAsync(s + plan.lay.cnt_off, 0, nbins * 4, stream));
+      if (OPS & B_PRESENT)
+        FH_CHECK(hipMemsetAsync(s + plan.lay.present_off, 0, nbins * 4, stream));
+      if (OPS & B_MIN)
+        FH_CHECK(hipMemsetAsync(s + plan.lay.minmax_off, 0xFF,
+                                nbins * sizeof(typename TR::Enc), stream));
+      if (OPS & B_MAX)
+        FH_CHECK(hipMemsetAsync(s + plan.lay.minmax_off, 0x00,
+                                nbins * sizeof(typename TR::Enc), stream));
+      if (OPS & B_NANFLAG)
+        FH_CHECK(hipMemsetAsync(s + plan.lay.nanflag_off, 0, nbins * 4, stream));
+    }
+  } else {
+    if (OPS & (B_SUM | B_SSD))
+      FH_CHECK(hipMemsetAsync(c->out_sum, 0, nbins * 8, stream));
+    if (OPS & B_PROD) {
+      int fb = (int)((nbins + 255) / 256);
+      if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
+        hipLaunchKernelGGL(k_fill_f64, dim3(fb), dim3(256), 0, stream,
+                           (double*)c->out_sum, nbins, 1.0);
+      else
+        hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                           (int64_t*)c->out_sum, nbins, (int64_t)1);
+      FH_CHECK(hipGetLastError());
+    }
+    if (OPS & B_CNT) FH_CHECK(hipMemsetAsync(c->out_count, 0, nbins * 8, stream));
+    if (OPS & B_PRESENT) FH_CHECK(hipMemsetAsync(c->out_present, 0, nbins * 4, stream));
+    if (OPS & B_MIN)
+      FH_CHECK(hipMemsetAsync(c->out_min, 0xFF, nbins * sizeof(typename TR::Enc), stream));
+    if (OPS & B_MAX)
+      FH_CHECK(hipMemsetAsync(c->out_max, 0x00, nbins * sizeof(typename TR::Enc), stream));
+    if (OPS & B_NANFLAG) FH_CHECK(hipMemsetAsync(c->out_nanflag, 0, nbins * 4, stream));
+  }
+
+  dim3 grid((uint32_t)plan.ncolblk, (uint32_t)plan.nchunks);
+  auto launch = [&](auto kern) -> int {
+    hipLaunchKernelGGL(kern, grid, dim3(COLS_BLOCK), 0, stream,
+                       (const V*)c->values, (const int*)c->labels,
+                       (const int*)c->perm, c->n, c->m, c->ldm, c->ngroups,
+                       c->means, skipnan, plan.chunk_rows, (char*)c->scratch,
+                       plan.lay, c->out_sum, c->out_count, c->out_present,
+                       c->out_min, c->out_max, c->out_nanflag);
+    return (int)hipGetLastError();
+  };
+  int rc;
+  if (slab_mode)
+    rc = plan.vc > 1 ? launch(k_reduce_cols<V, OPS, Traits<V>::VEC, true>)
+                     : launch(k_reduce_cols<V, OPS, 1, true>);
+  else
+    rc = plan.vc > 1 ? launch(k_reduce_cols<V, OPS, Traits<V>::VEC, false>)
+                     : launch(k_reduce_cols<V, OPS, 1, false>);
+  if (rc != 0) return rc + 1000;
+
+  if (slab_mode) {
+    /* fold the chunk partials; k_combine also decodes min/max */
+    int cb = (int)((nbins + 255) / 256);
+    hipLaunchKernelGGL((k_combine<V, OPS>), dim3(cb), dim3(256), 0, stream,
+                       (const char*)c->scratch, plan.nchunks, nbins, plan.lay,
+                       c->out_sum, c->out_count, c->out_present, c->out_min,
+                       c->out_max, c->out_nanflag);
+    FH_CHECK(hipGetLastError());
+  } else if (OPS & (B_MIN | B_MAX)) {
     int db = (int)((nbins + 255) / 256);
     hipLaunchKernelGGL((k_decode<V, OPS>), dim3(db), dim3(256), 0, stream,
                        nbins, c->out_min, c->out_max, c->out_count,
@@ -728,6 +860,17 @@ int64_t fh_scratch_bytes(const fh_call* c) {
     case FH_I64: per_block = bin_layout<int64_t>(bits, c->ngroups, 4).bytes; break;
     case FH_I32: per_block = bin_layout<int32_t>(bits, c->ngroups, 4).bytes; break;
     default: return -1;
+  }
+  if (c->m > 0) {
+    /* column path: per-chunk slab when the row range is split */
+    ColsPlan p;
+    switch (c->vdtype) {
+      case FH_F32: p = cols_plan<float>(c); break;
+      case FH_F64: p = cols_plan<double>(c); break;
+      case FH_I64: p = cols_plan<int64_t>(c); break;
+      default: p = cols_plan<int32_t>(c); break;
+    }
+    return p.nchunks > 1 ? (int64_t)p.nchunks * p.lay.bytes : 0;
   }
   if (per_block > LDS_MAX && !(c->flags & FH_FORCE_LDS)) return 0; /* atomic path */
   int blocks_per_cu = per_block * 2 <= LDS_MAX ? 2 : 1;
