@@ -69,6 +69,8 @@ class FhCall(ctypes.Structure):
         ("perm", ctypes.c_void_p),
         ("m", ctypes.c_int64),
         ("ldm", ctypes.c_int64),
+        ("chunk_offsets", ctypes.c_void_p),
+        ("nchunks", ctypes.c_int64),
     ]
 
 

@@ -367,6 +367,24 @@ def grouped_partials_cols(
     perm = perm.contiguous()
     c.labels = codes_sorted.data_ptr()
     c.perm = perm.data_ptr()
+    # group-aligned row chunks: enough workgroups to fill the chip with NO
+    # partials slab (each chunk owns disjoint groups since codes are sorted)
+    chunk_t = None
+    col_blocks = max(1, m // 1024)
+    desired = int(min(64, max(1, -(-2048 // col_blocks))))
+    if desired > 1 and n_t > 1:
+        sc = codes_sorted.cpu().numpy()
+        seg = np.flatnonzero(np.diff(sc)) + 1
+        target = max(1, -(-n_t // desired))
+        bounds = [0]
+        for b in seg:
+            if b - bounds[-1] >= target:
+                bounds.append(int(b))
+        bounds.append(int(n_t))
+        if len(bounds) > 2:
+            chunk_t = torch.tensor(bounds, dtype=torch.int64, device=dev)
+            c.chunk_offsets = chunk_t.data_ptr()
+            c.nchunks = len(bounds) - 1
     if means is not None:
         assert means.dtype == torch.float64 and means.is_cuda
         means = means.contiguous()
@@ -406,7 +424,7 @@ def grouped_partials_cols(
     c.stream = torch.cuda.current_stream(dev).cuda_stream
     _ffi.check(lib.fh_grouped_reduce_cols(ctypes.byref(c)))
     out["_path"] = c.path_used  # type: ignore[assignment]
-    for t in (values2d, codes_sorted, perm, means, scratch):
+    for t in (values2d, codes_sorted, perm, means, scratch, chunk_t):
         if isinstance(t, torch.Tensor):
             t.record_stream(torch.cuda.current_stream(dev))
     return out

@@ -433,7 +433,8 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
     const V* __restrict__ values, const int* __restrict__ codes_sorted,
     const int* __restrict__ perm, int64_t n_t, int64_t m, int64_t ldm,
     int64_t ngroups, const double* __restrict__ means, int skipnan,
-    int64_t chunk_rows, char* __restrict__ slab, BinLayout lay,
+    int64_t chunk_rows, const int64_t* __restrict__ chunk_offs,
+    char* __restrict__ slab, BinLayout lay,
     void* out_sum, int64_t* out_count, uint32_t* out_present, void* out_min,
     void* out_max, uint32_t* out_nanflag) {
   using TR = Traits<V>;
@@ -447,15 +448,21 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
 
   const int64_t c0 = ((int64_t)blockIdx.x * COLS_BLOCK + threadIdx.x) * VC;
   const bool full = c0 + VC <= m;
-  const int64_t t_begin = (int64_t)blockIdx.y * chunk_rows;
-  const int64_t t_end = (t_begin + chunk_rows < n_t) ? t_begin + chunk_rows : n_t;
+  int64_t t_begin, t_end;
+  if (chunk_offs) {
+    t_begin = chunk_offs[blockIdx.y];
+    t_end = chunk_offs[blockIdx.y + 1];
+  } else {
+    t_begin = (int64_t)blockIdx.y * chunk_rows;
+    t_end = (t_begin + chunk_rows < n_t) ? t_begin + chunk_rows : n_t;
+  }
 
   char* my_slab = SLAB ? slab + (int64_t)blockIdx.y * lay.bytes : nullptr;
 
   SumT acc[VC];
   uint32_t cnt[VC];
   Enc mn[VC], mx[VC];
-  uint32_t nanflag = 0;
+  uint32_t nanflag[VC];
   double mean_g[VC];
   int cur_g = -1;
 
@@ -466,8 +473,8 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
       cnt[k] = 0;
       mn[k] = (Enc)~(Enc)0;
       mx[k] = (Enc)0;
+      nanflag[k] = 0;
     }
-    nanflag = 0;
   };
   reset();
 
@@ -483,14 +490,14 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
         if (OPS & B_PRESENT) ((uint32_t*)(my_slab + lay.present_off))[o] = 1u;
         if (OPS & B_MIN) ((Enc*)(my_slab + lay.minmax_off))[o] = mn[k];
         if (OPS & B_MAX) ((Enc*)(my_slab + lay.minmax_off))[o] = mx[k];
-        if (OPS & B_NANFLAG) ((uint32_t*)(my_slab + lay.nanflag_off))[o] = nanflag;
+        if (OPS & B_NANFLAG) ((uint32_t*)(my_slab + lay.nanflag_off))[o] = nanflag[k];
       } else {
         if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)out_sum)[o] = acc[k];
         if (OPS & B_CNT) out_count[o] = (int64_t)cnt[k];
         if (OPS & B_PRESENT) out_present[o] = 1u;
         if (OPS & B_MIN) ((Enc*)out_min)[o] = mn[k];
         if (OPS & B_MAX) ((Enc*)out_max)[o] = mx[k];
-        if (OPS & B_NANFLAG) out_nanflag[o] = nanflag;
+        if (OPS & B_NANFLAG) out_nanflag[o] = nanflag[k];
       }
     }
   };
@@ -508,7 +515,7 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
     if (OPS & B_CNT) cnt[k] += vnan ? 0u : 1u;
     if (OPS & (B_MIN | B_MAX)) {
       if (vnan) {
-        if (OPS & B_NANFLAG) nanflag = 1u;
+        if (OPS & B_NANFLAG) nanflag[k] = 1u;
       } else {
         const Enc e = TR::enc(v);
         if (OPS & B_MIN) mn[k] = e < mn[k] ? e : mn[k];
@@ -734,8 +741,10 @@ int launch_cols(fh_call* c) {
   using TR = Traits<V>;
   hipStream_t stream = (hipStream_t)c->stream;
   const int64_t nbins = c->ngroups * c->m;
-  const ColsPlan plan = cols_plan<V>(c);
-  const bool slab_mode = plan.nchunks > 1;
+  ColsPlan plan = cols_plan<V>(c);
+  const bool aligned_chunks = c->chunk_offsets != nullptr && c->nchunks > 0;
+  if (aligned_chunks) plan.nchunks = (int)c->nchunks;
+  const bool slab_mode = !aligned_chunks && plan.nchunks > 1;
   const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
 
   if (slab_mode) {
@@ -790,13 +799,14 @@ int launch_cols(fh_call* c) {
   }
 
   dim3 grid((uint32_t)plan.ncolblk, (uint32_t)plan.nchunks);
+  const int64_t* offs = aligned_chunks ? (const int64_t*)c->chunk_offsets : nullptr;
   auto launch = [&](auto kern) -> int {
     hipLaunchKernelGGL(kern, grid, dim3(COLS_BLOCK), 0, stream,
                        (const V*)c->values, (const int*)c->labels,
                        (const int*)c->perm, c->n, c->m, c->ldm, c->ngroups,
-                       c->means, skipnan, plan.chunk_rows, (char*)c->scratch,
-                       plan.lay, c->out_sum, c->out_count, c->out_present,
-                       c->out_min, c->out_max, c->out_nanflag);
+                       c->means, skipnan, plan.chunk_rows, offs,
+                       (char*)c->scratch, plan.lay, c->out_sum, c->out_count,
+                       c->out_present, c->out_min, c->out_max, c->out_nanflag);
     return (int)hipGetLastError();
   };
   int rc;
@@ -862,6 +872,7 @@ int64_t fh_scratch_bytes(const fh_call* c) {
     default: return -1;
   }
   if (c->m > 0) {
+    if (c->chunk_offsets) return 0; /* group-aligned chunks write directly */
     /* column path: per-chunk slab when the row range is split */
     ColsPlan p;
     switch (c->vdtype) {

@@ -97,6 +97,11 @@ typedef struct fh_call {
    * permutation; outputs are (ngroups, m) group-major. */
   const void* perm;
   int64_t m, ldm;
+  /* optional: group-aligned row chunking (int64[nchunks+1] row offsets, each
+   * chunk covering disjoint groups -> chunks write final bins directly, no
+   * slab); when NULL the launcher row-splits with a partials slab instead */
+  const void* chunk_offsets;
+  int64_t nchunks;
 } fh_call;
 
 /* scratch requirement for this call (0 when the global-atomic path is used) */
