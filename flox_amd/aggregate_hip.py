@@ -19,6 +19,7 @@ library or a GPU is missing these raise.
 
 from __future__ import annotations
 
+import builtins
 import ctypes
 import math
 
@@ -370,12 +371,12 @@ def grouped_partials_cols(
     # group-aligned row chunks: enough workgroups to fill the chip with NO
     # partials slab (each chunk owns disjoint groups since codes are sorted)
     chunk_t = None
-    col_blocks = max(1, m // 1024)
-    desired = int(min(64, max(1, -(-2048 // col_blocks))))
+    col_blocks = builtins.max(1, m // 1024)
+    desired = int(builtins.min(64, builtins.max(1, -(-2048 // col_blocks))))
     if desired > 1 and n_t > 1:
         sc = codes_sorted.cpu().numpy()
         seg = np.flatnonzero(np.diff(sc)) + 1
-        target = max(1, -(-n_t // desired))
+        target = builtins.max(1, -(-n_t // desired))
         bounds = [0]
         for b in seg:
             if b - bounds[-1] >= target:
