@@ -1,6 +1,8 @@
 """GPU parity tests: the HIP engine vs the pinned oracle and the golden
 fixtures, through the product API and through the engine seam."""
 
+import zlib
+
 import numpy as np
 import pytest
 import torch
@@ -44,7 +46,7 @@ def _tol(func, dtype):
 @pytest.mark.parametrize("dtype", ["float32", "float64", "int64"])
 @pytest.mark.parametrize("ngroups", [7, 1000, 40_000])  # LDS path and global-atomic path
 def test_random_sweep_vs_oracle(func, dtype, ngroups):
-    rng = np.random.default_rng(hash((func, dtype, ngroups)) % 2**31)
+    rng = np.random.default_rng(zlib.crc32(f"{func}-{dtype}-{ngroups}".encode()))
     n = 200_000
     labels = rng.integers(0, ngroups, n)
     if np.dtype(dtype).kind == "f":
