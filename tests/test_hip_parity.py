@@ -39,7 +39,9 @@ def _tol(func, dtype):
         return dict(rtol=0, atol=0)
     if np.dtype(dtype).itemsize == 4:
         return dict(rtol=3e-6, atol=1e-6)
-    return dict(rtol=1e-12, atol=1e-14)
+    # atol absorbs fp-cancellation under differing summation order (group
+    # sums near zero from O(100)-magnitude addends)
+    return dict(rtol=1e-12, atol=1e-9)
 
 
 @pytest.mark.parametrize("func", FUNCS)
