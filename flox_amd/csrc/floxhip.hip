@@ -590,6 +590,374 @@ __global__ void k_decode(int64_t ngroups, void* out_min, void* out_max,
   }
 }
 
+
+/* ---- partition path (huge ngroups, e.g. 1e7): bucket scatter + per-bucket
+ * LDS reduce — the GPU analogue of the reference's sort-by-key + reduceat
+ * engine (aggregate_flox.py:9-23 _prepare_for_flox + :133-192
+ * _np_grouped_op), except only ONE radix digit is needed: rows are
+ * partitioned by the high bits of their group code into buckets whose bins
+ * fit in LDS, then each bucket is reduced like the LDS path. Traffic:
+ * labels (count) + values+labels (scatter read) + pairs (write+read) ~=
+ * 3.7x the algorithmic bytes, vs ~45x slower raw global atomics. */
+
+template <typename V> struct PairT;
+template <> struct PairT<float> { float v; uint32_t lc; };
+template <> struct PairT<int32_t> { int32_t v; uint32_t lc; };
+template <> struct PairT<double> { double v; uint32_t lc; uint32_t pad; };
+template <> struct PairT<int64_t> { int64_t v; uint32_t lc; uint32_t pad; };
+
+__device__ __forceinline__ int64_t code_of(int64_t l0, int64_t l1, bool twolab,
+                                           int64_t g0, int64_t g1, int64_t ngroups) {
+  if (twolab) {
+    if ((uint64_t)l0 >= (uint64_t)g0 || (uint64_t)l1 >= (uint64_t)g1) return -1;
+    return l0 * g1 + l1;
+  }
+  return ((uint64_t)l0 >= (uint64_t)ngroups) ? -1 : l0;
+}
+
+template <typename L>
+__launch_bounds__(256) __global__ void k_part_count(
+    const L* __restrict__ labels, const L* __restrict__ labels2, int64_t n,
+    int64_t ngroups, int64_t g0, int64_t g1, int shift, int B,
+    uint32_t* __restrict__ bucket_counts) {
+  extern __shared__ __attribute__((aligned(16))) char smem_pc[];
+  uint32_t* s_hist = (uint32_t*)smem_pc;
+  for (int i = threadIdx.x; i < B; i += blockDim.x) s_hist[i] = 0;
+  __syncthreads();
+  const bool twolab = labels2 != nullptr;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    int64_t code = code_of((int64_t)labels[i], twolab ? (int64_t)labels2[i] : 0,
+                           twolab, g0, g1, ngroups);
+    if (code >= 0) atomicAdd(&s_hist[(int)(code >> shift)], 1u);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < B; i += blockDim.x)
+    if (s_hist[i]) atomicAdd(&bucket_counts[i], s_hist[i]);
+}
+
+/* tile-staged scatter: per tile, histogram -> block scan -> global
+ * reservation (one returning atomic per nonempty bucket) -> bucket-ordered
+ * staging in LDS with per-slot destinations -> coalesced dump */
+constexpr int PART_BLOCK = 512;
+
+template <typename V, typename L>
+__launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
+    const V* __restrict__ values, const L* __restrict__ labels,
+    const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
+    int64_t g1, int shift, int B, int Bpad /* pow2 >= B */,
+    uint32_t* __restrict__ cursors, PairT<V>* __restrict__ pairs) {
+  constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+  constexpr int RPT = T / PART_BLOCK; /* rows per thread per tile */
+  extern __shared__ __attribute__((aligned(16))) char smem_ps[];
+  uint32_t* s_hist = (uint32_t*)smem_ps;              /* [Bpad] scanned in place */
+  uint32_t* s_off = s_hist + Bpad;                    /* [Bpad] exclusive offsets */
+  uint32_t* s_gbase = s_off + Bpad;                   /* [B] global base per bucket */
+  uint32_t* s_cur = s_gbase + Bpad;                   /* [B] running cursor */
+  uint32_t* s_dest = s_cur + Bpad;                    /* [T] per-slot global dest */
+  PairT<V>* s_stage = (PairT<V>*)(s_dest + T);        /* [T] */
+
+  const bool twolab = labels2 != nullptr;
+  const int tid = threadIdx.x;
+
+  for (int64_t tile = (int64_t)blockIdx.x * T; tile < n; tile += (int64_t)gridDim.x * T) {
+    const int nt = (int)((n - tile < T) ? (n - tile) : T);
+    for (int i = tid; i < Bpad; i += PART_BLOCK) s_hist[i] = 0;
+    __syncthreads();
+
+    V rv[RPT];
+    uint32_t rlc[RPT];
+    int rbk[RPT];
+#pragma unroll
+    for (int k = 0; k < RPT; ++k) {
+      const int idx = tid + k * PART_BLOCK;
+      rbk[k] = -1;
+      if (idx < nt) {
+        const int64_t i = tile + idx;
+        const int64_t code = code_of((int64_t)labels[i],
+                                     twolab ? (int64_t)labels2[i] : 0, twolab,
+                                     g0, g1, ngroups);
+        if (code >= 0) {
+          rv[k] = values[i];
+          rbk[k] = (int)(code >> shift);
+          rlc[k] = (uint32_t)(code - ((int64_t)rbk[k] << shift));
+          atomicAdd(&s_hist[rbk[k]], 1u);
+        }
+      }
+    }
+    __syncthreads();
+    /* exclusive block scan of s_hist[Bpad] (Hillis-Steele, in place) */
+    for (int d = 1; d < Bpad; d <<= 1) {
+      uint32_t add[16];
+      int na = 0;
+      for (int i = tid; i < Bpad; i += PART_BLOCK)
+        add[na++] = (i >= d) ? s_hist[i - d] : 0u;
+      __syncthreads();
+      na = 0;
+      for (int i = tid; i < Bpad; i += PART_BLOCK) s_hist[i] += add[na++];
+      __syncthreads();
+    }
+    /* s_hist now holds INCLUSIVE sums; derive exclusive + reserve globally */
+    for (int b = tid; b < B; b += PART_BLOCK) {
+      const uint32_t incl = s_hist[b];
+      const uint32_t excl = (b == 0) ? 0u : s_hist[b - 1];
+      const uint32_t cnt = incl - excl;
+      s_off[b] = excl;
+      s_cur[b] = excl;
+      if (cnt) s_gbase[b] = atomicAdd(&cursors[b], cnt);
+    }
+    __syncthreads();
+    const int valid = (int)s_hist[Bpad - 1];
+    /* bucket-ordered staging with per-slot destination */
+#pragma unroll
+    for (int k = 0; k < RPT; ++k) {
+      if (rbk[k] >= 0) {
+        const uint32_t pos = atomicAdd(&s_cur[rbk[k]], 1u);
+        s_stage[pos].v = rv[k];
+        s_stage[pos].lc = rlc[k];
+        s_dest[pos] = s_gbase[rbk[k]] + (pos - s_off[rbk[k]]);
+      }
+    }
+    __syncthreads();
+    for (int i = tid; i < valid; i += PART_BLOCK) pairs[s_dest[i]] = s_stage[i];
+    __syncthreads();
+  }
+}
+
+/* per-bucket LDS-binned reduce of the scattered pairs; flush with global
+ * atomics (a few chunks per bucket at most) into the FINAL bins */
+template <typename V, int OPS>
+__launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
+    const PairT<V>* __restrict__ pairs, const uint32_t* __restrict__ base,
+    int64_t chunk, int gpb, int shift, int64_t ngroups,
+    const double* __restrict__ means, int skipnan, BinLayout lay,
+    void* out_sum, int64_t* out_count, uint32_t* out_present, void* out_min,
+    void* out_max, uint32_t* out_nanflag) {
+  using TR = Traits<V>;
+  using Acc = typename TR::Acc;
+  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
+  using Enc = typename TR::Enc;
+  constexpr bool IS_PROD = (OPS & B_PROD) != 0;
+
+  const int b = blockIdx.y;
+  const int64_t bkt_begin = base[b], bkt_end = base[b + 1];
+  const int64_t start = bkt_begin + (int64_t)blockIdx.x * chunk;
+  if (start >= bkt_end) return;
+  const int64_t end = (start + chunk < bkt_end) ? start + chunk : bkt_end;
+  const int64_t gbase = (int64_t)b << shift;
+  const int ng_here = (int)(((gbase + gpb) <= ngroups) ? gpb : (ngroups - gbase));
+
+  extern __shared__ __attribute__((aligned(16))) char smem_rb[];
+  SumT* s_sum = (OPS & (B_SUM | B_SSD | B_PROD)) ? (SumT*)(smem_rb + lay.sum_off) : nullptr;
+  uint32_t* s_cnt = (OPS & B_CNT) ? (uint32_t*)(smem_rb + lay.cnt_off) : nullptr;
+  uint32_t* s_present = (OPS & B_PRESENT) ? (uint32_t*)(smem_rb + lay.present_off) : nullptr;
+  Enc* s_mm = (OPS & (B_MIN | B_MAX)) ? (Enc*)(smem_rb + lay.minmax_off) : nullptr;
+  uint32_t* s_nanflag = (OPS & B_NANFLAG) ? (uint32_t*)(smem_rb + lay.nanflag_off) : nullptr;
+
+  const int tid = threadIdx.x;
+  for (int g = tid; g < ng_here; g += blockDim.x) {
+    if (OPS & (B_SUM | B_SSD)) s_sum[g] = (SumT)0;
+    if (IS_PROD) s_sum[g] = (SumT)1;
+    if (OPS & B_CNT) s_cnt[g] = 0u;
+    if (OPS & B_PRESENT) s_present[g] = 0u;
+    if (OPS & B_MIN) s_mm[g] = (Enc)~(Enc)0;
+    if (OPS & B_MAX) s_mm[g] = (Enc)0;
+    if (OPS & B_NANFLAG) s_nanflag[g] = 0u;
+  }
+  __syncthreads();
+
+  for (int64_t i = start + tid; i < end; i += blockDim.x) {
+    const PairT<V> p = pairs[i];
+    const uint32_t lc = p.lc;
+    const V v = p.v;
+    const bool vnan = TR::isnan_(v);
+    if (OPS & B_PRESENT) s_present[lc] = 1u;
+    if (vnan && skipnan) continue;
+    if (OPS & B_SUM) acc_add(&s_sum[lc], (Acc)v);
+    if (IS_PROD) acc_mul(&s_sum[lc], (Acc)v);
+    if (OPS & B_SSD) {
+      const double d = (double)v - means[gbase + lc];
+      atomicAdd((double*)s_sum + lc, d * d);
+    }
+    if (OPS & B_CNT) {
+      if (!vnan) atomicAdd(&s_cnt[lc], 1u);
+    }
+    if (OPS & (B_MIN | B_MAX)) {
+      if (vnan) {
+        if (OPS & B_NANFLAG) s_nanflag[lc] = 1u;
+      } else {
+        if (OPS & B_MIN) enc_min(&s_mm[lc], TR::enc(v));
+        if (OPS & B_MAX) enc_max(&s_mm[lc], TR::enc(v));
+      }
+    }
+  }
+  __syncthreads();
+  /* flush into the final bins (out_min/out_max hold ENCODED values until
+   * k_decode, exactly like the global-atomic path) */
+  for (int g = tid; g < ng_here; g += blockDim.x) {
+    const int64_t o = gbase + g;
+    if ((OPS & (B_SUM | B_SSD)) && s_sum[g] != (SumT)0) acc_add(&((SumT*)out_sum)[o], s_sum[g]);
+    if (IS_PROD && s_sum[g] != (SumT)1) acc_mul(&((SumT*)out_sum)[o], s_sum[g]);
+    if ((OPS & B_CNT) && s_cnt[g])
+      atomicAdd(reinterpret_cast<unsigned long long*>(&out_count[o]), (unsigned long long)s_cnt[g]);
+    if ((OPS & B_PRESENT) && s_present[g]) out_present[o] = 1u;
+    if (OPS & B_MIN) enc_min(&((Enc*)out_min)[o], s_mm[g]);
+    if (OPS & B_MAX) enc_max(&((Enc*)out_max)[o], s_mm[g]);
+    if ((OPS & B_NANFLAG) && s_nanflag[g]) out_nanflag[o] = 1u;
+  }
+}
+
+
+/* ---- partition-path host plumbing ---------------------------------------- */
+template <typename V, int OPS>
+int init_outs(fh_call* c, int64_t nbins, hipStream_t stream) {
+  using TR = Traits<V>;
+  if (OPS & (B_SUM | B_SSD)) FH_CHECK(hipMemsetAsync(c->out_sum, 0, nbins * 8, stream));
+  if (OPS & B_PROD) {
+    int fb = (int)((nbins + 255) / 256);
+    if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
+      hipLaunchKernelGGL(k_fill_f64, dim3(fb), dim3(256), 0, stream,
+                         (double*)c->out_sum, nbins, 1.0);
+    else
+      hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                         (int64_t*)c->out_sum, nbins, (int64_t)1);
+    FH_CHECK(hipGetLastError());
+  }
+  if (OPS & B_CNT) FH_CHECK(hipMemsetAsync(c->out_count, 0, nbins * 8, stream));
+  if (OPS & B_PRESENT) FH_CHECK(hipMemsetAsync(c->out_present, 0, nbins * 4, stream));
+  if (OPS & B_MIN)
+    FH_CHECK(hipMemsetAsync(c->out_min, 0xFF, nbins * sizeof(typename TR::Enc), stream));
+  if (OPS & B_MAX)
+    FH_CHECK(hipMemsetAsync(c->out_max, 0x00, nbins * sizeof(typename TR::Enc), stream));
+  if (OPS & B_NANFLAG) FH_CHECK(hipMemsetAsync(c->out_nanflag, 0, nbins * 4, stream));
+  return 0;
+}
+
+struct PartPlan {
+  int shift, gpb, B, Bpad;
+  BinLayout lay;       /* per-bucket bins (gpb entries) */
+  int64_t pairs_off, counts_off, base_off, cursors_off, bytes;
+  int64_t scatter_lds;
+  bool feasible;
+};
+
+template <typename V>
+PartPlan part_plan(const fh_call* c) {
+  PartPlan p{};
+  p.feasible = false;
+  if (c->n >= ((int64_t)1 << 31) || c->ngroups <= 0) return p;
+  const int bits = set_bits(c->op_set);
+  int shift = 13;
+  while (shift > 8 && bin_layout<V>(bits, (int64_t)1 << shift, 4).bytes > 120 * 1024)
+    shift--;
+  BinLayout lay = bin_layout<V>(bits, (int64_t)1 << shift, 4);
+  if (lay.bytes > 120 * 1024) return p;
+  int64_t B64 = (c->ngroups + ((int64_t)1 << shift) - 1) >> shift;
+  if (B64 > 4096) return p;
+  p.B = (int)B64;
+  p.shift = shift;
+  p.gpb = 1 << shift;
+  p.lay = lay;
+  p.Bpad = 256;
+  while (p.Bpad < p.B) p.Bpad <<= 1;
+  constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+  p.scatter_lds = (int64_t)4 * p.Bpad * 4 + (int64_t)T * 4 + (int64_t)T * sizeof(PairT<V>);
+  if (p.scatter_lds > LDS_MAX) return p;
+  int64_t off = 0;
+  auto carve = [&](int64_t b) {
+    int64_t o = off;
+    off += ((b + 255) / 256) * 256;
+    return o;
+  };
+  p.pairs_off = carve(c->n * (int64_t)sizeof(PairT<V>));
+  p.counts_off = carve((int64_t)p.B * 4);
+  p.base_off = carve(((int64_t)p.B + 1) * 4);
+  p.cursors_off = carve((int64_t)p.B * 4);
+  p.bytes = off;
+  p.feasible = true;
+  return p;
+}
+
+template <typename V, typename L, int OPS>
+int launch_partition(fh_call* c, const PartPlan& pp) {
+  hipStream_t stream = (hipStream_t)c->stream;
+  const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
+  char* scr = (char*)c->scratch;
+  uint32_t* counts = (uint32_t*)(scr + pp.counts_off);
+  uint32_t* based = (uint32_t*)(scr + pp.base_off);
+  uint32_t* cursors = (uint32_t*)(scr + pp.cursors_off);
+  PairT<V>* pairs = (PairT<V>*)(scr + pp.pairs_off);
+
+  FH_CHECK(hipMemsetAsync(counts, 0, (int64_t)pp.B * 4, stream));
+  {
+    int64_t wb = (c->n + 255) / 256;
+    int nb0 = (int)(wb < 2048 ? (wb > 0 ? wb : 1) : 2048);
+    hipLaunchKernelGGL((k_part_count<L>), dim3(nb0), dim3(256), pp.Bpad * 4,
+                       stream, (const L*)c->labels, (const L*)c->labels2, c->n,
+                       c->ngroups, c->g0, c->g1, pp.shift, pp.B, counts);
+    FH_CHECK(hipGetLastError());
+  }
+  /* bucket directory on the host (one sync; B <= 4096 words) */
+  static thread_local uint32_t h_counts[4096], h_base[4097];
+  FH_CHECK(hipMemcpyAsync(h_counts, counts, (int64_t)pp.B * 4,
+                          hipMemcpyDeviceToHost, stream));
+  FH_CHECK(hipStreamSynchronize(stream));
+  uint32_t acc = 0, maxc = 0;
+  for (int i = 0; i < pp.B; ++i) {
+    h_base[i] = acc;
+    acc += h_counts[i];
+    if (h_counts[i] > maxc) maxc = h_counts[i];
+  }
+  h_base[pp.B] = acc;
+  FH_CHECK(hipMemcpyAsync(based, h_base, ((int64_t)pp.B + 1) * 4,
+                          hipMemcpyHostToDevice, stream));
+  FH_CHECK(hipMemcpyAsync(cursors, h_base, (int64_t)pp.B * 4,
+                          hipMemcpyHostToDevice, stream));
+
+  {
+    auto kern = k_part_scatter<V, L>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)pp.scatter_lds));
+    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    int64_t wb = (c->n + T - 1) / T;
+    int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
+    hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
+                       stream, (const V*)c->values, (const L*)c->labels,
+                       (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
+                       pp.shift, pp.B, pp.Bpad, cursors, pairs);
+    FH_CHECK(hipGetLastError());
+  }
+  {
+    int rc = init_outs<V, OPS>(c, c->ngroups, stream);
+    if (rc) return rc;
+  }
+  {
+    auto kern = k_reduce_bucket<V, OPS>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)pp.lay.bytes));
+    const int64_t chunk = 1 << 19;
+    int maxchunks = (int)((maxc + chunk - 1) / chunk);
+    if (maxchunks < 1) maxchunks = 1;
+    hipLaunchKernelGGL(kern, dim3(maxchunks, pp.B), dim3(BLOCK_LDS),
+                       pp.lay.bytes, stream, pairs, based, chunk, pp.gpb,
+                       pp.shift, c->ngroups, c->means, skipnan, pp.lay,
+                       c->out_sum, c->out_count, c->out_present, c->out_min,
+                       c->out_max, c->out_nanflag);
+    FH_CHECK(hipGetLastError());
+  }
+  if (OPS & (B_MIN | B_MAX)) {
+    int db = (int)((c->ngroups + 255) / 256);
+    hipLaunchKernelGGL((k_decode<V, OPS>), dim3(db), dim3(256), 0, stream,
+                       c->ngroups, c->out_min, c->out_max, c->out_count,
+                       c->out_present);
+    FH_CHECK(hipGetLastError());
+  }
+  c->path_used = 4;
+  return 0;
+}
+
 /* ---- host-side dispatch -------------------------------------------------- */
 
 template <typename V, typename L, int OPS>
@@ -631,6 +999,13 @@ int launch_typed(fh_call* c) {
     FH_CHECK(hipGetLastError());
     c->path_used = 1;
     return 0;
+  }
+
+  /* huge group counts: bucket-partition path when scratch allows */
+  if (!(c->flags & FH_FORCE_ATOMIC)) {
+    PartPlan pp = part_plan<V>(c);
+    if (pp.feasible && c->scratch_bytes >= pp.bytes)
+      return launch_partition<V, L, OPS>(c, pp);
   }
 
   /* global-atomic path: memset-style init of the bins */
@@ -883,7 +1258,18 @@ int64_t fh_scratch_bytes(const fh_call* c) {
     }
     return p.nchunks > 1 ? (int64_t)p.nchunks * p.lay.bytes : 0;
   }
-  if (per_block > LDS_MAX && !(c->flags & FH_FORCE_LDS)) return 0; /* atomic path */
+  if (per_block > LDS_MAX && !(c->flags & FH_FORCE_LDS)) {
+    if (c->flags & FH_FORCE_ATOMIC) return 0;
+    /* partition path wants the pairs buffer + bucket directory */
+    PartPlan p;
+    switch (c->vdtype) {
+      case FH_F32: p = part_plan<float>(c); break;
+      case FH_F64: p = part_plan<double>(c); break;
+      case FH_I64: p = part_plan<int64_t>(c); break;
+      default: p = part_plan<int32_t>(c); break;
+    }
+    return p.feasible ? p.bytes : 0;
+  }
   int blocks_per_cu = per_block * 2 <= LDS_MAX ? 2 : 1;
   return (int64_t)NUM_CU * blocks_per_cu * per_block;
 }

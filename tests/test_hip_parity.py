@@ -79,6 +79,20 @@ def test_both_kernel_paths_agree(force_path):
     np.testing.assert_allclose(p["sum"].cpu().numpy(), want_sum, rtol=1e-12)
 
 
+@pytest.mark.parametrize("func", ["sum", "nansum", "mean", "count", "min", "nanmax", "var"])
+def test_partition_path_many_groups(func):
+    """2e6 groups: the bucket-partition (sort) path."""
+    rng = np.random.default_rng(zlib.crc32(func.encode()))
+    n, ng = 1_000_000, 2_000_000
+    labels = rng.integers(0, ng, n)
+    vals = rng.standard_normal(n)
+    vals[rng.random(n) < 0.02] = np.nan
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    assert got.dtype == want.dtype
+    np.testing.assert_allclose(got, want, equal_nan=True, **_tol(func, want.dtype))
+
+
 def test_engine_seam_callables():
     """The reference-shaped seam: generic_aggregate(engine='hip', func=...)
     (reference flox/aggregations.py:60-133 signature)."""
