@@ -651,10 +651,10 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
   constexpr int RPT = T / PART_BLOCK; /* rows per thread per tile */
   extern __shared__ __attribute__((aligned(16))) char smem_ps[];
   uint32_t* s_hist = (uint32_t*)smem_ps;              /* [Bpad] scanned in place */
-  uint32_t* s_off = s_hist + Bpad;                    /* [Bpad] exclusive offsets */
-  uint32_t* s_gbase = s_off + Bpad;                   /* [B] global base per bucket */
-  uint32_t* s_cur = s_gbase + Bpad;                   /* [B] running cursor */
-  uint32_t* s_dest = s_cur + Bpad;                    /* [T] per-slot global dest */
+  uint32_t* s_gbase = s_hist + Bpad;                  /* [Bpad] gbase - excl per bucket */
+  uint32_t* s_cur = s_gbase + Bpad;                   /* [Bpad] running cursor */
+  uint32_t* s_tot = s_cur + Bpad;                     /* [PART_BLOCK] chunk totals */
+  uint32_t* s_dest = s_tot + PART_BLOCK;              /* [T] per-slot global dest */
   PairT<V>* s_stage = (PairT<V>*)(s_dest + T);        /* [T] */
 
   const bool twolab = labels2 != nullptr;
@@ -686,28 +686,44 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
       }
     }
     __syncthreads();
-    /* exclusive block scan of s_hist[Bpad] (Hillis-Steele, in place) */
-    for (int d = 1; d < Bpad; d <<= 1) {
-      uint32_t add[16];
-      int na = 0;
-      for (int i = tid; i < Bpad; i += PART_BLOCK)
-        add[na++] = (i >= d) ? s_hist[i - d] : 0u;
+    /* exclusive block scan of s_hist[Bpad]: per-thread contiguous chunk
+     * scans + one 512-wide Hillis-Steele over the chunk totals (no private
+     * arrays -> no scratch spill) */
+    {
+      const int C = Bpad / PART_BLOCK; /* >= 1 (Bpad is padded to >= 512) */
+      const int b0 = tid * C;
+      uint32_t run = 0;
+      for (int j = 0; j < C; ++j) {
+        const uint32_t x = s_hist[b0 + j];
+        s_hist[b0 + j] = run;
+        run += x;
+      }
+      s_tot[tid] = run;
       __syncthreads();
-      na = 0;
-      for (int i = tid; i < Bpad; i += PART_BLOCK) s_hist[i] += add[na++];
-      __syncthreads();
-    }
-    /* s_hist now holds INCLUSIVE sums; derive exclusive + reserve globally */
-    for (int b = tid; b < B; b += PART_BLOCK) {
-      const uint32_t incl = s_hist[b];
-      const uint32_t excl = (b == 0) ? 0u : s_hist[b - 1];
-      const uint32_t cnt = incl - excl;
-      s_off[b] = excl;
-      s_cur[b] = excl;
-      if (cnt) s_gbase[b] = atomicAdd(&cursors[b], cnt);
+      uint32_t v = s_tot[tid];
+      for (int d = 1; d < PART_BLOCK; d <<= 1) {
+        const uint32_t o = (tid >= d) ? s_tot[tid - d] : 0u;
+        __syncthreads();
+        v += o;
+        s_tot[tid] = v;
+        __syncthreads();
+      }
+      const uint32_t chunk_excl = v - run;
+      for (int j = 0; j < C; ++j) s_hist[b0 + j] += chunk_excl;
     }
     __syncthreads();
-    const int valid = (int)s_hist[Bpad - 1];
+    const uint32_t total = s_tot[PART_BLOCK - 1];
+    /* reserve each nonempty bucket's run globally; store gbase - excl so the
+     * staged slot index alone gives the global destination */
+    for (int b = tid; b < B; b += PART_BLOCK) {
+      const uint32_t excl = s_hist[b];
+      const uint32_t incl = (b + 1 < Bpad) ? s_hist[b + 1] : total;
+      const uint32_t cnt = incl - excl;
+      s_cur[b] = excl;
+      if (cnt) s_gbase[b] = atomicAdd(&cursors[b], cnt) - excl;
+    }
+    __syncthreads();
+    const int valid = (int)total;
     /* bucket-ordered staging with per-slot destination */
 #pragma unroll
     for (int k = 0; k < RPT; ++k) {
@@ -715,7 +731,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
         const uint32_t pos = atomicAdd(&s_cur[rbk[k]], 1u);
         s_stage[pos].v = rv[k];
         s_stage[pos].lc = rlc[k];
-        s_dest[pos] = s_gbase[rbk[k]] + (pos - s_off[rbk[k]]);
+        s_dest[pos] = s_gbase[rbk[k]] + pos;
       }
     }
     __syncthreads();
@@ -858,10 +874,11 @@ PartPlan part_plan(const fh_call* c) {
   p.shift = shift;
   p.gpb = 1 << shift;
   p.lay = lay;
-  p.Bpad = 256;
+  p.Bpad = 512; /* = PART_BLOCK, so every thread owns >= 1 scan chunk entry */
   while (p.Bpad < p.B) p.Bpad <<= 1;
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
-  p.scatter_lds = (int64_t)4 * p.Bpad * 4 + (int64_t)T * 4 + (int64_t)T * sizeof(PairT<V>);
+  p.scatter_lds = (int64_t)3 * p.Bpad * 4 + 512 * 4 + (int64_t)T * 4 +
+                  (int64_t)T * sizeof(PairT<V>);
   if (p.scatter_lds > LDS_MAX) return p;
   int64_t off = 0;
   auto carve = [&](int64_t b) {
