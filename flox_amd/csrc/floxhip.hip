@@ -740,6 +740,85 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
   }
 }
 
+
+/* second-level scatter: rows of one super-bucket (already contiguous in
+ * `in`) are partitioned into their <= 64 fine buckets. Same tile machinery
+ * as k_part_scatter, with a fixed 64-entry histogram. */
+constexpr int PART_SUB = 64;
+
+template <typename V>
+__launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
+    const PairT<V>* __restrict__ in, const uint32_t* __restrict__ baseA,
+    int shift /* fine-bucket shift */, uint32_t* __restrict__ cursors,
+    PairT<V>* __restrict__ out) {
+  constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+  constexpr int RPT = T / PART_BLOCK;
+  extern __shared__ __attribute__((aligned(16))) char smem_p2[];
+  uint32_t* s_hist = (uint32_t*)smem_p2;       /* [PART_SUB+1] scanned in place */
+  uint32_t* s_gbase = s_hist + PART_SUB + 1;   /* [PART_SUB] */
+  uint32_t* s_cur = s_gbase + PART_SUB;        /* [PART_SUB] */
+  uint32_t* s_dest = s_cur + PART_SUB;         /* [T] */
+  PairT<V>* s_stage = (PairT<V>*)(s_dest + T); /* [T] */
+
+  const int sb = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int64_t r0 = baseA[sb], r1 = baseA[sb + 1];
+  const uint32_t lmask = (1u << shift) - 1u;
+
+  for (int64_t tile = r0 + (int64_t)blockIdx.x * T; tile < r1;
+       tile += (int64_t)gridDim.x * T) {
+    const int nt = (int)((r1 - tile < T) ? (r1 - tile) : T);
+    for (int i = tid; i <= PART_SUB; i += PART_BLOCK) s_hist[i] = 0;
+    __syncthreads();
+    V rv[RPT];
+    uint32_t rlc[RPT];
+    int rbk[RPT];
+#pragma unroll
+    for (int k = 0; k < RPT; ++k) {
+      const int idx = tid + k * PART_BLOCK;
+      rbk[k] = -1;
+      if (idx < nt) {
+        const PairT<V> pr = in[tile + idx];
+        rv[k] = pr.v;
+        rbk[k] = (int)(pr.lc >> shift);
+        rlc[k] = pr.lc & lmask;
+        atomicAdd(&s_hist[rbk[k]], 1u);
+      }
+    }
+    __syncthreads();
+    /* tiny exclusive scan (64 entries) by thread 0 */
+    if (tid == 0) {
+      uint32_t run = 0;
+      for (int b = 0; b <= PART_SUB; ++b) {
+        const uint32_t x = s_hist[b];
+        s_hist[b] = run;
+        run += x;
+      }
+    }
+    __syncthreads();
+    for (int b = tid; b < PART_SUB; b += PART_BLOCK) {
+      const uint32_t excl = s_hist[b];
+      const uint32_t cnt = s_hist[b + 1] - excl;
+      s_cur[b] = excl;
+      if (cnt) s_gbase[b] = atomicAdd(&cursors[sb * PART_SUB + b], cnt) - excl;
+    }
+    __syncthreads();
+    const int valid = (int)s_hist[PART_SUB];
+#pragma unroll
+    for (int k = 0; k < RPT; ++k) {
+      if (rbk[k] >= 0) {
+        const uint32_t pos = atomicAdd(&s_cur[rbk[k]], 1u);
+        s_stage[pos].v = rv[k];
+        s_stage[pos].lc = rlc[k];
+        s_dest[pos] = s_gbase[rbk[k]] + pos;
+      }
+    }
+    __syncthreads();
+    for (int i = tid; i < valid; i += PART_BLOCK) out[s_dest[i]] = s_stage[i];
+    __syncthreads();
+  }
+}
+
 /* per-bucket LDS-binned reduce of the scattered pairs; flush with global
  * atomics (a few chunks per bucket at most) into the FINAL bins */
 template <typename V, int OPS>
@@ -851,9 +930,11 @@ int init_outs(fh_call* c, int64_t nbins, hipStream_t stream) {
 
 struct PartPlan {
   int shift, gpb, B, Bpad;
+  bool two_level;      /* B > 64: scatter via <= 64 super-buckets first */
+  int B1;              /* super-bucket count (two_level) */
   BinLayout lay;       /* per-bucket bins (gpb entries) */
-  int64_t pairs_off, counts_off, base_off, cursors_off, bytes;
-  int64_t scatter_lds;
+  int64_t pairs_off, pairs2_off, counts_off, base_off, baseA_off, cursors_off, bytes;
+  int64_t scatter_lds, scatter2_lds;
   bool feasible;
 };
 
@@ -874,11 +955,17 @@ PartPlan part_plan(const fh_call* c) {
   p.shift = shift;
   p.gpb = 1 << shift;
   p.lay = lay;
+  p.two_level = p.B > PART_SUB;
+  p.B1 = p.two_level ? (p.B + PART_SUB - 1) / PART_SUB : 0;
+  /* pass-A bucket count: B1 when two-level, else the fine B */
+  const int bA = p.two_level ? p.B1 : p.B;
   p.Bpad = 512; /* = PART_BLOCK, so every thread owns >= 1 scan chunk entry */
-  while (p.Bpad < p.B) p.Bpad <<= 1;
+  while (p.Bpad < bA) p.Bpad <<= 1;
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
   p.scatter_lds = (int64_t)3 * p.Bpad * 4 + 512 * 4 + (int64_t)T * 4 +
                   (int64_t)T * sizeof(PairT<V>);
+  p.scatter2_lds = (int64_t)(3 * (PART_SUB + 1)) * 4 + (int64_t)T * 4 +
+                   (int64_t)T * sizeof(PairT<V>);
   if (p.scatter_lds > LDS_MAX) return p;
   int64_t off = 0;
   auto carve = [&](int64_t b) {
@@ -887,8 +974,10 @@ PartPlan part_plan(const fh_call* c) {
     return o;
   };
   p.pairs_off = carve(c->n * (int64_t)sizeof(PairT<V>));
+  p.pairs2_off = p.two_level ? carve(c->n * (int64_t)sizeof(PairT<V>)) : p.pairs_off;
   p.counts_off = carve((int64_t)p.B * 4);
   p.base_off = carve(((int64_t)p.B + 1) * 4);
+  p.baseA_off = carve(((int64_t)p.B1 + 1) * 4);
   p.cursors_off = carve((int64_t)p.B * 4);
   p.bytes = off;
   p.feasible = true;
@@ -902,8 +991,10 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
   char* scr = (char*)c->scratch;
   uint32_t* counts = (uint32_t*)(scr + pp.counts_off);
   uint32_t* based = (uint32_t*)(scr + pp.base_off);
+  uint32_t* baseAd = (uint32_t*)(scr + pp.baseA_off);
   uint32_t* cursors = (uint32_t*)(scr + pp.cursors_off);
   PairT<V>* pairs = (PairT<V>*)(scr + pp.pairs_off);
+  PairT<V>* pairs2 = (PairT<V>*)(scr + pp.pairs2_off);
 
   FH_CHECK(hipMemsetAsync(counts, 0, (int64_t)pp.B * 4, stream));
   {
@@ -928,8 +1019,27 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
   h_base[pp.B] = acc;
   FH_CHECK(hipMemcpyAsync(based, h_base, ((int64_t)pp.B + 1) * 4,
                           hipMemcpyHostToDevice, stream));
-  FH_CHECK(hipMemcpyAsync(cursors, h_base, (int64_t)pp.B * 4,
-                          hipMemcpyHostToDevice, stream));
+  /* super-bucket directory: sums of PART_SUB fine buckets */
+  static thread_local uint32_t h_baseA[4097];
+  uint32_t maxA = 0;
+  if (pp.two_level) {
+    for (int sbi = 0; sbi <= pp.B1; ++sbi) {
+      const int fb = sbi * PART_SUB;
+      h_baseA[sbi] = h_base[fb < pp.B ? fb : pp.B];
+    }
+    for (int sbi = 0; sbi < pp.B1; ++sbi) {
+      const uint32_t cnt = h_baseA[sbi + 1] - h_baseA[sbi];
+      if (cnt > maxA) maxA = cnt;
+    }
+    FH_CHECK(hipMemcpyAsync(baseAd, h_baseA, ((int64_t)pp.B1 + 1) * 4,
+                            hipMemcpyHostToDevice, stream));
+    /* pass-A cursors: super-bucket bases (first B1 words of cursors) */
+    FH_CHECK(hipMemcpyAsync(cursors, h_baseA, (int64_t)pp.B1 * 4,
+                            hipMemcpyHostToDevice, stream));
+  } else {
+    FH_CHECK(hipMemcpyAsync(cursors, h_base, (int64_t)pp.B * 4,
+                            hipMemcpyHostToDevice, stream));
+  }
 
   {
     auto kern = k_part_scatter<V, L>;
@@ -939,10 +1049,29 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
     int64_t wb = (c->n + T - 1) / T;
     int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
+    const int shA = pp.two_level ? pp.shift + 6 : pp.shift;
+    const int bA = pp.two_level ? pp.B1 : pp.B;
     hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
                        stream, (const V*)c->values, (const L*)c->labels,
                        (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       pp.shift, pp.B, pp.Bpad, cursors, pairs);
+                       shA, bA, pp.Bpad, cursors, pairs);
+    FH_CHECK(hipGetLastError());
+  }
+  if (pp.two_level) {
+    /* fine-bucket cursors, then the in-super-bucket scatter */
+    FH_CHECK(hipMemcpyAsync(cursors, h_base, (int64_t)pp.B * 4,
+                            hipMemcpyHostToDevice, stream));
+    auto kern2 = k_part_scatter2<V>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern2,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)pp.scatter2_lds));
+    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    int tiles_x = (int)((maxA + T - 1) / T);
+    if (tiles_x > 64) tiles_x = 64;
+    if (tiles_x < 1) tiles_x = 1;
+    hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK),
+                       pp.scatter2_lds, stream, pairs, baseAd, pp.shift,
+                       cursors, pairs2);
     FH_CHECK(hipGetLastError());
   }
   {
@@ -958,10 +1087,10 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     int maxchunks = (int)((maxc + chunk - 1) / chunk);
     if (maxchunks < 1) maxchunks = 1;
     hipLaunchKernelGGL(kern, dim3(maxchunks, pp.B), dim3(BLOCK_LDS),
-                       pp.lay.bytes, stream, pairs, based, chunk, pp.gpb,
-                       pp.shift, c->ngroups, c->means, skipnan, pp.lay,
-                       c->out_sum, c->out_count, c->out_present, c->out_min,
-                       c->out_max, c->out_nanflag);
+                       pp.lay.bytes, stream, pp.two_level ? pairs2 : pairs,
+                       based, chunk, pp.gpb, pp.shift, c->ngroups, c->means,
+                       skipnan, pp.lay, c->out_sum, c->out_count,
+                       c->out_present, c->out_min, c->out_max, c->out_nanflag);
     FH_CHECK(hipGetLastError());
   }
   if (OPS & (B_MIN | B_MAX)) {
