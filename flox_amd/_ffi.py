@@ -28,6 +28,8 @@ SET_MAX_FULL = 5
 SET_MAX_COUNT = 6
 SET_SSD = 7
 SET_PROD = 8
+SET_IDXMIN = 9
+SET_IDXMAX = 10
 FLAG_SKIPNAN = 1
 FLAG_FORCE_LDS = 2
 FLAG_FORCE_ATOMIC = 4
@@ -71,6 +73,8 @@ class FhCall(ctypes.Structure):
         ("ldm", ctypes.c_int64),
         ("chunk_offsets", ctypes.c_void_p),
         ("nchunks", ctypes.c_int64),
+        ("target", ctypes.c_void_p),
+        ("row_offset", ctypes.c_int64),
     ]
 
 

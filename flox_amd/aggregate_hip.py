@@ -30,6 +30,8 @@ from . import _ffi
 from ._ffi import (
     FLAG_SKIPNAN,
     SET_COUNT,
+    SET_IDXMAX,
+    SET_IDXMIN,
     SET_MAX_COUNT,
     SET_MAX_FULL,
     SET_MIN_COUNT,
@@ -60,7 +62,12 @@ _SET_MEMBERS = {
     SET_MAX_COUNT: ("max", "count"),
     SET_SSD: ("sum",),
     SET_PROD: ("sum", "count", "present"),
+    SET_IDXMIN: ("idx", "count", "present"),
+    SET_IDXMAX: ("idx", "count", "present"),
 }
+
+IDX_SENTINEL_MIN = (1 << 63) - 1  # untouched IDXMIN bin
+IDX_SENTINEL_MAX = -1             # untouched IDXMAX bin
 
 
 def _acc_dtype(value_dtype: torch.dtype) -> torch.dtype:
@@ -85,6 +92,8 @@ def grouped_partials(
     labels2: torch.Tensor | None = None,
     grp_shape: tuple[int, int] | None = None,
     means: torch.Tensor | None = None,
+    target: torch.Tensor | None = None,
+    row_offset: int = 0,
     skipnan: bool = False,
     force_path: int = 0,
 ) -> dict[str, torch.Tensor]:
@@ -93,6 +102,7 @@ def grouped_partials(
     values, labels: contiguous 1-D CUDA tensors of equal length.
     labels2/grp_shape: fused 2-D groupby (reference factorize.py:102-108).
     means: f64[ngroups] for SET_SSD (var pass 2).
+    target/row_offset: SET_IDXMIN/IDXMAX (arg-reductions, first/last).
     force_path: 0 auto, 1 LDS-binned, 2 global-atomic (testing).
     """
     lib = _ffi.load_library()
@@ -102,6 +112,7 @@ def grouped_partials(
         dev0 = values.device
         acc = _acc_dtype(values.dtype)
         empty = {
+            "idx": torch.empty(0, dtype=torch.int64, device=dev0),
             "sum": torch.empty(0, dtype=acc, device=dev0),
             "count": torch.empty(0, dtype=torch.int64, device=dev0),
             "present": torch.empty(0, dtype=torch.int32, device=dev0),
@@ -142,9 +153,17 @@ def grouped_partials(
         assert means.dtype == torch.float64 and means.is_cuda
         means = means.contiguous()
         c.means = means.data_ptr()
+    if target is not None:
+        assert target.dtype == values.dtype and target.is_cuda
+        target = target.contiguous()
+        c.target = target.data_ptr()
+    c.row_offset = row_offset
 
     members = _SET_MEMBERS[op_set]
     out: dict[str, torch.Tensor] = {}
+    if "idx" in members:
+        out["idx"] = torch.empty(ngroups, dtype=torch.int64, device=dev)
+        c.out_sum = out["idx"].data_ptr()
     if "sum" in members:
         out["sum"] = torch.empty(ngroups, dtype=_acc_dtype(values.dtype), device=dev)
         if op_set == SET_SSD:
@@ -181,7 +200,7 @@ def grouped_partials(
     # keep tensors alive until the stream consumes them (torch caching allocator
     # ties lifetime to the stream via recorded events only for torch ops; we
     # record explicitly)
-    for t in (values, labels, labels2, means, scratch):
+    for t in (values, labels, labels2, means, target, scratch):
         if isinstance(t, torch.Tensor):
             t.record_stream(torch.cuda.current_stream(dev))
     return out

@@ -74,6 +74,37 @@ REDUCTIONS: dict[str, Aggregation] = {
     # var family: two fused passes (SUM_COUNT then SSD); see core._reduce_var
     "var": Aggregation("var", None),
     "nanvar": Aggregation("nanvar", None, skipnan=True),
+    # arg-reductions: extremum pass then min-index-among-matches pass
+    # (reference argmax/argmin/nanarg* recipes, aggregations.py:582-649)
+    "argmax": Aggregation("argmax", None),
+    "argmin": Aggregation("argmin", None),
+    "nanargmax": Aggregation("nanargmax", None, skipnan=True),
+    "nanargmin": Aggregation("nanargmin", None, skipnan=True),
+    # first/last: index extremum + gather (reference aggregations.py:635-649)
+    "first": Aggregation(
+        "first", _ffi.SET_IDXMIN,
+        combine={"idx": "min", "count": "sum", "present": "max"},
+        preserves_dtype=True,
+    ),
+    "last": Aggregation(
+        "last", _ffi.SET_IDXMAX,
+        combine={"idx": "max", "count": "sum", "present": "max"},
+        preserves_dtype=True,
+    ),
+    "nanfirst": Aggregation(
+        "nanfirst", _ffi.SET_IDXMIN, skipnan=True,
+        combine={"idx": "min", "count": "sum", "present": "max"},
+        preserves_dtype=True,
+    ),
+    "nanlast": Aggregation(
+        "nanlast", _ffi.SET_IDXMAX, skipnan=True,
+        combine={"idx": "max", "count": "sum", "present": "max"},
+        preserves_dtype=True,
+    ),
+    # bool reductions via min/max of the 0/1-cast input
+    # (reference aggregations.py:651-676 all_/any_)
+    "any": Aggregation("any", _ffi.SET_MAX_COUNT, combine={"max": "max", "count": "sum"}),
+    "all": Aggregation("all", _ffi.SET_MIN_COUNT, combine={"min": "min", "count": "sum"}),
     "std": Aggregation("std", None),
     "nanstd": Aggregation("nanstd", None, skipnan=True),
 }

@@ -148,6 +148,7 @@ def groupby_reduce(
     engine: str = "hip",
     finalize_kwargs: dict[str, Any] | None = None,
     distributed_combine: bool | None = None,
+    shard_row_offset: int = 0,
 ):
     """Grouped reduction with flox semantics on MI355X. Returns (result, *groups).
 
@@ -171,7 +172,8 @@ def groupby_reduce(
 
     return_numpy = not isinstance(array, torch.Tensor)
     arr = _as_device_tensor(array, device)
-    if arr.dtype == torch.bool:
+    was_bool = arr.dtype == torch.bool
+    if was_bool:
         arr = arr.to(torch.int64)  # reference core.py:916-917
     in_np_dtype = _np_dtype(arr.dtype)
     bys = tuple(_as_device_tensor(b, device) for b in by)
@@ -225,10 +227,11 @@ def groupby_reduce(
     ddof = (finalize_kwargs or {}).get("ddof", 0)
 
     if lead_M == 1:
-        def run_set(op_set, skipnan, means=None):
+        def run_set(op_set, skipnan, means=None, target=None):
             return grouped_partials(
                 op_set, vals, labels, ngroups, skipnan=skipnan,
                 labels2=labels2, grp_shape=grp_pair, means=means,
+                target=target, row_offset=shard_row_offset,
             )
     else:
         # column path: grouped dims to the front, lead dims flattened as
@@ -261,12 +264,62 @@ def groupby_reduce(
         scodes64, perm64 = torch.sort(codes_full, stable=True)
         scodes, perm = scodes64.to(torch.int32), perm64.to(torch.int32)
 
-        def run_set(op_set, skipnan, means=None):
+        if func in ("argmax", "argmin", "nanargmax", "nanargmin", "first",
+                    "last", "nanfirst", "nanlast"):
+            raise NotImplementedError(f"{func} with leading array dims: next row")
+
+        def run_set(op_set, skipnan, means=None, target=None):
             return grouped_partials_cols(
                 op_set, vt, scodes, perm, ngroups, skipnan=skipnan, means=means
             )
 
-    if func in ("var", "nanvar", "std", "nanstd"):
+    if func in ("argmax", "argmin", "nanargmax", "nanargmin"):
+        # pass 1: the per-group extremum; pass 2: the smallest row index whose
+        # value matches it (ties -> first occurrence, like np.argmax; a NaN
+        # target matches NaN rows, so non-skip arg* land on the first NaN)
+        skip = agg.skipnan
+        ismax = "max" in func
+        if skip:
+            p1 = run_set(_ffi.SET_MAX_COUNT if ismax else _ffi.SET_MIN_COUNT, True)
+        else:
+            p1 = run_set(_ffi.SET_MAX_FULL if ismax else _ffi.SET_MIN_FULL, False)
+        if dist_on:
+            distributed.all_reduce_(p1["max" if ismax else "min"], "max" if ismax else "min")
+            distributed.all_reduce_(p1["count"], "sum")
+            if "nanflag" in p1:
+                distributed.all_reduce_(p1["nanflag"], "max")
+        target = p1["max" if ismax else "min"]
+        if "nanflag" in p1 and arr.dtype.is_floating_point:
+            target = torch.where(p1["nanflag"] != 0, torch.full_like(target, float("nan")), target)
+        p2 = run_set(_ffi.SET_IDXMIN, skip, target=target)
+        if dist_on:
+            distributed.all_reduce_(p2["idx"], "min")
+        idx = p2["idx"]
+        sentinel = (1 << 63) - 1
+        result = idx
+        empty_mask = (p2["present"] == 0) | (idx == sentinel)
+        counts_for_mask = p2["count"]
+    elif func in ("first", "last", "nanfirst", "nanlast"):
+        p = run_set(agg.op_set, agg.skipnan)
+        if dist_on:
+            distributed.combine_partials(p, agg.combine)
+        idx = p["idx"]
+        sentinel = ((1 << 63) - 1) if func in ("first", "nanfirst") else -1
+        valid = idx != sentinel
+        local = idx - shard_row_offset
+        n_local = vals.numel()
+        safe = torch.clamp(local, 0, max(n_local - 1, 0))
+        if dist_on:
+            # the owning rank contributes the value; others contribute exact 0
+            owner = valid & (local >= 0) & (local < n_local)
+            gathered = torch.where(owner, vals[safe], torch.zeros((), dtype=vals.dtype, device=device))
+            distributed.all_reduce_(gathered, "sum")
+        else:
+            gathered = vals[safe]
+        result = gathered
+        empty_mask = ~valid
+        counts_for_mask = p["count"]
+    elif func in ("var", "nanvar", "std", "nanstd"):
         skip = agg.skipnan
         p1 = run_set(_ffi.SET_SUM_COUNT, skip)
         if dist_on:
@@ -311,6 +364,12 @@ def groupby_reduce(
                 nan_t = torch.full_like(result, float("nan"))
                 result = torch.where(p["nanflag"] != 0, nan_t, result)
             empty_mask = (p["present"] == 0) if "present" in p else (p["count"] == 0)
+            counts_for_mask = p["count"]
+        elif func in ("any", "all"):
+            if not was_bool:
+                raise NotImplementedError("any/all support bool input only")
+            result = (p["max" if func == "any" else "min"] == 1)
+            empty_mask = p["count"] == 0
             counts_for_mask = p["count"]
         else:  # pragma: no cover
             raise NotImplementedError(func)

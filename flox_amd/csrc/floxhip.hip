@@ -60,6 +60,8 @@ enum {
   B_NANFLAG = 32,
   B_SSD = 64,
   B_PROD = 128,
+  B_IDXMIN = 256,
+  B_IDXMAX = 512,
 };
 
 __host__ __device__ constexpr int set_bits(int op_set) {
@@ -73,6 +75,8 @@ __host__ __device__ constexpr int set_bits(int op_set) {
     case FH_SET_MAX_COUNT: return B_MAX | B_CNT;
     case FH_SET_SSD: return B_SSD;
     case FH_SET_PROD: return B_PROD | B_CNT | B_PRESENT;
+    case FH_SET_IDXMIN: return B_IDXMIN | B_CNT | B_PRESENT;
+    case FH_SET_IDXMAX: return B_IDXMAX | B_CNT | B_PRESENT;
     default: return 0;
   }
 }
@@ -151,6 +155,13 @@ __device__ __forceinline__ void enc_max(uint32_t* p, uint32_t v) { atomicMax(p, 
 __device__ __forceinline__ void enc_max(uint64_t* p, uint64_t v) {
   atomicMax(reinterpret_cast<unsigned long long*>(p), (unsigned long long)v);
 }
+__device__ __forceinline__ void idx_min(int64_t* p, int64_t v) {
+  atomicMin(reinterpret_cast<long long*>(p), (long long)v);
+}
+__device__ __forceinline__ void idx_max(int64_t* p, int64_t v) {
+  atomicMax(reinterpret_cast<long long*>(p), (long long)v);
+}
+
 /* CAS product (prod is rare; contention-tolerant CAS loop) */
 __device__ __forceinline__ void acc_mul(double* p, double v) {
   unsigned long long* u = reinterpret_cast<unsigned long long*>(p);
@@ -188,7 +199,7 @@ __host__ __device__ BinLayout bin_layout(int bits, int64_t ngroups, int64_t cnt_
     off += ((ngroups * elem + 255) / 256) * 256; /* 256-B aligned sections */
     return o;
   };
-  L.sum_off = (bits & (B_SUM | B_SSD | B_PROD)) ? carve(8) : -1;
+  L.sum_off = (bits & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ? carve(8) : -1;
   L.cnt_off = (bits & B_CNT) ? carve(cnt_elem_size) : -1;
   L.present_off = (bits & B_PRESENT) ? carve(4) : -1;
   L.minmax_off = (bits & (B_MIN | B_MAX)) ? carve(sizeof(typename Traits<V>::Enc)) : -1;
@@ -202,11 +213,14 @@ template <typename V, typename L, int OPS>
 __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
     const V* __restrict__ values, const L* __restrict__ labels,
     const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
-    int64_t g1, const double* __restrict__ means, int skipnan,
+    int64_t g1, const double* __restrict__ means,
+    const V* __restrict__ target, int64_t row_offset, int skipnan,
     char* __restrict__ slab, BinLayout lay) {
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
-  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
+  using SumT = typename std::conditional<
+      (OPS & B_SSD) != 0, double,
+      typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr int VEC = TR::VEC;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
@@ -222,6 +236,8 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
   for (int64_t g = tid; g < ngroups; g += blockDim.x) {
     if (OPS & (B_SUM | B_SSD)) s_sum[g] = (SumT)0;
     if (IS_PROD) s_sum[g] = (SumT)1;
+    if (OPS & B_IDXMIN) s_sum[g] = (SumT)INT64_MAX;
+    if (OPS & B_IDXMAX) s_sum[g] = (SumT)(-1);
     if (OPS & B_CNT) s_cnt[g] = 0u;
     if (OPS & B_PRESENT) s_present[g] = 0u;
     if (OPS & B_MIN) s_mm[g] = (Enc)~(Enc)0;
@@ -232,7 +248,7 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
 
   const bool twolab = labels2 != nullptr;
 
-  auto process = [&](V v, int64_t l0raw, int64_t l1raw) {
+  auto process = [&](V v, int64_t l0raw, int64_t l1raw, int64_t row) {
     uint64_t code;
     if (twolab) {
       if ((uint64_t)l0raw >= (uint64_t)g0 || (uint64_t)l1raw >= (uint64_t)g1) return;
@@ -249,6 +265,20 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
     if (OPS & B_SSD) {
       double d = (double)v - means[code];
       atomicAdd((double*)s_sum + code, d * d);
+    }
+    if (OPS & (B_IDXMIN | B_IDXMAX)) {
+      /* candidate: every surviving row, or rows matching the per-group
+       * target value (NaN target matches NaN rows: argmax over data with
+       * NaNs lands on the first NaN, as np.argmax does) */
+      bool match = true;
+      if (target) {
+        const V t = target[code];
+        match = vnan ? TR::isnan_(t) : (!TR::isnan_(t) && v == t);
+      }
+      if (match) {
+        if (OPS & B_IDXMIN) idx_min((int64_t*)s_sum + code, row + row_offset);
+        if (OPS & B_IDXMAX) idx_max((int64_t*)s_sum + code, row + row_offset);
+      }
     }
     if (OPS & B_CNT) {
       if (!vnan) atomicAdd(&s_cnt[code], 1u);
@@ -272,15 +302,16 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
     if (twolab) {
       Vec<L, VEC> lv2 = *reinterpret_cast<const Vec<L, VEC>*>(labels2 + i * VEC);
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], (int64_t)lv2.v[k]);
+      for (int k = 0; k < VEC; ++k)
+        process(vv.v[k], (int64_t)lv.v[k], (int64_t)lv2.v[k], i * VEC + k);
     } else {
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], 0);
+      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], 0, i * VEC + k);
     }
   }
   /* tail */
   for (int64_t i = nvec * VEC + gtid; i < n; i += stride) {
-    process(values[i], (int64_t)labels[i], twolab ? (int64_t)labels2[i] : 0);
+    process(values[i], (int64_t)labels[i], twolab ? (int64_t)labels2[i] : 0, i);
   }
 
   __syncthreads();
@@ -303,13 +334,17 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
                           void* out_min, void* out_max, uint32_t* out_nanflag) {
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
-  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
+  using SumT = typename std::conditional<
+      (OPS & B_SSD) != 0, double,
+      typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
   const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (g >= ngroups) return;
 
   SumT s = IS_PROD ? (SumT)1 : (SumT)0;
+  if (OPS & B_IDXMIN) s = (SumT)INT64_MAX;
+  if (OPS & B_IDXMAX) s = (SumT)(-1);
   int64_t c = 0;
   uint32_t p = 0, nf = 0;
   Enc mn = (Enc)~(Enc)0, mx = (Enc)0;
@@ -317,6 +352,14 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
     const char* blk = slab + (int64_t)b * lay.bytes;
     if (OPS & (B_SUM | B_SSD)) s += ((const SumT*)(blk + lay.sum_off))[g];
     if (IS_PROD) s *= ((const SumT*)(blk + lay.sum_off))[g];
+    if (OPS & B_IDXMIN) {
+      const SumT x = ((const SumT*)(blk + lay.sum_off))[g];
+      s = x < s ? x : s;
+    }
+    if (OPS & B_IDXMAX) {
+      const SumT x = ((const SumT*)(blk + lay.sum_off))[g];
+      s = x > s ? x : s;
+    }
     if (OPS & B_CNT) c += (int64_t)((const uint32_t*)(blk + lay.cnt_off))[g];
     if (OPS & B_PRESENT) p |= ((const uint32_t*)(blk + lay.present_off))[g];
     if (OPS & B_MIN) {
@@ -330,7 +373,7 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
     if (OPS & B_NANFLAG) nf |= ((const uint32_t*)(blk + lay.nanflag_off))[g];
   }
   const bool present = (OPS & B_PRESENT) ? (p != 0) : (c != 0);
-  if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)out_sum)[g] = s;
+  if (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ((SumT*)out_sum)[g] = s;
   if (OPS & B_CNT) out_count[g] = c;
   if (OPS & B_PRESENT) out_present[g] = p;
   if (OPS & B_MIN) ((V*)out_min)[g] = present ? TR::dec(mn) : TR::pos_inf();
@@ -345,9 +388,10 @@ template <typename V, typename L, int OPS>
 __launch_bounds__(BLOCK_ATOMIC) __global__ void k_reduce_atomic(
     const V* __restrict__ values, const L* __restrict__ labels,
     const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
-    int64_t g1, const double* __restrict__ means, int skipnan, void* out_sum,
-    int64_t* out_count, uint32_t* out_present, void* out_min, void* out_max,
-    uint32_t* out_nanflag) {
+    int64_t g1, const double* __restrict__ means,
+    const V* __restrict__ target, int64_t row_offset, int skipnan,
+    void* out_sum, int64_t* out_count, uint32_t* out_present, void* out_min,
+    void* out_max, uint32_t* out_nanflag) {
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
   using Enc = typename TR::Enc;
@@ -355,7 +399,7 @@ __launch_bounds__(BLOCK_ATOMIC) __global__ void k_reduce_atomic(
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
   const bool twolab = labels2 != nullptr;
 
-  auto process = [&](V v, int64_t l0raw, int64_t l1raw) {
+  auto process = [&](V v, int64_t l0raw, int64_t l1raw, int64_t row) {
     uint64_t code;
     if (twolab) {
       if ((uint64_t)l0raw >= (uint64_t)g0 || (uint64_t)l1raw >= (uint64_t)g1) return;
@@ -372,6 +416,17 @@ __launch_bounds__(BLOCK_ATOMIC) __global__ void k_reduce_atomic(
     if (OPS & B_SSD) {
       double d = (double)v - means[code];
       atomicAdd(&((double*)out_sum)[code], d * d);
+    }
+    if (OPS & (B_IDXMIN | B_IDXMAX)) {
+      bool match = true;
+      if (target) {
+        const V t = target[code];
+        match = vnan ? TR::isnan_(t) : (!TR::isnan_(t) && v == t);
+      }
+      if (match) {
+        if (OPS & B_IDXMIN) idx_min((int64_t*)out_sum + code, row + row_offset);
+        if (OPS & B_IDXMAX) idx_max((int64_t*)out_sum + code, row + row_offset);
+      }
     }
     if (OPS & B_CNT) {
       if (!vnan)
@@ -396,14 +451,15 @@ __launch_bounds__(BLOCK_ATOMIC) __global__ void k_reduce_atomic(
     if (twolab) {
       Vec<L, VEC> lv2 = *reinterpret_cast<const Vec<L, VEC>*>(labels2 + i * VEC);
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], (int64_t)lv2.v[k]);
+      for (int k = 0; k < VEC; ++k)
+        process(vv.v[k], (int64_t)lv.v[k], (int64_t)lv2.v[k], i * VEC + k);
     } else {
 #pragma unroll
-      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], 0);
+      for (int k = 0; k < VEC; ++k) process(vv.v[k], (int64_t)lv.v[k], 0, i * VEC + k);
     }
   }
   for (int64_t i = nvec * VEC + gtid; i < n; i += stride) {
-    process(values[i], (int64_t)labels[i], twolab ? (int64_t)labels2[i] : 0);
+    process(values[i], (int64_t)labels[i], twolab ? (int64_t)labels2[i] : 0, i);
   }
 }
 
@@ -439,7 +495,9 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
     void* out_max, uint32_t* out_nanflag) {
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
-  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
+  using SumT = typename std::conditional<
+      (OPS & B_SSD) != 0, double,
+      typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
 
@@ -830,7 +888,9 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
     void* out_max, uint32_t* out_nanflag) {
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
-  using SumT = typename std::conditional<(OPS & B_SSD) != 0, double, Acc>::type;
+  using SumT = typename std::conditional<
+      (OPS & B_SSD) != 0, double,
+      typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
 
@@ -853,6 +913,8 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
   for (int g = tid; g < ng_here; g += blockDim.x) {
     if (OPS & (B_SUM | B_SSD)) s_sum[g] = (SumT)0;
     if (IS_PROD) s_sum[g] = (SumT)1;
+    if (OPS & B_IDXMIN) s_sum[g] = (SumT)INT64_MAX;
+    if (OPS & B_IDXMAX) s_sum[g] = (SumT)(-1);
     if (OPS & B_CNT) s_cnt[g] = 0u;
     if (OPS & B_PRESENT) s_present[g] = 0u;
     if (OPS & B_MIN) s_mm[g] = (Enc)~(Enc)0;
@@ -944,6 +1006,7 @@ PartPlan part_plan(const fh_call* c) {
   p.feasible = false;
   if (c->n >= ((int64_t)1 << 31) || c->ngroups <= 0) return p;
   const int bits = set_bits(c->op_set);
+  if (bits & (B_IDXMIN | B_IDXMAX)) return p; /* pairs carry no row index */
   int shift = 13;
   while (shift > 8 && bin_layout<V>(bits, (int64_t)1 << shift, 4).bytes > 120 * 1024)
     shift--;
@@ -1135,7 +1198,8 @@ int launch_typed(fh_call* c) {
     hipLaunchKernelGGL(kern, dim3(nblocks), dim3(BLOCK_LDS), lay.bytes, stream,
                        (const V*)c->values, (const L*)c->labels,
                        (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       c->means, (int)skipnan, (char*)c->scratch, lay);
+                       c->means, (const V*)c->target, c->row_offset,
+                       (int)skipnan, (char*)c->scratch, lay);
     FH_CHECK(hipGetLastError());
     int cb = (int)((c->ngroups + 255) / 256);
     hipLaunchKernelGGL((k_combine<V, OPS>), dim3(cb), dim3(256), 0, stream,
@@ -1158,6 +1222,13 @@ int launch_typed(fh_call* c) {
   const int64_t ng = c->ngroups;
   if (OPS & (B_SUM | B_SSD))
     FH_CHECK(hipMemsetAsync(c->out_sum, 0, ng * 8, stream));
+  if (OPS & (B_IDXMIN | B_IDXMAX)) {
+    int fb = (int)((ng + 255) / 256);
+    hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                       (int64_t*)c->out_sum, ng,
+                       (OPS & B_IDXMIN) ? INT64_MAX : (int64_t)-1);
+    FH_CHECK(hipGetLastError());
+  }
   if (OPS & B_PROD) {
     int fb = (int)((ng + 255) / 256);
     if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
@@ -1181,7 +1252,8 @@ int launch_typed(fh_call* c) {
   hipLaunchKernelGGL((k_reduce_atomic<V, L, OPS>), dim3(nblocks),
                      dim3(BLOCK_ATOMIC), 0, stream, (const V*)c->values,
                      (const L*)c->labels, (const L*)c->labels2, c->n, c->ngroups,
-                     c->g0, c->g1, c->means, (int)skipnan, c->out_sum,
+                     c->g0, c->g1, c->means, (const V*)c->target, c->row_offset,
+                     (int)skipnan, c->out_sum,
                      c->out_count, c->out_present, c->out_min, c->out_max,
                      c->out_nanflag);
   FH_CHECK(hipGetLastError());
@@ -1211,6 +1283,10 @@ int dispatch_ops(fh_call* c) {
     case B_SSD: return launch_typed<V, L, B_SSD>(c);
     case B_PROD | B_CNT | B_PRESENT:
       return launch_typed<V, L, B_PROD | B_CNT | B_PRESENT>(c);
+    case B_IDXMIN | B_CNT | B_PRESENT:
+      return launch_typed<V, L, B_IDXMIN | B_CNT | B_PRESENT>(c);
+    case B_IDXMAX | B_CNT | B_PRESENT:
+      return launch_typed<V, L, B_IDXMAX | B_CNT | B_PRESENT>(c);
     default: return 4;
   }
 }

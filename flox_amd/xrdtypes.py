@@ -28,13 +28,19 @@ NAN_SKIP = {
 }
 
 
+ARG_FUNCS = {"argmax", "argmin", "nanargmax", "nanargmin"}
+BOOL_FUNCS = {"any", "all"}
+
+
 def final_dtype(func: str, array_dtype: np.dtype, dtype=None) -> np.dtype:
     """Output dtype (reference xrdtypes.py:153-186)."""
     array_dtype = np.dtype(array_dtype)
     if dtype is not None:
         return np.dtype(dtype)
-    if func == "count":
+    if func == "count" or func in ARG_FUNCS:
         return np.dtype(np.intp)
+    if func in BOOL_FUNCS:
+        return np.dtype(bool)
     if func in FLOAT_FUNCS:
         if array_dtype.kind in "fc":
             return array_dtype
@@ -56,6 +62,10 @@ def fill_default(func: str, out_dtype: np.dtype):
     out_dtype = np.dtype(out_dtype)
     if func == "count":
         return 0
+    if func in ARG_FUNCS:
+        return -1  # reference aggregations.py:589/617 final_fill_value=-1
+    if func in BOOL_FUNCS:
+        return False  # reference aggregations.py:651-676
     if func == "prod":
         return 1
     if out_dtype.kind in "fc":

@@ -46,6 +46,17 @@ enum fh_opset {
   FH_SET_MAX_COUNT = 6,         /* max + count: nanmax */
   FH_SET_SSD = 7,               /* sum of (x - mean[g])^2, f64: var pass 2 */
   FH_SET_PROD = 8,              /* product + count + present (CAS loop) */
+  /* index extremum + count + present: the second pass of arg-reductions
+   * (reference chunk_argreduce core.py:157-211 / argmax-argmin recipes
+   * aggregations.py:582-649) and of first/last/nanfirst/nanlast. With
+   * `target` set, only rows whose value equals target[group] (NaN matches
+   * NaN) are candidates; without it every (valid, and non-NaN when
+   * FH_SKIPNAN) row is. IDXMIN keeps the smallest row index, IDXMAX the
+   * largest; out_count/out_present as usual; the index (+ row_offset, for
+   * multi-GPU shards) lands in out_sum as int64, -2^63.. sentinel for
+   * untouched groups handled by init (IDXMIN: INT64_MAX, IDXMAX: -1). */
+  FH_SET_IDXMIN = 9,
+  FH_SET_IDXMAX = 10,
 };
 
 /* flags */
@@ -102,6 +113,10 @@ typedef struct fh_call {
    * slab); when NULL the launcher row-splits with a partials slab instead */
   const void* chunk_offsets;
   int64_t nchunks;
+  /* FH_SET_IDXMIN/IDXMAX: per-group match targets (value dtype, nullable)
+   * and the global row offset of this shard (multi-GPU arg-reductions) */
+  const void* target;
+  int64_t row_offset;
 } fh_call;
 
 /* scratch requirement for this call (0 when the global-atomic path is used) */

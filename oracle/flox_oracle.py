@@ -20,7 +20,22 @@ import math
 
 import numpy as np
 
+# arg/first/last/any/all parity anchors on plain numpy per-group semantics:
+# the reference pins these against getattr(np, func) on single-group data in
+# its own tests (test_core.py:222-385) and cannot run them through
+# engine="flox" (core.py:856-859 raises for argreductions; aggregate_flox has
+# no first/last/any/all and numpy_groupies is absent in this container).
 ALL_FUNCS = (
+    "argmax",
+    "argmin",
+    "nanargmax",
+    "nanargmin",
+    "first",
+    "last",
+    "nanfirst",
+    "nanlast",
+    "any",
+    "all",
     "count",
     "sum",
     "nansum",
@@ -39,10 +54,12 @@ ALL_FUNCS = (
 )
 
 _FLOAT_FUNCS = {"mean", "nanmean", "var", "nanvar", "std", "nanstd"}
+_ARG_FUNCS = {"argmax", "argmin", "nanargmax", "nanargmin"}
+_POS_FUNCS = {"first", "last", "nanfirst", "nanlast"}
 _NAN_SKIP = {"nansum", "nanprod", "nanmean", "nanvar", "nanstd", "nanmin", "nanmax", "count"}
 # funcs whose output dtype equals the input dtype (reference: preserves_dtype=True,
 # aggregations.py:529-546)
-_PRESERVES_DTYPE = {"min", "nanmin", "max", "nanmax"}
+_PRESERVES_DTYPE = {"min", "nanmin", "max", "nanmax", "first", "last", "nanfirst", "nanlast"}
 
 
 def _isnull(a: np.ndarray) -> np.ndarray:
@@ -108,8 +125,10 @@ def _final_dtype(func: str, array_dtype: np.dtype, dtype=None) -> np.dtype:
     """Output dtype rules: xrdtypes.py:153-186 (_normalize_dtype/_maybe_promote_int)."""
     if dtype is not None:
         return np.dtype(dtype)
-    if func == "count":
+    if func == "count" or func in _ARG_FUNCS:
         return np.dtype(np.intp)
+    if func in ("any", "all"):
+        return np.dtype(bool)
     if func in _FLOAT_FUNCS:
         # "mean, std, var always result in floating, preserving a floating input
         # dtype" (xrdtypes.py:161-167)
@@ -132,6 +151,10 @@ def _fill_default(func: str, out_dtype: np.dtype):
     ints per xrdtypes.py:188-209)."""
     if func == "count":
         return 0
+    if func in _ARG_FUNCS:
+        return -1
+    if func in ("any", "all"):
+        return False
     if func in ("prod",):
         return 1
     if out_dtype.kind in "fc":
@@ -299,6 +322,66 @@ def groupby_reduce(
             v = np.sqrt(v)
         result = v.astype(out_dtype)
         empty_mask = counts == 0
+    elif func in _ARG_FUNCS or func in _POS_FUNCS:
+        order = np.argsort(codes, kind="stable")
+        sc = codes[order]
+        starts = np.searchsorted(sc, np.arange(ngroups), side="left")
+        ends = np.searchsorted(sc, np.arange(ngroups), side="right")
+        # drop the invalid-code region (-1 sorts first)
+        if func in _ARG_FUNCS:
+            out = np.full((M, ngroups), -1, dtype=np.int64)
+        else:
+            out = np.zeros((M, ngroups), dtype=array.dtype)
+        seen = np.zeros((M, ngroups), dtype=bool)
+        for r in range(M):
+            row = vals2d[r]
+            for g in range(ngroups):
+                rows = order[starts[g] : ends[g]]
+                if rows.size == 0:
+                    continue
+                vg = row[rows]
+                if func == "argmax":
+                    out[r, g] = rows[np.argmax(vg)]
+                elif func == "argmin":
+                    out[r, g] = rows[np.argmin(vg)]
+                elif func in ("nanargmax", "nanargmin"):
+                    ok = ~_isnull(vg)
+                    if not ok.any():
+                        continue  # stays -1
+                    sub = vg[ok]
+                    pick = np.argmax(sub) if func == "nanargmax" else np.argmin(sub)
+                    out[r, g] = rows[ok][pick]
+                elif func == "first":
+                    out[r, g] = vg[0]
+                elif func == "last":
+                    out[r, g] = vg[-1]
+                elif func in ("nanfirst", "nanlast"):
+                    ok = ~_isnull(vg)
+                    if not ok.any():
+                        if array.dtype.kind in "fc":
+                            out[r, g] = np.nan
+                        continue
+                    out[r, g] = vg[ok][0] if func == "nanfirst" else vg[ok][-1]
+                seen[r, g] = True
+        if func in _ARG_FUNCS:
+            result = out.astype(out_dtype)
+            empty_mask = out == -1
+        else:
+            result = out.astype(out_dtype)
+            empty_mask = ~np.broadcast_to(present, (M, ngroups))
+    elif func in ("any", "all"):
+        assert array.dtype == np.int_ or array.dtype.kind in "iub", "any/all: bool input"
+        out = np.zeros((M, ngroups), dtype=bool)
+        for r in range(M):
+            m = valid_code
+            if func == "any":
+                np.logical_or.at(out[r], codes[m], vals2d[r][m] != 0)
+            else:
+                out[r] = True
+                np.logical_and.at(out[r], codes[m], vals2d[r][m] != 0)
+        out[:, ~present] = False
+        result = out
+        empty_mask = ~np.broadcast_to(present, (M, ngroups))
     elif func in ("min", "nanmin"):
         if array.dtype.kind in "fc":
             init = np.inf

@@ -79,6 +79,39 @@ def test_both_kernel_paths_agree(force_path):
     np.testing.assert_allclose(p["sum"].cpu().numpy(), want_sum, rtol=1e-12)
 
 
+EXT_FUNCS = ["argmax", "argmin", "nanargmax", "nanargmin", "first", "last", "nanfirst", "nanlast"]
+
+
+@pytest.mark.parametrize("func", EXT_FUNCS)
+@pytest.mark.parametrize("dtype", ["float32", "float64", "int64"])
+@pytest.mark.parametrize("ngroups", [7, 500, 20_000])
+def test_arg_first_last_vs_oracle(func, dtype, ngroups):
+    rng = np.random.default_rng(zlib.crc32(f"{func}-{dtype}-{ngroups}".encode()))
+    n = 60_000
+    labels = rng.integers(0, ngroups, n)
+    if np.dtype(dtype).kind == "f":
+        vals = (rng.standard_normal(n) * 100).astype(dtype)
+        vals[rng.random(n) < 0.03] = np.nan
+    else:
+        vals = rng.integers(-1000, 1000, n).astype(dtype)
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ngroups))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ngroups))
+    assert got.dtype == want.dtype, (got.dtype, want.dtype)
+    np.testing.assert_allclose(got, want, equal_nan=True, rtol=0, atol=0)
+
+
+def test_any_all_bool():
+    rng = np.random.default_rng(17)
+    n, ng = 100_000, 300
+    b = rng.random(n) < 0.02
+    labels = rng.integers(0, ng, n)
+    for func in ("any", "all"):
+        want, *_ = oracle_reduce(b, labels, func=func, expected_groups=np.arange(ng))
+        got, *_ = flox_amd.groupby_reduce(b, labels, func=func, expected_groups=np.arange(ng))
+        assert got.dtype == np.dtype(bool)
+        np.testing.assert_array_equal(got, want)
+
+
 @pytest.mark.parametrize("func", ["sum", "nansum", "mean", "count", "min", "nanmax", "var"])
 def test_partition_path_many_groups(func):
     """2e6 groups: the bucket-partition (sort) path."""
