@@ -226,7 +226,7 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  SumT* s_sum = (OPS & (B_SUM | B_SSD | B_PROD)) ? (SumT*)(smem + lay.sum_off) : nullptr;
+  SumT* s_sum = (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ? (SumT*)(smem + lay.sum_off) : nullptr;
   uint32_t* s_cnt = (OPS & B_CNT) ? (uint32_t*)(smem + lay.cnt_off) : nullptr;
   uint32_t* s_present = (OPS & B_PRESENT) ? (uint32_t*)(smem + lay.present_off) : nullptr;
   Enc* s_mm = (OPS & (B_MIN | B_MAX)) ? (Enc*)(smem + lay.minmax_off) : nullptr;
@@ -318,7 +318,7 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
   /* flush bins to this block's slab section with plain coalesced stores */
   char* my = slab + (int64_t)blockIdx.x * lay.bytes;
   for (int64_t g = tid; g < ngroups; g += blockDim.x) {
-    if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)(my + lay.sum_off))[g] = s_sum[g];
+    if (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ((SumT*)(my + lay.sum_off))[g] = s_sum[g];
     if (OPS & B_CNT) ((uint32_t*)(my + lay.cnt_off))[g] = s_cnt[g];
     if (OPS & B_PRESENT) ((uint32_t*)(my + lay.present_off))[g] = s_present[g];
     if (OPS & (B_MIN | B_MAX)) ((Enc*)(my + lay.minmax_off))[g] = s_mm[g];
