@@ -77,6 +77,12 @@ def main():
     ap.add_argument("--ngroups", type=int, default=NGROUPS)
     ap.add_argument("--func", default="mean")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument(
+        "--config",
+        default="2",
+        choices=["2", "3", "4", "5"],
+        help="BASELINE.json config to measure (default 2 = the headline 1e9-row/1e4-group mean)",
+    )
     args = ap.parse_args()
 
     import flox_amd
@@ -94,17 +100,69 @@ def main():
     device = torch.device("cuda", local_rank if world > 1 else 0)
     torch.cuda.set_device(device)
 
-    n, ng = args.rows, args.ngroups
-    log(f"generating {n:.0e} rows, {ng} groups on {device} (rank {rank}/{world})")
     gen = torch.Generator(device=device).manual_seed(1234 + rank)
-    vals = torch.rand(n, generator=gen, dtype=torch.float32, device=device)
-    labels = torch.randint(0, ng, (n,), generator=gen, dtype=torch.int64, device=device)
-    expected = np.arange(ng)
-    bytes_per_step_per_gpu = n * (4 + 8)
+    if args.config == "2":
+        n, ng = args.rows, args.ngroups
+        func = args.func
+        log(f"config 2: {n:.0e} rows, {ng} groups, {func} (rank {rank}/{world})")
+        vals = torch.rand(n, generator=gen, dtype=torch.float32, device=device)
+        labels = torch.randint(0, ng, (n,), generator=gen, dtype=torch.int64, device=device)
+        expected = np.arange(ng)
+        bytes_per_step_per_gpu = n * (4 + 8)
+        workload = "configs[1]: fp32 (1e9,) values, int64 labels, 1e4 uniform groups, func=mean, engine=hip"
+        cfg_extra = {"rows": n, "ngroups": ng, "func": func, "labels": "int64"}
 
-    def step():
-        res, _ = flox_amd.groupby_reduce(vals, labels, func=args.func, expected_groups=expected)
-        return res
+        def step():
+            res, _ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=expected)
+            return res
+    elif args.config == "3":
+        n, ng = args.rows, 10_000_000
+        func = "sum"
+        log(f"config 3: {n:.0e} rows, 1e7 groups, sum (rank {rank}/{world})")
+        vals = torch.rand(n, generator=gen, dtype=torch.float32, device=device)
+        labels = torch.randint(0, ng, (n,), generator=gen, dtype=torch.int64, device=device)
+        expected = np.arange(ng)
+        bytes_per_step_per_gpu = n * (4 + 8)
+        workload = "configs[2]: fp32 (1e9,) values, 1e7 groups, func=sum (bucket-partition path)"
+        cfg_extra = {"rows": n, "ngroups": ng, "func": func, "labels": "int64"}
+
+        def step():
+            res, _ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=expected)
+            return res
+    elif args.config == "4":
+        n_t, y, x = 8760, 720, 1440
+        func = args.func if args.func in ("sum", "mean", "var") else "mean"
+        log(f"config 4: ({n_t},{y},{x}) by hour-of-day, {func} (rank {rank}/{world})")
+        # weak scaling: each rank holds its own year of hourly data
+        arr = torch.rand((n_t, y, x), generator=gen, dtype=torch.float32, device=device)
+        hours = (torch.arange(n_t, device=device) % 24).to(torch.int64)
+        view = arr.permute(1, 2, 0)
+        expected = np.arange(24)
+        bytes_per_step_per_gpu = arr.numel() * 4
+        workload = "configs[3]: fp32 (8760,720,1440) by hour-of-day (24 groups, axis 0), RCCL combine of (24,720,1440) partials"
+        cfg_extra = {"shape": [n_t, y, x], "ngroups": 24, "func": func}
+
+        def step():
+            res, _ = flox_amd.groupby_reduce(view, hours, func=func, expected_groups=expected)
+            return res
+    else:  # config 5
+        n = 100_000_000
+        func = "nanmean"
+        log(f"config 5: fp64 (1e8,) 2-D groupby 12x180, nanmean (rank {rank}/{world})")
+        vals = torch.rand(n, generator=gen, dtype=torch.float64, device=device)
+        vals[torch.rand(n, generator=gen, device=device) < 0.05] = float("nan")
+        months = torch.randint(0, 12, (n,), generator=gen, dtype=torch.int64, device=device)
+        latbin = torch.randint(0, 180, (n,), generator=gen, dtype=torch.int64, device=device)
+        expected = (np.arange(12), np.arange(180))
+        bytes_per_step_per_gpu = n * (8 + 16)
+        workload = "configs[4]: fp64 (1e8,) ~5% NaN by (month, lat-bin) 12x180, func=nanmean, reindexed to expected"
+        cfg_extra = {"rows": n, "grp_shape": [12, 180], "func": func}
+
+        def step():
+            res, _ = flox_amd.groupby_reduce(
+                vals, months, latbin, func=func, expected_groups=expected
+            )
+            return res
 
     def barrier():
         if world > 1:
@@ -120,18 +178,23 @@ def main():
     # on the launch stream (kernel + its O(ngroups) slab-combine tail)
     events = []
     orig = aggregate_hip.grouped_partials
+    orig_cols = aggregate_hip.grouped_partials_cols
 
-    def timed_partials(*a, **kw):
-        e0 = torch.cuda.Event(enable_timing=True)
-        e1 = torch.cuda.Event(enable_timing=True)
-        e0.record()
-        out = orig(*a, **kw)
-        e1.record()
-        events.append((e0, e1))
-        return out
+    def _timed(fn):
+        def wrapper(*a, **kw):
+            e0 = torch.cuda.Event(enable_timing=True)
+            e1 = torch.cuda.Event(enable_timing=True)
+            e0.record()
+            out = fn(*a, **kw)
+            e1.record()
+            events.append((e0, e1))
+            return out
+        return wrapper
 
-    aggregate_hip.grouped_partials = timed_partials
-    fa_core.grouped_partials = timed_partials  # core binds the name directly
+    aggregate_hip.grouped_partials = _timed(orig)
+    fa_core.grouped_partials = aggregate_hip.grouped_partials  # core binds the name
+    aggregate_hip.grouped_partials_cols = _timed(orig_cols)
+    fa_core.grouped_partials_cols = aggregate_hip.grouped_partials_cols
 
     barrier()
     t0 = time.perf_counter()
@@ -141,6 +204,8 @@ def main():
     elapsed = time.perf_counter() - t0
     aggregate_hip.grouped_partials = orig
     fa_core.grouped_partials = orig
+    aggregate_hip.grouped_partials_cols = orig_cols
+    fa_core.grouped_partials_cols = orig_cols
 
     if world > 1:
         t = torch.tensor([elapsed], device=device)
@@ -154,7 +219,9 @@ def main():
     if rank == 0:
         total_bytes = bytes_per_step_per_gpu * world * args.steps
         value = total_bytes / elapsed / 1e9
-        cpu = None if args.no_cpu_baseline else cpu_baseline_leg(min(n, 30_000_000), ng)
+        cpu = None
+        if args.config == "2" and not args.no_cpu_baseline:
+            cpu = cpu_baseline_leg(min(args.rows, 30_000_000), args.ngroups)
         out = {
             "metric": "grouped-reduce GB/s (input bytes/s) + fraction of HBM peak, 1e9 rows",
             "value": round(value, 2),
@@ -166,14 +233,11 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # reference publishes no number
-            "dtype": "f32",  # values f32, accumulated in f64 (npg contract)
+            "dtype": "f64" if args.config == "5" else "f32",  # accumulation is f64 (npg contract)
             "data": "synthetic",
             "config": {
-                "workload": "configs[1]: fp32 (1e9,) values, int64 labels, 1e4 uniform groups, func=mean, engine=hip",
-                "rows": n,
-                "ngroups": ng,
-                "func": args.func,
-                "labels": "int64",
+                "workload": workload,
+                **cfg_extra,
                 "parallelism": f"dp{world}" if world > 1 else "single",
             },
             "roofline": {
@@ -182,7 +246,7 @@ def main():
                 "peak": HBM_PEAK_GBPS,
                 "unit": "GB/s",
                 "frac": round(achieved_gbps / HBM_PEAK_GBPS, 4) if achieved_gbps else None,
-                "traffic": read_traffic(),
+                "traffic": read_traffic() if args.config == "2" else None,
             },
             "cpu_baseline": cpu,
         }

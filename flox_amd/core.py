@@ -375,23 +375,43 @@ def groupby_reduce(
             raise NotImplementedError(func)
 
     # --- finalize: masking + fills + final dtype (reference core.py:410-475) ---
+    # fills are applied with an unconditional where (no host sync); the one
+    # case that must inspect the mask on the host is a NaN fill on an
+    # integer result, which promotes the output dtype (reference
+    # core.py:446-449 maybe_promote)
     t_out_dtype = _torch_dtype(out_dtype)
+    is_float_out = torch.empty(0, dtype=t_out_dtype).is_floating_point()
     if min_count_ > 0:
         mask = counts_for_mask < min_count_
-        if bool(mask.any().item()):
-            if fill_value is None:
-                raise ValueError("Filling is required but fill_value is None.")
-            if not torch.is_floating_point(torch.empty(0, dtype=t_out_dtype)) and isinstance(
-                fill_value, float
-            ) and math.isnan(fill_value):
+        nan_fill = isinstance(fill_value, float) and math.isnan(fill_value)
+        if fill_value is None or (nan_fill and not is_float_out):
+            if bool(mask.any().item()):
+                if fill_value is None:
+                    raise ValueError("Filling is required but fill_value is None.")
                 t_out_dtype = torch.float64
+                result = result.to(t_out_dtype)
+                result = torch.where(
+                    mask, torch.tensor(fill_value, dtype=t_out_dtype, device=device), result
+                )
+        else:
             result = result.to(t_out_dtype)
-            result = torch.where(mask, torch.tensor(fill_value, dtype=t_out_dtype, device=device), result)
+            result = torch.where(
+                mask, torch.tensor(fill_value, dtype=t_out_dtype, device=device), result
+            )
     else:
         fv = fill_value if fill_value is not None else xrdtypes.fill_default(func, out_dtype)
-        if bool(empty_mask.any().item()):
+        if isinstance(fv, float) and math.isnan(fv) and not is_float_out:
+            if bool(empty_mask.any().item()):
+                result = result.to(torch.float64)
+                t_out_dtype = torch.float64
+                result = torch.where(
+                    empty_mask, torch.tensor(fv, dtype=t_out_dtype, device=device), result
+                )
+        else:
             result = result.to(t_out_dtype)
-            result = torch.where(empty_mask, torch.tensor(fv, dtype=t_out_dtype, device=device), result)
+            result = torch.where(
+                empty_mask, torch.tensor(fv, dtype=t_out_dtype, device=device), result
+            )
 
     result = result.to(t_out_dtype)
     if lead_M > 1:
