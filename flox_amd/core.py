@@ -82,6 +82,14 @@ def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
             codes = torch.where(torch.isnan(flat), torch.full_like(codes, -1), codes)
         return _FactorizedBy(codes, uniq.cpu().numpy(), direct=False)
 
+    if isinstance(expect, range):
+        # O(1) RangeIndex-style fast path (reference factorize.py:44-53)
+        if expect.step != 1 or expect.start != 0:
+            expect = np.asarray(expect)
+        elif flat.dtype.is_floating_point:
+            expect = np.asarray(expect)
+        else:
+            return _FactorizedBy(flat, np.arange(len(expect)), direct=True)
     expect_np = np.asarray(expect)
     n = len(expect_np)
     if (

@@ -76,6 +76,8 @@ def _factorize_single(flat: np.ndarray, expect) -> tuple[np.ndarray, np.ndarray]
     fast path (44-53), searchsorted-vs-expected (84-94), unique/factorize (96).
     Returns (codes int64, found_group_values).
     """
+    if isinstance(expect, range):
+        expect = np.asarray(expect)
     if expect is None:
         # hash/factorize path (reference line 96: pd.factorize(flat, sort=True))
         nanmask = _isnull(flat)
