@@ -89,7 +89,9 @@ def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
         elif flat.dtype.is_floating_point:
             expect = np.asarray(expect)
         else:
-            return _FactorizedBy(flat, np.arange(len(expect)), direct=True)
+            # groups echo the caller's range (materializing np.arange(1e7)
+            # costs ~10 ms per call)
+            return _FactorizedBy(flat, expect, direct=True)
     expect_np = np.asarray(expect)
     n = len(expect_np)
     if (
