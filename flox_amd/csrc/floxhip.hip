@@ -326,7 +326,11 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
   }
 }
 
-/* ---- kernel 2: combine per-block slabs (fixed order over blocks) --------- */
+/* ---- kernel 2: combine per-block slabs ----------------------------------
+ * grid.y splits the slab-block range so small group counts still fill the
+ * chip; with gridDim.y == 1 final bins are written directly (min/max
+ * decoded inline), otherwise each chunk folds its share and merges into
+ * memset-initialized bins with atomics (min/max stay encoded for k_decode). */
 template <typename V, int OPS>
 __global__ void k_combine(const char* __restrict__ slab, int nblocks,
                           int64_t ngroups, BinLayout lay, void* out_sum,
@@ -341,6 +345,12 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
   const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (g >= ngroups) return;
+  const int S = gridDim.y;
+  const int chunk = (nblocks + S - 1) / S;
+  const int b0 = blockIdx.y * chunk;
+  int b1 = b0 + chunk;
+  if (b1 > nblocks) b1 = nblocks;
+  if (b0 >= b1) return;
 
   SumT s = IS_PROD ? (SumT)1 : (SumT)0;
   if (OPS & B_IDXMIN) s = (SumT)INT64_MAX;
@@ -348,7 +358,7 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
   int64_t c = 0;
   uint32_t p = 0, nf = 0;
   Enc mn = (Enc)~(Enc)0, mx = (Enc)0;
-  for (int b = 0; b < nblocks; ++b) {
+  for (int b = b0; b < b1; ++b) {
     const char* blk = slab + (int64_t)b * lay.bytes;
     if (OPS & (B_SUM | B_SSD)) s += ((const SumT*)(blk + lay.sum_off))[g];
     if (IS_PROD) s *= ((const SumT*)(blk + lay.sum_off))[g];
@@ -372,13 +382,28 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
     }
     if (OPS & B_NANFLAG) nf |= ((const uint32_t*)(blk + lay.nanflag_off))[g];
   }
-  const bool present = (OPS & B_PRESENT) ? (p != 0) : (c != 0);
-  if (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ((SumT*)out_sum)[g] = s;
-  if (OPS & B_CNT) out_count[g] = c;
-  if (OPS & B_PRESENT) out_present[g] = p;
-  if (OPS & B_MIN) ((V*)out_min)[g] = present ? TR::dec(mn) : TR::pos_inf();
-  if (OPS & B_MAX) ((V*)out_max)[g] = present ? TR::dec(mx) : (V)-TR::pos_inf();
-  if (OPS & B_NANFLAG) out_nanflag[g] = nf;
+  if (S == 1) {
+    const bool present = (OPS & B_PRESENT) ? (p != 0) : (c != 0);
+    if (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ((SumT*)out_sum)[g] = s;
+    if (OPS & B_CNT) out_count[g] = c;
+    if (OPS & B_PRESENT) out_present[g] = p;
+    if (OPS & B_MIN) ((V*)out_min)[g] = present ? TR::dec(mn) : TR::pos_inf();
+    if (OPS & B_MAX) ((V*)out_max)[g] = present ? TR::dec(mx) : (V)-TR::pos_inf();
+    if (OPS & B_NANFLAG) out_nanflag[g] = nf;
+  } else {
+    if (OPS & (B_SUM | B_SSD)) acc_add((SumT*)out_sum + g, s);
+    if (IS_PROD) {
+      if (s != (SumT)1) acc_mul((SumT*)out_sum + g, s);
+    }
+    if (OPS & B_IDXMIN) idx_min((int64_t*)out_sum + g, (int64_t)s);
+    if (OPS & B_IDXMAX) idx_max((int64_t*)out_sum + g, (int64_t)s);
+    if ((OPS & B_CNT) && c)
+      atomicAdd(reinterpret_cast<unsigned long long*>(&out_count[g]), (unsigned long long)c);
+    if ((OPS & B_PRESENT) && p) out_present[g] = 1u;
+    if (OPS & B_MIN) enc_min(&((Enc*)out_min)[g], mn);
+    if (OPS & B_MAX) enc_max(&((Enc*)out_max)[g], mx);
+    if ((OPS & B_NANFLAG) && nf) out_nanflag[g] = 1u;
+  }
 }
 
 /* ---- global-atomic path -------------------------------------------------- */
@@ -1202,11 +1227,28 @@ int launch_typed(fh_call* c) {
                        (int)skipnan, (char*)c->scratch, lay);
     FH_CHECK(hipGetLastError());
     int cb = (int)((c->ngroups + 255) / 256);
-    hipLaunchKernelGGL((k_combine<V, OPS>), dim3(cb), dim3(256), 0, stream,
+    /* fill the chip even for tiny group counts: split the slab-block fold */
+    int S = 1;
+    if (c->ngroups < 65536) {
+      S = (int)((524288 + c->ngroups - 1) / c->ngroups);
+      if (S > nblocks) S = nblocks;
+      if (S < 1) S = 1;
+    }
+    if (S > 1) {
+      int rc = init_outs<V, OPS>(c, c->ngroups, stream);
+      if (rc) return rc;
+    }
+    hipLaunchKernelGGL((k_combine<V, OPS>), dim3(cb, S), dim3(256), 0, stream,
                        (const char*)c->scratch, nblocks, c->ngroups, lay,
                        c->out_sum, c->out_count, c->out_present, c->out_min,
                        c->out_max, c->out_nanflag);
     FH_CHECK(hipGetLastError());
+    if (S > 1 && (OPS & (B_MIN | B_MAX))) {
+      hipLaunchKernelGGL((k_decode<V, OPS>), dim3(cb), dim3(256), 0, stream,
+                         c->ngroups, c->out_min, c->out_max, c->out_count,
+                         c->out_present);
+      FH_CHECK(hipGetLastError());
+    }
     c->path_used = 1;
     return 0;
   }
@@ -1416,9 +1458,9 @@ int launch_cols(fh_call* c) {
   if (rc != 0) return rc + 1000;
 
   if (slab_mode) {
-    /* fold the chunk partials; k_combine also decodes min/max */
+    /* fold the chunk partials; k_combine decodes min/max (gridDim.y==1) */
     int cb = (int)((nbins + 255) / 256);
-    hipLaunchKernelGGL((k_combine<V, OPS>), dim3(cb), dim3(256), 0, stream,
+    hipLaunchKernelGGL((k_combine<V, OPS>), dim3(cb, 1), dim3(256), 0, stream,
                        (const char*)c->scratch, plan.nchunks, nbins, plan.lay,
                        c->out_sum, c->out_count, c->out_present, c->out_min,
                        c->out_max, c->out_nanflag);
