@@ -995,6 +995,13 @@ template <typename V, int OPS>
 int init_outs(fh_call* c, int64_t nbins, hipStream_t stream) {
   using TR = Traits<V>;
   if (OPS & (B_SUM | B_SSD)) FH_CHECK(hipMemsetAsync(c->out_sum, 0, nbins * 8, stream));
+  if (OPS & (B_IDXMIN | B_IDXMAX)) {
+    int fb = (int)((nbins + 255) / 256);
+    hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                       (int64_t*)c->out_sum, nbins,
+                       (OPS & B_IDXMIN) ? INT64_MAX : (int64_t)-1);
+    FH_CHECK(hipGetLastError());
+  }
   if (OPS & B_PROD) {
     int fb = (int)((nbins + 255) / 256);
     if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
