@@ -67,6 +67,30 @@ class _FactorizedBy:
         self.direct = direct
 
 
+def _factorize_bins(flat: torch.Tensor, edges: np.ndarray) -> _FactorizedBy:
+    """Bin-edge grouping (isbin=True): right-closed intervals (edges[i-1],
+    edges[i]], values outside every interval -> invalid (restates reference
+    factorize.py:55-82: np.digitize(right=True) - 1). Groups are the
+    pandas IntervalIndex the reference returns."""
+    import pandas as pd
+
+    edges = np.asarray(edges)
+    assert edges.ndim == 1 and len(edges) >= 2, "bin edges need >= 2 values"
+    nbins = len(edges) - 1
+    fl = flat
+    edges_t = torch.from_numpy(np.ascontiguousarray(edges)).to(flat.device)
+    if edges_t.dtype != fl.dtype:
+        common = torch.promote_types(edges_t.dtype, fl.dtype)
+        edges_t = edges_t.to(common)
+        fl = fl.to(common)
+    codes = torch.searchsorted(edges_t, fl, right=False) - 1
+    bad = (codes < 0) | (codes >= nbins)
+    if fl.dtype.is_floating_point:
+        bad |= torch.isnan(fl)
+    codes = torch.where(bad, torch.full_like(codes, -1), codes)
+    return _FactorizedBy(codes, pd.IntervalIndex.from_breaks(edges), direct=False)
+
+
 def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
     """labels -> codes in [0, ngroups), invalid/NaN -> out-of-range
     (restates reference _factorize_single, factorize.py:42-99)."""
@@ -150,7 +174,7 @@ def groupby_reduce(
     func: str,
     expected_groups=None,
     sort: bool = True,
-    isbin: bool = False,
+    isbin=False,
     axis=None,
     fill_value=None,
     dtype=None,
@@ -170,8 +194,6 @@ def groupby_reduce(
     """
     if engine != "hip":
         raise ValueError(f"flox_amd implements engine='hip' only (got {engine!r})")
-    if isbin:
-        raise NotImplementedError("isbin=True (bin edges) is not implemented yet")
     if func not in REDUCTIONS:
         raise NotImplementedError(f"reduction {func!r}")
     agg = REDUCTIONS[func]
@@ -208,8 +230,14 @@ def groupby_reduce(
     if expected_groups is None:
         expected_groups = (None,) * nby
     provided_expected = any(e is not None for e in expected_groups)
+    if isbin and not provided_expected:
+        raise ValueError("isbin=True requires expected_groups (the bin edges)")
 
-    facs = [_factorize_device(b.reshape(-1), e, sort) for b, e in zip(bys, expected_groups)]
+    isbins = isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby
+    facs = [
+        _factorize_bins(b.reshape(-1), e) if ib else _factorize_device(b.reshape(-1), e, sort)
+        for b, e, ib in zip(bys, expected_groups, isbins)
+    ]
     grp_shape = tuple(f.ngroups for f in facs)
     ngroups = math.prod(grp_shape)
     labels, labels2, grp_pair = _combined_codes(facs)

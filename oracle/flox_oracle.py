@@ -165,6 +165,19 @@ def _fill_default(func: str, out_dtype: np.dtype):
     return np.iinfo(out_dtype).min
 
 
+def _factorize_bins_np(flat, edges):
+    """right-closed binning, reference factorize.py:55-82."""
+    import pandas as pd
+
+    edges = np.asarray(edges)
+    nbins = len(edges) - 1
+    codes = np.digitize(flat, bins=edges, right=True) - 1
+    within = flat <= edges.max()
+    codes[(codes < 0) | (codes >= nbins) | ~within] = -1
+    codes[_isnull(np.asarray(flat, dtype=float)) if flat.dtype.kind in "fc" else np.zeros(len(flat), bool)] = -1
+    return codes.astype(np.int64), pd.IntervalIndex.from_breaks(edges)
+
+
 def groupby_reduce(
     array,
     *by,
@@ -174,6 +187,7 @@ def groupby_reduce(
     fill_value=None,
     dtype=None,
     min_count=None,
+    isbin=False,
     finalize_kwargs=None,
 ):
     """Eager grouped reduction with flox semantics. Returns (result, *groups).
@@ -207,9 +221,13 @@ def groupby_reduce(
         array = array.astype(np.int_)
 
     # --- factorize (early, like reference core.py:943-949) ---
+    isbins = isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby
     codes_list, found = [], []
-    for b, e in zip(bys, expected_groups):
-        c, f = _factorize_single(b.reshape(-1), e)
+    for b, e, ib in zip(bys, expected_groups, isbins):
+        if ib:
+            c, f = _factorize_bins_np(b.reshape(-1), e)
+        else:
+            c, f = _factorize_single(b.reshape(-1), e)
         codes_list.append(c)
         found.append(f)
     grp_shape = tuple(len(f) for f in found)

@@ -148,6 +148,22 @@ def gen_cases():
                 expected_groups=(np.arange(4), np.arange(6)),
             ),
         )
+    # isbin: values grouped into right-closed bins (reference factorize.py:55-82)
+    bvals = rng.standard_normal(400)
+    bby = rng.standard_normal(400) * 3
+    edges = np.array([-4.0, -1.5, 0.0, 0.5, 2.0, 5.0])
+    for func in ["sum", "mean", "count", "nanmax", "var"]:
+        yield (
+            f"{func}_isbin",
+            dict(array=bvals, by=bby, func=func, expected_groups=edges, isbin=True),
+        )
+    yield "mean_isbin_int", dict(
+        array=rng.standard_normal(300),
+        by=rng.integers(0, 50, 300).astype(np.float64),
+        func="mean",
+        expected_groups=np.array([0, 10, 20, 30, 40, 50]).astype(np.float64),
+        isbin=True,
+    )
     # empty groups at the tail of the range
     yield "mean_sparse_groups", dict(
         array=rng.standard_normal(50),
@@ -187,7 +203,9 @@ def main():
         arr = None
         out[f"{name}::result"] = np.asarray(result)
         for i, g in enumerate(groups):
-            out[f"{name}::groups{i}"] = np.asarray(g)
+            ga = np.asarray(g)
+            if ga.dtype != object:  # interval groups are pinned via the edges
+                out[f"{name}::groups{i}"] = ga
         n_done += 1
     # re-run to also store the inputs (gen_cases is deterministic)
     for name, kw in gen_cases():
@@ -203,6 +221,8 @@ def main():
             egs = eg if isinstance(eg, tuple) else (eg,)
             for i, e in enumerate(egs):
                 out[f"{name}::expected{i}"] = np.asarray(e)
+        if kw.get("isbin"):
+            out[f"{name}::isbin"] = np.asarray(True)
         if kw.get("fill_value") is not None:
             out[f"{name}::fill_value"] = np.asarray(kw["fill_value"])
         if kw.get("finalize_kwargs"):
