@@ -450,6 +450,59 @@ def grouped_partials_cols(
     return out
 
 
+def grouped_quantile(
+    values: torch.Tensor,
+    labels: torch.Tensor,
+    ngroups: int,
+    q,
+    *,
+    skipnan: bool = False,
+    labels2: torch.Tensor | None = None,
+    grp_shape: tuple[int, int] | None = None,
+) -> torch.Tensor:
+    """Grouped linear-interpolation quantiles via the sorted path
+    (fh_grouped_quantile). Returns f64 (nq, ngroups); NaN for empty groups,
+    all-NaN groups (skipnan) and NaN-containing groups (not skipnan)."""
+    lib = _ffi.load_library()
+    _require_gpu_tensor(values, "values")
+    _require_gpu_tensor(labels, "labels")
+    values = values.contiguous()
+    labels = labels.contiguous()
+    if labels.dtype not in _TORCH_LDTYPE:
+        labels = labels.to(torch.int64)
+    dev = values.device
+    q_t = torch.as_tensor(np.atleast_1d(np.asarray(q, dtype=np.float64)), device=dev)
+    nq = q_t.numel()
+    out = torch.empty((nq, ngroups), dtype=torch.float64, device=dev)
+
+    c = FhCall()
+    c.vdtype = _TORCH_VDTYPE[values.dtype]
+    c.ldtype = _TORCH_LDTYPE[labels.dtype]
+    c.flags = FLAG_SKIPNAN if skipnan else 0
+    c.n = values.numel()
+    c.ngroups = ngroups
+    c.values = values.data_ptr()
+    c.labels = labels.data_ptr()
+    if labels2 is not None:
+        labels2 = labels2.contiguous()
+        if labels2.dtype != labels.dtype:
+            labels2 = labels2.to(labels.dtype)
+        c.labels2 = labels2.data_ptr()
+        c.g0, c.g1 = grp_shape
+    c.means = q_t.contiguous().data_ptr()
+    c.out_sum = out.data_ptr()
+    nscratch = lib.fh_quantile_scratch_bytes(ctypes.byref(c))
+    scratch = torch.empty(max(int(nscratch), 1), dtype=torch.uint8, device=dev)
+    c.scratch = scratch.data_ptr()
+    c.scratch_bytes = nscratch
+    c.stream = torch.cuda.current_stream(dev).cuda_stream
+    _ffi.check(lib.fh_grouped_quantile(ctypes.byref(c), nq))
+    for t in (values, labels, labels2, q_t, scratch, out):
+        if isinstance(t, torch.Tensor):
+            t.record_stream(torch.cuda.current_stream(dev))
+    return out
+
+
 def var_partials(
     group_idx, array, *, skipnan, size, labels2=None, grp_shape=None,
     global_counts=None, global_sums=None,

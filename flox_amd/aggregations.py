@@ -101,6 +101,12 @@ REDUCTIONS: dict[str, Aggregation] = {
         combine={"idx": "max", "count": "sum", "present": "max"},
         preserves_dtype=True,
     ),
+    # quantile family: sorted path (blockwise-only in the reference too,
+    # aggregations.py:672-712); median = quantile(q=0.5)
+    "quantile": Aggregation("quantile", None),
+    "nanquantile": Aggregation("nanquantile", None, skipnan=True),
+    "median": Aggregation("median", None),
+    "nanmedian": Aggregation("nanmedian", None, skipnan=True),
     # bool reductions via min/max of the 0/1-cast input
     # (reference aggregations.py:651-676 all_/any_)
     "any": Aggregation("any", _ffi.SET_MAX_COUNT, combine={"max": "max", "count": "sum"}),

@@ -311,6 +311,48 @@ def groupby_reduce(
                 op_set, vt, scodes, perm, ngroups, skipnan=skipnan, means=means
             )
 
+    if func in ("quantile", "nanquantile", "median", "nanmedian"):
+        from .aggregate_hip import grouped_quantile
+
+        if lead_M != 1:
+            raise NotImplementedError(f"{func} with leading array dims: next row")
+        if func in ("quantile", "nanquantile"):
+            if not finalize_kwargs or "q" not in finalize_kwargs:
+                raise ValueError("Please pass `q` for quantile calculations.")
+            q = finalize_kwargs["q"]
+        else:
+            q = 0.5
+        q_arr = np.atleast_1d(np.asarray(q, dtype=np.float64))
+        scalar_q = np.isscalar(q) or np.ndim(q) == 0
+        resq = grouped_quantile(
+            vals, labels, ngroups, q_arr, skipnan=agg.skipnan,
+            labels2=labels2, grp_shape=grp_pair,
+        )
+        result = resq[0] if scalar_q else resq
+        if min_count_ > 0:
+            pc = grouped_partials(
+                _ffi.SET_COUNT, vals, labels, ngroups, skipnan=True,
+                labels2=labels2, grp_shape=grp_pair,
+            )
+            if dist_on:
+                distributed.all_reduce_(pc["count"], "sum")
+            counts_for_mask = pc["count"]
+        else:
+            counts_for_mask = None
+        if dist_on:
+            raise NotImplementedError("distributed quantiles need a global sort: next row")
+        t_out = _torch_dtype(out_dtype)
+        if min_count_ > 0 and fill_value is not None:
+            mask = counts_for_mask < min_count_
+            result = torch.where(mask, torch.tensor(float(fill_value), dtype=result.dtype, device=device), result)
+        result = result.to(t_out)
+        new_shape = ((len(q_arr),) if not scalar_q else ()) + lead_shape + grp_shape
+        result = result.reshape(new_shape)
+        groups = tuple(f.groups for f in facs)
+        if return_numpy:
+            return (result.cpu().numpy(), *groups)
+        return (result, *groups)
+
     if func in ("argmax", "argmin", "nanargmax", "nanargmin"):
         # pass 1: the per-group extremum; pass 2: the smallest row index whose
         # value matches it (ties -> first occurrence, like np.argmax; a NaN

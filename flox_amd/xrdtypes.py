@@ -11,7 +11,10 @@ from __future__ import annotations
 import numpy as np
 
 # funcs that always produce floating point, preserving a floating input dtype
-FLOAT_FUNCS = {"mean", "nanmean", "var", "nanvar", "std", "nanstd"}
+FLOAT_FUNCS = {
+    "mean", "nanmean", "var", "nanvar", "std", "nanstd",
+    "quantile", "nanquantile", "median", "nanmedian",
+}
 # funcs whose output dtype equals the input dtype (preserves_dtype=True)
 PRESERVES_DTYPE = {"min", "nanmin", "max", "nanmax", "first", "nanfirst", "last", "nanlast"}
 NAN_SKIP = {

@@ -26,6 +26,10 @@ import numpy as np
 # engine="flox" (core.py:856-859 raises for argreductions; aggregate_flox has
 # no first/last/any/all and numpy_groupies is absent in this container).
 ALL_FUNCS = (
+    "quantile",
+    "nanquantile",
+    "median",
+    "nanmedian",
     "argmax",
     "argmin",
     "nanargmax",
@@ -53,7 +57,11 @@ ALL_FUNCS = (
     "nanmax",
 )
 
-_FLOAT_FUNCS = {"mean", "nanmean", "var", "nanvar", "std", "nanstd"}
+_FLOAT_FUNCS = {
+    "mean", "nanmean", "var", "nanvar", "std", "nanstd",
+    "quantile", "nanquantile", "median", "nanmedian",
+}
+_Q_FUNCS = {"quantile", "nanquantile", "median", "nanmedian"}
 _ARG_FUNCS = {"argmax", "argmin", "nanargmax", "nanargmin"}
 _POS_FUNCS = {"first", "last", "nanfirst", "nanlast"}
 _NAN_SKIP = {"nansum", "nanprod", "nanmean", "nanvar", "nanstd", "nanmin", "nanmax", "count"}
@@ -342,6 +350,48 @@ def groupby_reduce(
             v = np.sqrt(v)
         result = v.astype(out_dtype)
         empty_mask = counts == 0
+    elif func in _Q_FUNCS:
+        if func in ("quantile", "nanquantile"):
+            q = (finalize_kwargs or {}).get("q")
+            assert q is not None, "Please pass `q` for quantile calculations."
+        else:
+            q = 0.5
+        q_arr = np.atleast_1d(np.asarray(q, dtype=np.float64))
+        scalar_q = np.isscalar(q) or np.ndim(q) == 0
+        skipq = func in ("nanquantile", "nanmedian")
+        order = np.argsort(codes, kind="stable")
+        sc = codes[order]
+        starts = np.searchsorted(sc, np.arange(ngroups), side="left")
+        ends = np.searchsorted(sc, np.arange(ngroups), side="right")
+        out = np.full((len(q_arr), M, ngroups), np.nan)
+        for r in range(M):
+            row = vals2d[r]
+            for g in range(ngroups):
+                rows = order[starts[g] : ends[g]]
+                if rows.size == 0:
+                    continue
+                vg = row[rows]
+                nanmask = _isnull(vg)
+                if skipq:
+                    vv = vg[~nanmask]
+                    if vv.size == 0:
+                        continue
+                    out[:, r, g] = np.quantile(vv, q_arr, method="linear")
+                else:
+                    if nanmask.any():
+                        continue  # stays NaN (reference quantile_ masks)
+                    out[:, r, g] = np.quantile(vg, q_arr, method="linear")
+        result = out.astype(out_dtype)
+        if scalar_q:
+            result = result[0]
+        counts_q = counts
+        # fills/min_count for quantiles: NaN already encodes missing; user
+        # fill via min_count mask
+        if min_count_ > 0 and fill_value is not None:
+            mask = counts_q < min_count_
+            result = np.where(np.broadcast_to(mask, result.shape), fill_value, result)
+        newshape = (() if scalar_q else (len(q_arr),)) + lead_shape + grp_shape
+        return (result.reshape(newshape), *found)
     elif func in _ARG_FUNCS or func in _POS_FUNCS:
         order = np.argsort(codes, kind="stable")
         sc = codes[order]

@@ -148,6 +148,36 @@ def gen_cases():
                 expected_groups=(np.arange(4), np.arange(6)),
             ),
         )
+    # quantile family (reference aggregate_flox.py:50-130)
+    qvals = rng.standard_normal(500) * 10
+    qby = rng.integers(0, 9, 500)
+    qnan = qvals.copy()
+    qnan[rng.random(500) < 0.1] = np.nan
+    yield "median_basic", dict(array=qvals, by=qby, func="median", expected_groups=np.arange(9))
+    yield "nanmedian_nan", dict(array=qnan, by=qby, func="nanmedian", expected_groups=np.arange(9))
+    yield "quantile_q25", dict(
+        array=qvals, by=qby, func="quantile", expected_groups=np.arange(9),
+        finalize_kwargs={"q": 0.25},
+    )
+    yield "quantile_multi", dict(
+        array=qvals, by=qby, func="quantile", expected_groups=np.arange(9),
+        finalize_kwargs={"q": [0.1, 0.5, 0.9]},
+    )
+    yield "nanquantile_q75", dict(
+        array=qnan, by=qby, func="nanquantile", expected_groups=np.arange(9),
+        finalize_kwargs={"q": 0.75},
+    )
+    yield "quantile_withnan", dict(
+        array=qnan, by=qby, func="quantile", expected_groups=np.arange(9),
+        finalize_kwargs={"q": 0.5},
+    )
+    yield "median_f32", dict(
+        array=qvals.astype(np.float32), by=qby, func="median", expected_groups=np.arange(9)
+    )
+    yield "median_int", dict(
+        array=rng.integers(-100, 100, 500).astype(np.int64), by=qby, func="median",
+        expected_groups=np.arange(9),
+    )
     # isbin: values grouped into right-closed bins (reference factorize.py:55-82)
     bvals = rng.standard_normal(400)
     bby = rng.standard_normal(400) * 3
@@ -226,7 +256,12 @@ def main():
         if kw.get("fill_value") is not None:
             out[f"{name}::fill_value"] = np.asarray(kw["fill_value"])
         if kw.get("finalize_kwargs"):
-            out[f"{name}::ddof"] = np.asarray(kw["finalize_kwargs"]["ddof"])
+            fk = kw["finalize_kwargs"]
+            if "ddof" in fk:
+                out[f"{name}::ddof"] = np.asarray(fk["ddof"])
+            if "q" in fk:
+                out[f"{name}::q"] = np.asarray(fk["q"])
+                out[f"{name}::q_scalar"] = np.asarray(np.isscalar(fk["q"]))
     np.savez_compressed(OUT, **out)
     print(f"wrote {OUT}: {n_done} cases ({n_skip} skipped)")
 

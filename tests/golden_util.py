@@ -43,6 +43,9 @@ def load_golden_cases():
         ddof = get("ddof")
         if ddof is not None:
             kw["finalize_kwargs"] = {"ddof": int(ddof)}
+        qv = get("q")
+        if qv is not None:
+            kw["finalize_kwargs"] = {"q": qv.item() if get("q_scalar") else qv.tolist()}
         if get("isbin") is not None:
             kw["isbin"] = True
         yield name, dict(array=get("array"), by=tuple(bys), func=func, **kw), get("result"), groups

@@ -100,6 +100,25 @@ def test_arg_first_last_vs_oracle(func, dtype, ngroups):
     np.testing.assert_allclose(got, want, equal_nan=True, rtol=0, atol=0)
 
 
+@pytest.mark.parametrize("func", ["median", "nanmedian", "quantile", "nanquantile"])
+@pytest.mark.parametrize("dtype", ["float32", "float64", "int64"])
+def test_quantile_family_vs_oracle(func, dtype):
+    rng = np.random.default_rng(zlib.crc32(f"q-{func}-{dtype}".encode()))
+    n, ng = 100_000, 257
+    labels = rng.integers(0, ng, n)
+    if np.dtype(dtype).kind == "f":
+        vals = (rng.standard_normal(n) * 100).astype(dtype)
+        vals[rng.random(n) < 0.02] = np.nan
+    else:
+        vals = rng.integers(-1000, 1000, n).astype(dtype)
+    fk = {"q": [0.1, 0.5, 0.9]} if "quantile" in func else None
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng), finalize_kwargs=fk)
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng), finalize_kwargs=fk)
+    assert got.dtype == want.dtype and got.shape == want.shape
+    tol = dict(rtol=3e-6, atol=1e-5) if np.dtype(dtype).itemsize == 4 and np.dtype(dtype).kind == "f" else dict(rtol=1e-12, atol=1e-12)
+    np.testing.assert_allclose(got, want, equal_nan=True, **tol)
+
+
 def test_any_all_bool():
     rng = np.random.default_rng(17)
     n, ng = 100_000, 300
