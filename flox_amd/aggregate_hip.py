@@ -492,7 +492,7 @@ def grouped_quantile(
     c.means = q_t.contiguous().data_ptr()
     c.out_sum = out.data_ptr()
     nscratch = lib.fh_quantile_scratch_bytes(ctypes.byref(c))
-    scratch = torch.empty(max(int(nscratch), 1), dtype=torch.uint8, device=dev)
+    scratch = torch.empty(builtins.max(int(nscratch), 1), dtype=torch.uint8, device=dev)
     c.scratch = scratch.data_ptr()
     c.scratch_bytes = nscratch
     c.stream = torch.cuda.current_stream(dev).cuda_stream
