@@ -503,6 +503,52 @@ def grouped_quantile(
     return out
 
 
+def grouped_mode(
+    values: torch.Tensor,
+    labels: torch.Tensor,
+    ngroups: int,
+    *,
+    skipnan: bool = False,
+    labels2: torch.Tensor | None = None,
+    grp_shape: tuple[int, int] | None = None,
+) -> torch.Tensor:
+    """Grouped mode via the sorted path (scipy.stats.mode semantics: most
+    frequent value, ties -> smallest; NaN propagates unless skipnan)."""
+    lib = _ffi.load_library()
+    _require_gpu_tensor(values, "values")
+    values = values.contiguous()
+    labels = labels.contiguous()
+    if labels.dtype not in _TORCH_LDTYPE:
+        labels = labels.to(torch.int64)
+    dev = values.device
+    out = torch.empty(ngroups, dtype=values.dtype, device=dev)
+    c = FhCall()
+    c.vdtype = _TORCH_VDTYPE[values.dtype]
+    c.ldtype = _TORCH_LDTYPE[labels.dtype]
+    c.flags = FLAG_SKIPNAN if skipnan else 0
+    c.n = values.numel()
+    c.ngroups = ngroups
+    c.values = values.data_ptr()
+    c.labels = labels.data_ptr()
+    if labels2 is not None:
+        labels2 = labels2.contiguous()
+        if labels2.dtype != labels.dtype:
+            labels2 = labels2.to(labels.dtype)
+        c.labels2 = labels2.data_ptr()
+        c.g0, c.g1 = grp_shape
+    c.out_sum = out.data_ptr()
+    nscratch = lib.fh_quantile_scratch_bytes(ctypes.byref(c))
+    scratch = torch.empty(builtins.max(int(nscratch), 1), dtype=torch.uint8, device=dev)
+    c.scratch = scratch.data_ptr()
+    c.scratch_bytes = nscratch
+    c.stream = torch.cuda.current_stream(dev).cuda_stream
+    _ffi.check(lib.fh_grouped_quantile(ctypes.byref(c), 0))
+    for t in (values, labels, labels2, scratch, out):
+        if isinstance(t, torch.Tensor):
+            t.record_stream(torch.cuda.current_stream(dev))
+    return out
+
+
 def var_partials(
     group_idx, array, *, skipnan, size, labels2=None, grp_shape=None,
     global_counts=None, global_sums=None,

@@ -107,6 +107,8 @@ REDUCTIONS: dict[str, Aggregation] = {
     "nanquantile": Aggregation("nanquantile", None, skipnan=True),
     "median": Aggregation("median", None),
     "nanmedian": Aggregation("nanmedian", None, skipnan=True),
+    "mode": Aggregation("mode", None, preserves_dtype=True),
+    "nanmode": Aggregation("nanmode", None, skipnan=True, preserves_dtype=True),
     # bool reductions via min/max of the 0/1-cast input
     # (reference aggregations.py:651-676 all_/any_)
     "any": Aggregation("any", _ffi.SET_MAX_COUNT, combine={"max": "max", "count": "sum"}),

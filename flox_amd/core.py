@@ -353,7 +353,29 @@ def groupby_reduce(
             return (result.cpu().numpy(), *groups)
         return (result, *groups)
 
-    if func in ("argmax", "argmin", "nanargmax", "nanargmin"):
+    if func in ("mode", "nanmode"):
+        from .aggregate_hip import grouped_mode
+
+        if lead_M != 1:
+            raise NotImplementedError(f"{func} with leading array dims: next row")
+        if dist_on:
+            raise NotImplementedError("distributed mode needs a global sort: next row")
+        result = grouped_mode(
+            vals, labels, ngroups, skipnan=agg.skipnan, labels2=labels2, grp_shape=grp_pair
+        )
+        p = grouped_partials(
+            _ffi.SET_COUNT, vals, labels, ngroups, skipnan=True,
+            labels2=labels2, grp_shape=grp_pair,
+        )
+        counts_for_mask = p["count"]
+        # float results carry NaN for empty/propagated groups already;
+        # integer results need the empty fill
+        empty_mask = (
+            torch.zeros_like(counts_for_mask, dtype=torch.bool)
+            if arr.dtype.is_floating_point
+            else counts_for_mask == 0
+        )
+    elif func in ("argmax", "argmin", "nanargmax", "nanargmin"):
         # pass 1: the per-group extremum; pass 2: the smallest row index whose
         # value matches it (ties -> first occurrence, like np.argmax; a NaN
         # target matches NaN rows, so non-skip arg* land on the first NaN)

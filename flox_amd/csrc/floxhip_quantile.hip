@@ -186,6 +186,75 @@ __global__ void k_quantile(const uint64_t* __restrict__ keys, int64_t n,
   out[(int64_t)qi * ngroups + g] = res;
 }
 
+/* mode: longest equal-value run in each group's sorted span; sorted order
+ * makes scipy's tie rule (smallest of the modes) the first-found run.
+ * nan_policy: "propagate" (mode) -> NaN result when the group has NaNs;
+ * "omit" (nanmode) -> NaN tail ignored (reference aggregate_npg.py:185-215). */
+template <typename V, bool PACKED>
+__global__ void k_mode(const uint64_t* __restrict__ keys, int64_t n,
+                       const int64_t* __restrict__ off, int64_t ngroups,
+                       int skipna, V* __restrict__ out) {
+  const int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (g >= ngroups) return;
+  const int64_t start = off[g], end = off[g + 1];
+  const bool is_f = std::is_same<V, float>::value || std::is_same<V, double>::value;
+  const double NAN_ = __longlong_as_double(0x7FF8000000000000ll);
+  auto write_nan_or_zero = [&]() { out[g] = is_f ? (V)NAN_ : (V)0; };
+  if (end <= start) {
+    write_nan_or_zero();
+    return;
+  }
+  int64_t nan_start = end;
+  if (is_f) {
+    int64_t lo = start, hi = end;
+    const uint64_t nan_key = PACKED ? ((((uint64_t)(uint32_t)g) << 32) | 0xFFFFFFFFu) : ~0ull;
+    while (lo < hi) {
+      const int64_t mid = (lo + hi) >> 1;
+      if (keys[mid] < nan_key)
+        lo = mid + 1;
+      else
+        hi = mid;
+    }
+    nan_start = lo;
+  }
+  if (!skipna && nan_start != end) {
+    write_nan_or_zero(); /* propagate: any NaN -> NaN */
+    return;
+  }
+  if (nan_start == start) {
+    write_nan_or_zero(); /* all NaN under omit */
+    return;
+  }
+  uint64_t best = keys[start], cur = keys[start];
+  int64_t best_n = 1, cur_n = 1;
+  for (int64_t i = start + 1; i < nan_start; ++i) {
+    const uint64_t k = keys[i];
+    if (k == cur) {
+      ++cur_n;
+    } else {
+      if (cur_n > best_n) {
+        best = cur;
+        best_n = cur_n;
+      }
+      cur = k;
+      cur_n = 1;
+    }
+  }
+  if (cur_n > best_n) best = cur;
+  if (PACKED) {
+    const uint32_t e = (uint32_t)best;
+    if (std::is_same<V, float>::value)
+      out[g] = (V)dec32f(e);
+    else
+      out[g] = (V)dec32i((int32_t)e);
+  } else {
+    if (std::is_same<V, double>::value)
+      out[g] = (V)dec64f(best);
+    else
+      out[g] = (V)dec64i(best);
+  }
+}
+
 template <typename V, typename L>
 int run_quantile(fh_call* c, const double* q_dev, int nq, double* out) {
   hipStream_t stream = (hipStream_t)c->stream;
@@ -219,6 +288,13 @@ int run_quantile(fh_call* c, const double* q_dev, int nq, double* out) {
     hipLaunchKernelGGL((k_qoffsets<true>), dim3(ob), dim3(256), 0, stream, kout,
                        (const uint32_t*)nullptr, n, ngroups, off);
     FHQ_CHECK(hipGetLastError());
+    if (nq == 0) { /* mode */
+      int mb = (int)((ngroups + 255) / 256);
+      hipLaunchKernelGGL((k_mode<V, true>), dim3(mb), dim3(256), 0, stream,
+                         kout, n, off, ngroups, skipna, (V*)out);
+      FHQ_CHECK(hipGetLastError());
+      return 0;
+    }
     int qb = (int)((ngroups * nq + 255) / 256);
     hipLaunchKernelGGL((k_quantile<V, true>), dim3(qb), dim3(256), 0, stream,
                        kout, n, off, ngroups, q_dev, nq, skipna, out);
@@ -248,6 +324,13 @@ int run_quantile(fh_call* c, const double* q_dev, int nq, double* out) {
   hipLaunchKernelGGL((k_qoffsets<false>), dim3(ob), dim3(256), 0, stream, e_in,
                      c_in, n, ngroups, off);
   FHQ_CHECK(hipGetLastError());
+  if (nq == 0) { /* mode */
+    int mb = (int)((ngroups + 255) / 256);
+    hipLaunchKernelGGL((k_mode<V, false>), dim3(mb), dim3(256), 0, stream,
+                       e_in, n, off, ngroups, skipna, (V*)out);
+    FHQ_CHECK(hipGetLastError());
+    return 0;
+  }
   int qb = (int)((ngroups * nq + 255) / 256);
   hipLaunchKernelGGL((k_quantile<V, false>), dim3(qb), dim3(256), 0, stream,
                      e_in, n, off, ngroups, q_dev, nq, skipna, out);
@@ -287,7 +370,9 @@ int64_t fh_quantile_scratch_bytes(const fh_call* c) {
  * out_sum receives f64[nq*ngroups] results (NaN for empty groups, all-NaN
  * groups under skipna, and NaN-containing groups without skipna) */
 int fh_grouped_quantile(fh_call* c, int nq) {
-  if (!c || !c->values || !c->labels || !c->means || !c->out_sum) return 6;
+  /* nq == 0: grouped mode — out_sum then holds value-dtype[ngroups] */
+  if (!c || !c->values || !c->labels || !c->out_sum) return 6;
+  if (nq > 0 && !c->means) return 6;
   const double* q_dev = c->means;
   double* out = (double*)c->out_sum;
   switch (c->vdtype) {

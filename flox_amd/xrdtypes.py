@@ -16,7 +16,10 @@ FLOAT_FUNCS = {
     "quantile", "nanquantile", "median", "nanmedian",
 }
 # funcs whose output dtype equals the input dtype (preserves_dtype=True)
-PRESERVES_DTYPE = {"min", "nanmin", "max", "nanmax", "first", "nanfirst", "last", "nanlast"}
+PRESERVES_DTYPE = {
+    "min", "nanmin", "max", "nanmax", "first", "nanfirst", "last", "nanlast",
+    "mode", "nanmode",
+}
 NAN_SKIP = {
     "nansum",
     "nanprod",

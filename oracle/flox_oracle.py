@@ -26,6 +26,8 @@ import numpy as np
 # engine="flox" (core.py:856-859 raises for argreductions; aggregate_flox has
 # no first/last/any/all and numpy_groupies is absent in this container).
 ALL_FUNCS = (
+    "mode",
+    "nanmode",
     "quantile",
     "nanquantile",
     "median",
@@ -67,7 +69,10 @@ _POS_FUNCS = {"first", "last", "nanfirst", "nanlast"}
 _NAN_SKIP = {"nansum", "nanprod", "nanmean", "nanvar", "nanstd", "nanmin", "nanmax", "count"}
 # funcs whose output dtype equals the input dtype (reference: preserves_dtype=True,
 # aggregations.py:529-546)
-_PRESERVES_DTYPE = {"min", "nanmin", "max", "nanmax", "first", "last", "nanfirst", "nanlast"}
+_PRESERVES_DTYPE = {
+    "min", "nanmin", "max", "nanmax", "first", "last", "nanfirst", "nanlast",
+    "mode", "nanmode",
+}
 
 
 def _isnull(a: np.ndarray) -> np.ndarray:
@@ -350,6 +355,34 @@ def groupby_reduce(
             v = np.sqrt(v)
         result = v.astype(out_dtype)
         empty_mask = counts == 0
+    elif func in ("mode", "nanmode"):
+        from scipy.stats import mode as scipy_mode
+
+        order = np.argsort(codes, kind="stable")
+        sc = codes[order]
+        starts = np.searchsorted(sc, np.arange(ngroups), side="left")
+        ends = np.searchsorted(sc, np.arange(ngroups), side="right")
+        out = np.zeros((M, ngroups), dtype=array.dtype)
+        if array.dtype.kind in "fc":
+            out[:] = np.nan
+        for r in range(M):
+            row = vals2d[r]
+            for g in range(ngroups):
+                rows = order[starts[g] : ends[g]]
+                if rows.size == 0:
+                    continue
+                m_ = scipy_mode(
+                    row[rows],
+                    nan_policy="omit" if func == "nanmode" else "propagate",
+                    axis=-1,
+                    keepdims=True,
+                ).mode
+                val = np.asarray(m_).reshape(-1)
+                out[r, g] = val[0] if val.size else (np.nan if array.dtype.kind in "fc" else 0)
+        result = out.astype(out_dtype)
+        empty_mask = np.broadcast_to(
+            ~present if array.dtype.kind not in "fc" else np.zeros(ngroups, bool), (M, ngroups)
+        )
     elif func in _Q_FUNCS:
         if func in ("quantile", "nanquantile"):
             q = (finalize_kwargs or {}).get("q")

@@ -97,3 +97,102 @@ def test_gloo_world2_combine():
         errs.append(fail_q.get())
     assert not errs, errs
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
+# ---------------------------------------------------------------------------
+# full groupby_reduce distributed branch on CPU: stub the kernel layer with a
+# numpy partials producer so core's all-reduce paths execute under gloo
+# ---------------------------------------------------------------------------
+
+
+def _fake_partials_factory(np_mod, torch_mod):
+    import flox_amd._ffi as F
+
+    def fake(op_set, values, labels, ngroups, *, labels2=None, grp_shape=None,
+             means=None, target=None, row_offset=0, skipnan=False, force_path=0):
+        v = values.numpy() if hasattr(values, "numpy") else np_mod.asarray(values)
+        l = labels.numpy() if hasattr(labels, "numpy") else np_mod.asarray(labels)
+        valid = (l >= 0) & (l < ngroups)
+        nan = np_mod.isnan(v) if v.dtype.kind == "f" else np_mod.zeros(len(v), bool)
+        m = valid & ~nan if skipnan else valid
+        out = {}
+        if op_set in (F.SET_SUM_COUNT, F.SET_SUM_COUNT_PRESENT):
+            out["sum"] = torch_mod.tensor(
+                np_mod.bincount(l[m], weights=v[m].astype("f8"), minlength=ngroups))
+            out["count"] = torch_mod.tensor(
+                np_mod.bincount(l[valid & ~nan], minlength=ngroups))
+            if op_set == F.SET_SUM_COUNT_PRESENT:
+                out["present"] = torch_mod.tensor(
+                    (np_mod.bincount(l[valid], minlength=ngroups) > 0).astype("i4"))
+        elif op_set == F.SET_SSD:
+            mu = means.numpy()
+            d = v[m].astype("f8") - mu[l[m]]
+            out["sum"] = torch_mod.tensor(np_mod.bincount(l[m], weights=d * d, minlength=ngroups))
+        elif op_set == F.SET_COUNT:
+            out["count"] = torch_mod.tensor(np_mod.bincount(l[valid & ~nan], minlength=ngroups))
+        else:
+            raise NotImplementedError(op_set)
+        return out
+
+    return fake
+
+
+def _worker_full_branch(rank, world, port, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import numpy as np
+        import torch
+        import flox_amd.core as core
+        from oracle import groupby_reduce as oracle_reduce
+
+        rng = np.random.default_rng(7)
+        n, ng = 20_000, 37
+        vals = rng.standard_normal(n)
+        vals[rng.random(n) < 0.05] = np.nan
+        labels = rng.integers(0, ng, n)
+        sl = slice(rank * n // world, (rank + 1) * n // world)
+
+        fake = _fake_partials_factory(np, torch)
+        orig_cuda = torch.cuda.is_available
+        torch.cuda.is_available = lambda: True  # let core proceed on CPU
+        torch.cuda.current_device = lambda: 0
+        core.grouped_partials = fake
+        core._as_device_tensor = lambda x, d: (
+            x if isinstance(x, torch.Tensor) else torch.from_numpy(np.ascontiguousarray(x)))
+        _orig_device = torch.device
+        core.torch.device = lambda *a, **k: _orig_device("cpu")
+        try:
+            for func in ["nanmean", "sum", "count", "var"]:
+                got, *_ = core.groupby_reduce(
+                    vals[sl], labels[sl], func=func, expected_groups=np.arange(ng))
+                want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+                np.testing.assert_allclose(
+                    got.numpy() if hasattr(got, "numpy") else got, want,
+                    equal_nan=True, rtol=1e-10, atol=1e-10)
+        finally:
+            torch.cuda.is_available = orig_cuda
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}\n{traceback.format_exc()}")
+        raise
+
+
+def test_gloo_full_groupby_reduce_branch():
+    """The exact distributed code in core.groupby_reduce (partial all-reduce,
+    global-mean var, finalize) at world_size 2, with the kernel layer stubbed
+    by numpy — shard results must equal the whole-data oracle."""
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_full_branch, args=(r, 2, 29519, fail_q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    errs = []
+    while not fail_q.empty():
+        errs.append(fail_q.get())
+    assert not errs, errs[0]
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]

@@ -119,6 +119,25 @@ def test_quantile_family_vs_oracle(func, dtype):
     np.testing.assert_allclose(got, want, equal_nan=True, **tol)
 
 
+@pytest.mark.parametrize("func", ["mode", "nanmode"])
+@pytest.mark.parametrize("dtype", ["float32", "float64", "int64"])
+def test_mode_vs_oracle(func, dtype):
+    """Oracle calls scipy.stats.mode — the exact function the reference's
+    mode wraps (aggregate_npg.py:185-215)."""
+    rng = np.random.default_rng(zlib.crc32(f"m-{func}-{dtype}".encode()))
+    n, ng = 50_000, 97
+    labels = rng.integers(0, ng, n)
+    if np.dtype(dtype).kind == "f":
+        vals = rng.integers(-20, 20, n).astype(dtype)  # repeats so modes exist
+        vals[rng.random(n) < 0.02] = np.nan
+    else:
+        vals = rng.integers(-20, 20, n).astype(dtype)
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    assert got.dtype == want.dtype
+    np.testing.assert_allclose(got, want, equal_nan=True, rtol=0, atol=0)
+
+
 def test_any_all_bool():
     rng = np.random.default_rng(17)
     n, ng = 100_000, 300
