@@ -217,12 +217,10 @@ __global__ void k_mode(const uint64_t* __restrict__ keys, int64_t n,
     }
     nan_start = lo;
   }
-  if (!skipna && nan_start != end) {
-    write_nan_or_zero(); /* propagate: any NaN -> NaN */
-    return;
-  }
   if (nan_start == start) {
-    write_nan_or_zero(); /* all NaN under omit */
+    /* no non-NaN values: NaN either way (scipy omit warns and returns NaN;
+     * propagate's most-common value IS NaN) */
+    write_nan_or_zero();
     return;
   }
   uint64_t best = keys[start], cur = keys[start];
@@ -240,7 +238,16 @@ __global__ void k_mode(const uint64_t* __restrict__ keys, int64_t n,
       cur_n = 1;
     }
   }
-  if (cur_n > best_n) best = cur;
+  if (cur_n > best_n) {
+    best = cur;
+    best_n = cur_n;
+  }
+  if (!skipna && (end - nan_start) > best_n) {
+    /* scipy propagate counts NaNs as a value: a strictly-most-frequent NaN
+     * run wins (ties resolve to the smallest, i.e. any non-NaN) */
+    write_nan_or_zero();
+    return;
+  }
   if (PACKED) {
     const uint32_t e = (uint32_t)best;
     if (std::is_same<V, float>::value)
