@@ -8,5 +8,6 @@ with an RCCL-over-xGMI combine across GPUs. See DESIGN.md.
 
 from .aggregations import REDUCTIONS, Aggregation, generic_aggregate  # noqa: F401
 from .core import groupby_reduce  # noqa: F401
+from .scan import groupby_scan  # noqa: F401
 
 __version__ = "0.1.0"

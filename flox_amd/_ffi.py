@@ -102,6 +102,10 @@ def load_library():
     lib.fh_quantile_scratch_bytes.restype = ctypes.c_int64
     lib.fh_grouped_quantile.argtypes = [ctypes.POINTER(FhCall), ctypes.c_int]
     lib.fh_grouped_quantile.restype = ctypes.c_int
+    lib.fh_scan_scratch_bytes.argtypes = [ctypes.POINTER(FhCall)]
+    lib.fh_scan_scratch_bytes.restype = ctypes.c_int64
+    lib.fh_grouped_scan.argtypes = [ctypes.POINTER(FhCall), ctypes.c_int]
+    lib.fh_grouped_scan.restype = ctypes.c_int
     lib.fh_error_string.argtypes = [ctypes.c_int]
     lib.fh_error_string.restype = ctypes.c_char_p
     lib.fh_version.argtypes = []

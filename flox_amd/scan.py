@@ -1,0 +1,109 @@
+"""groupby_scan: grouped scans with flox semantics on MI355X.
+
+Mirrors the reference's eager scan path (flox/scan.py:101-352 groupby_scan
+-> chunk_scan -> aggregate_flox._np_grouped_scan / ffill,
+aggregate_flox.py:269-325): cumulative sums and forward/backward fills
+within each group, in the original row order, over the trailing axis.
+"""
+
+from __future__ import annotations
+
+import builtins
+import ctypes
+
+import numpy as np
+import torch
+
+from . import _ffi
+from ._ffi import FhCall
+from .core import _as_device_tensor, _combined_codes, _factorize_device, _np_dtype
+
+SCAN_OPS = {"cumsum": 0, "nancumsum": 1, "ffill": 2, "bfill": 3}
+
+
+def groupby_scan(array, *by, func: str, expected_groups=None, axis=None, dtype=None):
+    """Grouped scan. Returns an array shaped like ``array``.
+
+    cumsum/nancumsum: per-group running sum in row order (np.cumsum /
+    np.nancumsum semantics, incl. integer promotion to the platform int).
+    ffill/bfill: carry the last/next non-NaN value within the group.
+    Rows with labels outside ``expected_groups`` scan as their own group
+    (the reference's NaN-sentinel group, factorize.py:201-210).
+    """
+    if func not in SCAN_OPS:
+        raise NotImplementedError(f"scan {func!r}")
+    if not torch.cuda.is_available():
+        raise RuntimeError("flox_amd.groupby_scan requires a GPU (engine='hip')")
+    device = torch.device("cuda", torch.cuda.current_device())
+    return_numpy = not isinstance(array, torch.Tensor)
+    arr = _as_device_tensor(array, device)
+    bys = tuple(_as_device_tensor(b, device) for b in by)
+    if len(bys) == 0:
+        raise ValueError("need at least one by array")
+    by_shape = bys[0].shape
+    if tuple(arr.shape[arr.ndim - len(by_shape):]) != tuple(by_shape):
+        raise ValueError("by must align with trailing dims of array")
+    if arr.ndim != bys[0].ndim:
+        raise NotImplementedError("scan with leading array dims: next row")
+    if axis is not None:
+        ax = axis if isinstance(axis, (tuple, list)) else (axis,)
+        ax = tuple(a % arr.ndim for a in ax)
+        if tuple(sorted(ax)) != tuple(range(arr.ndim - len(by_shape), arr.ndim)):
+            raise NotImplementedError("scan over an axis subset: next row")
+
+    orig_shape = arr.shape
+    vals = arr.reshape(-1)
+    # integer promotion like np.cumsum (sub-platform ints accumulate in intp)
+    if func in ("cumsum", "nancumsum") and vals.dtype in (torch.int32, torch.bool):
+        vals = vals.to(torch.int64)
+    if vals.dtype == torch.bool:
+        vals = vals.to(torch.int64)
+
+    if expected_groups is not None and not isinstance(expected_groups, tuple):
+        expected_groups = (expected_groups,)
+    if expected_groups is None:
+        expected_groups = (None,) * len(bys)
+    facs = [_factorize_device(b.reshape(-1), e, True) for b, e in zip(bys, expected_groups)]
+    ngroups = 1
+    for f in facs:
+        ngroups *= f.ngroups
+    labels, labels2, grp_pair = _combined_codes(facs)
+
+    lib = _ffi.load_library()
+    vals = vals.contiguous()
+    labels = labels.contiguous()
+    if labels.dtype not in (torch.int64, torch.int32):
+        labels = labels.to(torch.int64)
+    out = torch.empty_like(vals)
+    c = FhCall()
+    c.vdtype = {torch.float32: _ffi.F32, torch.float64: _ffi.F64,
+                torch.int64: _ffi.I64, torch.int32: _ffi.I32}[vals.dtype]
+    c.ldtype = _ffi.L_I64 if labels.dtype == torch.int64 else _ffi.L_I32
+    c.n = vals.numel()
+    c.ngroups = ngroups
+    c.values = vals.data_ptr()
+    c.labels = labels.data_ptr()
+    if labels2 is not None:
+        labels2 = labels2.contiguous()
+        if labels2.dtype != labels.dtype:
+            labels2 = labels2.to(labels.dtype)
+        c.labels2 = labels2.data_ptr()
+        c.g0, c.g1 = grp_pair
+    c.out_sum = out.data_ptr()
+    nscratch = lib.fh_scan_scratch_bytes(ctypes.byref(c))
+    scratch = torch.empty(builtins.max(int(nscratch), 1), dtype=torch.uint8, device=device)
+    c.scratch = scratch.data_ptr()
+    c.scratch_bytes = nscratch
+    c.stream = torch.cuda.current_stream(device).cuda_stream
+    _ffi.check(lib.fh_grouped_scan(ctypes.byref(c), SCAN_OPS[func]))
+    for t in (vals, labels, labels2, scratch, out):
+        if isinstance(t, torch.Tensor):
+            t.record_stream(torch.cuda.current_stream(device))
+
+    if dtype is not None:
+        td = torch.from_numpy(np.empty(0, dtype=np.dtype(dtype))).dtype
+        out = out.to(td)
+    out = out.reshape(orig_shape)
+    if return_numpy:
+        return out.cpu().numpy()
+    return out

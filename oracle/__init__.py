@@ -15,4 +15,4 @@ reference's own implementation (see tests/golden/generate.py, runnable only in
 the build container where /root/reference is mounted) — tests/test_oracle.py.
 """
 
-from .flox_oracle import groupby_reduce, ALL_FUNCS  # noqa: F401
+from .flox_oracle import ALL_FUNCS, groupby_reduce, groupby_scan  # noqa: F401

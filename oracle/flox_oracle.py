@@ -529,3 +529,63 @@ def groupby_reduce(
 
     result = result.reshape(lead_shape + grp_shape)
     return (result, *found)
+
+
+def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
+    """Grouped scans (reference scan.py:101-352 + aggregate_flox.py:269-325):
+    stable sort to group order, scan, undo the permutation. Rows with labels
+    outside expected_groups form their own trailing group (the reference's
+    NaN-sentinel group)."""
+    array = np.asarray(array)
+    bys = tuple(np.asarray(b) for b in by)
+    if expected_groups is not None and not isinstance(expected_groups, tuple):
+        expected_groups = (expected_groups,)
+    if expected_groups is None:
+        expected_groups = (None,) * len(bys)
+    codes_list, found = [], []
+    for b, e in zip(bys, expected_groups):
+        c, f = _factorize_single(b.reshape(-1), e)
+        codes_list.append(c)
+        found.append(f)
+    grp_shape = tuple(len(f) for f in found)
+    ngroups = math.prod(grp_shape)
+    codes = _ravel_codes(codes_list, grp_shape) if len(bys) > 1 else codes_list[0]
+    codes = codes.copy()
+    codes[codes < 0] = ngroups  # sentinel group scans together, like factorize_:201-210
+
+    flat = array.reshape(-1)
+    if func in ("cumsum", "nancumsum") and flat.dtype.kind in "iub" and flat.dtype.itemsize < 8:
+        flat = flat.astype(np.int64)
+    perm = np.argsort(codes, kind="stable")
+    sv = flat[perm].astype(flat.dtype)
+    sc = codes[perm]
+    seg_start = np.concatenate(([True], sc[1:] != sc[:-1]))
+    out_sorted = np.empty_like(sv, dtype=np.result_type(sv.dtype))
+    starts = np.flatnonzero(seg_start)
+    ends = np.append(starts[1:], len(sv))
+    for s0, s1 in zip(starts, ends):
+        seg = sv[s0:s1]
+        if func == "cumsum":
+            acc = np.nancumsum(seg)
+            if seg.dtype.kind in "fc":
+                nanpos = np.cumsum(_isnull(seg)) > 0
+                acc = acc.astype(float)
+                acc[nanpos] = np.nan
+            out_sorted[s0:s1] = acc
+        elif func == "nancumsum":
+            out_sorted[s0:s1] = np.nancumsum(seg)
+        elif func in ("ffill", "bfill"):
+            seg2 = seg[::-1] if func == "bfill" else seg.copy()
+            mask = _isnull(seg2)
+            idx = np.where(mask, 0, np.arange(len(seg2)))
+            np.maximum.accumulate(idx, out=idx)
+            filled = seg2[idx]
+            filled[np.cumsum(~mask) == 0] = np.nan
+            out_sorted[s0:s1] = filled[::-1] if func == "bfill" else filled
+        else:
+            raise NotImplementedError(func)
+    inv = np.argsort(perm, kind="stable")
+    out = out_sorted[inv]
+    if dtype is not None:
+        out = out.astype(dtype)
+    return out.reshape(array.shape)

@@ -210,8 +210,27 @@ def gen_cases():
     )
 
 
+def gen_scan_cases():
+    rng = np.random.default_rng(77)
+    vals = rng.standard_normal(200)
+    vals[rng.random(200) < 0.15] = np.nan
+    by = rng.integers(0, 7, 200)
+    for func in ["cumsum", "nancumsum", "ffill", "bfill"]:
+        yield f"scan_{func}_f64", dict(array=vals, by=by, func=func)
+        yield f"scan_{func}_f32", dict(array=vals.astype(np.float32), by=by, func=func)
+    ints = rng.integers(-50, 50, 200).astype(np.int32)
+    yield "scan_cumsum_i32", dict(array=ints, by=by, func="cumsum")
+    # NaN labels -> sentinel group scans together
+    nby = by.astype(float)
+    nby[rng.random(200) < 0.1] = np.nan
+    yield "scan_nancumsum_nanby", dict(array=vals, by=nby, func="nancumsum")
+
+
 def main():
     core = load_reference()
+    import importlib
+
+    refscan = importlib.import_module("floxref.scan")
     out = {}
     n_done, n_skip = 0, 0
     for name, kw in gen_cases():
@@ -262,6 +281,22 @@ def main():
             if "q" in fk:
                 out[f"{name}::q"] = np.asarray(fk["q"])
                 out[f"{name}::q_scalar"] = np.asarray(np.isscalar(fk["q"]))
+    for name, kw in gen_scan_cases():
+        by = kw.pop("by")
+        bys = by if isinstance(by, tuple) else (by,)
+        arr = kw.pop("array")
+        try:
+            result = refscan.groupby_scan(arr, *bys, **kw)
+        except Exception as e:  # pragma: no cover
+            print(f"SKIP {name}: {type(e).__name__}: {e}")
+            n_skip += 1
+            continue
+        out[f"{name}::result"] = np.asarray(result)
+        out[f"{name}::array"] = np.asarray(arr)
+        for i, b in enumerate(bys):
+            out[f"{name}::by{i}"] = np.asarray(b)
+        out[f"{name}::scan"] = np.asarray(True)
+        n_done += 1
     np.savez_compressed(OUT, **out)
     print(f"wrote {OUT}: {n_done} cases ({n_skip} skipped)")
 

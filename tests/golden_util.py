@@ -48,6 +48,9 @@ def load_golden_cases():
             kw["finalize_kwargs"] = {"q": qv.item() if get("q_scalar") else qv.tolist()}
         if get("isbin") is not None:
             kw["isbin"] = True
+        if get("scan") is not None:
+            kw["_scan"] = True
+            func = name.split("_")[1]
         yield name, dict(array=get("array"), by=tuple(bys), func=func, **kw), get("result"), groups
 
 
@@ -60,6 +63,11 @@ def tolerance_for(name, result_dtype):
     if result_dtype.kind in "iub":
         return dict(rtol=0, atol=0)
     if result_dtype.itemsize == 4:
+        if name.startswith("scan_"):
+            # the reference computes segment scans as (global cumsum) minus
+            # (previous groups' cumsum) in fp32 — cancellation differs from a
+            # direct per-segment sum by ~1 ulp of the global running sum
+            return dict(rtol=2e-6, atol=1e-5)
         return dict(rtol=2e-6, atol=1e-7)
     if "var" in name or "std" in name:
         return dict(rtol=1e-12, atol=1e-14)

@@ -21,7 +21,11 @@ def test_golden_parity(name, inputs, expected, groups):
     kw = dict(inputs)
     arr = kw.pop("array")
     bys = kw.pop("by")
-    result, *found = flox_amd.groupby_reduce(arr, *bys, **kw)
+    if kw.pop("_scan", False):
+        result = flox_amd.groupby_scan(arr, *bys, **kw)
+        found = []
+    else:
+        result, *found = flox_amd.groupby_reduce(arr, *bys, **kw)
     assert result.shape == expected.shape
     assert result.dtype == expected.dtype, (result.dtype, expected.dtype)
     tol = tolerance_for(name, expected.dtype)
@@ -136,6 +140,28 @@ def test_mode_vs_oracle(func, dtype):
     got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
     assert got.dtype == want.dtype
     np.testing.assert_allclose(got, want, equal_nan=True, rtol=0, atol=0)
+
+
+@pytest.mark.parametrize("func", ["cumsum", "nancumsum", "ffill", "bfill"])
+@pytest.mark.parametrize("dtype", ["float32", "float64", "int64"])
+def test_scan_vs_oracle(func, dtype):
+    from oracle import groupby_scan as oracle_scan
+
+    rng = np.random.default_rng(zlib.crc32(f"s-{func}-{dtype}".encode()))
+    n, ng = 200_000, 512
+    labels = rng.integers(0, ng, n)
+    if np.dtype(dtype).kind == "f":
+        vals = (rng.standard_normal(n)).astype(dtype)
+        vals[rng.random(n) < 0.05] = np.nan
+    else:
+        vals = rng.integers(-100, 100, n).astype(dtype)
+    want = oracle_scan(vals, labels, func=func, expected_groups=np.arange(ng))
+    got = flox_amd.groupby_scan(vals, labels, func=func, expected_groups=np.arange(ng))
+    assert got.dtype == want.dtype, (got.dtype, want.dtype)
+    tol = dict(rtol=0, atol=0) if np.dtype(dtype).kind in "iu" else (
+        dict(rtol=3e-6, atol=1e-5) if np.dtype(dtype).itemsize == 4 else dict(rtol=1e-12, atol=1e-12)
+    )
+    np.testing.assert_allclose(got, want, equal_nan=True, **tol)
 
 
 def test_any_all_bool():

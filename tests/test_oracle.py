@@ -5,6 +5,7 @@ import numpy as np
 import pytest
 
 from oracle import groupby_reduce as oracle_reduce
+from oracle import groupby_scan as oracle_scan
 from tests.golden_util import load_golden_cases, tolerance_for
 
 CASES = list(load_golden_cases())
@@ -15,7 +16,11 @@ def test_oracle_matches_reference(name, inputs, expected, groups):
     kw = dict(inputs)
     arr = kw.pop("array")
     bys = kw.pop("by")
-    result, *found = oracle_reduce(arr, *bys, **kw)
+    if kw.pop("_scan", False):
+        result = oracle_scan(arr, *bys, **kw)
+        found = []
+    else:
+        result, *found = oracle_reduce(arr, *bys, **kw)
     assert result.shape == expected.shape, (result.shape, expected.shape)
     assert result.dtype == expected.dtype, (result.dtype, expected.dtype)
     tol = tolerance_for(name, expected.dtype)
