@@ -580,7 +580,8 @@ def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
             idx = np.where(mask, 0, np.arange(len(seg2)))
             np.maximum.accumulate(idx, out=idx)
             filled = seg2[idx]
-            filled[np.cumsum(~mask) == 0] = np.nan
+            if filled.dtype.kind in "fc":
+                filled[np.cumsum(~mask) == 0] = np.nan
             out_sorted[s0:s1] = filled[::-1] if func == "bfill" else filled
         else:
             raise NotImplementedError(func)
