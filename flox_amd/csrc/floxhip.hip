@@ -719,25 +719,29 @@ __launch_bounds__(256) __global__ void k_part_count(
     if (s_hist[i]) atomicAdd(&bucket_counts[i], s_hist[i]);
 }
 
-/* tile-staged scatter: per tile, histogram -> block scan -> global
+/* tile-staged scatter: per tile, histogram -> tiny scan -> global
  * reservation (one returning atomic per nonempty bucket) -> bucket-ordered
- * staging in LDS with per-slot destinations -> coalesced dump */
+ * staging in LDS with per-slot destinations -> coalesced dump.
+ * With the two-level scheme this pass NEVER sees more than PART_SUB=64
+ * buckets (pass A uses the <= 64 super-buckets; single-level runs only when
+ * the fine bucket count is <= 64), so the histogram/scan state is tiny and
+ * three blocks fit per CU. */
 constexpr int PART_BLOCK = 512;
 
 template <typename V, typename L>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     const V* __restrict__ values, const L* __restrict__ labels,
     const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
-    int64_t g1, int shift, int B, int Bpad /* pow2 >= B */,
+    int64_t g1, int shift, int B, int Bpad /* unused, <= 64 buckets */,
     uint32_t* __restrict__ cursors, PairT<V>* __restrict__ pairs) {
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
-  constexpr int RPT = T / PART_BLOCK; /* rows per thread per tile */
+  constexpr int RPT = T / PART_BLOCK;
+  constexpr int NB = 64;
   extern __shared__ __attribute__((aligned(16))) char smem_ps[];
-  uint32_t* s_hist = (uint32_t*)smem_ps;              /* [Bpad] scanned in place */
-  uint32_t* s_gbase = s_hist + Bpad;                  /* [Bpad] gbase - excl per bucket */
-  uint32_t* s_cur = s_gbase + Bpad;                   /* [Bpad] running cursor */
-  uint32_t* s_tot = s_cur + Bpad;                     /* [PART_BLOCK] chunk totals */
-  uint32_t* s_dest = s_tot + PART_BLOCK;              /* [T] per-slot global dest */
+  uint32_t* s_hist = (uint32_t*)smem_ps;              /* [NB+1] scanned in place */
+  uint32_t* s_gbase = s_hist + NB + 1;                /* [NB] gbase - excl */
+  uint32_t* s_cur = s_gbase + NB;                     /* [NB] running cursor */
+  uint32_t* s_dest = s_cur + NB + 3;                  /* [T]; pads stage to 16 B */
   PairT<V>* s_stage = (PairT<V>*)(s_dest + T);        /* [T] */
 
   const bool twolab = labels2 != nullptr;
@@ -745,7 +749,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
 
   for (int64_t tile = (int64_t)blockIdx.x * T; tile < n; tile += (int64_t)gridDim.x * T) {
     const int nt = (int)((n - tile < T) ? (n - tile) : T);
-    for (int i = tid; i < Bpad; i += PART_BLOCK) s_hist[i] = 0;
+    for (int i = tid; i <= NB; i += PART_BLOCK) s_hist[i] = 0;
     __syncthreads();
 
     V rv[RPT];
@@ -769,45 +773,23 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
       }
     }
     __syncthreads();
-    /* exclusive block scan of s_hist[Bpad]: per-thread contiguous chunk
-     * scans + one 512-wide Hillis-Steele over the chunk totals (no private
-     * arrays -> no scratch spill) */
-    {
-      const int C = Bpad / PART_BLOCK; /* >= 1 (Bpad is padded to >= 512) */
-      const int b0 = tid * C;
+    if (tid == 0) {
       uint32_t run = 0;
-      for (int j = 0; j < C; ++j) {
-        const uint32_t x = s_hist[b0 + j];
-        s_hist[b0 + j] = run;
+      for (int b = 0; b <= NB; ++b) {
+        const uint32_t x = s_hist[b];
+        s_hist[b] = run;
         run += x;
       }
-      s_tot[tid] = run;
-      __syncthreads();
-      uint32_t v = s_tot[tid];
-      for (int d = 1; d < PART_BLOCK; d <<= 1) {
-        const uint32_t o = (tid >= d) ? s_tot[tid - d] : 0u;
-        __syncthreads();
-        v += o;
-        s_tot[tid] = v;
-        __syncthreads();
-      }
-      const uint32_t chunk_excl = v - run;
-      for (int j = 0; j < C; ++j) s_hist[b0 + j] += chunk_excl;
     }
     __syncthreads();
-    const uint32_t total = s_tot[PART_BLOCK - 1];
-    /* reserve each nonempty bucket's run globally; store gbase - excl so the
-     * staged slot index alone gives the global destination */
     for (int b = tid; b < B; b += PART_BLOCK) {
       const uint32_t excl = s_hist[b];
-      const uint32_t incl = (b + 1 < Bpad) ? s_hist[b + 1] : total;
-      const uint32_t cnt = incl - excl;
+      const uint32_t cnt = s_hist[b + 1] - excl;
       s_cur[b] = excl;
       if (cnt) s_gbase[b] = atomicAdd(&cursors[b], cnt) - excl;
     }
     __syncthreads();
-    const int valid = (int)total;
-    /* bucket-ordered staging with per-slot destination */
+    const int valid = (int)s_hist[NB];
 #pragma unroll
     for (int k = 0; k < RPT; ++k) {
       if (rbk[k] >= 0) {
@@ -822,7 +804,6 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     __syncthreads();
   }
 }
-
 
 /* second-level scatter: rows of one super-bucket (already contiguous in
  * `in`) are partitioned into their <= 64 fine buckets. Same tile machinery
@@ -1054,10 +1035,10 @@ PartPlan part_plan(const fh_call* c) {
   p.B1 = p.two_level ? (p.B + PART_SUB - 1) / PART_SUB : 0;
   /* pass-A bucket count: B1 when two-level, else the fine B */
   const int bA = p.two_level ? p.B1 : p.B;
-  p.Bpad = 512; /* = PART_BLOCK, so every thread owns >= 1 scan chunk entry */
-  while (p.Bpad < bA) p.Bpad <<= 1;
+  if (bA > PART_SUB) return p; /* scatter passes handle <= 64 buckets */
+  p.Bpad = PART_SUB;
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
-  p.scatter_lds = (int64_t)3 * p.Bpad * 4 + 512 * 4 + (int64_t)T * 4 +
+  p.scatter_lds = (int64_t)(3 * (PART_SUB + 2)) * 4 + (int64_t)T * 4 +
                   (int64_t)T * sizeof(PairT<V>);
   p.scatter2_lds = (int64_t)(3 * (PART_SUB + 1)) * 4 + (int64_t)T * 4 +
                    (int64_t)T * sizeof(PairT<V>);
