@@ -1076,7 +1076,9 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
   {
     int64_t wb = (c->n + 255) / 256;
     int nb0 = (int)(wb < 2048 ? (wb > 0 ? wb : 1) : 2048);
-    hipLaunchKernelGGL((k_part_count<L>), dim3(nb0), dim3(256), pp.Bpad * 4,
+    /* the count pass histograms the FINE buckets (up to 4096) */
+    const int64_t hist_lds = ((int64_t)pp.B + 63) / 64 * 64 * 4;
+    hipLaunchKernelGGL((k_part_count<L>), dim3(nb0), dim3(256), hist_lds,
                        stream, (const L*)c->labels, (const L*)c->labels2, c->n,
                        c->ngroups, c->g0, c->g1, pp.shift, pp.B, counts);
     FH_CHECK(hipGetLastError());
