@@ -15,3 +15,10 @@ for _ in range(3):
     p = grouped_partials(_ffi.SET_SUM_COUNT_PRESENT, vals, labels, ng)
 torch.cuda.synchronize()
 print(f"{(time.perf_counter()-t0)/3*1e3:.2f} ms, path={p['_path']}")
+# correctness invariants at full scale
+assert int(p["count"].sum().item()) == n, int(p["count"].sum().item())
+tot = vals.sum(dtype=torch.float64).item()
+got = p["sum"].sum().item()
+assert abs(got - tot) < 1e-4 * abs(tot) + 1e-3, (got, tot)
+assert int(p["present"].min().item()) >= 0
+print("invariants OK")
