@@ -728,6 +728,23 @@ __launch_bounds__(256) __global__ void k_part_count(
  * three blocks fit per CU. */
 constexpr int PART_BLOCK = 512;
 
+/* per-wave-histogram tile scatter machinery, shared by both scatter passes:
+ * each of the NW waves owns its own 64-entry histogram slice (layout
+ * [bucket][wave]) so LDS atomic contention is divided by NW, the exclusive
+ * scan over the 512 (bucket, wave) slots runs as wave shuffle scans plus
+ * one tiny cross-wave combine (2 barriers instead of 18), and the scanned
+ * slots double as the staging cursors. */
+constexpr int PART_NW = PART_BLOCK / 64;
+
+__device__ __forceinline__ uint32_t wave_incl_scan(uint32_t v) {
+#pragma unroll
+  for (int d = 1; d < 64; d <<= 1) {
+    const uint32_t o = __shfl_up((int)v, d);
+    if ((threadIdx.x & 63) >= d) v += o;
+  }
+  return v;
+}
+
 template <typename V, typename L>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     const V* __restrict__ values, const L* __restrict__ labels,
@@ -737,19 +754,21 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
   constexpr int RPT = T / PART_BLOCK;
   constexpr int NB = 64;
+  constexpr int NE = NB * PART_NW; /* = PART_BLOCK slots */
   extern __shared__ __attribute__((aligned(16))) char smem_ps[];
-  uint32_t* s_hist = (uint32_t*)smem_ps;              /* [NB+1] scanned in place */
-  uint32_t* s_gbase = s_hist + NB + 1;                /* [NB] gbase - excl */
-  uint32_t* s_cur = s_gbase + NB;                     /* [NB] running cursor */
-  uint32_t* s_dest = s_cur + NB + 3;                  /* [T]; pads stage to 16 B */
+  uint32_t* s_slot = (uint32_t*)smem_ps;              /* [NE] hist -> excl -> cursor */
+  uint32_t* s_wtot = s_slot + NE;                     /* [PART_NW + 1] wave totals */
+  uint32_t* s_gbase = s_wtot + PART_NW + 1;           /* [NB] gbase - bucket excl */
+  uint32_t* s_dest = s_gbase + NB + 3;                /* [T]; pads stage to 16 B */
   PairT<V>* s_stage = (PairT<V>*)(s_dest + T);        /* [T] */
 
   const bool twolab = labels2 != nullptr;
   const int tid = threadIdx.x;
+  const int wid = tid >> 6;
 
   for (int64_t tile = (int64_t)blockIdx.x * T; tile < n; tile += (int64_t)gridDim.x * T) {
     const int nt = (int)((n - tile < T) ? (n - tile) : T);
-    for (int i = tid; i <= NB; i += PART_BLOCK) s_hist[i] = 0;
+    s_slot[tid] = 0;
     __syncthreads();
 
     V rv[RPT];
@@ -768,39 +787,49 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
           rv[k] = values[i];
           rbk[k] = (int)(code >> shift);
           rlc[k] = (uint32_t)(code - ((int64_t)rbk[k] << shift));
-          atomicAdd(&s_hist[rbk[k]], 1u);
+          atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
         }
       }
     }
     __syncthreads();
-    if (tid == 0) {
-      uint32_t run = 0;
-      for (int b = 0; b <= NB; ++b) {
-        const uint32_t x = s_hist[b];
-        s_hist[b] = run;
-        run += x;
+    /* exclusive scan over the NE slots: wave shuffle scans + wave totals */
+    {
+      const uint32_t mine = s_slot[tid];
+      const uint32_t incl = wave_incl_scan(mine);
+      if ((tid & 63) == 63) s_wtot[wid] = incl;
+      __syncthreads();
+      if (tid == 0) {
+        uint32_t run = 0;
+        for (int w = 0; w < PART_NW; ++w) {
+          const uint32_t x = s_wtot[w];
+          s_wtot[w] = run;
+          run += x;
+        }
+        s_wtot[PART_NW] = run;
       }
+      __syncthreads();
+      s_slot[tid] = incl - mine + s_wtot[wid];
     }
     __syncthreads();
+    const uint32_t total = s_wtot[PART_NW];
     for (int b = tid; b < B; b += PART_BLOCK) {
-      const uint32_t excl = s_hist[b];
-      const uint32_t cnt = s_hist[b + 1] - excl;
-      s_cur[b] = excl;
+      const uint32_t excl = s_slot[b * PART_NW];
+      const uint32_t nxt = (b + 1 < NB) ? s_slot[(b + 1) * PART_NW] : total;
+      const uint32_t cnt = nxt - excl;
       if (cnt) s_gbase[b] = atomicAdd(&cursors[b], cnt) - excl;
     }
     __syncthreads();
-    const int valid = (int)s_hist[NB];
 #pragma unroll
     for (int k = 0; k < RPT; ++k) {
       if (rbk[k] >= 0) {
-        const uint32_t pos = atomicAdd(&s_cur[rbk[k]], 1u);
+        const uint32_t pos = atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
         s_stage[pos].v = rv[k];
         s_stage[pos].lc = rlc[k];
         s_dest[pos] = s_gbase[rbk[k]] + pos;
       }
     }
     __syncthreads();
-    for (int i = tid; i < valid; i += PART_BLOCK) pairs[s_dest[i]] = s_stage[i];
+    for (int i = tid; i < (int)total; i += PART_BLOCK) pairs[s_dest[i]] = s_stage[i];
     __syncthreads();
   }
 }
@@ -817,22 +846,25 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     PairT<V>* __restrict__ out) {
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
   constexpr int RPT = T / PART_BLOCK;
+  constexpr int NB = PART_SUB;
+  constexpr int NE = NB * PART_NW;
   extern __shared__ __attribute__((aligned(16))) char smem_p2[];
-  uint32_t* s_hist = (uint32_t*)smem_p2;       /* [PART_SUB+1] scanned in place */
-  uint32_t* s_gbase = s_hist + PART_SUB + 1;   /* [PART_SUB] */
-  uint32_t* s_cur = s_gbase + PART_SUB;        /* [PART_SUB] */
-  uint32_t* s_dest = s_cur + PART_SUB;         /* [T] */
-  PairT<V>* s_stage = (PairT<V>*)(s_dest + T); /* [T] */
+  uint32_t* s_slot = (uint32_t*)smem_p2;
+  uint32_t* s_wtot = s_slot + NE;
+  uint32_t* s_gbase = s_wtot + PART_NW + 1;
+  uint32_t* s_dest = s_gbase + NB + 3;
+  PairT<V>* s_stage = (PairT<V>*)(s_dest + T);
 
   const int sb = blockIdx.y;
   const int tid = threadIdx.x;
+  const int wid = tid >> 6;
   const int64_t r0 = baseA[sb], r1 = baseA[sb + 1];
   const uint32_t lmask = (1u << shift) - 1u;
 
   for (int64_t tile = r0 + (int64_t)blockIdx.x * T; tile < r1;
        tile += (int64_t)gridDim.x * T) {
     const int nt = (int)((r1 - tile < T) ? (r1 - tile) : T);
-    for (int i = tid; i <= PART_SUB; i += PART_BLOCK) s_hist[i] = 0;
+    s_slot[tid] = 0;
     __syncthreads();
     V rv[RPT];
     uint32_t rlc[RPT];
@@ -846,39 +878,47 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
         rv[k] = pr.v;
         rbk[k] = (int)(pr.lc >> shift);
         rlc[k] = pr.lc & lmask;
-        atomicAdd(&s_hist[rbk[k]], 1u);
+        atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
       }
     }
     __syncthreads();
-    /* tiny exclusive scan (64 entries) by thread 0 */
-    if (tid == 0) {
-      uint32_t run = 0;
-      for (int b = 0; b <= PART_SUB; ++b) {
-        const uint32_t x = s_hist[b];
-        s_hist[b] = run;
-        run += x;
+    {
+      const uint32_t mine = s_slot[tid];
+      const uint32_t incl = wave_incl_scan(mine);
+      if ((tid & 63) == 63) s_wtot[wid] = incl;
+      __syncthreads();
+      if (tid == 0) {
+        uint32_t run = 0;
+        for (int w = 0; w < PART_NW; ++w) {
+          const uint32_t x = s_wtot[w];
+          s_wtot[w] = run;
+          run += x;
+        }
+        s_wtot[PART_NW] = run;
       }
+      __syncthreads();
+      s_slot[tid] = incl - mine + s_wtot[wid];
     }
     __syncthreads();
-    for (int b = tid; b < PART_SUB; b += PART_BLOCK) {
-      const uint32_t excl = s_hist[b];
-      const uint32_t cnt = s_hist[b + 1] - excl;
-      s_cur[b] = excl;
+    const uint32_t total = s_wtot[PART_NW];
+    for (int b = tid; b < NB; b += PART_BLOCK) {
+      const uint32_t excl = s_slot[b * PART_NW];
+      const uint32_t nxt = (b + 1 < NB) ? s_slot[(b + 1) * PART_NW] : total;
+      const uint32_t cnt = nxt - excl;
       if (cnt) s_gbase[b] = atomicAdd(&cursors[sb * PART_SUB + b], cnt) - excl;
     }
     __syncthreads();
-    const int valid = (int)s_hist[PART_SUB];
 #pragma unroll
     for (int k = 0; k < RPT; ++k) {
       if (rbk[k] >= 0) {
-        const uint32_t pos = atomicAdd(&s_cur[rbk[k]], 1u);
+        const uint32_t pos = atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
         s_stage[pos].v = rv[k];
         s_stage[pos].lc = rlc[k];
         s_dest[pos] = s_gbase[rbk[k]] + pos;
       }
     }
     __syncthreads();
-    for (int i = tid; i < valid; i += PART_BLOCK) out[s_dest[i]] = s_stage[i];
+    for (int i = tid; i < (int)total; i += PART_BLOCK) out[s_dest[i]] = s_stage[i];
     __syncthreads();
   }
 }
@@ -1038,10 +1078,10 @@ PartPlan part_plan(const fh_call* c) {
   if (bA > PART_SUB) return p; /* scatter passes handle <= 64 buckets */
   p.Bpad = PART_SUB;
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
-  p.scatter_lds = (int64_t)(3 * (PART_SUB + 2)) * 4 + (int64_t)T * 4 +
-                  (int64_t)T * sizeof(PairT<V>);
-  p.scatter2_lds = (int64_t)(3 * (PART_SUB + 1)) * 4 + (int64_t)T * 4 +
-                   (int64_t)T * sizeof(PairT<V>);
+  const int64_t tile_lds = (int64_t)(PART_SUB * PART_NW + PART_NW + 1 + PART_SUB + 3) * 4 +
+                           (int64_t)T * 4 + (int64_t)T * sizeof(PairT<V>);
+  p.scatter_lds = tile_lds;
+  p.scatter2_lds = tile_lds;
   if (p.scatter_lds > LDS_MAX) return p;
   int64_t off = 0;
   auto carve = [&](int64_t b) {
