@@ -774,20 +774,52 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     V rv[RPT];
     uint32_t rlc[RPT];
     int rbk[RPT];
+    /* per-thread CONTIGUOUS rows -> 16-B vector loads (G13) */
+    const int base = tid * RPT;
+    if (base + RPT <= nt) {
+      constexpr int VW = 16 / (int)sizeof(V);
 #pragma unroll
-    for (int k = 0; k < RPT; ++k) {
-      const int idx = tid + k * PART_BLOCK;
-      rbk[k] = -1;
-      if (idx < nt) {
-        const int64_t i = tile + idx;
-        const int64_t code = code_of((int64_t)labels[i],
-                                     twolab ? (int64_t)labels2[i] : 0, twolab,
+      for (int k = 0; k < RPT; k += VW) {
+        const Vec<V, VW> vv =
+            *reinterpret_cast<const Vec<V, VW>*>(values + tile + base + k);
+#pragma unroll
+        for (int j = 0; j < VW; ++j) rv[k + j] = vv.v[j];
+      }
+      Vec<L, 2> lv[RPT / 2 > 0 ? RPT / 2 : 1];
+      if (sizeof(L) == 8) {
+#pragma unroll
+        for (int k = 0; k < RPT; k += 2)
+          lv[k / 2] = *reinterpret_cast<const Vec<L, 2>*>(labels + tile + base + k);
+      }
+#pragma unroll
+      for (int k = 0; k < RPT; ++k) {
+        const int64_t i = tile + base + k;
+        const int64_t l0 = sizeof(L) == 8 ? (int64_t)lv[k / 2].v[k & 1] : (int64_t)labels[i];
+        const int64_t code = code_of(l0, twolab ? (int64_t)labels2[i] : 0, twolab,
                                      g0, g1, ngroups);
+        rbk[k] = -1;
         if (code >= 0) {
-          rv[k] = values[i];
           rbk[k] = (int)(code >> shift);
           rlc[k] = (uint32_t)(code - ((int64_t)rbk[k] << shift));
           atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int k = 0; k < RPT; ++k) {
+        const int idx = base + k;
+        rbk[k] = -1;
+        if (idx < nt) {
+          const int64_t i = tile + idx;
+          const int64_t code = code_of((int64_t)labels[i],
+                                       twolab ? (int64_t)labels2[i] : 0, twolab,
+                                       g0, g1, ngroups);
+          if (code >= 0) {
+            rv[k] = values[i];
+            rbk[k] = (int)(code >> shift);
+            rlc[k] = (uint32_t)(code - ((int64_t)rbk[k] << shift));
+            atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
+          }
         }
       }
     }
@@ -869,9 +901,10 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     V rv[RPT];
     uint32_t rlc[RPT];
     int rbk[RPT];
+    const int base = tid * RPT;
 #pragma unroll
     for (int k = 0; k < RPT; ++k) {
-      const int idx = tid + k * PART_BLOCK;
+      const int idx = base + k;
       rbk[k] = -1;
       if (idx < nt) {
         const PairT<V> pr = in[tile + idx];
