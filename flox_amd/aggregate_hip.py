@@ -573,3 +573,166 @@ def var_partials(
     means = sums.to(torch.float64) / counts
     p2 = grouped_partials(SET_SSD, array, group_idx, ng, skipnan=skipnan, means=means, **kw)
     return p2["sum"], sums, counts
+
+
+# ---------------------------------------------------------------------------
+# remaining seam callables: var/std, arg-reductions, first/last, quantiles,
+# mode, any/all — so every reference reduction name resolves through
+# generic_aggregate(engine="hip") (reference aggregations.py:60-133)
+# ---------------------------------------------------------------------------
+
+
+def _var_like(group_idx, array, *, skipnan, std, size=None, fill_value=None, dtype=None, ddof=0):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    ssd, sums, counts = var_partials(group_idx, array, skipnan=skipnan, size=ng)
+    den = counts.to(torch.float64) - ddof
+    res = ssd / den
+    nan_t = torch.full_like(res, float("nan"))
+    res = torch.where((den < 0) | (counts == 0), nan_t, res)
+    if std:
+        res = torch.sqrt(res)
+    if array.dtype.is_floating_point and dtype is None:
+        res = res.to(array.dtype)
+    res = _fill(res, counts == 0, fill_value)
+    return _to_dtype(res, dtype)
+
+
+def var(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, ddof=0, **kw):
+    return _var_like(group_idx, array, skipnan=False, std=False, size=size, fill_value=fill_value, dtype=dtype, ddof=ddof)
+
+
+def nanvar(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, ddof=0, **kw):
+    return _var_like(group_idx, array, skipnan=True, std=False, size=size, fill_value=fill_value, dtype=dtype, ddof=ddof)
+
+
+def std(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, ddof=0, **kw):
+    return _var_like(group_idx, array, skipnan=False, std=True, size=size, fill_value=fill_value, dtype=dtype, ddof=ddof)
+
+
+def nanstd(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, ddof=0, **kw):
+    return _var_like(group_idx, array, skipnan=True, std=True, size=size, fill_value=fill_value, dtype=dtype, ddof=ddof)
+
+
+def _arg_like(group_idx, array, *, ismax, skipnan, size=None, fill_value=None, dtype=None):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    if skipnan:
+        p1 = grouped_partials(SET_MAX_COUNT if ismax else SET_MIN_COUNT, array, group_idx, ng, skipnan=True)
+    else:
+        p1 = grouped_partials(SET_MAX_FULL if ismax else SET_MIN_FULL, array, group_idx, ng, skipnan=False)
+    target = p1["max" if ismax else "min"]
+    if "nanflag" in p1 and array.dtype.is_floating_point:
+        target = torch.where(p1["nanflag"] != 0, torch.full_like(target, float("nan")), target)
+    p2 = grouped_partials(SET_IDXMIN, array, group_idx, ng, skipnan=skipnan, target=target)
+    idx = p2["idx"]
+    sentinel = (1 << 63) - 1
+    missing = (p2["present"] == 0) | (idx == sentinel)
+    res = torch.where(missing, torch.full_like(idx, -1 if fill_value is None else int(fill_value)), idx)
+    return _to_dtype(res, dtype if dtype is not None else np.intp)
+
+
+def argmax(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _arg_like(group_idx, array, ismax=True, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def argmin(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _arg_like(group_idx, array, ismax=False, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanargmax(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _arg_like(group_idx, array, ismax=True, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanargmin(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _arg_like(group_idx, array, ismax=False, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def _pos_like(group_idx, array, *, last, skipnan, size=None, fill_value=None, dtype=None):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_IDXMAX if last else SET_IDXMIN, array, group_idx, ng, skipnan=skipnan)
+    idx = p["idx"]
+    sentinel = -1 if last else (1 << 63) - 1
+    valid = idx != sentinel
+    safe = torch.clamp(idx, 0, builtins.max(array.numel() - 1, 0))
+    res = array[safe]
+    if fill_value is not None:
+        res = _fill(res, ~valid, fill_value)
+    elif array.dtype.is_floating_point:
+        res = torch.where(~valid, torch.full_like(res, float("nan")), res)
+    return _to_dtype(res, dtype)
+
+
+def first(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _pos_like(group_idx, array, last=False, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def last(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _pos_like(group_idx, array, last=True, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanfirst(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _pos_like(group_idx, array, last=False, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanlast(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _pos_like(group_idx, array, last=True, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def _quantile_like(group_idx, array, *, q, skipnan, size=None, fill_value=None, dtype=None):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    q_arr = np.atleast_1d(np.asarray(q, dtype=np.float64))
+    res = grouped_quantile(array, group_idx, ng, q_arr, skipnan=skipnan)
+    if np.isscalar(q) or np.ndim(q) == 0:
+        res = res[0]
+    if array.dtype.is_floating_point and dtype is None:
+        res = res.to(array.dtype)
+    return _to_dtype(res, dtype)
+
+
+def quantile(group_idx, array, *, q, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _quantile_like(group_idx, array, q=q, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanquantile(group_idx, array, *, q, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _quantile_like(group_idx, array, q=q, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def median(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _quantile_like(group_idx, array, q=0.5, skipnan=False, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def nanmedian(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _quantile_like(group_idx, array, q=0.5, skipnan=True, size=size, fill_value=fill_value, dtype=dtype)
+
+
+def mode(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    return _to_dtype(grouped_mode(array, group_idx, ng, skipnan=False), dtype)
+
+
+def nanmode(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    return _to_dtype(grouped_mode(array, group_idx, ng, skipnan=True), dtype)
+
+
+def any_(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    group_idx, array = _prep(group_idx, array)
+    if not array.dtype.is_floating_point:
+        array = array.to(torch.int64)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_MAX_COUNT, array.to(torch.int64) if array.dtype.is_floating_point else array, group_idx, ng)
+    return (p["max"] != 0) & (p["count"] > 0)
+
+
+def all_(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    group_idx, array = _prep(group_idx, array)
+    if not array.dtype.is_floating_point:
+        array = array.to(torch.int64)
+    ng = _size_of(group_idx, size)
+    p = grouped_partials(SET_MIN_COUNT, array.to(torch.int64) if array.dtype.is_floating_point else array, group_idx, ng)
+    return (p["min"] != 0) & (p["count"] > 0)

@@ -140,7 +140,8 @@ def generic_aggregate(
         )
     from . import aggregate_hip
 
-    method = getattr(aggregate_hip, func, None)
+    seam_name = {"any": "any_", "all": "all_"}.get(func, func)
+    method = getattr(aggregate_hip, seam_name, None)
     if method is None:
         raise NotImplementedError(f"engine='hip' does not implement {func!r} yet")
     return method(group_idx, array, axis=axis, size=size, fill_value=fill_value, dtype=dtype, **kwargs)

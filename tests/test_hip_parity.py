@@ -208,6 +208,35 @@ def test_engine_seam_callables():
     np.testing.assert_array_equal(got_c.cpu().numpy(), np.bincount(g, minlength=ng))
 
 
+ALL_SEAM_FUNCS = [
+    "count", "sum", "nansum", "prod", "nanprod", "mean", "nanmean", "var",
+    "nanvar", "std", "nanstd", "min", "nanmin", "max", "nanmax", "argmax",
+    "nanargmax", "argmin", "nanargmin", "first", "nanfirst", "last",
+    "nanlast", "median", "nanmedian", "mode", "nanmode", "any", "all",
+    "nanlen",
+]
+
+
+def test_engine_seam_covers_every_reference_reduction():
+    """generic_aggregate(engine='hip', func=...) resolves and runs for every
+    name in the reference registry (aggregations.py:881-913) + quantiles."""
+    from flox_amd import generic_aggregate
+
+    rng = np.random.default_rng(23)
+    n, ng = 5_000, 17
+    g = rng.integers(0, ng, n)
+    a = rng.standard_normal(n)
+    b = rng.random(n) < 0.5
+    for func in ALL_SEAM_FUNCS:
+        arr = b if func in ("any", "all") else a
+        out = generic_aggregate(g, arr, engine="hip", func=func, size=ng)
+        assert out.shape[-1] == ng, func
+    outq = generic_aggregate(g, a, engine="hip", func="quantile", size=ng, q=0.25)
+    assert outq.shape[-1] == ng
+    outq2 = generic_aggregate(g, a, engine="hip", func="nanquantile", size=ng, q=[0.2, 0.8])
+    assert outq2.shape == (2, ng)
+
+
 def test_multi_by_fused_ravel():
     """2-D groupby goes through the fused labels2 kernel path."""
     rng = np.random.default_rng(11)
