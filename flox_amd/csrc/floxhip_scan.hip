@@ -208,15 +208,20 @@ int64_t fh_scan_scratch_bytes(const fh_call* c) {
   auto al = [](int64_t b) { return ((b + 255) / 256) * 256; };
   const int64_t vsz = (c->vdtype == FH_F64 || c->vdtype == FH_I64) ? 8 : 4;
   const int64_t psz = (c->vdtype == FH_F64 || c->vdtype == FH_I64) ? 16 : 8;
-  size_t ts = 0, t2 = 0;
+  size_t ts = 0, t2 = 0, t3 = 0;
   rocprim::radix_sort_pairs(nullptr, ts, (const uint32_t*)nullptr, (uint32_t*)nullptr,
                             (const uint32_t*)nullptr, (uint32_t*)nullptr, (size_t)n,
                             0, 32, 0);
   rocprim::inclusive_scan_by_key(nullptr, t2, (const uint32_t*)nullptr,
                                  (const double*)nullptr, (double*)nullptr, (size_t)n,
                                  rocprim::plus<double>(), rocprim::equal_to<uint32_t>(), 0);
+  /* the fill ops scan 16-byte pairs, which need more rocprim temp */
+  rocprim::inclusive_scan_by_key(nullptr, t3, (const uint32_t*)nullptr,
+                                 (const FillPair<double>*)nullptr, (FillPair<double>*)nullptr,
+                                 (size_t)n, FillOp<double>(), rocprim::equal_to<uint32_t>(), 0);
   /* 4 x u32 arrays + reversed keys + 2 scan buffers (pair-sized upper bound) */
-  return 5 * al(n * 4) + 2 * al(n * psz) + al((int64_t)std::max(ts, t2)) + 2 * al(n * vsz);
+  return 5 * al(n * 4) + 2 * al(n * psz) +
+         al((int64_t)std::max(ts, std::max(t2, t3))) + 2 * al(n * vsz);
 }
 
 /* grouped scan; op: 0 cumsum, 1 nancumsum, 2 ffill, 3 bfill.
