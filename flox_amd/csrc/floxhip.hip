@@ -615,28 +615,50 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
       s_perm[i] = perm[t0 + i];
     }
     __syncthreads();
-    for (int i = 0; i < nt; ++i) {
+    const bool col_act = c0 < m;
+    int i = 0;
+    while (i < nt) {
       const int g = s_code[i];
       if (g != cur_g) {
         flush(cur_g);
         reset();
         cur_g = g;
-        if ((OPS & B_SSD) && g >= 0) {
+        if ((OPS & B_SSD) && g >= 0 && col_act) {
 #pragma unroll
           for (int k = 0; k < VC; ++k)
             if (c0 + k < m) mean_g[k] = means[(int64_t)g * m + c0 + k];
         }
       }
-      if (g < 0 || c0 >= m) continue;
-      const int64_t row = (int64_t)s_perm[i] * ldm;
+      if (g < 0 || !col_act) {
+        ++i;
+        continue;
+      }
+      /* two rows of the same segment per iteration: halves the LDS reads,
+       * compares and loop control per element (segment boundaries are rare
+       * relative to rows) */
       if (VC > 1 && full) {
-        Vec<V, VC> vv = *reinterpret_cast<const Vec<V, VC>*>(values + row + c0);
+        if (i + 1 < nt && s_code[i + 1] == g) {
+          const int64_t rowa = (int64_t)s_perm[i] * ldm;
+          const int64_t rowb = (int64_t)s_perm[i + 1] * ldm;
+          Vec<V, VC> va = *reinterpret_cast<const Vec<V, VC>*>(values + rowa + c0);
+          Vec<V, VC> vb = *reinterpret_cast<const Vec<V, VC>*>(values + rowb + c0);
+#pragma unroll
+          for (int k = 0; k < VC; ++k) consume(va.v[k], k, true);
+#pragma unroll
+          for (int k = 0; k < VC; ++k) consume(vb.v[k], k, true);
+          i += 2;
+          continue;
+        }
+        Vec<V, VC> vv = *reinterpret_cast<const Vec<V, VC>*>(values + (int64_t)s_perm[i] * ldm + c0);
 #pragma unroll
         for (int k = 0; k < VC; ++k) consume(vv.v[k], k, true);
+        ++i;
       } else {
+        const int64_t row = (int64_t)s_perm[i] * ldm;
 #pragma unroll
         for (int k = 0; k < VC; ++k)
           consume((c0 + k < m) ? values[row + c0 + k] : (V)0, k, c0 + k < m);
+        ++i;
       }
     }
   }
