@@ -107,7 +107,7 @@ def main():
         log(f"config 2: {n:.0e} rows, {ng} groups, {func} (rank {rank}/{world})")
         vals = torch.rand(n, generator=gen, dtype=torch.float32, device=device)
         labels = torch.randint(0, ng, (n,), generator=gen, dtype=torch.int64, device=device)
-        expected = np.arange(ng)
+        expected = range(ng)
         bytes_per_step_per_gpu = n * (4 + 8)
         workload = "configs[1]: fp32 (1e9,) values, int64 labels, 1e4 uniform groups, func=mean, engine=hip"
         cfg_extra = {"rows": n, "ngroups": ng, "func": func, "labels": "int64"}

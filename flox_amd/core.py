@@ -502,7 +502,11 @@ def groupby_reduce(
             )
     else:
         fv = fill_value if fill_value is not None else xrdtypes.fill_default(func, out_dtype)
-        if isinstance(fv, float) and math.isnan(fv) and not is_float_out:
+        nan_fv = isinstance(fv, float) and math.isnan(fv)
+        if nan_fv and func in ("mean", "nanmean", "var", "nanvar", "std", "nanstd"):
+            # 0/0 division already produced NaN exactly where the fill goes
+            pass
+        elif nan_fv and not is_float_out:
             if bool(empty_mask.any().item()):
                 result = result.to(torch.float64)
                 t_out_dtype = torch.float64
