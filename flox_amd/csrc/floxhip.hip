@@ -665,6 +665,11 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
   flush(cur_g);
 }
 
+__global__ void k_init_cursors(uint32_t* cur, int n, uint32_t cap) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) cur[i] = (uint32_t)i * cap;
+}
+
 /* init product bins to 1 (memset cannot) */
 __global__ void k_fill_f64(double* p, int64_t n, double v) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
@@ -772,7 +777,8 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     const V* __restrict__ values, const L* __restrict__ labels,
     const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
     int64_t g1, int shift, int B, int Bpad /* unused, <= 64 buckets */,
-    uint32_t* __restrict__ cursors, PairT<V>* __restrict__ pairs) {
+    uint32_t* __restrict__ cursors, uint32_t cap /* 0 = exact bases */,
+    uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs) {
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
   constexpr int RPT = T / PART_BLOCK;
   constexpr int NB = 64;
@@ -870,7 +876,16 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
       const uint32_t excl = s_slot[b * PART_NW];
       const uint32_t nxt = (b + 1 < NB) ? s_slot[(b + 1) * PART_NW] : total;
       const uint32_t cnt = nxt - excl;
-      if (cnt) s_gbase[b] = atomicAdd(&cursors[b], cnt) - excl;
+      if (cnt) {
+        uint32_t gb = atomicAdd(&cursors[b], cnt);
+        if (cap && gb + cnt > (uint32_t)(b + 1) * cap) {
+          /* optimistic region overflow: flag it and keep writes in-bounds
+           * (results are discarded and recomputed by the exact path) */
+          *overflow = 1u;
+          gb = (uint32_t)b * cap;
+        }
+        s_gbase[b] = gb - excl;
+      }
     }
     __syncthreads();
 #pragma unroll
@@ -896,7 +911,9 @@ constexpr int PART_SUB = 64;
 template <typename V>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     const PairT<V>* __restrict__ in, const uint32_t* __restrict__ baseA,
+    uint32_t capA /* 0: baseA[sb]..baseA[sb+1]; else sb*capA..baseA[sb] */,
     int shift /* fine-bucket shift */, uint32_t* __restrict__ cursors,
+    uint32_t cap2 /* 0 = exact fine bases */, uint32_t* __restrict__ overflow,
     PairT<V>* __restrict__ out) {
   constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
   constexpr int RPT = T / PART_BLOCK;
@@ -912,7 +929,8 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
   const int sb = blockIdx.y;
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
-  const int64_t r0 = baseA[sb], r1 = baseA[sb + 1];
+  const int64_t r0 = capA ? (int64_t)(uint32_t)sb * capA : (int64_t)baseA[sb];
+  const int64_t r1 = capA ? (int64_t)baseA[sb] : (int64_t)baseA[sb + 1];
   const uint32_t lmask = (1u << shift) - 1u;
 
   for (int64_t tile = r0 + (int64_t)blockIdx.x * T; tile < r1;
@@ -960,7 +978,15 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
       const uint32_t excl = s_slot[b * PART_NW];
       const uint32_t nxt = (b + 1 < NB) ? s_slot[(b + 1) * PART_NW] : total;
       const uint32_t cnt = nxt - excl;
-      if (cnt) s_gbase[b] = atomicAdd(&cursors[sb * PART_SUB + b], cnt) - excl;
+      if (cnt) {
+        const uint32_t fb = (uint32_t)sb * PART_SUB + (uint32_t)b;
+        uint32_t gb = atomicAdd(&cursors[fb], cnt);
+        if (cap2 && gb + cnt > (fb + 1u) * cap2) {
+          *overflow = 1u;
+          gb = fb * cap2;
+        }
+        s_gbase[b] = gb - excl;
+      }
     }
     __syncthreads();
 #pragma unroll
@@ -983,6 +1009,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
 template <typename V, int OPS>
 __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
     const PairT<V>* __restrict__ pairs, const uint32_t* __restrict__ base,
+    uint32_t cap2 /* 0: base[b]..base[b+1]; else b*cap2..min(base[b],(b+1)*cap2) */,
     int64_t chunk, int gpb, int shift, int64_t ngroups,
     const double* __restrict__ means, int skipnan, BinLayout lay,
     void* out_sum, int64_t* out_count, uint32_t* out_present, void* out_min,
@@ -996,7 +1023,16 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
 
   const int b = blockIdx.y;
-  const int64_t bkt_begin = base[b], bkt_end = base[b + 1];
+  int64_t bkt_begin, bkt_end;
+  if (cap2) {
+    bkt_begin = (int64_t)(uint32_t)b * cap2;
+    bkt_end = base[b];
+    const int64_t lim = bkt_begin + cap2;
+    if (bkt_end > lim) bkt_end = lim;
+  } else {
+    bkt_begin = base[b];
+    bkt_end = base[b + 1];
+  }
   const int64_t start = bkt_begin + (int64_t)blockIdx.x * chunk;
   if (start >= bkt_end) return;
   const int64_t end = (start + chunk < bkt_end) ? start + chunk : bkt_end;
@@ -1104,6 +1140,9 @@ struct PartPlan {
   int B1;              /* super-bucket count (two_level) */
   BinLayout lay;       /* per-bucket bins (gpb entries) */
   int64_t pairs_off, pairs2_off, counts_off, base_off, baseA_off, cursors_off, bytes;
+  int64_t overflow_off;
+  uint32_t cap1, cap2; /* optimistic region strides (super / fine) */
+  bool optimistic;     /* capacity regions fit u32 cursors */
   int64_t scatter_lds, scatter2_lds;
   bool feasible;
 };
@@ -1138,28 +1177,43 @@ PartPlan part_plan(const fh_call* c) {
   p.scatter_lds = tile_lds;
   p.scatter2_lds = tile_lds;
   if (p.scatter_lds > LDS_MAX) return p;
+  /* optimistic capacity regions: uniform 25% + 8K rows of slack per bucket;
+   * overflow (extreme skew) falls back to the exact counted path */
+  auto cap_of = [&](int nb) {
+    const int64_t mean = (c->n + nb - 1) / nb;
+    return (uint32_t)(((mean + (mean >> 2) + 8192) + 63) / 64 * 64);
+  };
+  p.cap2 = cap_of(p.B);
+  p.cap1 = p.two_level ? cap_of(p.B1) : 0;
+  const int64_t reg2 = (int64_t)p.B * p.cap2;
+  const int64_t reg1 = p.two_level ? (int64_t)p.B1 * p.cap1 : reg2;
+  p.optimistic = reg1 < ((int64_t)1 << 31) && reg2 < ((int64_t)1 << 31);
   int64_t off = 0;
   auto carve = [&](int64_t b) {
     int64_t o = off;
     off += ((b + 255) / 256) * 256;
     return o;
   };
-  p.pairs_off = carve(c->n * (int64_t)sizeof(PairT<V>));
-  p.pairs2_off = p.two_level ? carve(c->n * (int64_t)sizeof(PairT<V>)) : p.pairs_off;
+  const int64_t p1elems = p.optimistic ? std::max(c->n, reg1) : c->n;
+  const int64_t p2elems = p.optimistic ? std::max(c->n, reg2) : c->n;
+  p.pairs_off = carve(p1elems * (int64_t)sizeof(PairT<V>));
+  p.pairs2_off = p.two_level ? carve(p2elems * (int64_t)sizeof(PairT<V>)) : p.pairs_off;
   p.counts_off = carve((int64_t)p.B * 4);
   p.base_off = carve(((int64_t)p.B + 1) * 4);
   p.baseA_off = carve(((int64_t)p.B1 + 1) * 4);
   p.cursors_off = carve((int64_t)p.B * 4);
+  p.overflow_off = carve(256);
   p.bytes = off;
   p.feasible = true;
   return p;
 }
 
 template <typename V, typename L, int OPS>
-int launch_partition(fh_call* c, const PartPlan& pp) {
+int launch_partition_exact(fh_call* c, const PartPlan& pp) {
   hipStream_t stream = (hipStream_t)c->stream;
   const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
   char* scr = (char*)c->scratch;
+  uint32_t* overflow = (uint32_t*)(scr + pp.overflow_off);
   uint32_t* counts = (uint32_t*)(scr + pp.counts_off);
   uint32_t* based = (uint32_t*)(scr + pp.base_off);
   uint32_t* baseAd = (uint32_t*)(scr + pp.baseA_off);
@@ -1227,7 +1281,7 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
                        stream, (const V*)c->values, (const L*)c->labels,
                        (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       shA, bA, pp.Bpad, cursors, pairs);
+                       shA, bA, pp.Bpad, cursors, 0u, overflow, pairs);
     FH_CHECK(hipGetLastError());
   }
   if (pp.two_level) {
@@ -1243,8 +1297,8 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     if (tiles_x > 64) tiles_x = 64;
     if (tiles_x < 1) tiles_x = 1;
     hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK),
-                       pp.scatter2_lds, stream, pairs, baseAd, pp.shift,
-                       cursors, pairs2);
+                       pp.scatter2_lds, stream, pairs, baseAd, 0u, pp.shift,
+                       cursors, 0u, overflow, pairs2);
     FH_CHECK(hipGetLastError());
   }
   {
@@ -1261,7 +1315,7 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     if (maxchunks < 1) maxchunks = 1;
     hipLaunchKernelGGL(kern, dim3(maxchunks, pp.B), dim3(BLOCK_LDS),
                        pp.lay.bytes, stream, pp.two_level ? pairs2 : pairs,
-                       based, chunk, pp.gpb, pp.shift, c->ngroups, c->means,
+                       based, 0u, chunk, pp.gpb, pp.shift, c->ngroups, c->means,
                        skipnan, pp.lay, c->out_sum, c->out_count,
                        c->out_present, c->out_min, c->out_max, c->out_nanflag);
     FH_CHECK(hipGetLastError());
@@ -1273,6 +1327,95 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
                        c->out_present);
     FH_CHECK(hipGetLastError());
   }
+  c->path_used = 4;
+  return 0;
+}
+
+/* optimistic partition: capacity regions replace the counting pre-pass and
+ * every host round trip except the final 4-byte overflow check */
+template <typename V, typename L, int OPS>
+int launch_partition(fh_call* c, const PartPlan& pp) {
+  if (!pp.optimistic) return launch_partition_exact<V, L, OPS>(c, pp);
+  hipStream_t stream = (hipStream_t)c->stream;
+  const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
+  char* scr = (char*)c->scratch;
+  uint32_t* cursorsA = (uint32_t*)(scr + pp.counts_off); /* reused slot */
+  uint32_t* cursors = (uint32_t*)(scr + pp.cursors_off);
+  uint32_t* overflow = (uint32_t*)(scr + pp.overflow_off);
+  PairT<V>* pairs = (PairT<V>*)(scr + pp.pairs_off);
+  PairT<V>* pairs2 = (PairT<V>*)(scr + pp.pairs2_off);
+
+  FH_CHECK(hipMemsetAsync(overflow, 0, 4, stream));
+  const int bA = pp.two_level ? pp.B1 : pp.B;
+  const uint32_t capA = pp.two_level ? pp.cap1 : pp.cap2;
+  hipLaunchKernelGGL(k_init_cursors, dim3(1), dim3(PART_SUB), 0, stream,
+                     cursorsA, bA, capA);
+  FH_CHECK(hipGetLastError());
+  {
+    auto kern = k_part_scatter<V, L>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)pp.scatter_lds));
+    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    int64_t wb = (c->n + T - 1) / T;
+    int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
+    const int shA = pp.two_level ? pp.shift + 6 : pp.shift;
+    hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
+                       stream, (const V*)c->values, (const L*)c->labels,
+                       (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
+                       shA, bA, pp.Bpad, cursorsA, capA, overflow, pairs);
+    FH_CHECK(hipGetLastError());
+  }
+  if (pp.two_level) {
+    hipLaunchKernelGGL(k_init_cursors, dim3((pp.B + 255) / 256), dim3(256), 0,
+                       stream, cursors, pp.B, pp.cap2);
+    FH_CHECK(hipGetLastError());
+    auto kern2 = k_part_scatter2<V>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern2,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)pp.scatter2_lds));
+    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    int tiles_x = (int)(((int64_t)pp.cap1 + T - 1) / T);
+    if (tiles_x > 64) tiles_x = 64;
+    if (tiles_x < 1) tiles_x = 1;
+    hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK),
+                       pp.scatter2_lds, stream, pairs, cursorsA, pp.cap1,
+                       pp.shift, cursors, pp.cap2, overflow, pairs2);
+    FH_CHECK(hipGetLastError());
+  }
+  {
+    int rc = init_outs<V, OPS>(c, c->ngroups, stream);
+    if (rc) return rc;
+  }
+  {
+    auto kern = k_reduce_bucket<V, OPS>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)pp.lay.bytes));
+    const int64_t chunk = 1 << 19;
+    const uint32_t capF = pp.cap2;
+    int maxchunks = (int)(((int64_t)capF + chunk - 1) / chunk);
+    if (maxchunks < 1) maxchunks = 1;
+    hipLaunchKernelGGL(kern, dim3(maxchunks, pp.B), dim3(BLOCK_LDS),
+                       pp.lay.bytes, stream,
+                       pp.two_level ? pairs2 : pairs,
+                       pp.two_level ? cursors : cursorsA, capF, chunk, pp.gpb,
+                       pp.shift, c->ngroups, c->means, skipnan, pp.lay,
+                       c->out_sum, c->out_count, c->out_present, c->out_min,
+                       c->out_max, c->out_nanflag);
+    FH_CHECK(hipGetLastError());
+  }
+  if (OPS & (B_MIN | B_MAX)) {
+    int db = (int)((c->ngroups + 255) / 256);
+    hipLaunchKernelGGL((k_decode<V, OPS>), dim3(db), dim3(256), 0, stream,
+                       c->ngroups, c->out_min, c->out_max, c->out_count,
+                       c->out_present);
+    FH_CHECK(hipGetLastError());
+  }
+  uint32_t h_ov = 0;
+  FH_CHECK(hipMemcpyAsync(&h_ov, overflow, 4, hipMemcpyDeviceToHost, stream));
+  FH_CHECK(hipStreamSynchronize(stream));
+  if (h_ov) return launch_partition_exact<V, L, OPS>(c, pp);
   c->path_used = 4;
   return 0;
 }
