@@ -179,7 +179,9 @@ def groupby_reduce(
     fill_value=None,
     dtype=None,
     min_count: int | None = None,
+    method: str | None = None,
     engine: str = "hip",
+    reindex=None,
     finalize_kwargs: dict[str, Any] | None = None,
     distributed_combine: bool | None = None,
     shard_row_offset: int = 0,

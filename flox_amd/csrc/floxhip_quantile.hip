@@ -283,7 +283,7 @@ int run_quantile(fh_call* c, const double* q_dev, int nq, double* out) {
     uint64_t* kout = (uint64_t*)(scr + carve(n * 8));
     int64_t* off = (int64_t*)(scr + carve((ngroups + 1) * 8));
     size_t temp_bytes = 0;
-    rocprim::radix_sort_keys(nullptr, temp_bytes, kin, kout, (size_t)n, 0, 64, stream);
+    (void)rocprim::radix_sort_keys(nullptr, temp_bytes, kin, kout, (size_t)n, 0, 64, stream);
     void* temp = scr + carve((int64_t)temp_bytes);
     if (off0 > c->scratch_bytes) return 3;
     hipLaunchKernelGGL((k_qpack32<V, L>), dim3(grid), dim3(256), 0, stream,
@@ -315,8 +315,8 @@ int run_quantile(fh_call* c, const double* q_dev, int nq, double* out) {
   uint32_t* c_out = (uint32_t*)(scr + carve(n * 4));
   int64_t* off = (int64_t*)(scr + carve((ngroups + 1) * 8));
   size_t t1 = 0, t2 = 0;
-  rocprim::radix_sort_pairs(nullptr, t1, e_in, e_out, c_in, c_out, (size_t)n, 0, 64, stream);
-  rocprim::radix_sort_pairs(nullptr, t2, c_out, c_in, e_out, e_in, (size_t)n, 0, 32, stream);
+  (void)rocprim::radix_sort_pairs(nullptr, t1, e_in, e_out, c_in, c_out, (size_t)n, 0, 64, stream);
+  (void)rocprim::radix_sort_pairs(nullptr, t2, c_out, c_in, e_out, e_in, (size_t)n, 0, 32, stream);
   void* temp = scr + carve((int64_t)std::max(t1, t2));
   if (off0 > c->scratch_bytes) return 3;
   hipLaunchKernelGGL((k_qpack64<V, L>), dim3(grid), dim3(256), 0, stream,
@@ -355,15 +355,15 @@ int64_t fh_quantile_scratch_bytes(const fh_call* c) {
   int64_t bytes = 0;
   auto al = [](int64_t b) { return ((b + 255) / 256) * 256; };
   if (c->vdtype == FH_F32 || c->vdtype == FH_I32) {
-    rocprim::radix_sort_keys(nullptr, temp, (const uint64_t*)nullptr,
+    (void)rocprim::radix_sort_keys(nullptr, temp, (const uint64_t*)nullptr,
                              (uint64_t*)nullptr, (size_t)n, 0, 64, 0);
     bytes = 2 * al(n * 8) + al((ngroups + 1) * 8) + al((int64_t)temp);
   } else {
     size_t t1 = 0, t2 = 0;
-    rocprim::radix_sort_pairs(nullptr, t1, (const uint64_t*)nullptr,
+    (void)rocprim::radix_sort_pairs(nullptr, t1, (const uint64_t*)nullptr,
                               (uint64_t*)nullptr, (const uint32_t*)nullptr,
                               (uint32_t*)nullptr, (size_t)n, 0, 64, 0);
-    rocprim::radix_sort_pairs(nullptr, t2, (const uint32_t*)nullptr,
+    (void)rocprim::radix_sort_pairs(nullptr, t2, (const uint32_t*)nullptr,
                               (uint32_t*)nullptr, (const uint64_t*)nullptr,
                               (uint64_t*)nullptr, (size_t)n, 0, 32, 0);
     bytes = 2 * al(n * 8) + 2 * al(n * 4) + al((ngroups + 1) * 8) +

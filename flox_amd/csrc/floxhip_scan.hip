@@ -149,12 +149,12 @@ int run_scan(fh_call* c, V* out) {
   if (OP == SCAN_BFILL) rkeys = (uint32_t*)(scr + carve(n * 4));
 
   size_t ts = 0, tscan = 0;
-  rocprim::radix_sort_pairs(nullptr, ts, codes, codes_s, idx, perm, (size_t)n, 0, 32, stream);
+  (void)rocprim::radix_sort_pairs(nullptr, ts, codes, codes_s, idx, perm, (size_t)n, 0, 32, stream);
   if (IS_FILL)
-    rocprim::inclusive_scan_by_key(nullptr, tscan, codes_s, sp, sp2, (size_t)n,
+    (void)rocprim::inclusive_scan_by_key(nullptr, tscan, codes_s, sp, sp2, (size_t)n,
                                    FillOp<V>(), rocprim::equal_to<uint32_t>(), stream);
   else
-    rocprim::inclusive_scan_by_key(nullptr, tscan, codes_s, sv, sv2, (size_t)n,
+    (void)rocprim::inclusive_scan_by_key(nullptr, tscan, codes_s, sv, sv2, (size_t)n,
                                    rocprim::plus<V>(), rocprim::equal_to<uint32_t>(), stream);
   void* temp = scr + carve((int64_t)std::max(ts, tscan));
   if (o > c->scratch_bytes) return 3;
@@ -209,14 +209,14 @@ int64_t fh_scan_scratch_bytes(const fh_call* c) {
   const int64_t vsz = (c->vdtype == FH_F64 || c->vdtype == FH_I64) ? 8 : 4;
   const int64_t psz = (c->vdtype == FH_F64 || c->vdtype == FH_I64) ? 16 : 8;
   size_t ts = 0, t2 = 0, t3 = 0;
-  rocprim::radix_sort_pairs(nullptr, ts, (const uint32_t*)nullptr, (uint32_t*)nullptr,
+  (void)rocprim::radix_sort_pairs(nullptr, ts, (const uint32_t*)nullptr, (uint32_t*)nullptr,
                             (const uint32_t*)nullptr, (uint32_t*)nullptr, (size_t)n,
                             0, 32, 0);
-  rocprim::inclusive_scan_by_key(nullptr, t2, (const uint32_t*)nullptr,
+  (void)rocprim::inclusive_scan_by_key(nullptr, t2, (const uint32_t*)nullptr,
                                  (const double*)nullptr, (double*)nullptr, (size_t)n,
                                  rocprim::plus<double>(), rocprim::equal_to<uint32_t>(), 0);
   /* the fill ops scan 16-byte pairs, which need more rocprim temp */
-  rocprim::inclusive_scan_by_key(nullptr, t3, (const uint32_t*)nullptr,
+  (void)rocprim::inclusive_scan_by_key(nullptr, t3, (const uint32_t*)nullptr,
                                  (const FillPair<double>*)nullptr, (FillPair<double>*)nullptr,
                                  (size_t)n, FillOp<double>(), rocprim::equal_to<uint32_t>(), 0);
   /* 4 x u32 arrays + reversed keys + 2 scan buffers (pair-sized upper bound) */
