@@ -30,6 +30,7 @@ SET_SSD = 7
 SET_PROD = 8
 SET_IDXMIN = 9
 SET_IDXMAX = 10
+SET_WELFORD = 11
 FLAG_SKIPNAN = 1
 FLAG_FORCE_LDS = 2
 FLAG_FORCE_ATOMIC = 4

@@ -40,6 +40,7 @@ from ._ffi import (
     SET_SSD,
     SET_SUM_COUNT,
     SET_SUM_COUNT_PRESENT,
+    SET_WELFORD,
     FhCall,
 )
 
@@ -64,6 +65,7 @@ _SET_MEMBERS = {
     SET_PROD: ("sum", "count", "present"),
     SET_IDXMIN: ("idx", "count", "present"),
     SET_IDXMAX: ("idx", "count", "present"),
+    SET_WELFORD: ("wssd", "wsum", "count"),  # cols path only
 }
 
 IDX_SENTINEL_MIN = (1 << 63) - 1  # untouched IDXMIN bin
@@ -413,6 +415,11 @@ def grouped_partials_cols(
     members = _SET_MEMBERS[op_set]
     out: dict[str, torch.Tensor] = {}
     shape = (ngroups, m)
+    if "wssd" in members:
+        out["wssd"] = torch.empty(shape, dtype=torch.float64, device=dev)
+        c.out_sum = out["wssd"].data_ptr()
+        out["wsum"] = torch.empty(shape, dtype=torch.float64, device=dev)
+        c.out_min = out["wsum"].data_ptr()  # sum-of-x rides the out_min slot
     if "sum" in members:
         dt = torch.float64 if op_set == SET_SSD else _acc_dtype(values2d.dtype)
         out["sum"] = torch.empty(shape, dtype=dt, device=dev)

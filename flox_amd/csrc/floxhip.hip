@@ -62,6 +62,7 @@ enum {
   B_PROD = 128,
   B_IDXMIN = 256,
   B_IDXMAX = 512,
+  B_WELFORD = 1024,
 };
 
 __host__ __device__ constexpr int set_bits(int op_set) {
@@ -77,6 +78,7 @@ __host__ __device__ constexpr int set_bits(int op_set) {
     case FH_SET_PROD: return B_PROD | B_CNT | B_PRESENT;
     case FH_SET_IDXMIN: return B_IDXMIN | B_CNT | B_PRESENT;
     case FH_SET_IDXMAX: return B_IDXMAX | B_CNT | B_PRESENT;
+    case FH_SET_WELFORD: return B_WELFORD | B_CNT;
     default: return 0;
   }
 }
@@ -186,7 +188,7 @@ template <typename T, int N> struct alignas(sizeof(T) * N) Vec { T v[N]; };
 
 /* ---- bin layout (shared between LDS carve, slab and combine) ------------ */
 struct BinLayout {
-  int64_t sum_off, cnt_off, present_off, minmax_off, nanflag_off;
+  int64_t sum_off, cnt_off, present_off, minmax_off, nanflag_off, sumx_off;
   int64_t bytes;  /* per block-copy of the bins */
 };
 
@@ -199,11 +201,12 @@ __host__ __device__ BinLayout bin_layout(int bits, int64_t ngroups, int64_t cnt_
     off += ((ngroups * elem + 255) / 256) * 256; /* 256-B aligned sections */
     return o;
   };
-  L.sum_off = (bits & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ? carve(8) : -1;
+  L.sum_off = (bits & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX | B_WELFORD)) ? carve(8) : -1;
   L.cnt_off = (bits & B_CNT) ? carve(cnt_elem_size) : -1;
   L.present_off = (bits & B_PRESENT) ? carve(4) : -1;
   L.minmax_off = (bits & (B_MIN | B_MAX)) ? carve(sizeof(typename Traits<V>::Enc)) : -1;
   L.nanflag_off = (bits & B_NANFLAG) ? carve(4) : -1;
+  L.sumx_off = (bits & B_WELFORD) ? carve(8) : -1;
   L.bytes = off;
   return L;
 }
@@ -219,7 +222,7 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_lds(
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
   using SumT = typename std::conditional<
-      (OPS & B_SSD) != 0, double,
+      (OPS & (B_SSD | B_WELFORD)) != 0, double,
       typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr int VEC = TR::VEC;
@@ -339,7 +342,7 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
   using SumT = typename std::conditional<
-      (OPS & B_SSD) != 0, double,
+      (OPS & (B_SSD | B_WELFORD)) != 0, double,
       typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
@@ -357,10 +360,21 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
   if (OPS & B_IDXMAX) s = (SumT)(-1);
   int64_t c = 0;
   uint32_t p = 0, nf = 0;
+  double w_sum = 0.0;
   Enc mn = (Enc)~(Enc)0, mx = (Enc)0;
   for (int b = b0; b < b1; ++b) {
     const char* blk = slab + (int64_t)b * lay.bytes;
     if (OPS & (B_SUM | B_SSD)) s += ((const SumT*)(blk + lay.sum_off))[g];
+    if (OPS & B_WELFORD) {
+      /* pairwise var merge: fold ssd_i + sum_i^2/n_i, subtract the global
+       * term after the loop (the reference's _var_combine adjustment,
+       * aggregations.py:392-451, in closed form) */
+      const double ssd_i = ((const double*)(blk + lay.sum_off))[g];
+      const double sum_i = ((const double*)(blk + lay.sumx_off))[g];
+      const double n_i = (double)((const uint32_t*)(blk + lay.cnt_off))[g];
+      s += (SumT)(ssd_i + (n_i > 0.0 ? (sum_i * sum_i) / n_i : 0.0));
+      w_sum += sum_i;
+    }
     if (IS_PROD) s *= ((const SumT*)(blk + lay.sum_off))[g];
     if (OPS & B_IDXMIN) {
       const SumT x = ((const SumT*)(blk + lay.sum_off))[g];
@@ -384,6 +398,13 @@ __global__ void k_combine(const char* __restrict__ slab, int nblocks,
   }
   if (S == 1) {
     const bool present = (OPS & B_PRESENT) ? (p != 0) : (c != 0);
+    if (OPS & B_WELFORD) {
+      const double total = (double)s - (c > 0 ? (w_sum * w_sum) / (double)c : 0.0);
+      ((double*)out_sum)[g] = c > 0 ? total : 0.0;
+      ((double*)out_min)[g] = w_sum;
+      out_count[g] = c;
+      return;
+    }
     if (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX)) ((SumT*)out_sum)[g] = s;
     if (OPS & B_CNT) out_count[g] = c;
     if (OPS & B_PRESENT) out_present[g] = p;
@@ -521,7 +542,7 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
   using SumT = typename std::conditional<
-      (OPS & B_SSD) != 0, double,
+      (OPS & (B_SSD | B_WELFORD)) != 0, double,
       typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
@@ -542,11 +563,15 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
 
   char* my_slab = SLAB ? slab + (int64_t)blockIdx.y * lay.bytes : nullptr;
 
+  constexpr bool IS_WEL = (OPS & B_WELFORD) != 0;
   SumT acc[VC];
   uint32_t cnt[VC];
   Enc mn[VC], mx[VC];
   uint32_t nanflag[VC];
   double mean_g[VC];
+  /* shifted-variance state: s1 = sum(x - x0), acc holds s2 = sum((x-x0)^2) */
+  double w_s1[VC], w_x0[VC];
+  uint32_t w_have[VC];
   int cur_g = -1;
 
   auto reset = [&]() {
@@ -557,6 +582,11 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
       mn[k] = (Enc)~(Enc)0;
       mx[k] = (Enc)0;
       nanflag[k] = 0;
+      if (IS_WEL) {
+        w_s1[k] = 0.0;
+        w_x0[k] = 0.0;
+        w_have[k] = 0;
+      }
     }
   };
   reset();
@@ -567,6 +597,23 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
     for (int k = 0; k < VC; ++k) {
       if (c0 + k >= m) break;
       const int64_t o = (int64_t)g * m + c0 + k;
+      if (IS_WEL) {
+        /* convert shifted sums to the var_chunk triple: sum = s1 + n*x0,
+         * ssd = s2 - s1^2/n (the mean-shifted form) */
+        const double nn = (double)cnt[k];
+        const double sum = w_s1[k] + nn * w_x0[k];
+        const double ssd = cnt[k] ? (double)acc[k] - (w_s1[k] * w_s1[k]) / nn : 0.0;
+        if (SLAB) {
+          ((double*)(my_slab + lay.sum_off))[o] = ssd;
+          ((double*)(my_slab + lay.sumx_off))[o] = sum;
+          ((uint32_t*)(my_slab + lay.cnt_off))[o] = cnt[k];
+        } else {
+          ((double*)out_sum)[o] = ssd;
+          ((double*)out_min)[o] = sum; /* sum-of-x rides the out_min slot */
+          out_count[o] = (int64_t)cnt[k];
+        }
+        continue;
+      }
       if (SLAB) {
         if (OPS & (B_SUM | B_SSD | B_PROD)) ((SumT*)(my_slab + lay.sum_off))[o] = acc[k];
         if (OPS & B_CNT) ((uint32_t*)(my_slab + lay.cnt_off))[o] = cnt[k];
@@ -589,6 +636,17 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
     if (!lanes_ok) return;
     const bool vnan = TR::isnan_(v);
     if (vnan && skipnan) return;
+    if (IS_WEL) {
+      if (!w_have[k]) {
+        w_x0[k] = (double)v; /* non-skip: a NaN first value poisons the group */
+        w_have[k] = 1;
+      }
+      const double d = (double)v - w_x0[k];
+      w_s1[k] += d;
+      acc[k] += (SumT)(d * d);
+      cnt[k] += vnan ? 0u : 1u;
+      return;
+    }
     if (OPS & B_SUM) acc[k] += (SumT)v;
     if (IS_PROD) acc[k] *= (SumT)v;
     if (OPS & B_SSD) {
@@ -1017,7 +1075,7 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
   using TR = Traits<V>;
   using Acc = typename TR::Acc;
   using SumT = typename std::conditional<
-      (OPS & B_SSD) != 0, double,
+      (OPS & (B_SSD | B_WELFORD)) != 0, double,
       typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
   using Enc = typename TR::Enc;
   constexpr bool IS_PROD = (OPS & B_PROD) != 0;
@@ -1619,8 +1677,10 @@ int launch_cols(fh_call* c) {
     /* identity-fill every chunk's slab sections (bins no segment touches) */
     for (int ch = 0; ch < plan.nchunks; ++ch) {
       char* s = (char*)c->scratch + (int64_t)ch * plan.lay.bytes;
-      if (OPS & (B_SUM | B_SSD))
+      if (OPS & (B_SUM | B_SSD | B_WELFORD))
         FH_CHECK(hipMemsetAsync(s + plan.lay.sum_off, 0, nbins * 8, stream));
+      if (OPS & B_WELFORD)
+        FH_CHECK(hipMemsetAsync(s + plan.lay.sumx_off, 0, nbins * 8, stream));
       if (OPS & B_PROD) {
         int fb = (int)((nbins + 255) / 256);
         if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
@@ -1644,8 +1704,10 @@ int launch_cols(fh_call* c) {
         FH_CHECK(hipMemsetAsync(s + plan.lay.nanflag_off, 0, nbins * 4, stream));
     }
   } else {
-    if (OPS & (B_SUM | B_SSD))
+    if (OPS & (B_SUM | B_SSD | B_WELFORD))
       FH_CHECK(hipMemsetAsync(c->out_sum, 0, nbins * 8, stream));
+    if (OPS & B_WELFORD)
+      FH_CHECK(hipMemsetAsync(c->out_min, 0, nbins * 8, stream));
     if (OPS & B_PROD) {
       int fb = (int)((nbins + 255) / 256);
       if (c->vdtype == FH_F32 || c->vdtype == FH_F64)
@@ -1718,6 +1780,7 @@ int dispatch_cols_ops(fh_call* c) {
       return launch_cols<V, B_MAX | B_CNT | B_PRESENT | B_NANFLAG>(c);
     case B_MAX | B_CNT: return launch_cols<V, B_MAX | B_CNT>(c);
     case B_SSD: return launch_cols<V, B_SSD>(c);
+    case B_WELFORD | B_CNT: return launch_cols<V, B_WELFORD | B_CNT>(c);
     case B_PROD | B_CNT | B_PRESENT:
       return launch_cols<V, B_PROD | B_CNT | B_PRESENT>(c);
     default: return 4;

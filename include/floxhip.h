@@ -57,6 +57,13 @@ enum fh_opset {
    * untouched groups handled by init (IDXMIN: INT64_MAX, IDXMAX: -1). */
   FH_SET_IDXMIN = 9,
   FH_SET_IDXMAX = 10,
+  /* single-pass variance partials for the COLUMN path only: per (group,
+   * column) the kernel accumulates shifted sums about the segment's first
+   * value (the stability device numpy_groupies also uses,
+   * aggregate_npg.py:112-126) and emits the var_chunk triple of the
+   * reference (aggregations.py:348-389): ssd -> out_sum (f64),
+   * sum -> out_min (f64, pointer reused), count -> out_count. */
+  FH_SET_WELFORD = 11,
 };
 
 /* flags */
