@@ -423,6 +423,29 @@ def groupby_reduce(
         result = gathered
         empty_mask = ~valid
         counts_for_mask = p["count"]
+    elif func in ("var", "nanvar", "std", "nanstd") and lead_M > 1:
+        # single fused pass on the column path: the kernel accumulates
+        # mean-shifted sums per (group, column) segment and emits the
+        # var_chunk triple (ssd, sum, count); cross-rank combine closes the
+        # reference's _var_combine adjustment in one formula
+        skip = agg.skipnan
+        p = run_set(_ffi.SET_WELFORD, skip)
+        ssd, sums, counts = p["wssd"], p["wsum"], p["count"]
+        if dist_on:
+            t = torch.where(counts > 0, sums * sums / counts, torch.zeros_like(sums))
+            a = ssd + t
+            distributed.all_reduce_(a, "sum")
+            distributed.all_reduce_(sums, "sum")
+            distributed.all_reduce_(counts, "sum")
+            ssd = a - torch.where(counts > 0, sums * sums / counts, torch.zeros_like(sums))
+        den_w = counts.to(torch.float64) - ddof
+        result = ssd / den_w
+        nan_t = torch.full_like(result, float("nan"))
+        result = torch.where((den_w < 0) | (counts == 0), nan_t, result)
+        if func in ("std", "nanstd"):
+            result = torch.sqrt(result)
+        counts_for_mask = counts
+        empty_mask = counts == 0
     elif func in ("var", "nanvar", "std", "nanstd"):
         skip = agg.skipnan
         p1 = run_set(_ffi.SET_SUM_COUNT, skip)
