@@ -90,6 +90,27 @@ def test_bfill_is_reversed_ffill(data):
 
 
 @settings(**SETTINGS)
+@given(
+    seed=st.integers(0, 2**31 - 1),
+    n_t=st.integers(1, 400),
+    m1=st.integers(1, 12),
+    m2=st.integers(1, 40),
+    func=st.sampled_from(["sum", "mean", "nanmean", "var", "nanvar", "count", "min", "nanmax"]),
+)
+def test_lead_dims_match_oracle(seed, n_t, m1, m2, func):
+    """Column path (leading array dims) vs oracle on random shapes."""
+    rng = np.random.default_rng(seed)
+    arr = rng.standard_normal((m1, m2, n_t))
+    arr[rng.random(arr.shape) < 0.05] = np.nan
+    ng = int(rng.integers(1, 30))
+    labels = rng.integers(0, ng, n_t)
+    want, *_ = oracle_reduce(arr, labels, func=func, expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(arr, labels, func=func, expected_groups=np.arange(ng))
+    assert got.shape == want.shape and got.dtype == want.dtype
+    np.testing.assert_allclose(got, want, equal_nan=True, rtol=1e-9, atol=1e-9)
+
+
+@settings(**SETTINGS)
 @given(data=array_strat)
 def test_single_group_equals_numpy(data):
     """reference test_properties.py:93-176: one group -> plain numpy."""
