@@ -178,6 +178,32 @@ def test_any_all_bool():
         np.testing.assert_array_equal(got, want)
 
 
+def test_partition_overflow_falls_back_exact():
+    """Extreme label skew overflows the optimistic capacity regions; the
+    exact counted path must kick in and produce correct results."""
+    rng = np.random.default_rng(99)
+    n, ng = 2_000_000, 2_000_000
+    labels = np.zeros(n, dtype=np.int64)          # everything in group 0
+    labels[: n // 100] = rng.integers(0, ng, n // 100)  # a sprinkle elsewhere
+    vals = rng.standard_normal(n)
+    want, *_ = oracle_reduce(vals, labels, func="sum", expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func="sum", expected_groups=np.arange(ng))
+    np.testing.assert_allclose(got, want, equal_nan=True, rtol=1e-10, atol=1e-6)
+
+
+def test_isbin_with_lead_dims():
+    """Binned grouping through the column path (binned climatology)."""
+    rng = np.random.default_rng(31)
+    n_t, y = 500, 64
+    arr = rng.standard_normal((y, n_t))
+    by = rng.standard_normal(n_t) * 2
+    edges = np.array([-3.0, -1.0, 0.0, 1.0, 3.0])
+    want, *_ = oracle_reduce(arr, by, func="mean", expected_groups=edges, isbin=True)
+    got, *_ = flox_amd.groupby_reduce(arr, by, func="mean", expected_groups=edges, isbin=True)
+    assert got.shape == (y, 4)
+    np.testing.assert_allclose(got, want, equal_nan=True, rtol=1e-10, atol=1e-12)
+
+
 @pytest.mark.parametrize("func", ["sum", "nansum", "mean", "count", "min", "nanmax", "var"])
 def test_partition_path_many_groups(func):
     """2e6 groups: the bucket-partition (sort) path."""
