@@ -1644,6 +1644,11 @@ template <typename V>
 ColsPlan cols_plan(const fh_call* c) {
   ColsPlan p{};
   p.vc = Traits<V>::VEC;
+  /* 4-byte dtypes: 8 columns per thread (two 16-B vectors) amortizes the
+   * per-row LDS reads, segment compare and address arithmetic further */
+  if (sizeof(V) == 4 && c->ldm % 8 == 0 && c->m >= 8 &&
+      ((uintptr_t)c->values % 32) == 0)
+    p.vc = 8;
   if (c->ldm % p.vc != 0 || ((uintptr_t)c->values % 16) != 0) p.vc = 1;
   p.ncolblk = (c->m + (int64_t)COLS_BLOCK * p.vc - 1) / ((int64_t)COLS_BLOCK * p.vc);
   if (p.ncolblk == 0) p.ncolblk = 1;
@@ -1739,6 +1744,14 @@ int launch_cols(fh_call* c) {
     return (int)hipGetLastError();
   };
   int rc;
+  if constexpr (sizeof(V) == 4) {
+    if (plan.vc == 8) {
+      rc = slab_mode ? launch(k_reduce_cols<V, OPS, 8, true>)
+                     : launch(k_reduce_cols<V, OPS, 8, false>);
+      if (rc != 0) return rc + 1000;
+      goto cols_launched;
+    }
+  }
   if (slab_mode)
     rc = plan.vc > 1 ? launch(k_reduce_cols<V, OPS, Traits<V>::VEC, true>)
                      : launch(k_reduce_cols<V, OPS, 1, true>);
@@ -1746,6 +1759,7 @@ int launch_cols(fh_call* c) {
     rc = plan.vc > 1 ? launch(k_reduce_cols<V, OPS, Traits<V>::VEC, false>)
                      : launch(k_reduce_cols<V, OPS, 1, false>);
   if (rc != 0) return rc + 1000;
+cols_launched:;
 
   if (slab_mode) {
     /* fold the chunk partials; k_combine decodes min/max (gridDim.y==1) */
