@@ -95,8 +95,6 @@ def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
     """labels -> codes in [0, ngroups), invalid/NaN -> out-of-range
     (restates reference _factorize_single, factorize.py:42-99)."""
     if expect is None:
-        if not sort:
-            raise NotImplementedError("sort=False without expected_groups")
         fl = flat
         if fl.dtype.is_floating_point:
             fl = fl[~torch.isnan(fl)]
@@ -551,7 +549,21 @@ def groupby_reduce(
         result = result.reshape(ngroups, lead_M).t().contiguous()
     result = result.reshape(lead_shape + grp_shape)
 
-    groups = tuple(f.groups for f in facs)
+    groups_list = [f.groups for f in facs]
+    if not sort and not provided_expected:
+        # groups in first-appearance order (reference pd.factorize(sort=False),
+        # factorize.py:96): find each group's first row with an index-min pass
+        # and permute the result bins — rows need no relabeling
+        if lead_M > 1:
+            raise NotImplementedError("sort=False with leading array dims: next row")
+        for ax_i, f in enumerate(facs):
+            pidx = grouped_partials(
+                _ffi.SET_IDXMIN, f.codes.to(torch.int64), f.codes, f.ngroups
+            )
+            order = torch.argsort(pidx["idx"], stable=True)
+            result = torch.index_select(result, len(lead_shape) + ax_i, order)
+            groups_list[ax_i] = np.asarray(groups_list[ax_i])[order.cpu().numpy()]
+    groups = tuple(groups_list)
     if return_numpy:
         return (result.cpu().numpy(), *groups)
     return (result, *groups)

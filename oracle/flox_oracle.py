@@ -201,6 +201,7 @@ def groupby_reduce(
     dtype=None,
     min_count=None,
     isbin=False,
+    sort=True,
     finalize_kwargs=None,
 ):
     """Eager grouped reduction with flox semantics. Returns (result, *groups).
@@ -528,6 +529,14 @@ def groupby_reduce(
         result = result.astype(out_dtype, copy=False)
 
     result = result.reshape(lead_shape + grp_shape)
+    if not sort and not provided_expected:
+        # groups in first-appearance order (reference pd.factorize(sort=False))
+        for ax_i, (c, f) in enumerate(zip(codes_list, found)):
+            first = np.full(len(f), np.iinfo(np.int64).max)
+            np.minimum.at(first, c[c >= 0], np.arange(len(c))[c >= 0])
+            order = np.argsort(first, kind="stable")
+            found[ax_i] = np.asarray(f)[order]
+            result = np.take(result, order, axis=len(lead_shape) + ax_i)
     return (result, *found)
 
 

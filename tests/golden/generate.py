@@ -194,6 +194,11 @@ def gen_cases():
         expected_groups=np.array([0, 10, 20, 30, 40, 50]).astype(np.float64),
         isbin=True,
     )
+    # sort=False: groups in first-appearance order (factorize.py:96)
+    ub = np.array([30, 5, 30, 17, 5, 2, 17, 30, 2, 9])
+    uv = rng.standard_normal(10)
+    for func in ["sum", "mean", "count"]:
+        yield f"{func}_nosort", dict(array=uv, by=ub, func=func, sort=False)
     # empty groups at the tail of the range
     yield "mean_sparse_groups", dict(
         array=rng.standard_normal(50),
@@ -272,6 +277,8 @@ def main():
                 out[f"{name}::expected{i}"] = np.asarray(e)
         if kw.get("isbin"):
             out[f"{name}::isbin"] = np.asarray(True)
+        if kw.get("sort") is False:
+            out[f"{name}::nosort"] = np.asarray(True)
         if kw.get("fill_value") is not None:
             out[f"{name}::fill_value"] = np.asarray(kw["fill_value"])
         if kw.get("finalize_kwargs"):

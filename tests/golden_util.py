@@ -48,6 +48,8 @@ def load_golden_cases():
             kw["finalize_kwargs"] = {"q": qv.item() if get("q_scalar") else qv.tolist()}
         if get("isbin") is not None:
             kw["isbin"] = True
+        if get("nosort") is not None:
+            kw["sort"] = False
         if get("scan") is not None:
             kw["_scan"] = True
             func = name.split("_")[1]
