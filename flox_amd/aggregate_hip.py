@@ -727,6 +727,30 @@ def nanmode(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None
     return _to_dtype(grouped_mode(array, group_idx, ng, skipnan=True), dtype)
 
 
+def _scan_like(group_idx, array, func, size=None, dtype=None):
+    from .scan import groupby_scan
+
+    group_idx, array = _prep(group_idx, array)
+    ng = _size_of(group_idx, size)
+    return groupby_scan(array, group_idx, func=func, expected_groups=range(ng))
+
+
+def cumsum(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _scan_like(group_idx, array, "cumsum", size=size, dtype=dtype)
+
+
+def nancumsum(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _scan_like(group_idx, array, "nancumsum", size=size, dtype=dtype)
+
+
+def ffill(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _scan_like(group_idx, array, "ffill", size=size, dtype=dtype)
+
+
+def bfill(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+    return _scan_like(group_idx, array, "bfill", size=size, dtype=dtype)
+
+
 def any_(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
     group_idx, array = _prep(group_idx, array)
     if not array.dtype.is_floating_point:
