@@ -217,11 +217,27 @@ def groupby_reduce(
         assert b.shape == by_shape
     if tuple(arr.shape[arr.ndim - len(by_shape) :]) != tuple(by_shape):
         raise ValueError(f"by {tuple(by_shape)} must align with trailing dims of array {tuple(arr.shape)}")
+    subset_keep_shape: tuple | None = None
     if axis is not None:
         ax = axis if isinstance(axis, (tuple, list)) else (axis,)
         ax = tuple(a % arr.ndim for a in ax)
-        if tuple(sorted(ax)) != tuple(range(arr.ndim - len(by_shape), arr.ndim)):
-            raise NotImplementedError("GPU path reduces over all dims of by (axis subset: next row)")
+        trailing = tuple(range(arr.ndim - len(by_shape), arr.ndim))
+        if tuple(sorted(ax)) != trailing:
+            # axis subset of by's dims (reference offset-labels case,
+            # factorize.py:24-39): move reduced dims last and fold the kept
+            # by-dims into the group codes below
+            if not set(ax) <= set(trailing):
+                raise ValueError(f"axis {axis} must address dims of by")
+            if arr.ndim != bys[0].ndim:
+                raise NotImplementedError(
+                    "axis subset with extra leading array dims: next row"
+                )
+            keep = [d for d in range(arr.ndim) if d not in ax]
+            perm = keep + sorted(ax)
+            arr = arr.permute(perm).contiguous()
+            bys = tuple(b.permute(perm).contiguous() for b in bys)
+            by_shape = bys[0].shape
+            subset_keep_shape = tuple(arr.shape[: len(keep)])
     lead_shape = tuple(arr.shape[: arr.ndim - len(by_shape)])
     lead_M = math.prod(lead_shape) if lead_shape else 1
 
@@ -241,13 +257,39 @@ def groupby_reduce(
     grp_shape = tuple(f.ngroups for f in facs)
     ngroups = math.prod(grp_shape)
     labels, labels2, grp_pair = _combined_codes(facs)
+
+    if subset_keep_shape is not None:
+        # offset the codes by the kept-dims slice index (reference
+        # offset_labels, factorize.py:24-39): slice s's groups become bins
+        # [s*ngroups, (s+1)*ngroups)
+        if labels2 is not None:
+            g0s, g1s = grp_pair
+            c0_, c1_ = labels.to(torch.int64), labels2.to(torch.int64)
+            bad_ = (c0_ < 0) | (c0_ >= g0s) | (c1_ < 0) | (c1_ >= g1s)
+            base_codes = torch.where(bad_, torch.full_like(c0_, -1), c0_ * g1s + c1_)
+        else:
+            c0_ = labels.to(torch.int64)
+            base_codes = torch.where(
+                (c0_ < 0) | (c0_ >= ngroups), torch.full_like(c0_, -1), c0_
+            )
+        inner_n = math.prod(by_shape[len(subset_keep_shape):])
+        lead_idx = torch.arange(base_codes.numel(), device=device) // inner_n
+        labels = torch.where(
+            base_codes < 0, torch.full_like(base_codes, -1),
+            lead_idx * ngroups + base_codes,
+        )
+        labels2, grp_pair = None, None
+        subset_ngroups = ngroups
+        ngroups = int(math.prod(subset_keep_shape)) * ngroups
     # only the 1-D path flattens the values (a no-op for contiguous input);
     # the column path reads the caller's strided view directly
     vals = arr.reshape(-1) if lead_M == 1 else None
 
     # min_count defaulting (reference core.py:1026-1038 + aggregations.py:997-1003)
     if min_count is None:
-        min_count_ = 1 if (fill_value is not None and provided_expected) else 0
+        min_count_ = 1 if (
+            (fill_value is not None and provided_expected) or subset_keep_shape is not None
+        ) else 0
     else:
         min_count_ = min_count
     out_dtype = xrdtypes.final_dtype(func, in_np_dtype, dtype)
@@ -547,10 +589,15 @@ def groupby_reduce(
         # column partials are (ngroups, M) group-major; the API result puts
         # the group dims last
         result = result.reshape(ngroups, lead_M).t().contiguous()
-    result = result.reshape(lead_shape + grp_shape)
+    if subset_keep_shape is not None:
+        result = result.reshape(subset_keep_shape + grp_shape)
+    else:
+        result = result.reshape(lead_shape + grp_shape)
 
     groups_list = [f.groups for f in facs]
     if not sort and not provided_expected:
+        if subset_keep_shape is not None:
+            raise NotImplementedError("sort=False with an axis subset: next row")
         # groups in first-appearance order (reference pd.factorize(sort=False),
         # factorize.py:96): find each group's first row with an index-min pass
         # and permute the result bins — rows need no relabeling

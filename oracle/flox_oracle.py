@@ -221,9 +221,47 @@ def groupby_reduce(
         assert b.shape == bys[0].shape
     assert array.shape[array.ndim - by_ndim :] == bys[0].shape, (array.shape, bys[0].shape)
     if axis is not None:
-        ax = axis if isinstance(axis, tuple) else (axis,)
+        ax = axis if isinstance(axis, (tuple, list)) else (axis,)
         ax = tuple(a % array.ndim for a in ax)
-        assert ax == tuple(range(array.ndim - by_ndim, array.ndim)), "oracle reduces over all by dims"
+        trailing = tuple(range(array.ndim - by_ndim, array.ndim))
+        if tuple(sorted(ax)) != trailing:
+            # axis subset of by's dims (reference offset-labels case,
+            # factorize.py:24-39 + core.py:1026-1032: min_count forced >= 1):
+            # restate as one full reduction per leading slice against the
+            # globally-found groups
+            assert set(ax) <= set(trailing) and array.ndim == by_ndim
+            keep = [d for d in range(array.ndim) if d not in ax]
+            perm = keep + sorted(ax)
+            arr_t = np.transpose(array, perm)
+            bys_t = [np.transpose(b, perm) for b in bys]
+            kshape = arr_t.shape[: len(keep)]
+            if expected_groups is not None and not isinstance(expected_groups, tuple):
+                expected_groups = (expected_groups,)
+            if expected_groups is None or all(e is None for e in expected_groups):
+                founds = []
+                for b in bys_t:
+                    _, f0 = _factorize_single(b.reshape(-1), None)
+                    founds.append(f0)
+                expected_groups = tuple(founds)
+            slices = []
+            groups_out = None
+            for idx in np.ndindex(*kshape):
+                r, *g = groupby_reduce(
+                    arr_t[idx],
+                    *[b[idx] for b in bys_t],
+                    func=func,
+                    expected_groups=expected_groups,
+                    fill_value=fill_value,
+                    dtype=dtype,
+                    min_count=min_count if min_count is not None else 1,
+                    isbin=isbin,
+                    sort=sort,
+                    finalize_kwargs=finalize_kwargs,
+                )
+                slices.append(r)
+                groups_out = g
+            out = np.stack(slices).reshape(kshape + slices[0].shape)
+            return (out, *groups_out)
 
     if expected_groups is not None and not isinstance(expected_groups, tuple):
         expected_groups = (expected_groups,)
@@ -262,6 +300,10 @@ def groupby_reduce(
             fill_value = _fill_default(func, out_dtype)
     if min_count_ > 0 and func in ("nansum", "nanprod") and fill_value is None:
         fill_value = np.nan  # reference core.py:1035-1038
+    if fill_value is not None:
+        # a concrete user fill promotes the output dtype unconditionally
+        # (reference xrdtypes.py:170-171)
+        out_dtype = np.result_type(out_dtype, fill_value)
 
     # --- chunk reduction over flattened group dims ---
     lead_shape = array.shape[: array.ndim - by_ndim]
@@ -524,9 +566,7 @@ def groupby_reduce(
         if empty_mask.any():
             result = np.where(empty_mask, fv, result)
 
-    result = np.asarray(result)
-    if result.dtype != out_dtype and (user_fill is None or np.asarray(user_fill).dtype.kind not in "fc" or out_dtype.kind in "fc"):
-        result = result.astype(out_dtype, copy=False)
+    result = np.asarray(result).astype(out_dtype, copy=False)
 
     result = result.reshape(lead_shape + grp_shape)
     if not sort and not provided_expected:

@@ -194,6 +194,19 @@ def gen_cases():
         expected_groups=np.array([0, 10, 20, 30, 40, 50]).astype(np.float64),
         isbin=True,
     )
+    # axis subset of by's dims (offset-labels, factorize.py:24-39)
+    a3 = rng.standard_normal((4, 5, 30))
+    b3 = rng.integers(0, 6, (4, 5, 30))
+    for func in ["sum", "mean", "count", "nanmax"]:
+        yield (
+            f"{func}_axis_subset",
+            dict(array=a3, by=b3, func=func, axis=(2,),
+                 expected_groups=np.arange(6), fill_value=-99.0),
+        )
+    yield "mean_axis_subset2", dict(
+        array=a3, by=b3, func="mean", axis=(1, 2),
+        expected_groups=np.arange(6), fill_value=-99.0,
+    )
     # sort=False: groups in first-appearance order (factorize.py:96)
     ub = np.array([30, 5, 30, 17, 5, 2, 17, 30, 2, 9])
     uv = rng.standard_normal(10)
@@ -279,6 +292,8 @@ def main():
             out[f"{name}::isbin"] = np.asarray(True)
         if kw.get("sort") is False:
             out[f"{name}::nosort"] = np.asarray(True)
+        if kw.get("axis") is not None:
+            out[f"{name}::axis"] = np.asarray(kw["axis"])
         if kw.get("fill_value") is not None:
             out[f"{name}::fill_value"] = np.asarray(kw["fill_value"])
         if kw.get("finalize_kwargs"):

@@ -50,6 +50,9 @@ def load_golden_cases():
             kw["isbin"] = True
         if get("nosort") is not None:
             kw["sort"] = False
+        axv = get("axis")
+        if axv is not None:
+            kw["axis"] = tuple(int(a) for a in np.atleast_1d(axv))
         if get("scan") is not None:
             kw["_scan"] = True
             func = name.split("_")[1]
