@@ -192,6 +192,10 @@ def groupby_reduce(
     initialized torch.distributed world (default: auto when initialized).
     Each rank passes its own row shard; results are full-group on every rank.
     """
+    from .aggregations import Aggregation as _Agg
+
+    if isinstance(func, _Agg):
+        func = func.name  # reference accepts Aggregation instances (core.py:934-948)
     if engine != "hip":
         raise ValueError(f"flox_amd implements engine='hip' only (got {engine!r})")
     if func not in REDUCTIONS:
@@ -203,6 +207,16 @@ def groupby_reduce(
     device = torch.device("cuda", torch.cuda.current_device())
 
     return_numpy = not isinstance(array, torch.Tensor)
+    # datetime64/timedelta64 values compute on their int64 view and
+    # dtype-preserving results view back (reference core.py:985-1001)
+    dt_dtype = None
+    if return_numpy:
+        arr_np = np.asarray(array)
+        if arr_np.dtype.kind in "Mm":
+            dt_dtype = arr_np.dtype
+            array = arr_np.view("i8")
+            if np.issubdtype(dt_dtype, np.datetime64) and (array == np.iinfo(np.int64).min).any():
+                raise NotImplementedError("NaT values: next row")
     arr = _as_device_tensor(array, device)
     was_bool = arr.dtype == torch.bool
     if was_bool:
@@ -245,6 +259,30 @@ def groupby_reduce(
         expected_groups = (expected_groups,)
     if expected_groups is None:
         expected_groups = (None,) * nby
+    # pd.IntervalIndex expected groups imply binning (reference
+    # _convert_expected_groups_to_index, core.py:931-933)
+    try:
+        import pandas as pd
+
+        new_eg, new_isbin = [], []
+        isbins0 = isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby
+        changed = False
+        for e, ib in zip(expected_groups, isbins0):
+            if isinstance(e, pd.IntervalIndex):
+                assert (e.left.to_numpy()[1:] == e.right.to_numpy()[:-1]).all(), (
+                    "IntervalIndex bins must be contiguous"
+                )
+                new_eg.append(np.append(e.left.to_numpy(), e.right.to_numpy()[-1]))
+                new_isbin.append(True)
+                changed = True
+            else:
+                new_eg.append(e)
+                new_isbin.append(ib)
+        if changed:
+            expected_groups = tuple(new_eg)
+            isbin = tuple(new_isbin)
+    except ImportError:
+        pass
     provided_expected = any(e is not None for e in expected_groups)
     if isbin and not provided_expected:
         raise ValueError("isbin=True requires expected_groups (the bin edges)")
@@ -612,5 +650,8 @@ def groupby_reduce(
             groups_list[ax_i] = np.asarray(groups_list[ax_i])[order.cpu().numpy()]
     groups = tuple(groups_list)
     if return_numpy:
-        return (result.cpu().numpy(), *groups)
+        out_np = result.cpu().numpy()
+        if dt_dtype is not None and func in xrdtypes.PRESERVES_DTYPE:
+            out_np = out_np.view(dt_dtype)
+        return (out_np, *groups)
     return (result, *groups)

@@ -210,12 +210,30 @@ def groupby_reduce(
     reduction over all dims of by (axis=None or the full trailing tuple).
     """
     array = np.asarray(array)
+    dt_dtype = None
+    if array.dtype.kind in "Mm":
+        dt_dtype = array.dtype
+        array = array.view("i8")
     bys = tuple(np.asarray(b) for b in by)
     nby = len(bys)
     if nby == 0:
         raise ValueError("need at least one by")
     if func not in ALL_FUNCS:
         raise NotImplementedError(func)
+    try:
+        import pandas as pd
+
+        eg0 = expected_groups if isinstance(expected_groups, tuple) else (
+            (expected_groups,) if expected_groups is not None else None)
+        if eg0 is not None and any(isinstance(e, pd.IntervalIndex) for e in eg0):
+            expected_groups = tuple(
+                np.append(e.left.to_numpy(), e.right.to_numpy()[-1])
+                if isinstance(e, pd.IntervalIndex) else e for e in eg0)
+            isbin = tuple(
+                True if isinstance(e, pd.IntervalIndex) else ib
+                for e, ib in zip(eg0, (isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby)))
+    except ImportError:
+        pass
     by_ndim = bys[0].ndim
     for b in bys:
         assert b.shape == bys[0].shape
@@ -567,6 +585,8 @@ def groupby_reduce(
             result = np.where(empty_mask, fv, result)
 
     result = np.asarray(result).astype(out_dtype, copy=False)
+    if dt_dtype is not None and func in _PRESERVES_DTYPE:
+        result = result.view(dt_dtype)
 
     result = result.reshape(lead_shape + grp_shape)
     if not sort and not provided_expected:

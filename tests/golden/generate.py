@@ -194,6 +194,19 @@ def gen_cases():
         expected_groups=np.array([0, 10, 20, 30, 40, 50]).astype(np.float64),
         isbin=True,
     )
+    # datetime64 values (viewed as int64 inside the reference, core.py:985-1001)
+    tvals = (np.datetime64("2020-01-01") + rng.integers(0, 10_000, 60).astype("timedelta64[m]"))
+    tby = rng.integers(0, 5, 60)
+    for func in ["min", "max", "count", "first", "nanmax", "last"]:
+        yield f"{func}_datetime", dict(array=tvals, by=tby, func=func, expected_groups=np.arange(5))
+    # pd.IntervalIndex expected_groups (binning without isbin=True)
+    import pandas as pd
+    iv_vals = rng.standard_normal(300)
+    iv_by = rng.standard_normal(300) * 2
+    yield "mean_intervalindex", dict(
+        array=iv_vals, by=iv_by, func="mean",
+        expected_groups=pd.IntervalIndex.from_breaks(np.array([-3.0, -1.0, 0.5, 3.0])),
+    )
     # axis subset of by's dims (offset-labels, factorize.py:24-39)
     a3 = rng.standard_normal((4, 5, 30))
     b3 = rng.integers(0, 6, (4, 5, 30))
@@ -287,7 +300,12 @@ def main():
         if eg is not None:
             egs = eg if isinstance(eg, tuple) else (eg,)
             for i, e in enumerate(egs):
-                out[f"{name}::expected{i}"] = np.asarray(e)
+                import pandas as pd
+                if isinstance(e, pd.IntervalIndex):
+                    out[f"{name}::expected{i}"] = np.append(e.left.to_numpy(), e.right.to_numpy()[-1])
+                    out[f"{name}::interval{i}"] = np.asarray(True)
+                else:
+                    out[f"{name}::expected{i}"] = np.asarray(e)
         if kw.get("isbin"):
             out[f"{name}::isbin"] = np.asarray(True)
         if kw.get("sort") is False:

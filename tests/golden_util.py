@@ -26,7 +26,12 @@ def load_golden_cases():
         expected = []
         i = 0
         while get(f"expected{i}") is not None:
-            expected.append(get(f"expected{i}"))
+            e = get(f"expected{i}")
+            if get(f"interval{i}") is not None:
+                import pandas as pd
+
+                e = pd.IntervalIndex.from_breaks(e)
+            expected.append(e)
             i += 1
         groups = []
         i = 0

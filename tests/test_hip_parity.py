@@ -28,8 +28,11 @@ def test_golden_parity(name, inputs, expected, groups):
         result, *found = flox_amd.groupby_reduce(arr, *bys, **kw)
     assert result.shape == expected.shape
     assert result.dtype == expected.dtype, (result.dtype, expected.dtype)
-    tol = tolerance_for(name, expected.dtype)
-    np.testing.assert_allclose(result, expected, equal_nan=True, **tol)
+    if expected.dtype.kind in "Mm":
+        np.testing.assert_array_equal(result, expected)
+    else:
+        tol = tolerance_for(name, expected.dtype)
+        np.testing.assert_allclose(result, expected, equal_nan=True, **tol)
 
 
 FUNCS = [
