@@ -57,14 +57,15 @@ def cpu_baseline_leg(rows_sample: int, ngroups: int) -> dict:
     }
 
 
-def read_traffic():
-    """Per-launch HBM bytes from the committed rocprof PMC summary, if any
-    (profiles/traffic.json, written by profiles/collect.sh on the GPU box)."""
+def read_traffic(config: str):
+    """Per-launch HBM bytes of the dominant kernel from the committed rocprof
+    PMC summary (profiles/traffic.json, written by profiles/collect.sh and
+    collect_pmc.sh on the GPU box)."""
     p = os.path.join(os.path.dirname(os.path.abspath(__file__)), "profiles", "traffic.json")
     if os.path.exists(p):
         with open(p) as f:
             d = json.load(f)
-        return d.get("bytes_per_launch")
+        return d.get("configs", {}).get(config, {}).get("bytes_per_launch")
     return None
 
 
@@ -246,7 +247,7 @@ def main():
                 "peak": HBM_PEAK_GBPS,
                 "unit": "GB/s",
                 "frac": round(achieved_gbps / HBM_PEAK_GBPS, 4) if achieved_gbps else None,
-                "traffic": read_traffic() if args.config == "2" else None,
+                "traffic": read_traffic(args.config),
             },
             "cpu_baseline": cpu,
         }
