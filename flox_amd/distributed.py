@@ -54,3 +54,76 @@ def all_reduce_(t: torch.Tensor, op: str = "sum") -> torch.Tensor:
     if is_active():
         dist.all_reduce(t, op=_ops()[op])
     return t
+
+
+def scan_carry_exchange(
+    out: torch.Tensor,
+    vals: torch.Tensor,
+    codes: torch.Tensor,
+    ngroups: int,
+    func: str,
+) -> torch.Tensor:
+    """Cross-rank carry for grouped scans: rank r holds rows [r·shard, …) of
+    the global row order, and its locally-scanned ``out`` is corrected with
+    the per-group state of earlier (later, for bfill) ranks.
+
+    This is the flat-rank form of the reference's Blelloch scan combine
+    (flox/aggregations.py:792-845 scan_binary_op, flox/dask.py:628-652):
+    the per-rank state is the per-group total (cumsum — apply_binary_op with
+    add) or the last/first valid value (ffill/bfill — concat_then_scan), the
+    tree is replaced by one all_gather of the (ngroups,) state vector and an
+    exclusive walk over ranks.
+
+    ``codes`` are the factorized group codes per row; rows with codes outside
+    [0, ngroups) (the reference's NaN-sentinel group, factorize.py:201-210)
+    carry locally only. Works on CPU (gloo) and GPU (RCCL) tensors alike.
+    """
+    if not is_active() or ngroups <= 0 or out.numel() == 0:
+        return out
+    world, rank = dist.get_world_size(), dist.get_rank()
+    codes = codes.to(torch.int64)
+    valid = (codes >= 0) & (codes < ngroups)
+    cidx = codes.clamp(0, ngroups - 1)
+
+    if func in ("cumsum", "nancumsum"):
+        # per-group total; for cumsum a NaN poisons the total (and so every
+        # later rank's rows of that group), matching np.cumsum propagation
+        v = vals
+        if func == "nancumsum" and v.is_floating_point():
+            v = torch.nan_to_num(v, nan=0.0)
+        totals = torch.zeros(ngroups, dtype=out.dtype, device=out.device)
+        totals.index_add_(0, cidx[valid], v[valid].to(out.dtype))
+        gathered = [torch.empty_like(totals) for _ in range(world)]
+        dist.all_gather(gathered, totals)
+        prefix = torch.zeros_like(totals)
+        for k in range(rank):
+            prefix += gathered[k]
+        add = prefix[cidx]
+        add = torch.where(valid, add, torch.zeros_like(add))
+        return out + add
+
+    if func in ("ffill", "bfill"):
+        if not out.is_floating_point():
+            return out  # integer fills have no missing rows to carry into
+        nan = torch.tensor(float("nan"), dtype=out.dtype, device=out.device)
+        idx = torch.arange(out.numel(), device=out.device)
+        pos = torch.full((ngroups,), -1, dtype=torch.int64, device=out.device)
+        # the edge row of each group in the local scan already holds that
+        # rank's state: last row -> last-valid (ffill), first row -> the
+        # first-valid seen scanning backwards (bfill)
+        red = "amax" if func == "ffill" else "amin"
+        if func == "bfill":
+            pos = torch.full_like(pos, out.numel())
+        pos.index_reduce_(0, cidx[valid], idx[valid], red, include_self=False)
+        has = (pos >= 0) & (pos < out.numel())
+        state = torch.where(has, out[pos.clamp(0, max(out.numel() - 1, 0))], nan)
+        gathered = [torch.empty_like(state) for _ in range(world)]
+        dist.all_gather(gathered, state)
+        carry = torch.full_like(state, float("nan"))
+        ranks = range(rank) if func == "ffill" else range(world - 1, rank, -1)
+        for k in ranks:  # nearest valid wins (forward for ffill, backward for bfill)
+            carry = torch.where(torch.isnan(gathered[k]), carry, gathered[k])
+        fill = torch.isnan(out) & valid
+        return torch.where(fill, carry[cidx], out)
+
+    raise NotImplementedError(f"distributed scan {func!r}")

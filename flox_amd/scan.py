@@ -14,14 +14,17 @@ import ctypes
 import numpy as np
 import torch
 
-from . import _ffi
+from . import _ffi, distributed
 from ._ffi import FhCall
 from .core import _as_device_tensor, _combined_codes, _factorize_device, _np_dtype
 
 SCAN_OPS = {"cumsum": 0, "nancumsum": 1, "ffill": 2, "bfill": 3}
 
 
-def groupby_scan(array, *by, func: str, expected_groups=None, axis=None, dtype=None):
+def groupby_scan(
+    array, *by, func: str, expected_groups=None, axis=None, dtype=None,
+    distributed_combine: bool | None = None,
+):
     """Grouped scan. Returns an array shaped like ``array``.
 
     cumsum/nancumsum: per-group running sum in row order (np.cumsum /
@@ -99,6 +102,16 @@ def groupby_scan(array, *by, func: str, expected_groups=None, axis=None, dtype=N
     for t in (vals, labels, labels2, scratch, out):
         if isinstance(t, torch.Tensor):
             t.record_stream(torch.cuda.current_stream(device))
+
+    if distributed_combine is None:
+        distributed_combine = distributed.is_active()
+    if distributed_combine and distributed.is_active():
+        if labels2 is not None:
+            codes = labels.to(torch.int64) * grp_pair[1] + labels2.to(torch.int64)
+            codes = torch.where((labels < 0) | (labels2 < 0), torch.full_like(codes, -1), codes)
+        else:
+            codes = labels
+        out = distributed.scan_carry_exchange(out, vals, codes, ngroups, func)
 
     if dtype is not None:
         td = torch.from_numpy(np.empty(0, dtype=np.dtype(dtype))).dtype

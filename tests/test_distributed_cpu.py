@@ -180,6 +180,69 @@ def _worker_full_branch(rank, world, port, fail_q):
         raise
 
 
+def _worker_scan(rank, world, port, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import numpy as np
+        import torch
+        import flox_amd.distributed as fdist
+        from oracle import groupby_scan as oracle_scan
+
+        rng = np.random.default_rng(11)
+        n, ng = 9_000, 23
+        fvals = rng.standard_normal(n)
+        fvals[rng.random(n) < 0.3] = np.nan  # big NaN runs so carries matter
+        ivals = rng.integers(-50, 50, n).astype(np.int64)
+        labels = rng.integers(0, ng, n)
+        eg = np.arange(ng)
+        sl = slice(rank * n // world, (rank + 1) * n // world)
+
+        for vals, funcs in [
+            (fvals, ["cumsum", "nancumsum", "ffill", "bfill"]),
+            (ivals, ["cumsum"]),
+        ]:
+            for func in funcs:
+                local = oracle_scan(vals[sl], labels[sl], func=func, expected_groups=eg)
+                got = fdist.scan_carry_exchange(
+                    torch.from_numpy(np.ascontiguousarray(local)),
+                    torch.from_numpy(np.ascontiguousarray(vals[sl])),
+                    torch.from_numpy(np.ascontiguousarray(labels[sl])),
+                    ng,
+                    func,
+                )
+                want = oracle_scan(vals, labels, func=func, expected_groups=eg)[sl]
+                np.testing.assert_allclose(
+                    got.numpy(), want, equal_nan=True, rtol=1e-10, atol=1e-10,
+                    err_msg=f"{func} {vals.dtype}",
+                )
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}\n{traceback.format_exc()}")
+        raise
+
+
+def test_gloo_scan_carry_exchange():
+    """Distributed grouped scans at world_size 2: each rank scans its shard
+    locally (via the oracle, standing in for the HIP scan kernel) and the
+    carry exchange must reproduce the whole-array scan — the flat-rank form
+    of the reference's scan_binary_op combine (flox/aggregations.py:792-845)."""
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_scan, args=(r, 2, 29521, fail_q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+    errs = []
+    while not fail_q.empty():
+        errs.append(fail_q.get())
+    assert not errs, errs[0]
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
 def test_gloo_full_groupby_reduce_branch():
     """The exact distributed code in core.groupby_reduce (partial all-reduce,
     global-mean var, finalize) at world_size 2, with the kernel layer stubbed
