@@ -226,9 +226,20 @@ def groupby_reduce(
     nby = len(bys)
     if nby == 0:
         raise ValueError("need at least one by array")
+    # size-1 by dims broadcast against the array's trailing dims (the
+    # dim=... case, reference core.py:300-309); multiple by arrays must
+    # share a shape, as in the reference
     by_shape = bys[0].shape
     for b in bys:
-        assert b.shape == by_shape
+        if b.shape != by_shape:
+            raise ValueError("by arrays must have the same shape")
+    if len(by_shape) <= arr.ndim:
+        trail = tuple(arr.shape[arr.ndim - len(by_shape):])
+        if trail != tuple(by_shape) and all(
+            bs in (1, ts) for bs, ts in zip(by_shape, trail)
+        ):
+            bys = tuple(b.broadcast_to(trail).contiguous() for b in bys)
+            by_shape = bys[0].shape
     if tuple(arr.shape[arr.ndim - len(by_shape) :]) != tuple(by_shape):
         raise ValueError(f"by {tuple(by_shape)} must align with trailing dims of array {tuple(arr.shape)}")
     subset_keep_shape: tuple | None = None

@@ -234,6 +234,14 @@ def groupby_reduce(
                 for e, ib in zip(eg0, (isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby)))
     except ImportError:
         pass
+    # size-1 by dims broadcast against the array's trailing dims
+    # (reference core.py:300-309)
+    if bys[0].ndim <= array.ndim:
+        trail = array.shape[array.ndim - bys[0].ndim:]
+        if trail != bys[0].shape and all(
+            bs in (1, ts) for bs, ts in zip(bys[0].shape, trail)
+        ):
+            bys = tuple(np.ascontiguousarray(np.broadcast_to(b, trail)) for b in bys)
     by_ndim = bys[0].ndim
     for b in bys:
         assert b.shape == bys[0].shape

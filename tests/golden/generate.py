@@ -239,6 +239,20 @@ def gen_cases():
         expected_groups=np.arange(10),
         fill_value=-1.0,
     )
+    # size-1 by dims broadcast against the array's trailing dims (the
+    # dim=... case, reference core.py:300-309)
+    ab = rng.standard_normal((30, 40))
+    bb = rng.integers(0, 5, (1, 40))
+    for func in ["sum", "mean", "nanmax"]:
+        yield f"{func}_by_size1_bcast", dict(
+            array=ab, by=bb, func=func, expected_groups=np.arange(5)
+        )
+    yield "mean_by_size1_bcast3d", dict(
+        array=rng.standard_normal((6, 10, 20)),
+        by=rng.integers(0, 4, (10, 1)),
+        func="mean",
+        expected_groups=np.arange(4),
+    )
 
 
 def gen_scan_cases():
