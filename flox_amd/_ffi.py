@@ -107,6 +107,11 @@ def load_library():
     lib.fh_scan_scratch_bytes.restype = ctypes.c_int64
     lib.fh_grouped_scan.argtypes = [ctypes.POINTER(FhCall), ctypes.c_int]
     lib.fh_grouped_scan.restype = ctypes.c_int
+    lib.fh_pack_argkeys.argtypes = [
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int64, ctypes.c_int64,
+        ctypes.c_int, ctypes.c_int, ctypes.c_void_p, ctypes.c_void_p,
+    ]
+    lib.fh_pack_argkeys.restype = ctypes.c_int
     lib.fh_error_string.argtypes = [ctypes.c_int]
     lib.fh_error_string.restype = ctypes.c_char_p
     lib.fh_version.argtypes = []

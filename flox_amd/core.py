@@ -23,6 +23,12 @@ from . import _ffi, distributed, xrdtypes
 from .aggregate_hip import grouped_partials, grouped_partials_cols
 from .aggregations import REDUCTIONS
 
+# arg-reductions switch from the two-pass (extremum, then index-match) form to
+# the packed-key single-pass form above this group count: the IDX bins cost
+# 20 B/group in LDS, so past ~8e3 groups the second pass would fall to the
+# global-atomic path. Tests lower this to exercise the packed form at small n.
+PACKED_ARG_THRESHOLD = 7000
+
 _TORCH_TO_NP = {
     torch.float32: np.dtype("float32"),
     torch.float64: np.dtype("float64"),
@@ -466,6 +472,38 @@ def groupby_reduce(
             if arr.dtype.is_floating_point
             else counts_for_mask == 0
         )
+    elif func in ("argmax", "argmin", "nanargmax", "nanargmin") and (
+        ngroups > PACKED_ARG_THRESHOLD
+        and vals.dtype in (torch.float32, torch.int32)
+        and not dist_on
+        and shard_row_offset + vals.numel() < (1 << 32)
+    ):
+        # huge group counts: pack (order-preserving 32-bit value encoding,
+        # row index) into one int64 key and take a single grouped MIN — the
+        # partition path then handles what the 20 B/group IDX bins cannot
+        # (LDS holds ~8e3 of them). Ties break to the smaller row, which is
+        # exactly np.argmin/argmax's first-occurrence rule.
+        skip = agg.skipnan
+        ismax = "max" in func
+        lib = _ffi.load_library()
+        key = torch.empty(vals.numel(), dtype=torch.int64, device=device)
+        vc = vals.contiguous()
+        _ffi.check(lib.fh_pack_argkeys(
+            vc.data_ptr(),
+            _ffi.F32 if vals.dtype == torch.float32 else _ffi.I32,
+            vc.numel(), shard_row_offset, int(ismax), int(skip),
+            key.data_ptr(),
+            torch.cuda.current_stream(device).cuda_stream,
+        ))
+        vc.record_stream(torch.cuda.current_stream(device))
+        p = grouped_partials(
+            _ffi.SET_MIN_COUNT, key, labels, ngroups,
+            skipnan=False, labels2=labels2, grp_shape=grp_pair,
+        )
+        kmin = p["min"]
+        empty_mask = kmin == ((1 << 63) - 1)
+        result = (kmin ^ (-(1 << 63))) & 0xFFFFFFFF
+        counts_for_mask = p["count"]
     elif func in ("argmax", "argmin", "nanargmax", "nanargmin"):
         # pass 1: the per-group extremum; pass 2: the smallest row index whose
         # value matches it (ties -> first occurrence, like np.argmax; a NaN

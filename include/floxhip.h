@@ -137,6 +137,15 @@ int fh_grouped_reduce(fh_call* c);
  * computed without offsetting labels: one segmented pass per column) */
 int fh_grouped_reduce_cols(fh_call* c);
 
+/* pack (order-preserving 32-bit value encoding, global row index) into one
+ * int64 key per row, so that a grouped MIN over the keys is argmin/argmax
+ * with np.argmin's first-occurrence tie-break (the packed form of the
+ * reference's argreduce, flox/aggregate_flox.py:151-187). FH_F32/FH_I32
+ * values only; requires n + row_offset < 2^32. out: int64[n]. */
+int fh_pack_argkeys(const void* values, int vdtype, int64_t n,
+                    int64_t row_offset, int ismax, int skipnan, void* out,
+                    void* stream);
+
 const char* fh_error_string(int code);
 int fh_version(void);
 

@@ -27,7 +27,12 @@ def test_cabi_library_loads_and_exports_declared_symbols():
     """include/floxhip.h declares the boundary; the built .so must export it."""
     assert os.path.exists(LIB), "libfloxhip.so not built (run __graft_entry__.build())"
     lib = ctypes.CDLL(LIB)
-    for sym in ["fh_grouped_reduce", "fh_scratch_bytes", "fh_error_string", "fh_version"]:
+    for sym in [
+        "fh_grouped_reduce", "fh_grouped_reduce_cols", "fh_scratch_bytes",
+        "fh_grouped_quantile", "fh_quantile_scratch_bytes", "fh_grouped_scan",
+        "fh_scan_scratch_bytes", "fh_pack_argkeys", "fh_error_string",
+        "fh_version",
+    ]:
         assert hasattr(lib, sym), sym
     lib.fh_version.restype = ctypes.c_int
     assert lib.fh_version() == 1

@@ -337,3 +337,48 @@ def test_column_path_large_ngroups():
     arr_t = torch.tensor(arr, device="cuda").permute(1, 0)
     got, _ = flox_amd.groupby_reduce(arr_t, torch.tensor(labels, device="cuda"), func="mean", expected_groups=np.arange(ng))
     np.testing.assert_allclose(got.cpu().numpy(), want, equal_nan=True, rtol=1e-12, atol=1e-14)
+
+
+def test_packed_arg_small():
+    """The packed-key arg form (one grouped i64 MIN over (enc(value), row)
+    keys) against the oracle, forced on at small sizes; heavy ties and NaN
+    runs exercise the first-occurrence rule and both NaN conventions."""
+    import flox_amd.core as core
+
+    old = core.PACKED_ARG_THRESHOLD
+    try:
+        core.PACKED_ARG_THRESHOLD = 1
+        rng = np.random.default_rng(33)
+        for dt in ["float32", "int32"]:
+            for n, ng in [(5000, 50), (200_000, 3000)]:
+                if dt == "float32":
+                    v = rng.standard_normal(n).astype(dt)
+                    v[rng.random(n) < 0.2] = np.nan
+                    v[rng.integers(0, n, n // 10)] = 1.5  # exact ties
+                else:
+                    v = rng.integers(-40, 40, n).astype(dt)  # many ties
+                labels = rng.integers(0, ng, n)
+                eg = np.arange(ng + 8)  # trailing empty groups
+                for func in ["argmin", "argmax", "nanargmin", "nanargmax"]:
+                    want, *_ = oracle_reduce(v, labels, func=func, expected_groups=eg)
+                    got, *_ = flox_amd.groupby_reduce(v, labels, func=func, expected_groups=eg)
+                    np.testing.assert_array_equal(
+                        np.asarray(got), want, err_msg=f"{dt} {func} {n}x{ng}"
+                    )
+    finally:
+        core.PACKED_ARG_THRESHOLD = old
+
+
+def test_packed_arg_partition_scale():
+    """arg-reductions at group counts far beyond LDS capacity: the packed
+    keys must route through the bucket-partition path and still match the
+    oracle (was the documented atomic-fallback gap)."""
+    rng = np.random.default_rng(44)
+    n, ng = 2_000_000, 200_000
+    v = rng.standard_normal(n).astype(np.float32)
+    v[rng.random(n) < 0.1] = np.nan
+    labels = rng.integers(0, ng, n)
+    for func in ["argmin", "nanargmax"]:
+        want, *_ = oracle_reduce(v, labels, func=func, expected_groups=np.arange(ng))
+        got, *_ = flox_amd.groupby_reduce(v, labels, func=func, expected_groups=range(ng))
+        np.testing.assert_array_equal(np.asarray(got), want, err_msg=func)
