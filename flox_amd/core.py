@@ -530,6 +530,41 @@ def groupby_reduce(
         result = idx
         empty_mask = (p2["present"] == 0) | (idx == sentinel)
         counts_for_mask = p2["count"]
+    elif func in ("first", "last", "nanfirst", "nanlast") and (
+        ngroups > PACKED_ARG_THRESHOLD
+        and not dist_on
+        and shard_row_offset + vals.numel() < (1 << 32)
+    ):
+        # huge group counts: first/last is a grouped MIN/MAX over the row
+        # indices themselves (any value dtype), so the partition path applies
+        # where the 20 B/group IDX bins cannot; NaN rows take the opposite
+        # extreme under nan* so all-NaN groups land on the empty sentinel
+        isfirst = func in ("first", "nanfirst")
+        rows_key = torch.arange(vals.numel(), device=device, dtype=torch.int64) + shard_row_offset
+        if agg.skipnan and vals.is_floating_point():
+            bad_v = (1 << 63) - 1 if isfirst else -1
+            rows_key = torch.where(torch.isnan(vals), torch.full_like(rows_key, bad_v), rows_key)
+        p = grouped_partials(
+            _ffi.SET_MIN_COUNT if isfirst else _ffi.SET_MAX_COUNT,
+            rows_key, labels, ngroups,
+            skipnan=False, labels2=labels2, grp_shape=grp_pair,
+        )
+        idx = p["min"] if isfirst else p["max"]
+        valid = (idx != ((1 << 63) - 1)) if isfirst else (idx >= 0)
+        local = idx - shard_row_offset
+        safe = torch.clamp(local, 0, max(vals.numel() - 1, 0))
+        result = vals[safe]
+        empty_mask = ~valid
+        if min_count_ > 0:
+            pc = grouped_partials(
+                _ffi.SET_COUNT, vals, labels, ngroups, skipnan=agg.skipnan,
+                labels2=labels2, grp_shape=grp_pair,
+            )
+            counts_for_mask = pc["count"]
+        else:
+            counts_for_mask = torch.where(
+                valid, torch.ones_like(idx), torch.zeros_like(idx)
+            )
     elif func in ("first", "last", "nanfirst", "nanlast"):
         p = run_set(agg.op_set, agg.skipnan)
         if dist_on:

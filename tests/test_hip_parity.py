@@ -359,7 +359,8 @@ def test_packed_arg_small():
                     v = rng.integers(-40, 40, n).astype(dt)  # many ties
                 labels = rng.integers(0, ng, n)
                 eg = np.arange(ng + 8)  # trailing empty groups
-                for func in ["argmin", "argmax", "nanargmin", "nanargmax"]:
+                for func in ["argmin", "argmax", "nanargmin", "nanargmax",
+                             "first", "last", "nanfirst", "nanlast"]:
                     want, *_ = oracle_reduce(v, labels, func=func, expected_groups=eg)
                     got, *_ = flox_amd.groupby_reduce(v, labels, func=func, expected_groups=eg)
                     np.testing.assert_array_equal(
@@ -378,7 +379,7 @@ def test_packed_arg_partition_scale():
     v = rng.standard_normal(n).astype(np.float32)
     v[rng.random(n) < 0.1] = np.nan
     labels = rng.integers(0, ng, n)
-    for func in ["argmin", "nanargmax"]:
+    for func in ["argmin", "nanargmax", "first", "nanlast"]:
         want, *_ = oracle_reduce(v, labels, func=func, expected_groups=np.arange(ng))
         got, *_ = flox_amd.groupby_reduce(v, labels, func=func, expected_groups=range(ng))
         np.testing.assert_array_equal(np.asarray(got), want, err_msg=func)
