@@ -29,6 +29,14 @@ from .aggregations import REDUCTIONS
 # global-atomic path. Tests lower this to exercise the packed form at small n.
 PACKED_ARG_THRESHOLD = 7000
 
+# order-dependent reductions that run with leading array dims by folding the
+# lead index into the group codes (see groupby_reduce's lead-fold block)
+_LEAD_FOLD_FUNCS = (
+    "quantile", "nanquantile", "median", "nanmedian", "mode", "nanmode",
+    "argmax", "argmin", "nanargmax", "nanargmin",
+    "first", "last", "nanfirst", "nanlast",
+)
+
 _TORCH_TO_NP = {
     torch.float32: np.dtype("float32"),
     torch.float64: np.dtype("float64"),
@@ -360,6 +368,7 @@ def groupby_reduce(
 
     dist_on = distributed.is_active() if distributed_combine is None else distributed_combine
     ddof = (finalize_kwargs or {}).get("ddof", 0)
+    lead_folded = False
 
     if lead_M == 1:
         def run_set(op_set, skipnan, means=None, target=None):
@@ -399,19 +408,46 @@ def groupby_reduce(
         scodes64, perm64 = torch.sort(codes_full, stable=True)
         scodes, perm = scodes64.to(torch.int32), perm64.to(torch.int32)
 
-        if func in ("argmax", "argmin", "nanargmax", "nanargmin", "first",
-                    "last", "nanfirst", "nanlast"):
-            raise NotImplementedError(f"{func} with leading array dims: next row")
-
         def run_set(op_set, skipnan, means=None, target=None):
             return grouped_partials_cols(
                 op_set, vt, scodes, perm, ngroups, skipnan=skipnan, means=means
             )
 
+        if func in _LEAD_FOLD_FUNCS:
+            # order-dependent reductions with leading dims: fold the lead
+            # index into the group codes (lead*ngroups + code) and run the
+            # 1-D machinery over the C-order flattened stream — the
+            # offset-labels trick of reference factorize.py:24-39
+            if subset_keep_shape is not None:
+                raise NotImplementedError(f"{func} with an axis subset: next row")
+            if dist_on:
+                raise NotImplementedError(f"distributed {func} with leading dims: next row")
+            if N * lead_M >= (1 << 31) or ngroups * lead_M >= (1 << 31):
+                raise NotImplementedError(f"{func} with leading dims at this size: next row")
+            vals = arr.reshape(-1)
+            lead_i = torch.arange(lead_M, device=device, dtype=torch.int64)
+            comp = lead_i[:, None] * ngroups + codes_full[None, :]
+            comp = torch.where(
+                (codes_full < 0)[None, :].expand_as(comp),
+                torch.full_like(comp, -1), comp,
+            )
+            labels = comp.reshape(-1)
+            labels2, grp_pair = None, None
+            arg_localize_N = N
+            base_ngroups = ngroups
+            ngroups = lead_M * ngroups
+            lead_folded = True
+
+            def run_set(op_set, skipnan, means=None, target=None):
+                return grouped_partials(
+                    op_set, vals, labels, ngroups, skipnan=skipnan,
+                    means=means, target=target, row_offset=0,
+                )
+
     if func in ("quantile", "nanquantile", "median", "nanmedian"):
         from .aggregate_hip import grouped_quantile
 
-        if lead_M != 1:
+        if lead_M != 1 and not lead_folded:
             raise NotImplementedError(f"{func} with leading array dims: next row")
         if func in ("quantile", "nanquantile"):
             if not finalize_kwargs or "q" not in finalize_kwargs:
@@ -453,7 +489,7 @@ def groupby_reduce(
     if func in ("mode", "nanmode"):
         from .aggregate_hip import grouped_mode
 
-        if lead_M != 1:
+        if lead_M != 1 and not lead_folded:
             raise NotImplementedError(f"{func} with leading array dims: next row")
         if dist_on:
             raise NotImplementedError("distributed mode needs a global sort: next row")
@@ -663,6 +699,14 @@ def groupby_reduce(
         else:  # pragma: no cover
             raise NotImplementedError(func)
 
+    if lead_folded and func in ("argmax", "argmin", "nanargmax", "nanargmin"):
+        # composite-group indices are flat (lead*N + t); the API returns the
+        # index within the reduced trailing axes (t), like np.argmax(axis=-1)
+        offs = (
+            torch.arange(lead_M, device=device, dtype=torch.int64) * arg_localize_N
+        ).repeat_interleave(base_ngroups)
+        result = result - offs  # empty slots are overwritten by the fill below
+
     # --- finalize: masking + fills + final dtype (reference core.py:410-475) ---
     # fills are applied with an unconditional where (no host sync); the one
     # case that must inspect the mask on the host is a NaN fill on an
@@ -707,7 +751,7 @@ def groupby_reduce(
             )
 
     result = result.to(t_out_dtype)
-    if lead_M > 1:
+    if lead_M > 1 and not lead_folded:
         # column partials are (ngroups, M) group-major; the API result puts
         # the group dims last
         result = result.reshape(ngroups, lead_M).t().contiguous()

@@ -46,8 +46,9 @@ def groupby_scan(
     by_shape = bys[0].shape
     if tuple(arr.shape[arr.ndim - len(by_shape):]) != tuple(by_shape):
         raise ValueError("by must align with trailing dims of array")
-    if arr.ndim != bys[0].ndim:
-        raise NotImplementedError("scan with leading array dims: next row")
+    lead_M = 1
+    for d in arr.shape[: arr.ndim - len(by_shape)]:
+        lead_M *= d
     if axis is not None:
         ax = axis if isinstance(axis, (tuple, list)) else (axis,)
         ax = tuple(a % arr.ndim for a in ax)
@@ -71,6 +72,26 @@ def groupby_scan(
     for f in facs:
         ngroups *= f.ngroups
     labels, labels2, grp_pair = _combined_codes(facs)
+
+    if lead_M > 1:
+        # leading array dims: fold the lead index into the group codes
+        # (lead*(ngroups+1) + code, +1 so every column keeps its own
+        # NaN-sentinel group) and scan the C-order flattened stream — the
+        # offset-labels trick of reference factorize.py:24-39 applied to scans
+        if labels2 is not None:
+            g0, g1 = grp_pair
+            c0, c1 = labels.to(torch.int64), labels2.to(torch.int64)
+            bad = (c0 < 0) | (c0 >= g0) | (c1 < 0) | (c1 >= g1)
+            base = torch.where(bad, torch.full_like(c0, -1), c0 * g1 + c1)
+        else:
+            base = labels.to(torch.int64)
+        ngs = ngroups + 1
+        valid = (base >= 0) & (base < ngroups)
+        col = torch.where(valid, base, torch.full_like(base, ngroups))
+        lead_idx = torch.arange(lead_M, device=device, dtype=torch.int64)
+        labels = (lead_idx[:, None] * ngs + col[None, :]).reshape(-1)
+        labels2, grp_pair = None, None
+        ngroups = lead_M * ngs
 
     lib = _ffi.load_library()
     vals = vals.contiguous()

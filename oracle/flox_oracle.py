@@ -631,6 +631,12 @@ def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
     codes[codes < 0] = ngroups  # sentinel group scans together, like factorize_:201-210
 
     flat = array.reshape(-1)
+    if flat.size != codes.size:
+        # leading array dims: each column scans independently — fold the lead
+        # index into the codes (stride ngroups+1 keeps per-column sentinels)
+        lead_M = flat.size // codes.size
+        ngs = ngroups + 1
+        codes = ((np.arange(lead_M) * ngs)[:, None] + codes[None, :]).reshape(-1)
     if func in ("cumsum", "nancumsum") and flat.dtype.kind in "iub" and flat.dtype.itemsize < 8:
         flat = flat.astype(np.int64)
     perm = np.argsort(codes, kind="stable")

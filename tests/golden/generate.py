@@ -239,6 +239,19 @@ def gen_cases():
         expected_groups=np.arange(10),
         fill_value=-1.0,
     )
+    # order-dependent reductions with leading array dims (the lead-fold path)
+    al = rng.standard_normal((3, 4, 60))
+    al[rng.random(al.shape) < 0.2] = np.nan
+    bl = rng.integers(0, 5, 60)
+    yield "median_lead", dict(array=al, by=bl, func="median", expected_groups=np.arange(5))
+    yield "nanquantile_lead", dict(
+        array=al, by=bl, func="nanquantile", expected_groups=np.arange(5),
+        finalize_kwargs={"q": 0.7},
+    )
+    yield "quantile_lead_vec", dict(
+        array=al, by=bl, func="quantile", expected_groups=np.arange(5),
+        finalize_kwargs={"q": [0.25, 0.75]},
+    )
     # size-1 by dims broadcast against the array's trailing dims (the
     # dim=... case, reference core.py:300-309)
     ab = rng.standard_normal((30, 40))
@@ -269,6 +282,12 @@ def gen_scan_cases():
     nby = by.astype(float)
     nby[rng.random(200) < 0.1] = np.nan
     yield "scan_nancumsum_nanby", dict(array=vals, by=nby, func="nancumsum")
+    # leading array dims: each column scans independently along the last axis
+    a2 = rng.standard_normal((4, 120))
+    a2[rng.random(a2.shape) < 0.25] = np.nan
+    b2 = rng.integers(0, 6, 120)
+    for func in ["cumsum", "ffill", "bfill"]:
+        yield f"scan_{func}_lead", dict(array=a2, by=b2, func=func)
 
 
 def main():

@@ -383,3 +383,44 @@ def test_packed_arg_partition_scale():
         want, *_ = oracle_reduce(v, labels, func=func, expected_groups=np.arange(ng))
         got, *_ = flox_amd.groupby_reduce(v, labels, func=func, expected_groups=range(ng))
         np.testing.assert_array_equal(np.asarray(got), want, err_msg=func)
+
+
+@pytest.mark.parametrize("func", [
+    "argmax", "nanargmin", "first", "nanlast", "median", "nanquantile",
+    "mode", "quantile",
+])
+def test_lead_fold_order_funcs(func):
+    """Order-dependent reductions with leading array dims (lead index folded
+    into the group codes) vs the oracle."""
+    rng = np.random.default_rng(91)
+    arr = rng.standard_normal((3, 4, 200))
+    arr[rng.random(arr.shape) < 0.2] = np.nan
+    labels = rng.integers(0, 7, 200)
+    eg = np.arange(9)  # two trailing empty groups
+    kw = {}
+    if func in ("quantile", "nanquantile"):
+        kw["finalize_kwargs"] = {"q": [0.3, 0.9] if func == "quantile" else 0.7}
+    want, *_ = oracle_reduce(arr, labels, func=func, expected_groups=eg, **kw)
+    got, *_ = flox_amd.groupby_reduce(arr, labels, func=func, expected_groups=eg, **kw)
+    assert np.asarray(got).shape == want.shape
+    if func.endswith(("argmax", "argmin")):
+        np.testing.assert_array_equal(np.asarray(got), want)
+    else:
+        np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
+                                   rtol=1e-6, atol=1e-6)
+
+
+@pytest.mark.parametrize("func", ["cumsum", "nancumsum", "ffill", "bfill"])
+def test_lead_dims_scan(func):
+    """Grouped scans with leading array dims vs the oracle."""
+    from oracle import groupby_scan as oracle_scan
+
+    rng = np.random.default_rng(92)
+    arr = rng.standard_normal((5, 300))
+    arr[rng.random(arr.shape) < 0.3] = np.nan
+    labels = rng.integers(0, 6, 300)
+    eg = np.arange(6)
+    want = oracle_scan(arr, labels, func=func, expected_groups=eg)
+    got = flox_amd.groupby_scan(arr, labels, func=func, expected_groups=eg)
+    np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
+                               rtol=1e-9, atol=1e-9)
