@@ -444,6 +444,11 @@ def groupby_reduce(
                     means=means, target=target, row_offset=0,
                 )
 
+    if func in ("quantile", "nanquantile", "median", "nanmedian", "mode", "nanmode") and (
+        not sort and not provided_expected
+    ):
+        raise NotImplementedError(f"sort=False with {func}: next row")
+
     if func in ("quantile", "nanquantile", "median", "nanmedian"):
         from .aggregate_hip import grouped_quantile
 
@@ -767,8 +772,6 @@ def groupby_reduce(
         # groups in first-appearance order (reference pd.factorize(sort=False),
         # factorize.py:96): find each group's first row with an index-min pass
         # and permute the result bins — rows need no relabeling
-        if lead_M > 1:
-            raise NotImplementedError("sort=False with leading array dims: next row")
         for ax_i, f in enumerate(facs):
             pidx = grouped_partials(
                 _ffi.SET_IDXMIN, f.codes.to(torch.int64), f.codes, f.ngroups

@@ -424,3 +424,14 @@ def test_lead_dims_scan(func):
     got = flox_amd.groupby_scan(arr, labels, func=func, expected_groups=eg)
     np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
                                rtol=1e-9, atol=1e-9)
+
+
+def test_nosort_lead_dims():
+    """sort=False (first-appearance group order) with leading array dims."""
+    rng = np.random.default_rng(17)
+    arr = rng.standard_normal((2, 3, 500))
+    labels = rng.choice([30, 5, 17, 2, 44, 9], 500)
+    want, wgrp = oracle_reduce(arr, labels, func="mean", sort=False)
+    got, ggrp = flox_amd.groupby_reduce(arr, labels, func="mean", sort=False)
+    np.testing.assert_array_equal(np.asarray(ggrp), wgrp)
+    np.testing.assert_allclose(np.asarray(got), want, rtol=1e-12, atol=1e-12)
