@@ -1118,13 +1118,12 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
   }
   __syncthreads();
 
-  for (int64_t i = start + tid; i < end; i += blockDim.x) {
-    const PairT<V> p = pairs[i];
+  auto body = [&](const PairT<V> p) {
     const uint32_t lc = p.lc;
     const V v = p.v;
     const bool vnan = TR::isnan_(v);
     if (OPS & B_PRESENT) s_present[lc] = 1u;
-    if (vnan && skipnan) continue;
+    if (vnan && skipnan) return;
     if (OPS & B_SUM) acc_add(&s_sum[lc], (Acc)v);
     if (IS_PROD) acc_mul(&s_sum[lc], (Acc)v);
     if (OPS & B_SSD) {
@@ -1142,6 +1141,25 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
         if (OPS & B_MAX) enc_max(&s_mm[lc], TR::enc(v));
       }
     }
+  };
+  if constexpr (sizeof(PairT<V>) == 8) {
+    /* two pairs per thread per trip (one 16-B load): doubles the per-wave
+     * outstanding-load count — this kernel is latency-parked, not
+     * bandwidth-bound (SQ_WAIT_ANY ~88% of WAVE_CYCLES before this) */
+    struct Pair2 { PairT<V> a, b; };
+    const int64_t s2 = (start + 1) & ~1LL;
+    if (start < s2 && start < end && tid == 0) body(pairs[start]);
+    const int64_t nv = (end - s2) / 2;
+    const Pair2* __restrict__ vp = (const Pair2*)(pairs + s2);
+    for (int64_t j = tid; j < nv; j += blockDim.x) {
+      const Pair2 q = vp[j];
+      body(q.a);
+      body(q.b);
+    }
+    const int64_t rem = s2 + nv * 2;
+    if (rem < end && tid == 0) body(pairs[rem]);
+  } else {
+    for (int64_t i = start + tid; i < end; i += blockDim.x) body(pairs[i]);
   }
   __syncthreads();
   /* flush into the final bins (out_min/out_max hold ENCODED values until
