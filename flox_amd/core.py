@@ -229,8 +229,12 @@ def groupby_reduce(
         if arr_np.dtype.kind in "Mm":
             dt_dtype = arr_np.dtype
             array = arr_np.view("i8")
-            if np.issubdtype(dt_dtype, np.datetime64) and (array == np.iinfo(np.int64).min).any():
-                raise NotImplementedError("NaT values: next row")
+            # NaT (= INT64_MIN) passes through the int64 view untouched, as
+            # in the reference (core.py:994-997): it wins min and loses max;
+            # count and nanfirst/nanlast skip NaT rows (handled below);
+            # empty groups fill with NaT (xrdtypes.py:54-61)
+            if func in xrdtypes.PRESERVES_DTYPE and fill_value is None:
+                fill_value = np.iinfo(np.int64).min
     arr = _as_device_tensor(array, device)
     was_bool = arr.dtype == torch.bool
     if was_bool:
@@ -320,6 +324,15 @@ def groupby_reduce(
     grp_shape = tuple(f.ngroups for f in facs)
     ngroups = math.prod(grp_shape)
     labels, labels2, grp_pair = _combined_codes(facs)
+
+    if dt_dtype is not None and func in ("count", "nanfirst", "nanlast"):
+        # NaT rows are missing for count (verified reference behavior) and
+        # for nanfirst/nanlast (xrutils.nanfirst's isnull, xrutils.py:389-397):
+        # drop them by invalidating their codes
+        if lead_M != 1:
+            raise NotImplementedError("datetime NaT skipping with leading dims: next row")
+        natm = arr.reshape(-1) == torch.iinfo(torch.int64).min
+        labels = torch.where(natm, torch.full_like(labels, -1), labels)
 
     if subset_keep_shape is not None:
         # offset the codes by the kept-dims slice index (reference
@@ -488,7 +501,10 @@ def groupby_reduce(
         result = result.reshape(new_shape)
         groups = tuple(f.groups for f in facs)
         if return_numpy:
-            return (result.cpu().numpy(), *groups)
+            out_np = result.cpu().numpy()
+            if dt_dtype is not None:
+                out_np = out_np.astype(dt_dtype)  # reference core.py:1209-1211
+            return (out_np, *groups)
         return (result, *groups)
 
     if func in ("mode", "nanmode"):
@@ -782,7 +798,10 @@ def groupby_reduce(
     groups = tuple(groups_list)
     if return_numpy:
         out_np = result.cpu().numpy()
-        if dt_dtype is not None and func in xrdtypes.PRESERVES_DTYPE:
-            out_np = out_np.view(dt_dtype)
+        if dt_dtype is not None and func not in ("count", "any", "all"):
+            # the reference casts every non-count result back to the datetime
+            # dtype (core.py:1209-1211): int64 counts reinterpret, float
+            # results truncate, NaN -> NaT
+            out_np = out_np.astype(dt_dtype)
         return (out_np, *groups)
     return (result, *groups)

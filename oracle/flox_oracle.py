@@ -214,6 +214,11 @@ def groupby_reduce(
     if array.dtype.kind in "Mm":
         dt_dtype = array.dtype
         array = array.view("i8")
+        # NaT = INT64_MIN passes through (reference core.py:994-997); empty
+        # groups fill with NaT (xrdtypes.py:54-61); count and nanfirst/nanlast
+        # skip NaT rows (verified reference behavior / xrutils.py:389-397)
+        if func in _PRESERVES_DTYPE and fill_value is None:
+            fill_value = np.iinfo(np.int64).min
     bys = tuple(np.asarray(b) for b in by)
     nby = len(bys)
     if nby == 0:
@@ -311,6 +316,13 @@ def groupby_reduce(
     grp_shape = tuple(len(f) for f in found)
     ngroups = math.prod(grp_shape)
     codes = _ravel_codes(codes_list, grp_shape) if nby > 1 else codes_list[0]
+
+    if dt_dtype is not None and func in ("count", "nanfirst", "nanlast"):
+        # NaT rows are missing for these (see datetime note above)
+        natm = array.reshape(-1) == np.iinfo(np.int64).min
+        if natm.size != codes.size:
+            raise NotImplementedError("datetime NaT skipping with leading dims")
+        codes = np.where(natm, -1, codes)
 
     # --- min_count defaulting (reference core.py:1026-1038) ---
     out_dtype = _final_dtype(func, array.dtype, dtype)
@@ -493,7 +505,10 @@ def groupby_reduce(
             mask = counts_q < min_count_
             result = np.where(np.broadcast_to(mask, result.shape), fill_value, result)
         newshape = (() if scalar_q else (len(q_arr),)) + lead_shape + grp_shape
-        return (result.reshape(newshape), *found)
+        result = result.reshape(newshape)
+        if dt_dtype is not None:
+            result = result.astype(dt_dtype)  # reference core.py:1209-1211
+        return (result, *found)
     elif func in _ARG_FUNCS or func in _POS_FUNCS:
         order = np.argsort(codes, kind="stable")
         sc = codes[order]
@@ -593,8 +608,11 @@ def groupby_reduce(
             result = np.where(empty_mask, fv, result)
 
     result = np.asarray(result).astype(out_dtype, copy=False)
-    if dt_dtype is not None and func in _PRESERVES_DTYPE:
-        result = result.view(dt_dtype)
+    if dt_dtype is not None and func not in ("count", "any", "all"):
+        # the reference casts every non-count result back to the datetime
+        # dtype (core.py:1209-1211) — astype reinterprets int64 counts and
+        # truncates float results (NaN -> NaT)
+        result = result.astype(dt_dtype)
 
     result = result.reshape(lead_shape + grp_shape)
     if not sort and not provided_expected:

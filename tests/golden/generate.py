@@ -199,6 +199,13 @@ def gen_cases():
     tby = rng.integers(0, 5, 60)
     for func in ["min", "max", "count", "first", "nanmax", "last"]:
         yield f"{func}_datetime", dict(array=tvals, by=tby, func=func, expected_groups=np.arange(5))
+    # NaT values (int64 min through the view) + an empty trailing group
+    tnat = tvals.copy()
+    tnat[rng.random(60) < 0.25] = np.datetime64("NaT")
+    for func in ["min", "max", "nanmin", "nanmax", "count", "median"]:
+        yield f"{func}_datetime_nat", dict(
+            array=tnat, by=tby, func=func, expected_groups=np.arange(6)
+        )
     # pd.IntervalIndex expected_groups (binning without isbin=True)
     import pandas as pd
     iv_vals = rng.standard_normal(300)

@@ -81,4 +81,7 @@ def tolerance_for(name, result_dtype):
         return dict(rtol=2e-6, atol=1e-7)
     if "var" in name or "std" in name:
         return dict(rtol=1e-12, atol=1e-14)
+    if "quantile" in name or "median" in name:
+        # the lerp t*(b-a)+a vs (1-t)*a+t*b forms differ by ~1 ulp
+        return dict(rtol=1e-12, atol=1e-14)
     return dict(rtol=1e-13, atol=1e-16)
