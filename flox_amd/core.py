@@ -271,16 +271,14 @@ def groupby_reduce(
             # by-dims into the group codes below
             if not set(ax) <= set(trailing):
                 raise ValueError(f"axis {axis} must address dims of by")
-            if arr.ndim != bys[0].ndim:
-                raise NotImplementedError(
-                    "axis subset with extra leading array dims: next row"
-                )
-            keep = [d for d in range(arr.ndim) if d not in ax]
+            nlead_s = arr.ndim - len(by_shape)
+            keep = [d for d in range(arr.ndim) if d not in ax]  # lead + kept by dims
             perm = keep + sorted(ax)
             arr = arr.permute(perm).contiguous()
-            bys = tuple(b.permute(perm).contiguous() for b in bys)
-            by_shape = bys[0].shape
-            subset_keep_shape = tuple(arr.shape[: len(keep)])
+            by_perm = [d - nlead_s for d in perm if d >= nlead_s]
+            bys = tuple(b.permute(by_perm).contiguous() for b in bys)
+            by_shape = bys[0].shape  # (kept by dims..., reduced dims...)
+            subset_keep_shape = tuple(arr.shape[nlead_s : len(keep)])
     lead_shape = tuple(arr.shape[: arr.ndim - len(by_shape)])
     lead_M = math.prod(lead_shape) if lead_shape else 1
 
@@ -777,7 +775,7 @@ def groupby_reduce(
         # the group dims last
         result = result.reshape(ngroups, lead_M).t().contiguous()
     if subset_keep_shape is not None:
-        result = result.reshape(subset_keep_shape + grp_shape)
+        result = result.reshape(lead_shape + subset_keep_shape + grp_shape)
     else:
         result = result.reshape(lead_shape + grp_shape)
 

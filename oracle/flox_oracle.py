@@ -260,11 +260,13 @@ def groupby_reduce(
             # factorize.py:24-39 + core.py:1026-1032: min_count forced >= 1):
             # restate as one full reduction per leading slice against the
             # globally-found groups
-            assert set(ax) <= set(trailing) and array.ndim == by_ndim
-            keep = [d for d in range(array.ndim) if d not in ax]
+            assert set(ax) <= set(trailing)
+            nlead = array.ndim - by_ndim
+            keep = [d for d in range(array.ndim) if d not in ax]  # lead + kept by
             perm = keep + sorted(ax)
             arr_t = np.transpose(array, perm)
-            bys_t = [np.transpose(b, perm) for b in bys]
+            by_perm = [d - nlead for d in perm if d >= nlead]
+            bys_t = [np.transpose(b, by_perm) for b in bys]
             kshape = arr_t.shape[: len(keep)]
             if expected_groups is not None and not isinstance(expected_groups, tuple):
                 expected_groups = (expected_groups,)
@@ -279,7 +281,7 @@ def groupby_reduce(
             for idx in np.ndindex(*kshape):
                 r, *g = groupby_reduce(
                     arr_t[idx],
-                    *[b[idx] for b in bys_t],
+                    *[b[idx[nlead:]] for b in bys_t],
                     func=func,
                     expected_groups=expected_groups,
                     fill_value=fill_value,

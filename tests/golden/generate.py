@@ -227,6 +227,14 @@ def gen_cases():
         array=a3, by=b3, func="mean", axis=(1, 2),
         expected_groups=np.arange(6), fill_value=-99.0,
     )
+    # axis subset with extra leading array dims (array.ndim > by.ndim)
+    a4 = rng.standard_normal((2, 4, 5, 30))
+    b4 = rng.integers(0, 6, (4, 5, 30))
+    for func, axs in [("mean", (3,)), ("sum", (2, 3)), ("nanmax", (3,))]:
+        yield f"{func}_lead_axis_subset{len(axs)}", dict(
+            array=a4, by=b4, func=func, axis=axs,
+            expected_groups=np.arange(6), fill_value=-99.0,
+        )
     # sort=False: groups in first-appearance order (factorize.py:96)
     ub = np.array([30, 5, 30, 17, 5, 2, 17, 30, 2, 9])
     uv = rng.standard_normal(10)

@@ -435,3 +435,21 @@ def test_nosort_lead_dims():
     got, ggrp = flox_amd.groupby_reduce(arr, labels, func="mean", sort=False)
     np.testing.assert_array_equal(np.asarray(ggrp), wgrp)
     np.testing.assert_allclose(np.asarray(got), want, rtol=1e-12, atol=1e-12)
+
+
+@pytest.mark.parametrize("axs", [(3,), (2, 3)])
+@pytest.mark.parametrize("func", ["mean", "sum", "var", "count"])
+def test_lead_axis_subset(func, axs):
+    """Axis subset of by's dims with extra leading array dims: offset codes
+    carry the kept by-dims, the column path carries the lead dims."""
+    rng = np.random.default_rng(zlib.crc32(f"las-{func}-{axs}".encode()))
+    arr = rng.standard_normal((2, 4, 5, 40))
+    arr[rng.random(arr.shape) < 0.05] = np.nan
+    by = rng.integers(0, 6, (4, 5, 40))
+    want, *_ = oracle_reduce(arr, by, func=func, axis=axs,
+                             expected_groups=np.arange(6), fill_value=-9.0)
+    got, *_ = flox_amd.groupby_reduce(arr, by, func=func, axis=axs,
+                                      expected_groups=np.arange(6), fill_value=-9.0)
+    assert np.asarray(got).shape == want.shape
+    np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
+                               rtol=1e-9, atol=1e-9)
