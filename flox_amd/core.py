@@ -473,10 +473,26 @@ def groupby_reduce(
             q = 0.5
         q_arr = np.atleast_1d(np.asarray(q, dtype=np.float64))
         scalar_q = np.isscalar(q) or np.ndim(q) == 0
-        resq = grouped_quantile(
-            vals, labels, ngroups, q_arr, skipnan=agg.skipnan,
-            labels2=labels2, grp_shape=grp_pair,
-        )
+        if dist_on:
+            # exact cross-rank quantiles by radix selection over grouped
+            # counts — histograms cross the wire, values never do
+            from .dist_quantile import distributed_grouped_quantile
+
+            if labels2 is not None:
+                g0d, g1d = grp_pair
+                cd0, cd1 = labels.to(torch.int64), labels2.to(torch.int64)
+                badd = (cd0 < 0) | (cd0 >= g0d) | (cd1 < 0) | (cd1 >= g1d)
+                codes_d = torch.where(badd, torch.full_like(cd0, -1), cd0 * g1d + cd1)
+            else:
+                codes_d = labels.to(torch.int64)
+            resq = distributed_grouped_quantile(
+                vals, codes_d, ngroups, q_arr, skipnan=agg.skipnan
+            )
+        else:
+            resq = grouped_quantile(
+                vals, labels, ngroups, q_arr, skipnan=agg.skipnan,
+                labels2=labels2, grp_shape=grp_pair,
+            )
         result = resq[0] if scalar_q else resq
         if min_count_ > 0:
             pc = grouped_partials(
@@ -488,8 +504,6 @@ def groupby_reduce(
             counts_for_mask = pc["count"]
         else:
             counts_for_mask = None
-        if dist_on:
-            raise NotImplementedError("distributed quantiles need a global sort: next row")
         t_out = _torch_dtype(out_dtype)
         if min_count_ > 0 and fill_value is not None:
             mask = counts_for_mask < min_count_

@@ -243,6 +243,71 @@ def test_gloo_scan_carry_exchange():
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
 
 
+def _worker_quantile(rank, world, port, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import numpy as np
+        import torch
+        from flox_amd.dist_quantile import distributed_grouped_quantile
+        from oracle import groupby_reduce as oracle_reduce
+
+        def np_count(values_i64, labels, nbins):
+            l = labels.numpy()
+            return torch.tensor(np.bincount(l[l >= 0], minlength=nbins), dtype=torch.int64)
+
+        rng = np.random.default_rng(13)
+        n, ng = 30_000, 41
+        for dtype, skipnan in [("float32", True), ("float64", True), ("float64", False), ("int64", True)]:
+            if np.dtype(dtype).kind == "f":
+                vals = (rng.standard_normal(n) * 50).astype(dtype)
+                vals[rng.random(n) < 0.08] = np.nan
+            else:
+                vals = rng.integers(-(2**45), 2**45, n).astype(dtype)
+            labels = rng.integers(0, ng, n)
+            q = np.array([0.1, 0.5, 0.95])
+            sl = slice(rank * n // world, (rank + 1) * n // world)
+            got = distributed_grouped_quantile(
+                torch.from_numpy(np.ascontiguousarray(vals[sl])),
+                torch.from_numpy(np.ascontiguousarray(labels[sl])),
+                ng, q, skipnan=skipnan, count_fn=np_count,
+            )
+            func = "nanquantile" if skipnan else "quantile"
+            want, *_ = oracle_reduce(vals, labels, func=func,
+                                     expected_groups=np.arange(ng),
+                                     finalize_kwargs={"q": list(q)})
+            # the selection is exact in f64; the oracle lerps f32 inputs in
+            # f32 (np.quantile preserves float dtype) — compare at the
+            # output dtype's precision
+            tol = dict(rtol=3e-6, atol=1e-5) if want.dtype.itemsize == 4 else dict(rtol=1e-12, atol=1e-12)
+            np.testing.assert_allclose(got.numpy().astype(want.dtype), want,
+                                       equal_nan=True, err_msg=f"{dtype} skipnan={skipnan}", **tol)
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}\n{traceback.format_exc()}")
+        raise
+
+
+def test_gloo_distributed_quantile():
+    """Exact cross-rank quantiles at world_size 2: sharded values, radix
+    selection with a numpy count stub (the HIP COUNT kernel is exercised by
+    the world-1 GPU test) — must equal the whole-data oracle bit-for-bit."""
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_quantile, args=(r, 2, 29523, fail_q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+    errs = []
+    while not fail_q.empty():
+        errs.append(fail_q.get())
+    assert not errs, errs[0]
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
 def test_gloo_full_groupby_reduce_branch():
     """The exact distributed code in core.groupby_reduce (partial all-reduce,
     global-mean var, finalize) at world_size 2, with the kernel layer stubbed

@@ -453,3 +453,35 @@ def test_lead_axis_subset(func, axs):
     assert np.asarray(got).shape == want.shape
     np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
                                rtol=1e-9, atol=1e-9)
+
+
+@pytest.mark.parametrize("dtype", ["float32", "float64", "int32", "int64"])
+@pytest.mark.parametrize("skipnan", [False, True])
+def test_radix_select_quantile_world1(dtype, skipnan):
+    """The distributed quantile machinery (radix selection over grouped
+    counts) at world_size 1 on the real COUNT kernel, vs the oracle."""
+    from flox_amd.dist_quantile import distributed_grouped_quantile
+
+    rng = np.random.default_rng(zlib.crc32(f"rs-{dtype}-{skipnan}".encode()))
+    n, ng = 80_000, 157
+    labels = rng.integers(0, ng, n)
+    if np.dtype(dtype).kind == "f":
+        vals = (rng.standard_normal(n) * 100).astype(dtype)
+        vals[rng.random(n) < 0.1] = np.nan
+    else:
+        vals = rng.integers(-(2**40) if dtype == "int64" else -1000,
+                            2**40 if dtype == "int64" else 1000, n).astype(dtype)
+    q = np.array([0.0, 0.25, 0.5, 0.9, 1.0])
+    func = "nanquantile" if skipnan else "quantile"
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng),
+                             finalize_kwargs={"q": list(q)})
+    got = distributed_grouped_quantile(
+        torch.tensor(vals, device="cuda"),
+        torch.tensor(labels, device="cuda"),
+        ng, q, skipnan=skipnan,
+    )
+    # the selection is exact in f64; f32 oracle results are f32-rounded
+    # (np.quantile preserves float dtype) — compare at the output precision
+    tol = dict(rtol=3e-6, atol=1e-5) if want.dtype.itemsize == 4 else dict(rtol=1e-12, atol=1e-12)
+    np.testing.assert_allclose(got.cpu().numpy().astype(want.dtype), want,
+                               equal_nan=True, **tol)
