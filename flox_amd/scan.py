@@ -39,6 +39,15 @@ def groupby_scan(
         raise RuntimeError("flox_amd.groupby_scan requires a GPU (engine='hip')")
     device = torch.device("cuda", torch.cuda.current_device())
     return_numpy = not isinstance(array, torch.Tensor)
+    # datetime64/timedelta64 scan on the int64 view, NaT (= int64 min)
+    # passing through as a plain value — the reference's behavior (its ffill
+    # isnull sees no missing values on the int64 view)
+    dt_dtype = None
+    if return_numpy:
+        arr_np0 = np.asarray(array)
+        if arr_np0.dtype.kind in "Mm":
+            dt_dtype = arr_np0.dtype
+            array = arr_np0.view("i8")
     arr = _as_device_tensor(array, device)
     bys = tuple(_as_device_tensor(b, device) for b in by)
     if len(bys) == 0:
@@ -139,5 +148,8 @@ def groupby_scan(
         out = out.to(td)
     out = out.reshape(orig_shape)
     if return_numpy:
-        return out.cpu().numpy()
+        out_np = out.cpu().numpy()
+        if dt_dtype is not None:
+            out_np = out_np.astype(dt_dtype)  # int64 counts reinterpret
+        return out_np
     return out

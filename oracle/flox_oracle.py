@@ -634,6 +634,12 @@ def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
     outside expected_groups form their own trailing group (the reference's
     NaN-sentinel group)."""
     array = np.asarray(array)
+    dt_dtype = None
+    if array.dtype.kind in "Mm":
+        # int64 view; NaT passes through as a plain value (the reference's
+        # ffill isnull sees no missing values on the int64 view)
+        dt_dtype = array.dtype
+        array = array.view("i8")
     bys = tuple(np.asarray(b) for b in by)
     if expected_groups is not None and not isinstance(expected_groups, tuple):
         expected_groups = (expected_groups,)
@@ -690,6 +696,8 @@ def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
             raise NotImplementedError(func)
     inv = np.argsort(perm, kind="stable")
     out = out_sorted[inv]
+    if dt_dtype is not None:
+        out = out.astype(dt_dtype)
     if dtype is not None:
         out = out.astype(dtype)
     return out.reshape(array.shape)

@@ -297,6 +297,12 @@ def gen_scan_cases():
     nby = by.astype(float)
     nby[rng.random(200) < 0.1] = np.nan
     yield "scan_nancumsum_nanby", dict(array=vals, by=nby, func="nancumsum")
+    # datetime values: int64 view, NaT passes through as a plain value
+    tv = (np.datetime64("2021-06-01") + rng.integers(0, 5000, 120).astype("timedelta64[m]"))
+    tv[rng.random(120) < 0.2] = np.datetime64("NaT")
+    tb = rng.integers(0, 5, 120)
+    for func in ["ffill", "bfill", "cumsum"]:
+        yield f"scan_{func}_datetime", dict(array=tv, by=tb, func=func)
     # leading array dims: each column scans independently along the last axis
     a2 = rng.standard_normal((4, 120))
     a2[rng.random(a2.shape) < 0.25] = np.nan
