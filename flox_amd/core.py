@@ -279,6 +279,9 @@ def groupby_reduce(
             bys = tuple(b.permute(by_perm).contiguous() for b in bys)
             by_shape = bys[0].shape  # (kept by dims..., reduced dims...)
             subset_keep_shape = tuple(arr.shape[nlead_s : len(keep)])
+            # for sort=False: first-appearance order is defined on the
+            # ORIGINAL by layout — keep the inverse permutation
+            subset_by_inv = tuple(int(i) for i in np.argsort(by_perm))
     lead_shape = tuple(arr.shape[: arr.ndim - len(by_shape)])
     lead_M = math.prod(lead_shape) if lead_shape else 1
 
@@ -455,11 +458,6 @@ def groupby_reduce(
                     means=means, target=target, row_offset=0,
                 )
 
-    if func in ("quantile", "nanquantile", "median", "nanmedian", "mode", "nanmode") and (
-        not sort and not provided_expected
-    ):
-        raise NotImplementedError(f"sort=False with {func}: next row")
-
     if func in ("quantile", "nanquantile", "median", "nanmedian"):
         from .aggregate_hip import grouped_quantile
 
@@ -511,7 +509,17 @@ def groupby_reduce(
         result = result.to(t_out)
         new_shape = ((len(q_arr),) if not scalar_q else ()) + lead_shape + grp_shape
         result = result.reshape(new_shape)
-        groups = tuple(f.groups for f in facs)
+        groups_list_q = [f.groups for f in facs]
+        if not sort and not provided_expected:
+            # first-appearance group order (as in the common finalize below)
+            for ax_i, f in enumerate(facs):
+                pidx = grouped_partials(
+                    _ffi.SET_IDXMIN, f.codes.to(torch.int64), f.codes, f.ngroups
+                )
+                order = torch.argsort(pidx["idx"], stable=True)
+                result = torch.index_select(result, result.dim() - len(facs) + ax_i, order)
+                groups_list_q[ax_i] = np.asarray(groups_list_q[ax_i])[order.cpu().numpy()]
+        groups = tuple(groups_list_q)
         if return_numpy:
             out_np = result.cpu().numpy()
             if dt_dtype is not None:
@@ -795,17 +803,22 @@ def groupby_reduce(
 
     groups_list = [f.groups for f in facs]
     if not sort and not provided_expected:
-        if subset_keep_shape is not None:
-            raise NotImplementedError("sort=False with an axis subset: next row")
         # groups in first-appearance order (reference pd.factorize(sort=False),
         # factorize.py:96): find each group's first row with an index-min pass
-        # and permute the result bins — rows need no relabeling
+        # and permute the result bins — rows need no relabeling; the group
+        # dims are always the trailing len(facs) axes of the result
         for ax_i, f in enumerate(facs):
+            fcodes = f.codes
+            if subset_keep_shape is not None:
+                # restore the original by layout before taking first rows
+                fcodes = (
+                    fcodes.reshape(by_shape).permute(subset_by_inv).reshape(-1)
+                ).contiguous()
             pidx = grouped_partials(
-                _ffi.SET_IDXMIN, f.codes.to(torch.int64), f.codes, f.ngroups
+                _ffi.SET_IDXMIN, fcodes.to(torch.int64), fcodes, f.ngroups
             )
             order = torch.argsort(pidx["idx"], stable=True)
-            result = torch.index_select(result, len(lead_shape) + ax_i, order)
+            result = torch.index_select(result, result.dim() - len(facs) + ax_i, order)
             groups_list[ax_i] = np.asarray(groups_list[ax_i])[order.cpu().numpy()]
     groups = tuple(groups_list)
     if return_numpy:

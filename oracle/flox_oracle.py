@@ -270,7 +270,8 @@ def groupby_reduce(
             kshape = arr_t.shape[: len(keep)]
             if expected_groups is not None and not isinstance(expected_groups, tuple):
                 expected_groups = (expected_groups,)
-            if expected_groups is None or all(e is None for e in expected_groups):
+            was_discovered = expected_groups is None or all(e is None for e in expected_groups)
+            if was_discovered:
                 founds = []
                 for b in bys_t:
                     _, f0 = _factorize_single(b.reshape(-1), None)
@@ -294,6 +295,21 @@ def groupby_reduce(
                 slices.append(r)
                 groups_out = g
             out = np.stack(slices).reshape(kshape + slices[0].shape)
+            if not sort and was_discovered:
+                # first-appearance order over the ORIGINAL by layout
+                groups_out = list(groups_out)
+                for ax_i, (b, f) in enumerate(zip(bys, groups_out)):
+                    fl = np.asarray(b).reshape(-1)
+                    fs = np.asarray(f)
+                    srt = np.argsort(fs, kind="stable")
+                    pos = np.searchsorted(fs[srt], fl)
+                    pos = pos.clip(0, len(fs) - 1)
+                    c = np.where(fs[srt][pos] == fl, srt[pos], -1)
+                    first = np.full(len(fs), np.iinfo(np.int64).max)
+                    np.minimum.at(first, c[c >= 0], np.arange(len(c))[c >= 0])
+                    order = np.argsort(first, kind="stable")
+                    groups_out[ax_i] = fs[order]
+                    out = np.take(out, order, axis=out.ndim - len(groups_out) + ax_i)
             return (out, *groups_out)
 
     if expected_groups is not None and not isinstance(expected_groups, tuple):
@@ -510,6 +526,13 @@ def groupby_reduce(
         result = result.reshape(newshape)
         if dt_dtype is not None:
             result = result.astype(dt_dtype)  # reference core.py:1209-1211
+        if not sort and not provided_expected:
+            for ax_i, (c, f) in enumerate(zip(codes_list, found)):
+                first = np.full(len(f), np.iinfo(np.int64).max)
+                np.minimum.at(first, c[c >= 0], np.arange(len(c))[c >= 0])
+                order = np.argsort(first, kind="stable")
+                found[ax_i] = np.asarray(f)[order]
+                result = np.take(result, order, axis=result.ndim - len(found) + ax_i)
         return (result, *found)
     elif func in _ARG_FUNCS or func in _POS_FUNCS:
         order = np.argsort(codes, kind="stable")
@@ -618,13 +641,14 @@ def groupby_reduce(
 
     result = result.reshape(lead_shape + grp_shape)
     if not sort and not provided_expected:
-        # groups in first-appearance order (reference pd.factorize(sort=False))
+        # groups in first-appearance order (reference pd.factorize(sort=False));
+        # the group dims are the trailing len(found) axes
         for ax_i, (c, f) in enumerate(zip(codes_list, found)):
             first = np.full(len(f), np.iinfo(np.int64).max)
             np.minimum.at(first, c[c >= 0], np.arange(len(c))[c >= 0])
             order = np.argsort(first, kind="stable")
             found[ax_i] = np.asarray(f)[order]
-            result = np.take(result, order, axis=len(lead_shape) + ax_i)
+            result = np.take(result, order, axis=result.ndim - len(found) + ax_i)
     return (result, *found)
 
 

@@ -240,6 +240,17 @@ def gen_cases():
     uv = rng.standard_normal(10)
     for func in ["sum", "mean", "count"]:
         yield f"{func}_nosort", dict(array=uv, by=ub, func=func, sort=False)
+    # sort=False with an axis subset (first-appearance order on the
+    # original by layout) and with discovered-group quantiles
+    ub3 = rng.choice([7, 3, 9, 1], (4, 5, 30))
+    yield "mean_nosort_subset", dict(
+        array=rng.standard_normal((4, 5, 30)), by=ub3, func="mean",
+        axis=(1,), sort=False, fill_value=-7.0,
+    )
+    yield "quantile_nosort", dict(
+        array=rng.standard_normal(80), by=rng.choice([7, 3, 9, 1], 80),
+        func="quantile", sort=False, finalize_kwargs={"q": 0.5},
+    )
     # empty groups at the tail of the range
     yield "mean_sparse_groups", dict(
         array=rng.standard_normal(50),

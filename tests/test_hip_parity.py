@@ -485,3 +485,21 @@ def test_radix_select_quantile_world1(dtype, skipnan):
     tol = dict(rtol=3e-6, atol=1e-5) if want.dtype.itemsize == 4 else dict(rtol=1e-12, atol=1e-12)
     np.testing.assert_allclose(got.cpu().numpy().astype(want.dtype), want,
                                equal_nan=True, **tol)
+
+
+def test_nosort_subset_and_mode():
+    """sort=False with axis subsets and with discovered-group mode."""
+    rng = np.random.default_rng(71)
+    arr = rng.standard_normal((4, 5, 60))
+    by = rng.choice([7, 3, 9, 1, 12], (4, 5, 60))
+    want, wg = oracle_reduce(arr, by, func="sum", axis=(1,), sort=False, fill_value=-7.0)
+    got, gg = flox_amd.groupby_reduce(arr, by, func="sum", axis=(1,), sort=False, fill_value=-7.0)
+    np.testing.assert_array_equal(np.asarray(gg), wg)
+    np.testing.assert_allclose(np.asarray(got), want, rtol=1e-12, atol=1e-12)
+
+    v = rng.integers(-5, 5, 400).astype(np.float64)
+    b = rng.choice([7, 3, 9, 1], 400)
+    want, wg = oracle_reduce(v, b, func="mode", sort=False)
+    got, gg = flox_amd.groupby_reduce(v, b, func="mode", sort=False)
+    np.testing.assert_array_equal(np.asarray(gg), wg)
+    np.testing.assert_array_equal(np.asarray(got), want)
