@@ -1098,7 +1098,8 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
   const int ng_here = (int)(((gbase + gpb) <= ngroups) ? gpb : (ngroups - gbase));
 
   extern __shared__ __attribute__((aligned(16))) char smem_rb[];
-  SumT* s_sum = (OPS & (B_SUM | B_SSD | B_PROD)) ? (SumT*)(smem_rb + lay.sum_off) : nullptr;
+  SumT* s_sum = (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX))
+                    ? (SumT*)(smem_rb + lay.sum_off) : nullptr;
   uint32_t* s_cnt = (OPS & B_CNT) ? (uint32_t*)(smem_rb + lay.cnt_off) : nullptr;
   uint32_t* s_present = (OPS & B_PRESENT) ? (uint32_t*)(smem_rb + lay.present_off) : nullptr;
   Enc* s_mm = (OPS & (B_MIN | B_MAX)) ? (Enc*)(smem_rb + lay.minmax_off) : nullptr;
