@@ -123,3 +123,70 @@ def test_single_group_equals_numpy(data):
         want = npf(vals.astype(np.float64 if vals.dtype.kind == "f" else vals.dtype))
         atol = 1e-9 * (float(np.nansum(np.abs(vals.astype("f8")))) + 1)
         np.testing.assert_allclose(float(got[0]), float(want), rtol=1e-6, atol=atol, equal_nan=True)
+
+
+@settings(**SETTINGS)
+@given(data=array_strat, func=st.sampled_from(
+    ["argmin", "argmax", "nanargmin", "nanargmax", "first", "last",
+     "nanfirst", "nanlast"]))
+def test_packed_equals_two_pass(data, func):
+    """The packed-key form and the two-pass (extremum + index-match) form of
+    the order reductions must agree bit-for-bit on any input."""
+    import flox_amd.core as core
+
+    vals, labels, ng = data
+    if vals.dtype == np.int64 and func.startswith(("arg", "nanarg")):
+        vals = vals.astype(np.int32)  # packed args cover f32/i32
+    old = core.PACKED_ARG_THRESHOLD
+    try:
+        core.PACKED_ARG_THRESHOLD = 1
+        a, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+        core.PACKED_ARG_THRESHOLD = 1 << 62
+        b, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    finally:
+        core.PACKED_ARG_THRESHOLD = old
+    np.testing.assert_array_equal(np.asarray(a), np.asarray(b))
+
+
+@settings(**SETTINGS)
+@given(
+    seed=st.integers(0, 2**31 - 1),
+    n=st.integers(1, 3000),
+    natfrac=st.sampled_from([0.0, 0.2, 0.9]),
+    func=st.sampled_from(["min", "max", "nanmin", "nanmax", "count",
+                          "first", "nanlast", "median"]),
+)
+def test_datetime_matches_oracle(seed, n, natfrac, func):
+    rng = np.random.default_rng(seed)
+    vals = np.datetime64("2020-01-01") + rng.integers(0, 10**6, n).astype("timedelta64[s]")
+    vals[rng.random(n) < natfrac] = np.datetime64("NaT")
+    ng = int(rng.integers(1, 20))
+    labels = rng.integers(0, ng, n)
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    assert got.dtype == want.dtype
+    np.testing.assert_array_equal(got, want)
+
+
+@settings(**SETTINGS)
+@given(
+    seed=st.integers(0, 2**31 - 1),
+    m=st.integers(1, 8),
+    n=st.integers(1, 800),
+    func=st.sampled_from(["argmax", "nanargmin", "first", "nanlast", "median", "cumsum", "ffill"]),
+)
+def test_lead_dims_order_funcs_match_oracle(seed, m, n, func):
+    rng = np.random.default_rng(seed)
+    arr = rng.standard_normal((m, n))
+    arr[rng.random(arr.shape) < 0.2] = np.nan
+    ng = int(rng.integers(1, 15))
+    labels = rng.integers(0, ng, n)
+    eg = np.arange(ng)
+    if func in ("cumsum", "ffill"):
+        want = oracle_scan(arr, labels, func=func, expected_groups=eg)
+        got = flox_amd.groupby_scan(arr, labels, func=func, expected_groups=eg)
+    else:
+        want, *_ = oracle_reduce(arr, labels, func=func, expected_groups=eg)
+        got, *_ = flox_amd.groupby_reduce(arr, labels, func=func, expected_groups=eg)
+    atol = 1e-9 * (float(np.nansum(np.abs(arr))) + 1)
+    np.testing.assert_allclose(np.asarray(got), want, equal_nan=True, rtol=1e-9, atol=atol)
