@@ -133,6 +133,25 @@ def gen_cases():
                 dict(array=vals_f8, by=labels_basic, func=func, finalize_kwargs={"ddof": 1}),
             )
 
+    # min_count masking and explicit output dtype
+    vmc = rng.standard_normal(60)
+    vmc[rng.random(60) < 0.5] = np.nan
+    bmc = rng.integers(0, 4, 60)
+    yield "nansum_min_count10", dict(
+        array=vmc, by=bmc, func="nansum", expected_groups=np.arange(6),
+        min_count=10, fill_value=np.nan,
+    )
+    yield "mean_dtype_f32", dict(
+        array=vmc, by=bmc, func="mean", expected_groups=np.arange(4),
+        dtype=np.float32,
+    )
+    # timedelta64 values
+    tdv = rng.integers(0, 1000, 60).astype("timedelta64[s]")
+    for func in ["sum", "min", "max", "count"]:
+        yield f"{func}_timedelta", dict(
+            array=tdv, by=bmc, func=func, expected_groups=np.arange(4)
+        )
+
     # multi-by (2-D groupby, like BASELINE config 5)
     by_a = rng.integers(0, 4, 300)
     by_b = rng.integers(0, 6, 300)
@@ -379,6 +398,10 @@ def main():
             out[f"{name}::axis"] = np.asarray(kw["axis"])
         if kw.get("fill_value") is not None:
             out[f"{name}::fill_value"] = np.asarray(kw["fill_value"])
+        if kw.get("min_count") is not None:
+            out[f"{name}::min_count"] = np.asarray(kw["min_count"])
+        if kw.get("dtype") is not None:
+            out[f"{name}::dtype_s"] = np.asarray(np.dtype(kw["dtype"]).str)
         if kw.get("finalize_kwargs"):
             fk = kw["finalize_kwargs"]
             if "ddof" in fk:

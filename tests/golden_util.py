@@ -45,6 +45,12 @@ def load_golden_cases():
         fv = get("fill_value")
         if fv is not None:
             kw["fill_value"] = fv.item()
+        mc = get("min_count")
+        if mc is not None:
+            kw["min_count"] = int(mc)
+        ds = get("dtype_s")
+        if ds is not None:
+            kw["dtype"] = np.dtype(str(ds))
         ddof = get("ddof")
         if ddof is not None:
             kw["finalize_kwargs"] = {"ddof": int(ddof)}
