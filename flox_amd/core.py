@@ -533,14 +533,28 @@ def groupby_reduce(
         if lead_M != 1 and not lead_folded:
             raise NotImplementedError(f"{func} with leading array dims: next row")
         if dist_on:
-            raise NotImplementedError("distributed mode needs a global sort: next row")
-        result = grouped_mode(
-            vals, labels, ngroups, skipnan=agg.skipnan, labels2=labels2, grp_shape=grp_pair
-        )
+            # exact cross-rank mode: run-length-encoded (group, value, count)
+            # runs are all_gathered and merged on every rank
+            from .dist_quantile import distributed_grouped_mode
+
+            if labels2 is not None:
+                g0m, g1m = grp_pair
+                cm0, cm1 = labels.to(torch.int64), labels2.to(torch.int64)
+                badm = (cm0 < 0) | (cm0 >= g0m) | (cm1 < 0) | (cm1 >= g1m)
+                codes_m = torch.where(badm, torch.full_like(cm0, -1), cm0 * g1m + cm1)
+            else:
+                codes_m = labels.to(torch.int64)
+            result = distributed_grouped_mode(vals, codes_m, ngroups, agg.skipnan)
+        else:
+            result = grouped_mode(
+                vals, labels, ngroups, skipnan=agg.skipnan, labels2=labels2, grp_shape=grp_pair
+            )
         p = grouped_partials(
             _ffi.SET_COUNT, vals, labels, ngroups, skipnan=True,
             labels2=labels2, grp_shape=grp_pair,
         )
+        if dist_on:
+            distributed.all_reduce_(p["count"], "sum")
         counts_for_mask = p["count"]
         # float results carry NaN for empty/propagated groups already;
         # integer results need the empty fill

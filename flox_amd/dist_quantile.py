@@ -184,3 +184,119 @@ def distributed_grouped_quantile(
     nan = torch.tensor(float("nan"), dtype=torch.float64, device=device)
     bad = (n_g == 0) if skipnan else ((n_g == 0) | (nan_g > 0))
     return torch.where(bad[None, :], nan, res)
+
+
+def _enc_full(vals):
+    """Full-width order-preserving integer key per value (int64), with ALL
+    NaNs mapped to one canonical largest key (sorts last, loses count ties
+    — scipy.stats.mode's ordering). Returns (keys, decode-to-vals.dtype)."""
+    if vals.dtype == torch.float32:
+        u = vals.view(torch.int32).to(torch.int64) & _M32
+        enc = torch.where(u & 0x80000000 != 0, (~u) & _M32, u | 0x80000000)
+        enc = torch.where(torch.isnan(vals), torch.full_like(enc, _M32), enc)
+
+        def dec(e):
+            u = torch.where(e & 0x80000000 != 0, e ^ 0x80000000, (~e) & _M32)
+            return u.to(torch.int32).view(torch.float32)
+
+        return enc, dec
+    if vals.dtype == torch.float64:
+        u = vals.view(torch.int64)
+        key = torch.where(u < 0, ~u, u ^ (-(1 << 63))) ^ (-(1 << 63))
+        key = torch.where(torch.isnan(vals), torch.full_like(key, (1 << 63) - 1), key)
+
+        def dec(e):
+            k2 = e ^ (-(1 << 63))
+            u = torch.where(k2 < 0, k2 ^ (-(1 << 63)), ~k2)
+            return u.view(torch.float64)
+
+        return key, dec
+    if vals.dtype in (torch.int32, torch.int64):
+        t = vals.dtype
+
+        def dec(e):
+            return e.to(t)
+
+        return vals.to(torch.int64), dec
+    raise NotImplementedError(f"distributed mode for {vals.dtype}")
+
+
+def _rle(keys, codes):
+    """Run-length encode sorted-(code, key) pairs -> (code, key, count)."""
+    if keys.numel() == 0:
+        z = torch.zeros(0, dtype=torch.int64, device=keys.device)
+        return z, z.clone(), z.clone()
+    sk, si = torch.sort(keys)
+    sc = codes[si]
+    sc, sj = torch.sort(sc, stable=True)
+    sk = sk[sj]
+    new = torch.ones_like(sc, dtype=torch.bool)
+    new[1:] = (sc[1:] != sc[:-1]) | (sk[1:] != sk[:-1])
+    starts = torch.nonzero(new).squeeze(1)
+    ends = torch.cat([starts[1:], torch.tensor([sc.numel()], device=sc.device)])
+    return sc[starts], sk[starts], ends - starts
+
+
+def distributed_grouped_mode(vals, codes, ngroups, skipnan):
+    """Exact grouped mode across ranks: each rank run-length encodes its
+    (group, value) pairs, the compressed runs are all_gathered, and every
+    rank merges counts and picks scipy.stats.mode's answer (most frequent;
+    ties -> smallest value; NaN competes as one value sorting last, so it
+    wins only when strictly most frequent — aggregate_npg.py:185-215).
+    Returns a tensor (ngroups,) in vals.dtype (NaN/0 for empty groups;
+    core's finalize fills integer empties)."""
+    import torch.distributed as dist
+
+    device = vals.device
+    codes = codes.to(torch.int64)
+    ok = (codes >= 0) & (codes < ngroups)
+    if skipnan and vals.is_floating_point():
+        ok &= ~torch.isnan(vals)
+    keys, dec = _enc_full(vals[ok])
+    rc, rk, rn = _rle(keys, codes[ok])
+
+    if distributed.is_active():
+        world = dist.get_world_size()
+        n_local = torch.tensor([rc.numel()], dtype=torch.int64, device=device)
+        sizes = [torch.zeros_like(n_local) for _ in range(world)]
+        dist.all_gather(sizes, n_local)
+        mx = int(max(int(s.item()) for s in sizes))
+        packed = torch.zeros((3, max(mx, 1)), dtype=torch.int64, device=device)
+        packed[0, : rc.numel()] = rc
+        packed[1, : rc.numel()] = rk
+        packed[2, : rc.numel()] = rn
+        gathered = [torch.empty_like(packed) for _ in range(world)]
+        dist.all_gather(gathered, packed)
+        parts = [g[:, : int(s.item())] for g, s in zip(gathered, sizes)]
+        allp = torch.cat(parts, dim=1)
+        # merge duplicate (code, key) runs from different ranks
+        sk, si = torch.sort(allp[1])
+        sc = allp[0][si]
+        sn = allp[2][si]
+        sc, sj = torch.sort(sc, stable=True)
+        sk, sn = sk[sj], sn[sj]
+        new = torch.ones_like(sc, dtype=torch.bool)
+        new[1:] = (sc[1:] != sc[:-1]) | (sk[1:] != sk[:-1])
+        seg = torch.cumsum(new.to(torch.int64), 0) - 1
+        nseg = int(seg[-1].item()) + 1 if sc.numel() else 0
+        cnt = torch.zeros(max(nseg, 1), dtype=torch.int64, device=device)
+        cnt.index_add_(0, seg, sn)
+        starts = torch.nonzero(new).squeeze(1)
+        rc, rk, rn = sc[starts], sk[starts], cnt[:nseg]
+
+    # per group: max count, ties -> first run (runs are value-ascending)
+    nan = float("nan") if vals.is_floating_point() else 0
+    out = torch.full((ngroups,), nan, dtype=vals.dtype, device=device)
+    if rc.numel():
+        maxc = torch.zeros(ngroups, dtype=torch.int64, device=device)
+        maxc.index_reduce_(0, rc, rn, "amax", include_self=False)
+        best = rn == maxc[rc]
+        big = rc.numel() + 1
+        firstpos = torch.full((ngroups,), big, dtype=torch.int64, device=device)
+        pos = torch.arange(rc.numel(), device=device)
+        firstpos.index_reduce_(0, rc[best], pos[best], "amin", include_self=False)
+        has = firstpos < big
+        sel = firstpos.clamp(max=rc.numel() - 1)
+        chosen = dec(rk[sel])
+        out = torch.where(has, chosen.to(vals.dtype), out)
+    return out

@@ -324,3 +324,53 @@ def test_gloo_full_groupby_reduce_branch():
         errs.append(fail_q.get())
     assert not errs, errs[0]
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
+def _worker_mode(rank, world, port, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import numpy as np
+        import torch
+        from flox_amd.dist_quantile import distributed_grouped_mode
+        from oracle import groupby_reduce as oracle_reduce
+
+        rng = np.random.default_rng(29)
+        n, ng = 20_000, 33
+        for dtype, skipnan in [("float64", False), ("float32", True), ("int64", False)]:
+            vals = rng.integers(-10, 10, n).astype(dtype)
+            if np.dtype(dtype).kind == "f":
+                vals[rng.random(n) < 0.15] = np.nan
+            labels = rng.integers(0, ng, n)
+            sl = slice(rank * n // world, (rank + 1) * n // world)
+            got = distributed_grouped_mode(
+                torch.from_numpy(np.ascontiguousarray(vals[sl])),
+                torch.from_numpy(np.ascontiguousarray(labels[sl])),
+                ng, skipnan)
+            func = "nanmode" if skipnan else "mode"
+            want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+            np.testing.assert_array_equal(got.numpy().astype(want.dtype), want,
+                                          err_msg=f"{dtype} skipnan={skipnan}")
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}\n{traceback.format_exc()}")
+        raise
+
+
+def test_gloo_distributed_mode():
+    """Exact cross-rank mode at world_size 2: compressed (group, value,
+    count) runs are exchanged and merged — must equal the whole-data oracle."""
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_mode, args=(r, 2, 29525, fail_q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+    errs = []
+    while not fail_q.empty():
+        errs.append(fail_q.get())
+    assert not errs, errs[0]
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]

@@ -503,3 +503,23 @@ def test_nosort_subset_and_mode():
     got, gg = flox_amd.groupby_reduce(v, b, func="mode", sort=False)
     np.testing.assert_array_equal(np.asarray(gg), wg)
     np.testing.assert_array_equal(np.asarray(got), want)
+
+
+@pytest.mark.parametrize("dtype", ["float32", "float64", "int64"])
+@pytest.mark.parametrize("skipnan", [False, True])
+def test_distributed_mode_world1(dtype, skipnan):
+    """The distributed-mode merge at world_size 1 vs the oracle (scipy.stats.mode)."""
+    from flox_amd.dist_quantile import distributed_grouped_mode
+
+    rng = np.random.default_rng(zlib.crc32(f"dm-{dtype}-{skipnan}".encode()))
+    n, ng = 40_000, 61
+    labels = rng.integers(0, ng, n)
+    vals = rng.integers(-15, 15, n).astype(dtype)
+    if np.dtype(dtype).kind == "f":
+        vals[rng.random(n) < 0.1] = np.nan
+    func = "nanmode" if skipnan else "mode"
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got = distributed_grouped_mode(
+        torch.tensor(vals, device="cuda"), torch.tensor(labels, device="cuda"),
+        ng, skipnan)
+    np.testing.assert_array_equal(got.cpu().numpy().astype(want.dtype), want)
