@@ -123,6 +123,14 @@ def grouped_partials(
             "nanflag": torch.empty(0, dtype=torch.int32, device=dev0),
         }
         return {k: empty[k] for k in _SET_MEMBERS[op_set]}
+    if values.numel() == 0:
+        # zero rows: the kernel rejects null device pointers, so feed one
+        # dummy row with an out-of-range code (dropped by the bounds check);
+        # the init paths then produce the correct empty bins
+        values = torch.zeros(1, dtype=values.dtype, device=values.device)
+        labels = torch.full((1,), -1, dtype=labels.dtype, device=labels.device)
+        if labels2 is not None:
+            labels2 = torch.full((1,), -1, dtype=labels2.dtype, device=labels2.device)
     assert values.ndim == 1 and labels.ndim == 1 and values.numel() == labels.numel()
     values = values.contiguous()
     labels = labels.contiguous()

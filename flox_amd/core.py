@@ -471,7 +471,10 @@ def groupby_reduce(
             q = 0.5
         q_arr = np.atleast_1d(np.asarray(q, dtype=np.float64))
         scalar_q = np.isscalar(q) or np.ndim(q) == 0
-        if dist_on:
+        if vals.numel() == 0 and not dist_on:
+            resq = torch.full((len(q_arr), ngroups), float("nan"),
+                              dtype=torch.float64, device=device)
+        elif dist_on:
             # exact cross-rank quantiles by radix selection over grouped
             # counts — histograms cross the wire, values never do
             from .dist_quantile import distributed_grouped_quantile
@@ -545,6 +548,12 @@ def groupby_reduce(
             else:
                 codes_m = labels.to(torch.int64)
             result = distributed_grouped_mode(vals, codes_m, ngroups, agg.skipnan)
+        elif vals.numel() == 0:
+            result = torch.full(
+                (ngroups,),
+                float("nan") if vals.is_floating_point() else 0,
+                dtype=vals.dtype, device=device,
+            )
         else:
             result = grouped_mode(
                 vals, labels, ngroups, skipnan=agg.skipnan, labels2=labels2, grp_shape=grp_pair
@@ -567,6 +576,7 @@ def groupby_reduce(
         ngroups > PACKED_ARG_THRESHOLD
         and vals.dtype in (torch.float32, torch.int32)
         and not dist_on
+        and 0 < vals.numel()
         and shard_row_offset + vals.numel() < (1 << 32)
     ):
         # huge group counts: pack (order-preserving 32-bit value encoding,
@@ -624,6 +634,7 @@ def groupby_reduce(
     elif func in ("first", "last", "nanfirst", "nanlast") and (
         ngroups > PACKED_ARG_THRESHOLD
         and not dist_on
+        and 0 < vals.numel()
         and shard_row_offset + vals.numel() < (1 << 32)
     ):
         # huge group counts: first/last is a grouped MIN/MAX over the row

@@ -66,6 +66,12 @@ def groupby_scan(
 
     orig_shape = arr.shape
     vals = arr.reshape(-1)
+    if vals.numel() == 0:
+        out0 = torch.empty_like(vals).reshape(orig_shape)
+        if return_numpy:
+            o = out0.cpu().numpy()
+            return o.astype(dt_dtype) if dt_dtype is not None else o
+        return out0
     # integer promotion like np.cumsum (sub-platform ints accumulate in intp)
     if func in ("cumsum", "nancumsum") and vals.dtype in (torch.int32, torch.bool):
         vals = vals.to(torch.int64)

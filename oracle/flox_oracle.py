@@ -508,11 +508,16 @@ def groupby_reduce(
                     vv = vg[~nanmask]
                     if vv.size == 0:
                         continue
+                    # int inputs lerp in f64, as the reference's quantile_
+                    # does (np.quantile on raw int64 wraps in b-a)
+                    if vv.dtype.kind in "iu":
+                        vv = vv.astype(np.float64)
                     out[:, r, g] = np.quantile(vv, q_arr, method="linear")
                 else:
                     if nanmask.any():
                         continue  # stays NaN (reference quantile_ masks)
-                    out[:, r, g] = np.quantile(vg, q_arr, method="linear")
+                    vq = vg.astype(np.float64) if vg.dtype.kind in "iu" else vg
+                    out[:, r, g] = np.quantile(vq, q_arr, method="linear")
         result = out.astype(out_dtype)
         if scalar_q:
             result = result[0]

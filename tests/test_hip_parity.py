@@ -523,3 +523,39 @@ def test_distributed_mode_world1(dtype, skipnan):
         torch.tensor(vals, device="cuda"), torch.tensor(labels, device="cuda"),
         ng, skipnan)
     np.testing.assert_array_equal(got.cpu().numpy().astype(want.dtype), want)
+
+
+def test_empty_and_single_row_inputs():
+    """Empty arrays fill everywhere; single-row groups match the oracle."""
+    e = np.array([], dtype=np.float64)
+    el = np.array([], dtype=np.int64)
+    want, *_ = oracle_reduce(e, el, func="sum", expected_groups=np.arange(4), fill_value=0.0)
+    got, *_ = flox_amd.groupby_reduce(e, el, func="sum", expected_groups=np.arange(4), fill_value=0.0)
+    np.testing.assert_array_equal(np.asarray(got), want)
+    want, *_ = oracle_reduce(e, el, func="mean", expected_groups=np.arange(3))
+    got, *_ = flox_amd.groupby_reduce(e, el, func="mean", expected_groups=np.arange(3))
+    np.testing.assert_array_equal(np.asarray(got), want)  # all NaN
+    want, *_ = oracle_reduce(np.array([5.0]), np.array([1]), func="var", expected_groups=np.arange(3))
+    got, *_ = flox_amd.groupby_reduce(np.array([5.0]), np.array([1]), func="var", expected_groups=np.arange(3))
+    np.testing.assert_array_equal(np.asarray(got), want)  # [nan, 0, nan]
+
+
+def test_atomic_fallback_extreme_ngroups():
+    """Group counts beyond the partition planner's reach (> 33M fine
+    buckets) fall to the global-atomic path and stay correct."""
+    rng = np.random.default_rng(55)
+    n, ng = 1_000_000, 40_000_000
+    labels = rng.integers(0, ng, n)
+    vals = rng.standard_normal(n).astype(np.float32)
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func="sum", expected_groups=range(ng))
+    got = np.asarray(got)
+    # spot-check against numpy on the rows present
+    order = np.argsort(labels, kind="stable")
+    ul, st = np.unique(labels[order], return_index=True)
+    sums = np.add.reduceat(vals[order].astype(np.float64), st)
+    np.testing.assert_allclose(got[ul], sums, rtol=1e-6, atol=1e-6)
+    present = np.zeros(ng, bool)
+    present[ul] = True
+    # missing groups fill with the dtype NA (NaN), as the reference does for
+    # plain sum with expected_groups and no user fill
+    assert np.isnan(got[~present]).all()

@@ -2,6 +2,8 @@
 randomized arrays/labels through the product API vs the pinned oracle, plus
 structural identities that need no oracle."""
 
+import os
+
 import numpy as np
 import pytest
 from hypothesis import HealthCheck, given, settings
@@ -14,7 +16,7 @@ from oracle import groupby_scan as oracle_scan
 pytestmark = pytest.mark.gpu
 
 SETTINGS = dict(
-    max_examples=25,
+    max_examples=int(os.environ.get("HYP_EXAMPLES", "25")),
     deadline=None,
     suppress_health_check=[HealthCheck.too_slow, HealthCheck.data_too_large],
 )
