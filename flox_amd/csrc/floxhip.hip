@@ -837,7 +837,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     int64_t g1, int shift, int B, int Bpad /* unused, <= 64 buckets */,
     uint32_t* __restrict__ cursors, uint32_t cap /* 0 = exact bases */,
     uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs) {
-  constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+  constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
   constexpr int RPT = T / PART_BLOCK;
   constexpr int NB = 64;
   constexpr int NE = NB * PART_NW; /* = PART_BLOCK slots */
@@ -973,7 +973,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     int shift /* fine-bucket shift */, uint32_t* __restrict__ cursors,
     uint32_t cap2 /* 0 = exact fine bases */, uint32_t* __restrict__ overflow,
     PairT<V>* __restrict__ out) {
-  constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+  constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
   constexpr int RPT = T / PART_BLOCK;
   constexpr int NB = PART_SUB;
   constexpr int NE = NB * PART_NW;
@@ -1248,7 +1248,7 @@ PartPlan part_plan(const fh_call* c) {
   const int bA = p.two_level ? p.B1 : p.B;
   if (bA > PART_SUB) return p; /* scatter passes handle <= 64 buckets */
   p.Bpad = PART_SUB;
-  constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+  constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
   const int64_t tile_lds = (int64_t)(PART_SUB * PART_NW + PART_NW + 1 + PART_SUB + 3) * 4 +
                            (int64_t)T * 4 + (int64_t)T * sizeof(PairT<V>);
   p.scatter_lds = tile_lds;
@@ -1350,7 +1350,7 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
     FH_CHECK(hipFuncSetAttribute((const void*)kern,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter_lds));
-    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
     int64_t wb = (c->n + T - 1) / T;
     int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
     const int shA = pp.two_level ? pp.shift + 6 : pp.shift;
@@ -1369,7 +1369,7 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
     FH_CHECK(hipFuncSetAttribute((const void*)kern2,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter2_lds));
-    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
     int tiles_x = (int)((maxA + T - 1) / T);
     if (tiles_x > 64) tiles_x = 64;
     if (tiles_x < 1) tiles_x = 1;
@@ -1433,7 +1433,7 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     FH_CHECK(hipFuncSetAttribute((const void*)kern,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter_lds));
-    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
     int64_t wb = (c->n + T - 1) / T;
     int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
     const int shA = pp.two_level ? pp.shift + 6 : pp.shift;
@@ -1451,7 +1451,7 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     FH_CHECK(hipFuncSetAttribute((const void*)kern2,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter2_lds));
-    constexpr int T = sizeof(V) == 4 ? 4096 : 2048;
+    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
     int tiles_x = (int)(((int64_t)pp.cap1 + T - 1) / T);
     if (tiles_x > 64) tiles_x = 64;
     if (tiles_x < 1) tiles_x = 1;
