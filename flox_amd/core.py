@@ -510,7 +510,12 @@ def groupby_reduce(
             mask = counts_for_mask < min_count_
             result = torch.where(mask, torch.tensor(float(fill_value), dtype=result.dtype, device=device), result)
         result = result.to(t_out)
-        new_shape = ((len(q_arr),) if not scalar_q else ()) + lead_shape + grp_shape
+        new_shape = (
+            ((len(q_arr),) if not scalar_q else ())
+            + lead_shape
+            + (subset_keep_shape if subset_keep_shape is not None else ())
+            + grp_shape
+        )
         result = result.reshape(new_shape)
         groups_list_q = [f.groups for f in facs]
         if not sort and not provided_expected:

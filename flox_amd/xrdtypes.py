@@ -47,6 +47,10 @@ def final_dtype(func: str, array_dtype: np.dtype, dtype=None) -> np.dtype:
         return np.dtype(np.intp)
     if func in BOOL_FUNCS:
         return np.dtype(bool)
+    if func in ("quantile", "nanquantile"):
+        # quantile's final_dtype is ALWAYS float64 (reference
+        # aggregations.py:695-710); median preserves a floating input dtype
+        return np.dtype("float64")
     if func in FLOAT_FUNCS:
         if array_dtype.kind in "fc":
             return array_dtype

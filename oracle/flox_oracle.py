@@ -144,6 +144,10 @@ def _final_dtype(func: str, array_dtype: np.dtype, dtype=None) -> np.dtype:
         return np.dtype(np.intp)
     if func in ("any", "all"):
         return np.dtype(bool)
+    if func in ("quantile", "nanquantile"):
+        # quantile's final_dtype is ALWAYS float64 (reference
+        # aggregations.py:695-710); median preserves a floating input dtype
+        return np.dtype("float64")
     if func in _FLOAT_FUNCS:
         # "mean, std, var always result in floating, preserving a floating input
         # dtype" (xrdtypes.py:161-167)
@@ -261,6 +265,14 @@ def groupby_reduce(
             # restate as one full reduction per leading slice against the
             # globally-found groups
             assert set(ax) <= set(trailing)
+            if dt_dtype is not None and func in ("nanfirst", "nanlast"):
+                raise NotImplementedError("datetime NaT skipping with an axis subset")
+            if dt_dtype is not None and func == "count":
+                # count must skip NaT: recurse on a float view with NaT -> NaN
+                # (count ignores magnitudes, so the f64 cast is exact for it)
+                array = np.where(
+                    array == np.iinfo(np.int64).min, np.nan, array.astype(np.float64)
+                )
             nlead = array.ndim - by_ndim
             keep = [d for d in range(array.ndim) if d not in ax]  # lead + kept by
             perm = keep + sorted(ax)
@@ -295,6 +307,13 @@ def groupby_reduce(
                 slices.append(r)
                 groups_out = g
             out = np.stack(slices).reshape(kshape + slices[0].shape)
+            q_kw = (finalize_kwargs or {}).get("q")
+            if func in ("quantile", "nanquantile") and q_kw is not None and np.ndim(q_kw) > 0:
+                # vector-q: the q dim leads the result (reference
+                # quantile_new_dims_func, aggregations.py:688-702)
+                out = np.moveaxis(out, len(kshape), 0)
+            if dt_dtype is not None and func not in ("count", "any", "all"):
+                out = out.astype(dt_dtype)  # reference core.py:1209-1211
             if not sort and was_discovered:
                 # first-appearance order over the ORIGINAL by layout
                 groups_out = list(groups_out)
@@ -381,6 +400,10 @@ def groupby_reduce(
             m = mask2d[r] if mask2d is not None else valid_code
             if weights2d is None:
                 out[r] = np.bincount(codes[m], minlength=ngroups).astype(acc_dtype)
+            elif acc_dtype == np.int64:
+                # integer sums accumulate (and wrap) in int64, like the
+                # reference's reduceat — np.bincount would force float64
+                np.add.at(out[r], codes[m], weights2d[r][m].astype(np.int64))
             else:
                 out[r] = np.bincount(codes[m], weights=weights2d[r][m].astype(acc_dtype), minlength=ngroups)
         return out

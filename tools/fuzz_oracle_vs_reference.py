@@ -1,0 +1,167 @@
+"""Fuzz the oracle against the shimmed reference (build container only).
+
+Generates random groupby_reduce / groupby_scan configurations over the
+funcs the reference's engine="flox" can run here, and checks the oracle
+reproduces the reference bit-for-bit (or within fp tolerance). Any
+mismatch is an oracle bug: the oracle is the parity anchor for the GPU
+tests, so this closes the chain  product == oracle == reference.
+
+Usage: python tools/fuzz_oracle_vs_reference.py [n_cases] [seed]
+"""
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "tests", "golden"))
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+
+from generate import load_reference  # noqa: E402
+
+FUNCS = [
+    "sum", "nansum", "prod", "nanprod", "mean", "nanmean", "var", "nanvar",
+    "std", "nanstd", "min", "nanmin", "max", "nanmax", "count",
+    "median", "nanmedian", "quantile", "nanquantile",
+]
+SCANS = ["cumsum", "nancumsum", "ffill", "bfill"]
+
+
+def make_case(rng):
+    n = int(rng.integers(1, 400))
+    dt = rng.choice(["float64", "float32", "int64", "int32", "datetime"])
+    if dt == "datetime":
+        vals = np.datetime64("2020-01-01") + rng.integers(0, 10**6, n).astype("timedelta64[s]")
+        if rng.random() < 0.5:
+            vals[rng.random(n) < 0.3] = np.datetime64("NaT")
+    elif dt in ("float64", "float32"):
+        vals = (rng.standard_normal(n) * 10 ** rng.integers(0, 4)).astype(dt)
+        if rng.random() < 0.6:
+            vals[rng.random(n) < rng.choice([0.05, 0.5, 0.95])] = np.nan
+    else:
+        vals = rng.integers(-1000, 1000, n).astype(dt)
+    ng = int(rng.integers(1, 25))
+    shape_kind = rng.choice(["1d", "lead", "multiby", "subset"])
+    kw = {}
+    if shape_kind == "1d":
+        by = rng.integers(0, ng, n)
+        arr = vals
+    elif shape_kind == "lead":
+        m = int(rng.integers(1, 5))
+        arr = np.repeat(vals[None, :], m, axis=0).copy()
+        arr[1:] = arr[1:][:, ::-1]
+        by = rng.integers(0, ng, n)
+    elif shape_kind == "multiby":
+        by = (rng.integers(0, 4, n), rng.integers(0, max(ng // 3, 1), n))
+        arr = vals
+    else:  # axis subset
+        a, b = int(rng.integers(1, 4)), int(rng.integers(2, 6))
+        n2 = a * b * max(n // (a * b), 1)
+        arr = np.resize(vals, (a, b, n2 // (a * b)))
+        by = rng.integers(0, ng, arr.shape)
+        kw["axis"] = (2,) if rng.random() < 0.5 else (1, 2)
+        kw["fill_value"] = -7.0
+    if isinstance(by, tuple):
+        kw["expected_groups"] = (np.arange(4), np.arange(max(ng // 3, 1)))
+    elif rng.random() < 0.7:
+        kw["expected_groups"] = np.arange(ng + int(rng.integers(0, 3)))
+    else:
+        kw["sort"] = bool(rng.random() < 0.7)
+    if rng.random() < 0.2 and "fill_value" not in kw and kw.get("expected_groups") is not None:
+        kw["fill_value"] = float(rng.integers(-99, 99))
+    if rng.random() < 0.15 and kw.get("expected_groups") is not None:
+        kw["min_count"] = int(rng.integers(1, 5))
+        kw.setdefault("fill_value", np.nan)
+    return arr, by, kw
+
+
+def main():
+    ncases = int(sys.argv[1]) if len(sys.argv) > 1 else 300
+    seed = int(sys.argv[2]) if len(sys.argv) > 2 else 12345
+    core = load_reference()
+    import importlib
+
+    refscan = importlib.import_module("floxref.scan")
+    from oracle import groupby_reduce as oreduce
+    from oracle import groupby_scan as oscan
+
+    rng = np.random.default_rng(seed)
+    n_ok = n_skip = n_bad = 0
+    for i in range(ncases):
+        arr, by, kw = make_case(rng)
+        is_scan = rng.random() < 0.2 and "axis" not in kw and "min_count" not in kw
+        if is_scan:
+            func = str(rng.choice(SCANS))
+            kw.pop("fill_value", None)
+            kw.pop("sort", None)
+        else:
+            func = str(rng.choice(FUNCS))
+            if func in ("quantile", "nanquantile"):
+                q = [0.25, 0.9] if rng.random() < 0.5 else float(rng.random())
+                kw["finalize_kwargs"] = {"q": q}
+            if func in ("var", "nanvar", "std", "nanstd") and rng.random() < 0.3:
+                kw["finalize_kwargs"] = {"ddof": 1}
+        if np.asarray(arr).dtype.kind in "Mm" and func not in (
+            "min", "nanmin", "max", "nanmax", "count",
+            "median", "nanmedian", "quantile", "nanquantile",
+        ) and not is_scan:
+            # datetime sum/mean/var with NaT is int64-min wrap garbage no
+            # caller or reference test exercises; parity is scoped to the
+            # meaningful datetime ops (see DESIGN.md numerics notes)
+            n_skip += 1
+            continue
+        if func in ("prod", "nanprod") and np.asarray(arr).dtype.kind in "iuMm":
+            # wrapped integer products are order-dependent; keep them in
+            # {-1, 0, 1} so overflow cannot occur (datetime prod is skipped)
+            if np.asarray(arr).dtype.kind in "Mm":
+                n_skip += 1
+                continue
+            arr = np.clip(arr, -1, 1)
+        bys = by if isinstance(by, tuple) else (by,)
+        try:
+            if is_scan:
+                want = refscan.groupby_scan(arr, *bys, func=func, **kw)
+            else:
+                want, *wg = core.groupby_reduce(arr, *bys, func=func, engine="flox", **kw)
+        except Exception:
+            n_skip += 1
+            continue
+        try:
+            if is_scan:
+                got = oscan(arr, *bys, func=func, **kw)
+            else:
+                got, *gg = oreduce(arr, *bys, func=func, **kw)
+        except Exception as e:
+            print(f"[{i}] ORACLE RAISED {type(e).__name__}: {e} | func={func} kw={list(kw)} "
+                  f"shape={np.shape(arr)} dt={np.asarray(arr).dtype}")
+            n_bad += 1
+            continue
+        want = np.asarray(want)
+        got = np.asarray(got)
+        try:
+            assert got.shape == want.shape, (got.shape, want.shape)
+            assert got.dtype == want.dtype, (got.dtype, want.dtype)
+            if want.dtype.kind in "Mm" and func in ("quantile", "nanquantile", "median", "nanmedian", "mean", "nanmean", "var", "nanvar", "std", "nanstd"):
+                # lerp/mean on NaT (=int64 min) magnitudes: the two f64 lerp
+                # forms differ by ~1 ulp of 9.2e18 — compare relatively
+                wi, gi = want.view("i8").astype("f8"), got.view("i8").astype("f8")
+                np.testing.assert_array_equal(want.view("i8") == np.iinfo(np.int64).min,
+                                              got.view("i8") == np.iinfo(np.int64).min)
+                np.testing.assert_allclose(gi, wi, rtol=1e-9, atol=1.0)
+            elif want.dtype.kind in "iubMm":
+                np.testing.assert_array_equal(got, want)
+            else:
+                rtol = 2e-5 if want.dtype.itemsize == 4 else 1e-11
+                atol = (1e-4 if want.dtype.itemsize == 4 else 1e-9) * (
+                    1 + float(np.nanmax(np.abs(want[np.isfinite(want)]), initial=0)))
+                np.testing.assert_allclose(got, want, equal_nan=True, rtol=rtol, atol=atol)
+            n_ok += 1
+        except AssertionError as e:
+            n_bad += 1
+            print(f"[{i}] MISMATCH func={func} scan={is_scan} kw={ {k: (v if not isinstance(v, np.ndarray) else v.shape) for k, v in kw.items()} } "
+                  f"shape={arr.shape} dt={arr.dtype}: {str(e)[:300]}")
+    print(f"{n_ok} ok, {n_skip} reference-skipped, {n_bad} mismatches of {ncases}")
+    sys.exit(1 if n_bad else 0)
+
+
+if __name__ == "__main__":
+    main()
