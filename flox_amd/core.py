@@ -329,11 +329,20 @@ def groupby_reduce(
     if dt_dtype is not None and func in ("count", "nanfirst", "nanlast"):
         # NaT rows are missing for count (verified reference behavior) and
         # for nanfirst/nanlast (xrutils.nanfirst's isnull, xrutils.py:389-397):
-        # drop them by invalidating their codes
+        # drop them by invalidating their codes; with leading dims NaT-ness
+        # is per (lead, row), so count recounts on a float view instead
+        # (NaT -> NaN; exact, count ignores magnitudes)
         if lead_M != 1:
-            raise NotImplementedError("datetime NaT skipping with leading dims: next row")
-        natm = arr.reshape(-1) == torch.iinfo(torch.int64).min
-        labels = torch.where(natm, torch.full_like(labels, -1), labels)
+            if func != "count":
+                raise NotImplementedError("datetime NaT skipping with leading dims: next row")
+            arr = torch.where(
+                arr == torch.iinfo(torch.int64).min,
+                torch.tensor(float("nan"), dtype=torch.float64, device=device),
+                arr.to(torch.float64),
+            )
+        else:
+            natm = arr.reshape(-1) == torch.iinfo(torch.int64).min
+            labels = torch.where(natm, torch.full_like(labels, -1), labels)
 
     if subset_keep_shape is not None:
         # offset the codes by the kept-dims slice index (reference

@@ -358,8 +358,16 @@ def groupby_reduce(
         # NaT rows are missing for these (see datetime note above)
         natm = array.reshape(-1) == np.iinfo(np.int64).min
         if natm.size != codes.size:
-            raise NotImplementedError("datetime NaT skipping with leading dims")
-        codes = np.where(natm, -1, codes)
+            if func == "count":
+                # per-(lead, row) NaT-ness: recount on a float view with
+                # NaT -> NaN (exact — count ignores magnitudes)
+                array = np.where(
+                    array == np.iinfo(np.int64).min, np.nan, array.astype(np.float64)
+                )
+            else:
+                raise NotImplementedError("datetime NaT skipping with leading dims")
+        else:
+            codes = np.where(natm, -1, codes)
 
     # --- min_count defaulting (reference core.py:1026-1038) ---
     out_dtype = _final_dtype(func, array.dtype, dtype)
