@@ -738,9 +738,18 @@ def nanmode(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None
 def _scan_like(group_idx, array, func, size=None, dtype=None):
     from .scan import groupby_scan
 
-    group_idx, array = _prep(group_idx, array)
-    ng = _size_of(group_idx, size)
-    return groupby_scan(array, group_idx, func=func, expected_groups=range(ng))
+    # keep the array's leading dims — groupby_scan folds them natively
+    if not isinstance(array, torch.Tensor):
+        array = torch.as_tensor(np.ascontiguousarray(array))
+    if not array.is_cuda:
+        if not torch.cuda.is_available():
+            raise RuntimeError("engine='hip' requires a GPU; none is available")
+        array = array.cuda()
+    gi = group_idx if isinstance(group_idx, torch.Tensor) else torch.as_tensor(
+        np.ascontiguousarray(group_idx))
+    gi = gi.to(array.device).reshape(-1)
+    ng = _size_of(gi, size)
+    return groupby_scan(array, gi, func=func, expected_groups=range(ng))
 
 
 def cumsum(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
@@ -775,3 +784,81 @@ def all_(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, *
     ng = _size_of(group_idx, size)
     p = grouped_partials(SET_MIN_COUNT, array.to(torch.int64) if array.dtype.is_floating_point else array, group_idx, ng)
     return (p["min"] != 0) & (p["count"] > 0)
+
+
+# ---------------------------------------------------------------------------
+# leading array dims through the engine seam: the reference's engine
+# callables accept (lead..., n) arrays with (n,) group_idx, reducing over
+# axis=-1 (aggregate_flox.py:133-192 operates on the trailing axis). Fold
+# the lead index into composite codes (lead*ng + code) and reshape back.
+# ---------------------------------------------------------------------------
+
+
+def _fold_lead(group_idx, array, size):
+    if not isinstance(array, torch.Tensor):
+        array = torch.as_tensor(np.ascontiguousarray(array))
+    if not isinstance(group_idx, torch.Tensor):
+        group_idx = torch.as_tensor(np.ascontiguousarray(group_idx))
+    if not array.is_cuda:
+        if not torch.cuda.is_available():
+            raise RuntimeError("engine='hip' requires a GPU; none is available")
+        array = array.cuda()
+    gi = group_idx.to(array.device).reshape(-1).to(torch.int64)
+    n = gi.numel()
+    if array.shape[-1] != n:
+        raise ValueError(f"group_idx ({n},) must match array's trailing axis {tuple(array.shape)}")
+    lead_shape = tuple(array.shape[:-1])
+    M = 1
+    for d in lead_shape:
+        M *= d
+    ng = _size_of(gi, size)
+    lead = torch.arange(M, device=array.device, dtype=torch.int64)
+    comp = lead[:, None] * ng + gi[None, :]
+    comp = torch.where((gi < 0)[None, :].expand_as(comp), torch.full_like(comp, -1), comp)
+    return comp.reshape(-1), array.reshape(-1), M, ng, lead_shape, n
+
+
+def _lead_wrap(kind):
+    import functools
+
+    def deco(f):
+        @functools.wraps(f)
+        def wrapper(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
+            a_nd = getattr(array, "ndim", None)
+            if a_nd is None:
+                a_nd = np.ndim(array)
+            g_nd = getattr(group_idx, "ndim", None)
+            if g_nd is None:
+                g_nd = np.ndim(group_idx)
+            if a_nd <= builtins.max(g_nd, 1):
+                return f(group_idx, array, axis=axis, size=size,
+                         fill_value=fill_value, dtype=dtype, **kw)
+            comp, flat, M, ng, lead_shape, n = _fold_lead(group_idx, array, size)
+            out = f(comp, flat, axis=-1, size=M * ng,
+                    fill_value=fill_value, dtype=dtype, **kw)
+            if kind == "q" and out.ndim == 2:  # (nq, M*ng)
+                return out.reshape((out.shape[0],) + lead_shape + (ng,))
+            out = out.reshape(lead_shape + (ng,))
+            if kind == "arg":
+                # composite indices are flat (lead*n + t); the API returns t
+                offs = (torch.arange(M, device=out.device, dtype=out.dtype) * n).reshape(
+                    lead_shape + (1,))
+                out = torch.where(out >= 0, out - offs, out)
+            return out
+        return wrapper
+    return deco
+
+
+_LEAD_KINDS = {
+    "sum": "reduce", "nansum": "reduce", "nanlen": "reduce", "count": "reduce",
+    "prod": "reduce", "nanprod": "reduce", "mean": "reduce", "nanmean": "reduce",
+    "min": "reduce", "nanmin": "reduce", "max": "reduce", "nanmax": "reduce",
+    "var": "reduce", "nanvar": "reduce", "std": "reduce", "nanstd": "reduce",
+    "argmax": "arg", "argmin": "arg", "nanargmax": "arg", "nanargmin": "arg",
+    "first": "reduce", "last": "reduce", "nanfirst": "reduce", "nanlast": "reduce",
+    "median": "reduce", "nanmedian": "reduce", "quantile": "q", "nanquantile": "q",
+    "mode": "reduce", "nanmode": "reduce", "any_": "reduce", "all_": "reduce",
+}
+for _n, _k in _LEAD_KINDS.items():
+    globals()[_n] = _lead_wrap(_k)(globals()[_n])
+del _n, _k

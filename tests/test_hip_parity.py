@@ -576,3 +576,23 @@ def test_quantile_axis_subset(q):
     assert np.asarray(got).shape == want.shape
     np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
                                rtol=1e-9, atol=1e-9)
+
+
+def test_engine_seam_2d_array_and_fill():
+    """Seam callables accept 2-D arrays (reduce over axis=-1, as
+    chunk_reduce calls them) and honor fill_value for absent groups."""
+    from flox_amd import generic_aggregate
+
+    rng = np.random.default_rng(5)
+    g = rng.integers(0, 7, 500)
+    a = rng.standard_normal((4, 500))
+    out = np.asarray(generic_aggregate(g, a, engine="hip", func="mean", size=7).cpu())
+    assert out.shape == (4, 7)
+    for r in range(4):
+        want = np.bincount(g, weights=a[r], minlength=7) / np.bincount(g, minlength=7)
+        np.testing.assert_allclose(out[r], want, rtol=1e-12, atol=1e-12)
+    # absent group (size > max label) gets fill_value
+    out2 = np.asarray(generic_aggregate(g, a[0], engine="hip", func="sum",
+                                        size=9, fill_value=-5.0).cpu())
+    assert out2.shape == (9,)
+    np.testing.assert_allclose(out2[7:], [-5.0, -5.0])
