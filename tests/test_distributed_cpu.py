@@ -374,3 +374,20 @@ def test_gloo_distributed_mode():
         errs.append(fail_q.get())
     assert not errs, errs[0]
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
+def test_gloo_world3_scan_quantile_mode():
+    """world_size=3: middle ranks both receive and forward carries/counts."""
+    ctx = mp.get_context("spawn")
+    for worker, port in [(_worker_scan, 29531), (_worker_quantile, 29533), (_worker_mode, 29535)]:
+        fail_q = ctx.Queue()
+        procs = [ctx.Process(target=worker, args=(r, 3, port, fail_q)) for r in range(3)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=240)
+        errs = []
+        while not fail_q.empty():
+            errs.append(fail_q.get())
+        assert not errs, errs[0]
+        assert all(p.exitcode == 0 for p in procs), (worker.__name__, [p.exitcode for p in procs])
