@@ -6,7 +6,12 @@ hand-written HIP/CDNA4 kernels behind flox's own engine-plugin boundary,
 with an RCCL-over-xGMI combine across GPUs. See DESIGN.md.
 """
 
-from .aggregations import REDUCTIONS, Aggregation, generic_aggregate  # noqa: F401
+from .aggregations import (  # noqa: F401
+    REDUCTIONS,
+    Aggregation,
+    CustomAggregation,
+    generic_aggregate,
+)
 from .core import groupby_reduce  # noqa: F401
 from .scan import groupby_scan  # noqa: F401
 

@@ -145,3 +145,42 @@ def generic_aggregate(
     if method is None:
         raise NotImplementedError(f"engine='hip' does not implement {func!r} yet")
     return method(group_idx, array, axis=axis, size=size, fill_value=fill_value, dtype=dtype, **kwargs)
+
+
+class CustomAggregation:
+    """Reference-compatible custom aggregation (flox.Aggregation,
+    reference aggregations.py:161-301 / docs "Custom Aggregations"):
+    compose existing chunk reductions with a user ``finalize``.
+
+    Supported ``chunk`` names: sum nansum count nanlen min nanmin max nanmax
+    prod nanprod. ``finalize`` receives one torch tensor per chunk
+    intermediate (use array operators, not numpy functions). ``combine``
+    names drive the cross-GPU all-reduce of each intermediate.
+    """
+
+    def __init__(
+        self,
+        name,
+        *,
+        numpy=None,
+        chunk,
+        combine,
+        preprocess=None,
+        aggregate=None,
+        finalize=None,
+        fill_value=None,
+        final_fill_value=float("nan"),
+        dtypes=None,
+        final_dtype=None,
+        reduction_type="reduce",
+    ):
+        if preprocess is not None:
+            raise NotImplementedError("CustomAggregation.preprocess")
+        self.name = name
+        self.numpy = numpy
+        self.chunk = (chunk,) if isinstance(chunk, str) else tuple(chunk)
+        self.combine = (combine,) if isinstance(combine, str) else tuple(combine)
+        self.finalize = finalize
+        self.fill_value = fill_value
+        self.final_fill_value = final_fill_value
+        self.final_dtype = final_dtype

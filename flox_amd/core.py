@@ -207,6 +207,55 @@ def groupby_reduce(
     Each rank passes its own row shard; results are full-group on every rank.
     """
     from .aggregations import Aggregation as _Agg
+    from .aggregations import CustomAggregation as _CustomAgg
+
+    if isinstance(func, _CustomAgg):
+        # reference-style custom aggregation (docs "Custom Aggregations"):
+        # run each chunk reduction through the full machinery, then apply
+        # the user's finalize and the final fill for empty groups
+        _CHUNK_OK = {"sum", "nansum", "count", "nanlen", "min", "nanmin",
+                     "max", "nanmax", "prod", "nanprod"}
+        inner = dict(
+            expected_groups=expected_groups, sort=sort, isbin=isbin, axis=axis,
+            method=method, engine=engine, reindex=reindex,
+            distributed_combine=distributed_combine,
+            shard_row_offset=shard_row_offset,
+        )
+        inters, groups_c, counts_c = [], None, None
+        for cn in func.chunk:
+            cn2 = "count" if cn == "nanlen" else cn
+            if cn2 not in _CHUNK_OK:
+                raise NotImplementedError(f"custom aggregation chunk {cn!r}")
+            r, *g = groupby_reduce(array, *by, func=cn2, **inner)
+            inters.append(r)
+            groups_c = g
+            if cn2 == "count":
+                counts_c = r
+        if counts_c is None:
+            counts_c, *_ = groupby_reduce(array, *by, func="count", **inner)
+        res = func.finalize(*inters) if func.finalize is not None else inters[0]
+        ffv = fill_value if fill_value is not None else func.final_fill_value
+        mask = counts_c == 0
+        if isinstance(res, torch.Tensor):
+            if not isinstance(mask, torch.Tensor):
+                mask = torch.as_tensor(np.asarray(mask), device=res.device)
+            if ffv is not None and bool(mask.any().item()):
+                nan_on_int = (
+                    isinstance(ffv, float) and math.isnan(ffv)
+                    and not res.dtype.is_floating_point
+                )
+                if nan_on_int:
+                    res = res.to(torch.float64)
+                res = torch.where(mask, torch.tensor(ffv, dtype=res.dtype, device=res.device), res)
+            if func.final_dtype is not None:
+                res = res.to(_torch_dtype(np.dtype(func.final_dtype)))
+        else:
+            res = np.asarray(res)
+            if ffv is not None and bool(np.any(mask)):
+                res = np.where(mask, ffv, res)
+            if func.final_dtype is not None:
+                res = res.astype(func.final_dtype)
+        return (res, *groups_c)
 
     if isinstance(func, _Agg):
         func = func.name  # reference accepts Aggregation instances (core.py:934-948)

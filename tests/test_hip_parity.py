@@ -596,3 +596,33 @@ def test_engine_seam_2d_array_and_fill():
                                         size=9, fill_value=-5.0).cpu())
     assert out2.shape == (9,)
     np.testing.assert_allclose(out2[7:], [-5.0, -5.0])
+
+
+def test_custom_aggregation():
+    """Reference-style custom Aggregation (docs 'Custom Aggregations'): the
+    mean-from-parts example and a custom range = max - min."""
+    from flox_amd import CustomAggregation
+
+    mean_agg = CustomAggregation(
+        name="mean", numpy="mean", chunk=("sum", "nanlen"), combine=("sum", "sum"),
+        finalize=lambda sum_, count: sum_ / count,
+        fill_value=0, final_fill_value=np.nan,
+    )
+    rng = np.random.default_rng(31)
+    v = rng.standard_normal(5000)
+    labels = rng.integers(0, 9, 5000)
+    got, gg = flox_amd.groupby_reduce(v, labels, func=mean_agg, expected_groups=np.arange(11))
+    want, *_ = oracle_reduce(v, labels, func="mean", expected_groups=np.arange(11))
+    np.testing.assert_allclose(np.asarray(got), want, equal_nan=True, rtol=1e-12, atol=1e-12)
+
+    rng_agg = CustomAggregation(
+        name="range", chunk=("max", "min"), combine=("max", "min"),
+        finalize=lambda mx, mn: mx - mn, final_fill_value=-1.0,
+        final_dtype=np.float64,
+    )
+    got2, _ = flox_amd.groupby_reduce(v, labels, func=rng_agg, expected_groups=np.arange(11))
+    wmax, *_ = oracle_reduce(v, labels, func="max", expected_groups=np.arange(11))
+    wmin, *_ = oracle_reduce(v, labels, func="min", expected_groups=np.arange(11))
+    want2 = wmax - wmin
+    want2[9:] = -1.0
+    np.testing.assert_allclose(np.asarray(got2), want2, rtol=1e-12, atol=1e-12)
