@@ -25,9 +25,12 @@ from .aggregations import REDUCTIONS
 
 # arg-reductions switch from the two-pass (extremum, then index-match) form to
 # the packed-key single-pass form above this group count: the IDX bins cost
-# 20 B/group in LDS, so past ~8e3 groups the second pass would fall to the
-# global-atomic path. Tests lower this to exercise the packed form at small n.
-PACKED_ARG_THRESHOLD = 7000
+# 20 B/group in LDS, so past ~8e3 groups the second pass falls to the
+# global-atomic path. Measured crossover (tools/arg_threshold_prof.py,
+# 2e8 rows): two-pass wins to ~4e3 groups (0.94 vs 1.14 ms), packed wins
+# from ~8e3 (1.09 vs 1.35; 2.6 vs 10.0 ms at 16e3). Tests lower this to
+# exercise the packed form at small n.
+PACKED_ARG_THRESHOLD = 6000
 
 # order-dependent reductions that run with leading array dims by folding the
 # lead index into the group codes (see groupby_reduce's lead-fold block)
