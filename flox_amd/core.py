@@ -6,9 +6,14 @@ kernel layer replaced by fused HIP passes (aggregate_hip.grouped_partials)
 and the dask combine tree replaced by RCCL all-reduces of the per-group
 partial bins (distributed.py).
 
-Scope (round 1): reduction over ALL dims of ``by`` (axis=None or the full
-trailing tuple), ``array`` fully covered by ``by`` (no leading batch dims on
-the GPU path yet), dtypes f32/f64/i32/i64/bool, 1 or more ``by`` arrays.
+Scope: every reduction/scan the reference registers, over any mix of
+leading array dims, axis subsets of ``by``'s dims, 1+ ``by`` arrays
+(multi-dim groupby), isbin/IntervalIndex binning, sort=False
+first-appearance ordering, expected_groups/fill_value/min_count/dtype
+rules, custom Aggregation(chunk/combine/finalize) instances; dtypes
+f32/f64/i32/i64/bool/datetime64/timedelta64 (NaT as int64-min, like the
+reference). Multi-GPU: every op combines across ranks (partial-bin
+all-reduce / scan carries / quantile radix selection / mode run merge).
 """
 
 from __future__ import annotations
