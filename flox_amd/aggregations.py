@@ -35,7 +35,9 @@ class Aggregation:
     preserves_dtype: bool = False
 
 
-_SUMLIKE_COMBINE = {"sum": "sum", "count": "sum", "present": "max"}
+# present rides the same coalesced SUM all-reduce as sum/count: the
+# per-rank flag is {0,1} and every consumer tests present != 0
+_SUMLIKE_COMBINE = {"sum": "sum", "count": "sum", "present": "sum"}
 
 REDUCTIONS: dict[str, Aggregation] = {
     "count": Aggregation("count", _ffi.SET_COUNT, skipnan=True, combine={"count": "sum"}),
@@ -43,9 +45,9 @@ REDUCTIONS: dict[str, Aggregation] = {
     "nansum": Aggregation(
         "nansum", _ffi.SET_SUM_COUNT_PRESENT, skipnan=True, combine=dict(_SUMLIKE_COMBINE)
     ),
-    "prod": Aggregation("prod", _ffi.SET_PROD, combine={"sum": "prod", "count": "sum", "present": "max"}),
+    "prod": Aggregation("prod", _ffi.SET_PROD, combine={"sum": "prod", "count": "sum", "present": "sum"}),
     "nanprod": Aggregation(
-        "nanprod", _ffi.SET_PROD, skipnan=True, combine={"sum": "prod", "count": "sum", "present": "max"}
+        "nanprod", _ffi.SET_PROD, skipnan=True, combine={"sum": "prod", "count": "sum", "present": "sum"}
     ),
     "mean": Aggregation("mean", _ffi.SET_SUM_COUNT, combine={"sum": "sum", "count": "sum"}),
     "nanmean": Aggregation(
@@ -54,7 +56,7 @@ REDUCTIONS: dict[str, Aggregation] = {
     "min": Aggregation(
         "min",
         _ffi.SET_MIN_FULL,
-        combine={"min": "min", "count": "sum", "present": "max", "nanflag": "max"},
+        combine={"min": "min", "count": "sum", "present": "sum", "nanflag": "max"},
         preserves_dtype=True,
     ),
     "nanmin": Aggregation(
@@ -64,7 +66,7 @@ REDUCTIONS: dict[str, Aggregation] = {
     "max": Aggregation(
         "max",
         _ffi.SET_MAX_FULL,
-        combine={"max": "max", "count": "sum", "present": "max", "nanflag": "max"},
+        combine={"max": "max", "count": "sum", "present": "sum", "nanflag": "max"},
         preserves_dtype=True,
     ),
     "nanmax": Aggregation(
@@ -83,22 +85,22 @@ REDUCTIONS: dict[str, Aggregation] = {
     # first/last: index extremum + gather (reference aggregations.py:635-649)
     "first": Aggregation(
         "first", _ffi.SET_IDXMIN,
-        combine={"idx": "min", "count": "sum", "present": "max"},
+        combine={"idx": "min", "count": "sum", "present": "sum"},
         preserves_dtype=True,
     ),
     "last": Aggregation(
         "last", _ffi.SET_IDXMAX,
-        combine={"idx": "max", "count": "sum", "present": "max"},
+        combine={"idx": "max", "count": "sum", "present": "sum"},
         preserves_dtype=True,
     ),
     "nanfirst": Aggregation(
         "nanfirst", _ffi.SET_IDXMIN, skipnan=True,
-        combine={"idx": "min", "count": "sum", "present": "max"},
+        combine={"idx": "min", "count": "sum", "present": "sum"},
         preserves_dtype=True,
     ),
     "nanlast": Aggregation(
         "nanlast", _ffi.SET_IDXMAX, skipnan=True,
-        combine={"idx": "max", "count": "sum", "present": "max"},
+        combine={"idx": "max", "count": "sum", "present": "sum"},
         preserves_dtype=True,
     ),
     # quantile family: sorted path (blockwise-only in the reference too,

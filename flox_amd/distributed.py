@@ -38,15 +38,25 @@ def combine_partials(partials: dict[str, torch.Tensor], combine: dict[str, str])
 
     min/max bins hold +inf/-inf for groups a rank never saw (the reference's
     intermediate fill values, flox/aggregations.py:529-546), so the
-    collective needs no validity mask.
+    collective needs no validity mask. Partials sharing a combine op ride
+    ONE coalesced all-reduce (mean's sum/count/present = a single
+    collective per step instead of three — collective launch latency is
+    the scaling tax at small bin sizes).
     """
     if not is_active():
         return partials
+    by_op: dict[str, list[torch.Tensor]] = {}
     for name, op in combine.items():
         t = partials.get(name)
         if t is None:
             continue
-        dist.all_reduce(t, op=_ops()[op])
+        by_op.setdefault(op, []).append(t)
+    for op, tensors in by_op.items():
+        if len(tensors) == 1 or not hasattr(dist, "all_reduce_coalesced"):
+            for t in tensors:
+                dist.all_reduce(t, op=_ops()[op])
+        else:
+            dist.all_reduce_coalesced(tensors, op=_ops()[op])
     return partials
 
 
