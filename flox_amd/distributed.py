@@ -55,8 +55,23 @@ def combine_partials(partials: dict[str, torch.Tensor], combine: dict[str, str])
         if len(tensors) == 1 or not hasattr(dist, "all_reduce_coalesced"):
             for t in tensors:
                 dist.all_reduce(t, op=_ops()[op])
-        else:
+            continue
+        if len({t.dtype for t in tensors}) == 1:
             dist.all_reduce_coalesced(tensors, op=_ops()[op])
+        elif op == "sum" and all(
+            t.dtype in (torch.float64, torch.int64, torch.int32) for t in tensors
+        ):
+            # counts/presence ride the f64 sum collective exactly (values
+            # are far below 2^53); converting ngroups-sized bins costs
+            # less than a second collective launch
+            up = [t if t.dtype == torch.float64 else t.to(torch.float64) for t in tensors]
+            dist.all_reduce_coalesced(up, op=dist.ReduceOp.SUM)
+            for t, u in zip(tensors, up):
+                if u is not t:
+                    t.copy_(u.to(t.dtype))
+        else:
+            for t in tensors:
+                dist.all_reduce(t, op=_ops()[op])
     return partials
 
 
