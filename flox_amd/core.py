@@ -769,9 +769,7 @@ def groupby_reduce(
         if dist_on:
             t = torch.where(counts > 0, sums * sums / counts, torch.zeros_like(sums))
             a = ssd + t
-            distributed.all_reduce_(a, "sum")
-            distributed.all_reduce_(sums, "sum")
-            distributed.all_reduce_(counts, "sum")
+            distributed.all_reduce_sum_many([a, sums, counts])
             ssd = a - torch.where(counts > 0, sums * sums / counts, torch.zeros_like(sums))
         den_w = counts.to(torch.float64) - ddof
         result = ssd / den_w
@@ -785,8 +783,7 @@ def groupby_reduce(
         skip = agg.skipnan
         p1 = run_set(_ffi.SET_SUM_COUNT, skip)
         if dist_on:
-            distributed.all_reduce_(p1["sum"], "sum")
-            distributed.all_reduce_(p1["count"], "sum")
+            distributed.all_reduce_sum_many([p1["sum"], p1["count"]])
         counts = p1["count"]
         # deviations about the (global) per-group mean: the cross-rank ssd
         # combine is then a plain sum (the reference's _var_combine,

@@ -81,6 +81,15 @@ def all_reduce_(t: torch.Tensor, op: str = "sum") -> torch.Tensor:
     return t
 
 
+def all_reduce_sum_many(tensors: list[torch.Tensor]) -> None:
+    """Sum-all-reduce several bins in one coalesced collective (mixed
+    f64/i64/i32 ride as f64 — exact below 2^53), in place."""
+    if not is_active() or not tensors:
+        return
+    combine_partials({str(i): t for i, t in enumerate(tensors)},
+                     {str(i): "sum" for i in range(len(tensors))})
+
+
 def scan_carry_exchange(
     out: torch.Tensor,
     vals: torch.Tensor,
