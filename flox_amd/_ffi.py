@@ -34,6 +34,7 @@ SET_WELFORD = 11
 FLAG_SKIPNAN = 1
 FLAG_FORCE_LDS = 2
 FLAG_FORCE_ATOMIC = 4
+FLAG_SORTED_LABELS = 8
 
 VDTYPE_OF = {
     np.dtype("float32"): F32,

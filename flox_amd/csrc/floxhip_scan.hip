@@ -72,7 +72,9 @@ __global__ void k_spack(const L* __restrict__ labels, const L* __restrict__ labe
   }
 }
 
-/* gather values into sorted order, building the scan input */
+/* gather values into sorted order, building the scan input; perm == NULL
+ * means the labels were already sorted (FH_SORTED_LABELS) and rows are
+ * read in place */
 template <typename V, int OP>
 __global__ void k_sgather(const V* __restrict__ values, const uint32_t* __restrict__ perm,
                           int64_t n, V* __restrict__ sv, FillPair<V>* __restrict__ sp) {
@@ -80,7 +82,7 @@ __global__ void k_sgather(const V* __restrict__ values, const uint32_t* __restri
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     /* bfill scans the reversed sequence */
     const int64_t j = (OP == SCAN_BFILL) ? (n - 1 - i) : i;
-    const V v = values[perm[j]];
+    const V v = values[perm ? (int64_t)perm[j] : j];
     if (OP == SCAN_CUMSUM) {
       sv[i] = v;
     } else if (OP == SCAN_NANCUMSUM) {
@@ -109,11 +111,12 @@ __global__ void k_sscatter(const V* __restrict__ sv, const FillPair<V>* __restri
   const double NAN_ = __longlong_as_double(0x7FF8000000000000ll);
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const int64_t j = (OP == SCAN_BFILL) ? (n - 1 - i) : i;
+    const int64_t o = perm ? (int64_t)perm[j] : j;
     if (OP == SCAN_CUMSUM || OP == SCAN_NANCUMSUM) {
-      out[perm[j]] = sv[i];
+      out[o] = sv[i];
     } else {
       const FillPair<V> p = sp[i];
-      out[perm[j]] = p.valid ? p.v : (V)NAN_; /* leading gap stays NaN */
+      out[o] = p.valid ? p.v : (V)NAN_; /* leading gap stays NaN */
     }
   }
 }
@@ -160,12 +163,24 @@ int run_scan(fh_call* c, V* out) {
   if (o > c->scratch_bytes) return 3;
 
   const int grid = (int)std::min<int64_t>((n + 255) / 256, 2048) + 1;
-  hipLaunchKernelGGL((k_spack<L>), dim3(grid), dim3(256), 0, stream,
-                     (const L*)c->labels, (const L*)c->labels2, n, c->ngroups,
-                     c->g0, c->g1, codes, idx);
-  FHS_CHECK(hipGetLastError());
-  FHS_CHECK(rocprim::radix_sort_pairs(temp, ts, codes, codes_s, idx, perm,
-                                      (size_t)n, 0, 32, stream));
+  const bool presorted = (c->flags & FH_SORTED_LABELS) != 0;
+  if (presorted) {
+    /* caller guarantees nondecreasing in-range labels (the reference's
+     * issorted fast path, aggregate_flox.py:9-23): pack the u32 keys but
+     * skip the radix sort, gather and scatter in place */
+    hipLaunchKernelGGL((k_spack<L>), dim3(grid), dim3(256), 0, stream,
+                       (const L*)c->labels, (const L*)c->labels2, n, c->ngroups,
+                       c->g0, c->g1, codes_s, idx);
+    FHS_CHECK(hipGetLastError());
+    perm = nullptr;
+  } else {
+    hipLaunchKernelGGL((k_spack<L>), dim3(grid), dim3(256), 0, stream,
+                       (const L*)c->labels, (const L*)c->labels2, n, c->ngroups,
+                       c->g0, c->g1, codes, idx);
+    FHS_CHECK(hipGetLastError());
+    FHS_CHECK(rocprim::radix_sort_pairs(temp, ts, codes, codes_s, idx, perm,
+                                        (size_t)n, 0, 32, stream));
+  }
   hipLaunchKernelGGL((k_sgather<V, OP>), dim3(grid), dim3(256), 0, stream,
                      (const V*)c->values, perm, n, sv, sp);
   FHS_CHECK(hipGetLastError());

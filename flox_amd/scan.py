@@ -134,6 +134,15 @@ def groupby_scan(
     c.scratch = scratch.data_ptr()
     c.scratch_bytes = nscratch
     c.stream = torch.cuda.current_stream(device).cuda_stream
+    # sorted-labels fast path (the reference's issorted check,
+    # aggregate_flox.py:9-23): nondecreasing in-range codes skip the
+    # device radix sort entirely — the common time-ordered layout
+    if labels.numel() > 1:
+        in_range = bool(
+            ((labels >= 0) & (labels < ngroups)).all().item()
+        ) if labels2 is None else False
+        if in_range and bool((labels[1:] >= labels[:-1]).all().item()):
+            c.flags |= _ffi.FLAG_SORTED_LABELS
     _ffi.check(lib.fh_grouped_scan(ctypes.byref(c), SCAN_OPS[func]))
     for t in (vals, labels, labels2, scratch, out):
         if isinstance(t, torch.Tensor):

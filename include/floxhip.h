@@ -71,6 +71,10 @@ enum fh_flags {
   FH_SKIPNAN = 1 << 0,    /* nan* reductions: NaN values contribute nothing */
   FH_FORCE_LDS = 1 << 1,  /* testing: force the LDS-binned path */
   FH_FORCE_ATOMIC = 1 << 2, /* testing: force the global-atomic path */
+  FH_SORTED_LABELS = 1 << 3, /* caller guarantees labels are nondecreasing and
+                              * all in [0, ngroups): fh_grouped_scan skips the
+                              * radix sort (the reference's issorted fast path,
+                              * aggregate_flox.py:9-23) */
 };
 
 typedef struct fh_call {

@@ -626,3 +626,19 @@ def test_custom_aggregation():
     want2 = wmax - wmin
     want2[9:] = -1.0
     np.testing.assert_allclose(np.asarray(got2), want2, rtol=1e-12, atol=1e-12)
+
+
+@pytest.mark.parametrize("func", ["cumsum", "nancumsum", "ffill", "bfill"])
+def test_scan_sorted_labels_fast_path(func):
+    """Nondecreasing in-range labels take the sort-free scan path and must
+    match the oracle exactly like the sorted-free case."""
+    from oracle import groupby_scan as oracle_scan
+
+    rng = np.random.default_rng(zlib.crc32(f"ss-{func}".encode()))
+    n, ng = 150_000, 300
+    labels = np.sort(rng.integers(0, ng, n))
+    vals = rng.standard_normal(n)
+    vals[rng.random(n) < 0.2] = np.nan
+    want = oracle_scan(vals, labels, func=func, expected_groups=np.arange(ng))
+    got = flox_amd.groupby_scan(vals, labels, func=func, expected_groups=np.arange(ng))
+    np.testing.assert_allclose(np.asarray(got), want, equal_nan=True, rtol=1e-12, atol=1e-10)
