@@ -147,6 +147,22 @@ def grouped_partials(
     c.flags = (FLAG_SKIPNAN if skipnan else 0) | (
         _ffi.FLAG_FORCE_LDS if force_path == 1 else _ffi.FLAG_FORCE_ATOMIC if force_path == 2 else 0
     )
+    # sorted-labels direct path at huge group counts: skips every scatter
+    # pass (~44 -> 12 B/row). A 4K-pair sample rejects random labels for
+    # ~20 us; the full O(n) verification only runs when the sample passes
+    # (the time-ordered layout this targets).
+    if (
+        labels2 is None
+        and force_path == 0
+        and ngroups >= 8192
+        and values.numel() >= 1_000_000
+        and labels.numel() > 1
+    ):
+        idx = torch.randint(0, labels.numel() - 1, (4096,), device=labels.device)
+        if bool((labels[idx + 1] >= labels[idx]).all().item()):
+            ok = bool((labels[1:] >= labels[:-1]).all().item())
+            if ok and bool((labels[0] >= 0).item()) and bool((labels[-1] < ngroups).item()):
+                c.flags |= _ffi.FLAG_SORTED_LABELS
     c.n = values.numel()
     c.ngroups = ngroups
     c.values = values.data_ptr()

@@ -1062,6 +1062,161 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
   }
 }
 
+/* sorted-labels direct path: bucket b's rows are the contiguous range
+ * [base[b], base[b+1]) of the ORIGINAL arrays (no scatter passes at all —
+ * 12 B/row instead of the partition's ~44 B/row). base comes from
+ * k_bucket_bounds; the FH_SORTED_LABELS contract guarantees nondecreasing
+ * in-range labels. */
+template <typename L>
+__global__ void k_bucket_bounds(const L* __restrict__ labels, int64_t n,
+                                int nb, int shift,
+                                uint32_t* __restrict__ base) {
+  const int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b > nb) return;
+  if (b == nb) {
+    base[nb] = (uint32_t)n;
+    return;
+  }
+  const int64_t target = (int64_t)b << shift;
+  int64_t lo = 0, hi = n; /* lower_bound */
+  while (lo < hi) {
+    const int64_t mid = (lo + hi) >> 1;
+    if ((int64_t)labels[mid] < target) lo = mid + 1; else hi = mid;
+  }
+  base[b] = (uint32_t)lo;
+}
+
+template <typename V, typename L, int OPS>
+__launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket_direct(
+    const V* __restrict__ values, const L* __restrict__ labels,
+    const uint32_t* __restrict__ base, int64_t chunk, int gpb, int shift,
+    int64_t ngroups, const double* __restrict__ means, int skipnan,
+    BinLayout lay, void* out_sum, int64_t* out_count, uint32_t* out_present,
+    void* out_min, void* out_max, uint32_t* out_nanflag) {
+  using TR = Traits<V>;
+  using Acc = typename TR::Acc;
+  using SumT = typename std::conditional<
+      (OPS & (B_SSD | B_WELFORD)) != 0, double,
+      typename std::conditional<(OPS & (B_IDXMIN | B_IDXMAX)) != 0, int64_t, Acc>::type>::type;
+  using Enc = typename TR::Enc;
+  constexpr bool IS_PROD = (OPS & B_PROD) != 0;
+
+  const int b = blockIdx.y;
+  const int64_t bkt_begin = base[b];
+  const int64_t bkt_end = base[b + 1];
+  if (bkt_begin >= bkt_end) return;
+  const int64_t gbase = (int64_t)b << shift;
+  const int ng_here = (int)(((gbase + gpb) <= ngroups) ? gpb : (ngroups - gbase));
+
+  extern __shared__ __attribute__((aligned(16))) char smem_rd[];
+  SumT* s_sum = (OPS & (B_SUM | B_SSD | B_PROD | B_IDXMIN | B_IDXMAX))
+                    ? (SumT*)(smem_rd + lay.sum_off) : nullptr;
+  uint32_t* s_cnt = (OPS & B_CNT) ? (uint32_t*)(smem_rd + lay.cnt_off) : nullptr;
+  uint32_t* s_present = (OPS & B_PRESENT) ? (uint32_t*)(smem_rd + lay.present_off) : nullptr;
+  Enc* s_mm = (OPS & (B_MIN | B_MAX)) ? (Enc*)(smem_rd + lay.minmax_off) : nullptr;
+  uint32_t* s_nanflag = (OPS & B_NANFLAG) ? (uint32_t*)(smem_rd + lay.nanflag_off) : nullptr;
+
+  const int tid = threadIdx.x;
+  for (int g = tid; g < ng_here; g += blockDim.x) {
+    if (OPS & (B_SUM | B_SSD)) s_sum[g] = (SumT)0;
+    if (IS_PROD) s_sum[g] = (SumT)1;
+    if (OPS & B_CNT) s_cnt[g] = 0u;
+    if (OPS & B_PRESENT) s_present[g] = 0u;
+    if (OPS & B_MIN) s_mm[g] = (Enc)~(Enc)0;
+    if (OPS & B_MAX) s_mm[g] = (Enc)0;
+    if (OPS & B_NANFLAG) s_nanflag[g] = 0u;
+  }
+  __syncthreads();
+
+  /* grid-stride over the bucket's row chunks (handles any skew). Sorted
+   * labels put a wave's 64 lanes in at most a couple of groups, so naive
+   * per-lane LDS atomics serialize 64-way; instead each wave combines its
+   * run segments with a shuffle scan and only the last lane of each run
+   * touches LDS (one atomic per run per wave). */
+  const int lane = tid & 63;
+  constexpr uint32_t LC_INVALID = 0xFFFFFFFFu;
+  for (int64_t start = bkt_begin + (int64_t)blockIdx.x * chunk; start < bkt_end;
+       start += (int64_t)gridDim.x * chunk) {
+    const int64_t end = (start + chunk < bkt_end) ? start + chunk : bkt_end;
+    for (int64_t i0 = start + (int64_t)(tid & ~63); i0 < end; i0 += blockDim.x) {
+      const int64_t i = i0 + lane;
+      uint32_t lc = LC_INVALID;
+      V v = (V)0;
+      bool vnan = false;
+      if (i < end) {
+        lc = (uint32_t)((int64_t)labels[i] - gbase);
+        if (lc >= (uint32_t)ng_here) {
+          lc = LC_INVALID; /* defensive: contract says in-range */
+        } else {
+          v = values[i];
+          vnan = TR::isnan_(v);
+        }
+      }
+      const bool skiprow = vnan && skipnan;
+      SumT s = (SumT)0;
+      uint32_t cn = 0;
+      Enc mm = (OPS & B_MIN) ? (Enc)~(Enc)0 : (Enc)0; /* combine identity */
+      uint32_t nf = 0;
+      if (lc != LC_INVALID) {
+        if (OPS & B_SUM) s = skiprow ? (SumT)0 : (SumT)(Acc)v;
+        if (IS_PROD) s = skiprow ? (SumT)1 : (SumT)(Acc)v;
+        if (OPS & B_SSD) {
+          if (!skiprow) {
+            const double d = (double)v - means[gbase + lc];
+            s = (SumT)(d * d);
+          }
+        }
+        cn = vnan ? 0u : 1u;
+        if ((OPS & (B_MIN | B_MAX)) && !vnan) mm = TR::enc(v);
+        nf = vnan ? 1u : 0u;
+      } else if (IS_PROD) {
+        s = (SumT)1;
+      }
+#pragma unroll
+      for (int d = 1; d < 64; d <<= 1) {
+        const uint32_t olc = __shfl_up(lc, d, 64);
+        const SumT os = __shfl_up(s, d, 64);
+        const uint32_t ocn = __shfl_up(cn, d, 64);
+        const Enc omm = __shfl_up(mm, d, 64);
+        const uint32_t onf = __shfl_up(nf, d, 64);
+        if (lane >= d && lc != LC_INVALID && olc == lc) {
+          if (OPS & (B_SUM | B_SSD)) s += os;
+          if (IS_PROD) s *= os;
+          cn += ocn;
+          if (OPS & B_MIN) mm = omm < mm ? omm : mm;
+          if (OPS & B_MAX) mm = omm > mm ? omm : mm;
+          nf |= onf;
+        }
+      }
+      const uint32_t nlc = __shfl_down(lc, 1, 64);
+      const bool boundary = (lane == 63) || (nlc != lc);
+      if (boundary && lc != LC_INVALID) {
+        if (OPS & (B_SUM | B_SSD)) acc_add(&s_sum[lc], (Acc)s);
+        if (IS_PROD) acc_mul(&s_sum[lc], (Acc)s);
+        if ((OPS & B_CNT) && cn) atomicAdd(&s_cnt[lc], cn);
+        if (OPS & B_PRESENT) s_present[lc] = 1u;
+        /* gate on cn (count of non-NaN rows in the run): a real extreme can
+         * encode to the combine identity (e.g. INT_MAX under min) */
+        if ((OPS & B_MIN) && cn) enc_min(&s_mm[lc], mm);
+        if ((OPS & B_MAX) && cn) enc_max(&s_mm[lc], mm);
+        if ((OPS & B_NANFLAG) && nf) s_nanflag[lc] = 1u;
+      }
+    }
+  }
+  __syncthreads();
+  for (int g = tid; g < ng_here; g += blockDim.x) {
+    const int64_t o = gbase + g;
+    if ((OPS & (B_SUM | B_SSD)) && s_sum[g] != (SumT)0) acc_add(&((SumT*)out_sum)[o], s_sum[g]);
+    if (IS_PROD && s_sum[g] != (SumT)1) acc_mul(&((SumT*)out_sum)[o], s_sum[g]);
+    if ((OPS & B_CNT) && s_cnt[g])
+      atomicAdd(reinterpret_cast<unsigned long long*>(&out_count[o]), (unsigned long long)s_cnt[g]);
+    if ((OPS & B_PRESENT) && s_present[g]) out_present[o] = 1u;
+    if (OPS & B_MIN) enc_min(&((Enc*)out_min)[o], s_mm[g]);
+    if (OPS & B_MAX) enc_max(&((Enc*)out_max)[o], s_mm[g]);
+    if ((OPS & B_NANFLAG) && s_nanflag[g]) out_nanflag[o] = 1u;
+  }
+}
+
 /* per-bucket LDS-binned reduce of the scattered pairs; flush with global
  * atomics (a few chunks per bucket at most) into the FINAL bins */
 template <typename V, int OPS>
@@ -1556,6 +1711,45 @@ int launch_typed(fh_call* c) {
     }
     c->path_used = 1;
     return 0;
+  }
+
+  /* sorted-labels direct path: skip every scatter pass — each bucket's rows
+   * are already a contiguous range of the original arrays */
+  if ((c->flags & FH_SORTED_LABELS) && !c->labels2 &&
+      !(c->flags & FH_FORCE_ATOMIC) && !(OPS & (B_IDXMIN | B_IDXMAX))) {
+    PartPlan pp = part_plan<V>(c);
+    if (pp.feasible && c->scratch_bytes >= (int64_t)(pp.B + 1) * 4) {
+      uint32_t* base = (uint32_t*)c->scratch;
+      const int bb = (pp.B + 256) / 256;
+      hipLaunchKernelGGL((k_bucket_bounds<L>), dim3(bb), dim3(256), 0, stream,
+                         (const L*)c->labels, c->n, pp.B, pp.shift, base);
+      FH_CHECK(hipGetLastError());
+      int rc = init_outs<V, OPS>(c, c->ngroups, stream);
+      if (rc) return rc;
+      auto kern = k_reduce_bucket_direct<V, L, OPS>;
+      FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                   hipFuncAttributeMaxDynamicSharedMemorySize,
+                                   (int)pp.lay.bytes));
+      const int64_t chunk = 1 << 19;
+      int nchunks = (int)((c->n / pp.B + chunk) / chunk) + 1;
+      if (nchunks > 16) nchunks = 16;
+      hipLaunchKernelGGL(kern, dim3(nchunks, pp.B), dim3(BLOCK_LDS),
+                         pp.lay.bytes, stream, (const V*)c->values,
+                         (const L*)c->labels, base, chunk, pp.gpb, pp.shift,
+                         c->ngroups, c->means, skipnan, pp.lay, c->out_sum,
+                         c->out_count, c->out_present, c->out_min, c->out_max,
+                         c->out_nanflag);
+      FH_CHECK(hipGetLastError());
+      if (OPS & (B_MIN | B_MAX)) {
+        int db = (int)((c->ngroups + 255) / 256);
+        hipLaunchKernelGGL((k_decode<V, OPS>), dim3(db), dim3(256), 0, stream,
+                           c->ngroups, c->out_min, c->out_max, c->out_count,
+                           c->out_present);
+        FH_CHECK(hipGetLastError());
+      }
+      c->path_used = 5;
+      return 0;
+    }
   }
 
   /* huge group counts: bucket-partition path when scratch allows */

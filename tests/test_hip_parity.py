@@ -642,3 +642,18 @@ def test_scan_sorted_labels_fast_path(func):
     want = oracle_scan(vals, labels, func=func, expected_groups=np.arange(ng))
     got = flox_amd.groupby_scan(vals, labels, func=func, expected_groups=np.arange(ng))
     np.testing.assert_allclose(np.asarray(got), want, equal_nan=True, rtol=1e-12, atol=1e-10)
+
+
+@pytest.mark.parametrize("func", ["sum", "nanmean", "min", "nanmax", "count", "var", "prod"])
+def test_sorted_labels_direct_reduce(func):
+    """Sorted in-range labels at huge group counts take the scatter-free
+    direct bucket path and must match the oracle."""
+    rng = np.random.default_rng(zlib.crc32(f"sd-{func}".encode()))
+    n, ng = 2_000_000, 100_000
+    labels = np.sort(rng.integers(0, ng, n))
+    vals = rng.standard_normal(n)
+    vals[rng.random(n) < 0.05] = np.nan
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=range(ng))
+    tol = dict(rtol=0, atol=0) if want.dtype.kind in "iu" else dict(rtol=1e-11, atol=1e-11)
+    np.testing.assert_allclose(np.asarray(got), want, equal_nan=True, **tol)
