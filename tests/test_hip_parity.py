@@ -657,3 +657,47 @@ def test_sorted_labels_direct_reduce(func):
     got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=range(ng))
     tol = dict(rtol=0, atol=0) if want.dtype.kind in "iu" else dict(rtol=1e-11, atol=1e-11)
     np.testing.assert_allclose(np.asarray(got), want, equal_nan=True, **tol)
+
+
+@pytest.mark.parametrize("func", [
+    "sum", "nansum", "mean", "min", "nanmin", "max", "nanmax", "var",
+    "argmin", "argmax", "first", "median", "cumsum", "ffill",
+])
+def test_extreme_values(func):
+    """±inf, denormals, signed zeros, huge magnitudes and NaN mixed —
+    must match the oracle (inf-inf -> NaN sums, inf extremes, etc.)."""
+    from oracle import groupby_scan as oracle_scan
+
+    rng = np.random.default_rng(zlib.crc32(f"ex-{func}".encode()))
+    n, ng = 20_000, 37
+    vals = rng.standard_normal(n)
+    if func in ("sum", "nansum", "mean", "var", "cumsum"):
+        # accumulating funcs: overflow-created infs (1e308+1e308) make ANY
+        # parallel association diverge from the sequential one (inf + -inf
+        # lands at different positions) — inherent to fp, not a bug; keep
+        # the non-overflowing specials
+        specials = np.array([np.inf, -np.inf, np.nan, 0.0, -0.0, 5e-324])
+    else:
+        specials = np.array([np.inf, -np.inf, np.nan, 0.0, -0.0, 1e308,
+                             -1e308, 5e-324, np.finfo(np.float64).max])
+    pos = rng.integers(0, n, 3000)
+    vals[pos] = specials[rng.integers(0, len(specials), 3000)]
+    labels = rng.integers(0, ng, n)
+    eg = np.arange(ng)
+    if func in ("cumsum", "ffill"):
+        want = oracle_scan(vals, labels, func=func, expected_groups=eg)
+        got = flox_amd.groupby_scan(vals, labels, func=func, expected_groups=eg)
+    else:
+        want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=eg)
+        got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=eg)
+    got = np.asarray(got)
+    if func in ("argmin", "argmax"):
+        np.testing.assert_array_equal(got, want)
+    else:
+        # inf-dominated sums are exact; finite ones to fp tolerance
+        finite = np.isfinite(want)
+        np.testing.assert_array_equal(np.isfinite(got), finite)
+        np.testing.assert_array_equal(got[~finite].astype(np.float64),
+                                      want[~finite].astype(np.float64))
+        np.testing.assert_allclose(got[finite], want[finite], rtol=1e-9,
+                                   atol=1e-9 * (1 + np.abs(want[finite]).max(initial=0)))
