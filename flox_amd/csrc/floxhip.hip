@@ -970,7 +970,8 @@ template <typename V>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     const PairT<V>* __restrict__ in, const uint32_t* __restrict__ baseA,
     uint32_t capA /* 0: baseA[sb]..baseA[sb+1]; else sb*capA..baseA[sb] */,
-    int shift /* fine-bucket shift */, uint32_t* __restrict__ cursors,
+    int shift /* fine-bucket shift */, int bfine /* total fine buckets */,
+    uint32_t* __restrict__ cursors,
     uint32_t cap2 /* 0 = exact fine bases */, uint32_t* __restrict__ overflow,
     PairT<V>* __restrict__ out) {
   constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
@@ -988,7 +989,11 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int64_t r0 = capA ? (int64_t)(uint32_t)sb * capA : (int64_t)baseA[sb];
-  const int64_t r1 = capA ? (int64_t)baseA[sb] : (int64_t)baseA[sb + 1];
+  int64_t r1 = capA ? (int64_t)baseA[sb] : (int64_t)baseA[sb + 1];
+  /* pass-A overflow leaves its cursor beyond the capacity region; clamp so
+   * we never read the neighbouring region's (or uninitialized) pairs — the
+   * results are discarded and recomputed by the exact path anyway */
+  if (capA && r1 > r0 + (int64_t)capA) r1 = r0 + (int64_t)capA;
   const uint32_t lmask = (1u << shift) - 1u;
 
   for (int64_t tile = r0 + (int64_t)blockIdx.x * T; tile < r1;
@@ -1007,7 +1012,8 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
       if (idx < nt) {
         const PairT<V> pr = in[tile + idx];
         rv[k] = pr.v;
-        rbk[k] = (int)(pr.lc >> shift);
+        /* mask: stale pairs after a pass-A overflow can carry garbage lc */
+        rbk[k] = (int)((pr.lc >> shift) & (PART_SUB - 1));
         rlc[k] = pr.lc & lmask;
         atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
       }
@@ -1038,12 +1044,19 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
       const uint32_t cnt = nxt - excl;
       if (cnt) {
         const uint32_t fb = (uint32_t)sb * PART_SUB + (uint32_t)b;
-        uint32_t gb = atomicAdd(&cursors[fb], cnt);
-        if (cap2 && gb + cnt > (fb + 1u) * cap2) {
+        if (fb >= (uint32_t)bfine) {
+          /* only reachable via pass-A overflow garbage: flag, park the
+           * rows at the buffer start (results are discarded) */
           *overflow = 1u;
-          gb = fb * cap2;
+          s_gbase[b] = 0u - excl;
+        } else {
+          uint32_t gb = atomicAdd(&cursors[fb], cnt);
+          if (cap2 && gb + cnt > (fb + 1u) * cap2) {
+            *overflow = 1u;
+            gb = fb * cap2;
+          }
+          s_gbase[b] = gb - excl;
         }
-        s_gbase[b] = gb - excl;
       }
     }
     __syncthreads();
@@ -1530,7 +1543,7 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
     if (tiles_x < 1) tiles_x = 1;
     hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK),
                        pp.scatter2_lds, stream, pairs, baseAd, 0u, pp.shift,
-                       cursors, 0u, overflow, pairs2);
+                       pp.B, cursors, 0u, overflow, pairs2);
     FH_CHECK(hipGetLastError());
   }
   {
@@ -1612,7 +1625,7 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     if (tiles_x < 1) tiles_x = 1;
     hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK),
                        pp.scatter2_lds, stream, pairs, cursorsA, pp.cap1,
-                       pp.shift, cursors, pp.cap2, overflow, pairs2);
+                       pp.shift, pp.B, cursors, pp.cap2, overflow, pairs2);
     FH_CHECK(hipGetLastError());
   }
   {

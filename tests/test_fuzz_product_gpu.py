@@ -24,7 +24,9 @@ N_CASES = int(os.environ.get("FUZZ_CASES", "80"))
 def _run_case(i, rng):
     from fuzz_oracle_vs_reference import FUNCS, SCANS, make_case
 
-    arr, by, kw = make_case(rng)
+    # ~20% big cases exercise the partition / packed-arg / sorted-direct /
+    # atomic kernel paths (the small default stays on the LDS path)
+    arr, by, kw = make_case(rng, big=bool(rng.random() < 0.2))
     is_scan = rng.random() < 0.2 and "axis" not in kw and "min_count" not in kw
     if is_scan:
         func = str(rng.choice(SCANS))

@@ -701,3 +701,18 @@ def test_extreme_values(func):
                                       want[~finite].astype(np.float64))
         np.testing.assert_allclose(got[finite], want[finite], rtol=1e-9,
                                    atol=1e-9 * (1 + np.abs(want[finite]).max(initial=0)))
+
+
+def test_partition_pass_a_overflow_regression():
+    """Fuzz-found crash (case 160, seed 424242): a two-level partition whose
+    first super-bucket overflows its optimistic capacity region; pass B must
+    stay in bounds and the exact fallback must produce oracle parity."""
+    rng = np.random.default_rng(160)
+    n, ng = 63_420, 529_450  # ~99% of rows land in super-bucket 0
+    labels = rng.integers(0, ng, n)
+    vals = rng.standard_normal(n)
+    for func in ["sum", "nanvar", "min"]:
+        want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+        got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=range(ng))
+        np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
+                                   rtol=1e-10, atol=1e-10, err_msg=func)

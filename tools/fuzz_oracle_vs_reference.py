@@ -26,8 +26,11 @@ FUNCS = [
 SCANS = ["cumsum", "nancumsum", "ffill", "bfill"]
 
 
-def make_case(rng):
-    n = int(rng.integers(1, 400))
+def make_case(rng, big=False):
+    """big=True draws sizes/group counts that exercise the partition,
+    packed-arg and atomic kernel paths (used by the GPU product fuzz; the
+    CPU oracle fuzz keeps the small default so old seeds reproduce)."""
+    n = int(rng.integers(10_000, 300_000)) if big else int(rng.integers(1, 400))
     dt = rng.choice(["float64", "float32", "int64", "int32", "datetime"])
     if dt == "datetime":
         vals = np.datetime64("2020-01-01") + rng.integers(0, 10**6, n).astype("timedelta64[s]")
@@ -39,7 +42,7 @@ def make_case(rng):
             vals[rng.random(n) < rng.choice([0.05, 0.5, 0.95])] = np.nan
     else:
         vals = rng.integers(-1000, 1000, n).astype(dt)
-    ng = int(rng.integers(1, 25))
+    ng = int(rng.integers(2, 100_000)) if big else int(rng.integers(1, 25))
     shape_kind = rng.choice(["1d", "lead", "multiby", "subset"])
     kw = {}
     if shape_kind == "1d":
