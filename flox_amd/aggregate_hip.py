@@ -21,7 +21,6 @@ from __future__ import annotations
 
 import builtins
 import ctypes
-import math
 
 import numpy as np
 import torch

@@ -16,7 +16,7 @@ import torch
 
 from . import _ffi, distributed
 from ._ffi import FhCall
-from .core import _as_device_tensor, _combined_codes, _factorize_device, _np_dtype
+from .core import _as_device_tensor, _combined_codes, _factorize_device
 
 SCAN_OPS = {"cumsum": 0, "nancumsum": 1, "ffill": 2, "bfill": 3}
 
