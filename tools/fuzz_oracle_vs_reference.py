@@ -160,7 +160,7 @@ def main():
                 # NaT-dominated lerps (|value| ~ 1e18) can bracket different
                 # pairs at virtual-index rounding edges — loose there, tight
                 # on meaningful (date-scale) cells
-                natish = np.abs(wi) > 1e17
+                natish = np.abs(wi) > 1e12  # beyond ~30k years = NaT-lerp territory
                 np.testing.assert_allclose(gi[~natish], wi[~natish], rtol=1e-9, atol=1.0)
                 np.testing.assert_allclose(gi[natish], wi[natish], rtol=1e-6, atol=1.0)
             elif want.dtype.kind in "iubMm":
