@@ -213,6 +213,34 @@ def groupby_reduce(
     Supports: by aligned with the trailing ``by[0].ndim`` dims of ``array``,
     reduction over all dims of by (axis=None or the full trailing tuple).
     """
+    if hasattr(func, "chunk") and hasattr(func, "finalize"):
+        # reference-style custom Aggregation: run each chunk reduction and
+        # apply the user's finalize + final fill (docs "Custom Aggregations")
+        chunk = (func.chunk,) if isinstance(func.chunk, str) else tuple(func.chunk)
+        inters, groups_c, counts_c = [], None, None
+        for cn in chunk:
+            cn2 = "count" if cn == "nanlen" else cn
+            r, *g = groupby_reduce(
+                array, *by, func=cn2, expected_groups=expected_groups,
+                axis=axis, isbin=isbin, sort=sort)
+            inters.append(r)
+            groups_c = g
+            if cn2 == "count":
+                counts_c = r
+        if counts_c is None:
+            counts_c, *_ = groupby_reduce(
+                array, *by, func="count", expected_groups=expected_groups,
+                axis=axis, isbin=isbin, sort=sort)
+        res = func.finalize(*inters) if func.finalize is not None else inters[0]
+        ffv = fill_value if fill_value is not None else getattr(func, "final_fill_value", None)
+        res = np.asarray(res)
+        if ffv is not None and bool(np.any(counts_c == 0)):
+            res = np.where(counts_c == 0, ffv, res)
+        fd = getattr(func, "final_dtype", None)
+        if fd is not None:
+            res = res.astype(fd)
+        return (res, *groups_c)
+
     array = np.asarray(array)
     dt_dtype = None
     if array.dtype.kind in "Mm":

@@ -348,6 +348,30 @@ def main():
     refscan = importlib.import_module("floxref.scan")
     out = {}
     n_done, n_skip = 0, 0
+    # custom Aggregation (docs 'Custom Aggregations'): pin the reference's
+    # chunk/combine/finalize execution; the loader reconstructs the same
+    # CustomAggregation on the flox_amd side (tests/golden_util.py)
+    refaggs = importlib.import_module("floxref.aggregations")
+    c_rng = np.random.default_rng(99)
+    c_vals = c_rng.standard_normal(250)
+    c_by = c_rng.integers(0, 7, 250)
+    c_agg = refaggs.Aggregation(
+        name="custommean", numpy="mean", chunk=("sum", "nanlen"),
+        combine=("sum", "sum"), finalize=lambda s, c: s / c,
+        fill_value=0, final_fill_value=np.nan)
+    try:
+        c_res, c_grp = core.groupby_reduce(
+            c_vals, c_by, func=c_agg, engine="flox", expected_groups=np.arange(9))
+        out["customagg_mean::result"] = np.asarray(c_res)
+        out["customagg_mean::groups0"] = np.asarray(c_grp)
+        out["customagg_mean::array"] = c_vals
+        out["customagg_mean::by0"] = c_by
+        out["customagg_mean::expected0"] = np.arange(9)
+        out["customagg_mean::customagg"] = np.asarray(True)
+        n_done += 1
+    except Exception as e:  # pragma: no cover
+        print(f"SKIP customagg_mean: {type(e).__name__}: {e}")
+        n_skip += 1
     for name, kw in gen_cases():
         by = kw.pop("by")
         bys = by if isinstance(by, tuple) else (by,)

@@ -67,6 +67,14 @@ def load_golden_cases():
         if get("scan") is not None:
             kw["_scan"] = True
             func = name.split("_")[1]
+        if get("customagg") is not None:
+            # reconstruct the fixed custom Aggregation the generator pinned
+            import flox_amd
+
+            func = flox_amd.CustomAggregation(
+                name="custommean", numpy="mean", chunk=("sum", "nanlen"),
+                combine=("sum", "sum"), finalize=lambda s, c: s / c,
+                fill_value=0, final_fill_value=float("nan"))
         yield name, dict(array=get("array"), by=tuple(bys), func=func, **kw), get("result"), groups
 
 
