@@ -38,7 +38,11 @@ def log(msg):
 def cpu_baseline_leg(rows_sample: int, ngroups: int) -> dict:
     """Time the oracle (the CPU restatement of the reference algorithm,
     engine="numpy"/numpy_groupies semantics) on a bounded sample of the same
-    workload, on this box's host cores. Reported baseline, not the target."""
+    workload, on this box's host cores. Reported baseline, not the target.
+    Also times a restatement of the reference's OWN engine="flox" chunk path
+    (stable argsort + gather + np.add.reduceat, aggregate_flox.py:9-23 +
+    133-192 — its dominant cost on this shape) on a smaller
+    sample, since /root/reference itself is absent on the GPU box."""
     from oracle import groupby_reduce as oracle_reduce
 
     rng = np.random.default_rng(1)
@@ -48,12 +52,30 @@ def cpu_baseline_leg(rows_sample: int, ngroups: int) -> dict:
     oracle_reduce(vals, labels, func="mean", expected_groups=np.arange(ngroups))
     dt = time.perf_counter() - t0
     gbps = rows_sample * 12 / dt / 1e9
+
+    fs = min(rows_sample, 5_000_000)
+    fv, fl = vals[:fs], labels[:fs]
+    t1 = time.perf_counter()
+    order = np.argsort(fl, kind="stable")  # _prepare_for_flox
+    sv = fv[order]
+    sc = fl[order]
+    starts = np.flatnonzero(np.concatenate(([True], sc[1:] != sc[:-1])))
+    sums = np.add.reduceat(sv, starts)  # _np_grouped_op
+    counts = np.add.reduceat(np.ones_like(sv), starts)
+    _ = sums / counts
+    dt_flox = time.perf_counter() - t1
     return {
         "value": round(gbps, 4),
         "unit": "GB/s",
         "cores": 1,  # the oracle's numpy ops are single-threaded
         "kind": "port",
         "sample": f"{rows_sample:.0e} of {ROWS:.0e} rows, {dt:.1f}s",
+        "engine_flox_restatement": {
+            "value": round(fs * 12 / dt_flox / 1e9, 4),
+            "unit": "GB/s",
+            "cores": 1,
+            "sample": f"{fs:.0e} rows, {dt_flox:.1f}s (argsort+gather+reduceat)",
+        },
     }
 
 
