@@ -716,3 +716,18 @@ def test_partition_pass_a_overflow_regression():
         got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=range(ng))
         np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
                                    rtol=1e-10, atol=1e-10, err_msg=func)
+
+
+def test_engine_seam_scan_2d():
+    """Seam scan callables accept (..., n) arrays like the reduction ones."""
+    from flox_amd import generic_aggregate
+
+    rng = np.random.default_rng(9)
+    g = rng.integers(0, 6, 400)
+    a = rng.standard_normal((3, 400))
+    a[rng.random(a.shape) < 0.2] = np.nan
+    out = np.asarray(generic_aggregate(g, a, engine="hip", func="ffill", size=6).cpu())
+    assert out.shape == a.shape
+    from oracle import groupby_scan as oracle_scan
+    want = oracle_scan(a, g, func="ffill", expected_groups=np.arange(6))
+    np.testing.assert_allclose(out, want, equal_nan=True, rtol=1e-12, atol=1e-12)
