@@ -50,6 +50,10 @@ def _run_case(i, rng):
     if func in ("prod", "nanprod"):
         if a.dtype.kind in "iuMm":
             arr = np.clip(arr, -1, 1)
+        elif np.asarray(arr).size > 1000:
+            # big groups: fp products overflow/underflow at order-dependent
+            # points; sign-only values keep them exact
+            arr = np.sign(np.asarray(arr))
         else:
             arr = np.clip(arr, -2.0, 2.0)  # keep fp products bounded
     bys = by if isinstance(by, tuple) else (by,)

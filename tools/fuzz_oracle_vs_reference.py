@@ -127,6 +127,10 @@ def main():
                 n_skip += 1
                 continue
             arr = np.clip(arr, -1, 1)
+        elif func in ("prod", "nanprod") and np.asarray(arr).size > 1000:
+            # big groups: fp products overflow/underflow at order-dependent
+            # points; sign-only values keep them exact
+            arr = np.sign(np.asarray(arr))
         bys = by if isinstance(by, tuple) else (by,)
         try:
             if is_scan:
