@@ -783,21 +783,26 @@ def bfill(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, 
     return _scan_like(group_idx, array, "bfill", size=size, dtype=dtype)
 
 
+def _truthy_i64(array: torch.Tensor) -> torch.Tensor:
+    """numpy truthiness as int64: for floats, NaN is truthy (np.any of a
+    NaN-containing group is True), so (v != 0) | isnan(v) — a raw
+    float->int cast would turn NaN into 0 on the GPU."""
+    if array.dtype.is_floating_point:
+        return ((array != 0) | torch.isnan(array)).to(torch.int64)
+    return array.to(torch.int64)
+
+
 def any_(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
     group_idx, array = _prep(group_idx, array)
-    if not array.dtype.is_floating_point:
-        array = array.to(torch.int64)
     ng = _size_of(group_idx, size)
-    p = grouped_partials(SET_MAX_COUNT, array.to(torch.int64) if array.dtype.is_floating_point else array, group_idx, ng)
+    p = grouped_partials(SET_MAX_COUNT, _truthy_i64(array), group_idx, ng)
     return (p["max"] != 0) & (p["count"] > 0)
 
 
 def all_(group_idx, array, *, axis=-1, size=None, fill_value=None, dtype=None, **kw):
     group_idx, array = _prep(group_idx, array)
-    if not array.dtype.is_floating_point:
-        array = array.to(torch.int64)
     ng = _size_of(group_idx, size)
-    p = grouped_partials(SET_MIN_COUNT, array.to(torch.int64) if array.dtype.is_floating_point else array, group_idx, ng)
+    p = grouped_partials(SET_MIN_COUNT, _truthy_i64(array), group_idx, ng)
     return (p["min"] != 0) & (p["count"] > 0)
 
 

@@ -447,6 +447,17 @@ def groupby_reduce(
         out_dtype = np.result_type(out_dtype, fill_value)
 
     dist_on = distributed.is_active() if distributed_combine is None else distributed_combine
+    if dist_on and any(e is None for e in expected_groups):
+        # rank-local factorization (torch.unique of the local shard) yields
+        # per-rank code spaces: the all-reduced bins would be misaligned.
+        # The reference's dask path has the same constraint (its combine
+        # tree reindexes every block to the full expected_groups vector,
+        # dask.py:90-144); discovering groups globally needs a set-union
+        # collective that is not built — require expected_groups instead.
+        raise NotImplementedError(
+            "distributed_combine requires expected_groups for every `by` "
+            "(rank-local group discovery would misalign the combined bins)"
+        )
     ddof = (finalize_kwargs or {}).get("ddof", 0)
     lead_folded = False
 
@@ -696,7 +707,12 @@ def groupby_reduce(
             target = torch.where(p1["nanflag"] != 0, torch.full_like(target, float("nan")), target)
         p2 = run_set(_ffi.SET_IDXMIN, skip, target=target)
         if dist_on:
+            # present/count are rank-local (the kernel marks presence only
+            # for rows this rank holds): combine them too, or a rank with no
+            # rows of a group would fill -1 while others return the index
             distributed.all_reduce_(p2["idx"], "min")
+            distributed.all_reduce_(p2["present"], "max")
+            distributed.all_reduce_(p2["count"], "sum")
         idx = p2["idx"]
         sentinel = (1 << 63) - 1
         result = idx

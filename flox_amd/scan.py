@@ -155,8 +155,18 @@ def groupby_scan(
             codes = labels.to(torch.int64) * grp_pair[1] + labels2.to(torch.int64)
             codes = torch.where((labels < 0) | (labels2 < 0), torch.full_like(codes, -1), codes)
         else:
-            codes = labels
-        out = distributed.scan_carry_exchange(out, vals, codes, ngroups, func)
+            codes = labels.to(torch.int64)
+        # rows outside [0, ngroups) are the reference's NaN-sentinel group
+        # (factorize.py:201-210) — the kernel scans them as one group, so
+        # their carry must cross ranks too: give them a dedicated slot
+        # ngroups (every rank gathers ngroups+1 slots so the collective
+        # shapes agree even when only some ranks hold sentinel rows).
+        # The lead_M>1 layout already folded sentinels in-range; its codes
+        # pass through unchanged and the extra slot stays empty.
+        codes = torch.where(
+            (codes >= 0) & (codes < ngroups), codes, torch.full_like(codes, ngroups)
+        )
+        out = distributed.scan_carry_exchange(out, vals, codes, ngroups + 1, func)
 
     if dtype is not None:
         td = torch.from_numpy(np.empty(0, dtype=np.dtype(dtype))).dtype

@@ -130,6 +130,52 @@ def _fake_partials_factory(np_mod, torch_mod):
             out["sum"] = torch_mod.tensor(np_mod.bincount(l[m], weights=d * d, minlength=ngroups))
         elif op_set == F.SET_COUNT:
             out["count"] = torch_mod.tensor(np_mod.bincount(l[valid & ~nan], minlength=ngroups))
+        elif op_set in (F.SET_MIN_FULL, F.SET_MAX_FULL, F.SET_MIN_COUNT, F.SET_MAX_COUNT):
+            # kernel semantics (floxhip.hip process()): min/max over non-NaN
+            # valid rows (+/-inf when none), count = non-NaN valid rows,
+            # present = any valid row, nanflag = any NaN row (FULL sets)
+            ismin = op_set in (F.SET_MIN_FULL, F.SET_MIN_COUNT)
+            ext = np_mod.full(ngroups, np_mod.inf if ismin else -np_mod.inf)
+            mm = valid & ~nan
+            if ismin:
+                np_mod.minimum.at(ext, l[mm], v[mm].astype("f8"))
+            else:
+                np_mod.maximum.at(ext, l[mm], v[mm].astype("f8"))
+            out["min" if ismin else "max"] = torch_mod.tensor(ext)
+            out["count"] = torch_mod.tensor(np_mod.bincount(l[mm], minlength=ngroups))
+            if op_set in (F.SET_MIN_FULL, F.SET_MAX_FULL):
+                out["present"] = torch_mod.tensor(
+                    (np_mod.bincount(l[valid], minlength=ngroups) > 0).astype("i4"))
+                out["nanflag"] = torch_mod.tensor(
+                    (np_mod.bincount(l[valid & nan], minlength=ngroups) > 0).astype("i4"))
+        elif op_set in (F.SET_IDXMIN, F.SET_IDXMAX):
+            # idx = extreme global row index among candidate rows: valid,
+            # (skipnan -> non-NaN), and matching the per-group target when
+            # given (a NaN target matches NaN rows). Sentinels INT64_MAX / -1.
+            ismin = op_set == F.SET_IDXMIN
+            cand = valid & ~nan if skipnan else valid
+            if target is not None:
+                t = target.numpy()
+                tl = t[np_mod.clip(l, 0, ngroups - 1)]
+                if v.dtype.kind == "f":
+                    match = np_mod.where(
+                        np_mod.isnan(v), np_mod.isnan(tl),
+                        ~np_mod.isnan(tl) & (v == tl))
+                else:
+                    match = v == tl
+                cand = cand & match
+            rows = np_mod.arange(len(v), dtype="i8") + row_offset
+            sent = np_mod.iinfo("i8").max if ismin else -1
+            idx = np_mod.full(ngroups, sent, dtype="i8")
+            if ismin:
+                np_mod.minimum.at(idx, l[cand], rows[cand])
+            else:
+                np_mod.maximum.at(idx, l[cand], rows[cand])
+            out["idx"] = torch_mod.tensor(idx)
+            out["count"] = torch_mod.tensor(
+                np_mod.bincount(l[valid & ~nan], minlength=ngroups))
+            out["present"] = torch_mod.tensor(
+                (np_mod.bincount(l[valid], minlength=ngroups) > 0).astype("i4"))
         else:
             raise NotImplementedError(op_set)
         return out
@@ -151,7 +197,12 @@ def _worker_full_branch(rank, world, port, fail_q):
         n, ng = 20_000, 37
         vals = rng.standard_normal(n)
         vals[rng.random(n) < 0.05] = np.nan
-        labels = rng.integers(0, ng, n)
+        labels = rng.integers(0, ng - 2, n)
+        # groups 35/36 live on exactly one rank each: a rank holding no rows
+        # of a group must still return the group's global result (the
+        # p2-present/count combine in the arg-reduction path)
+        labels[:50] = ng - 2
+        labels[-50:] = ng - 1
         sl = slice(rank * n // world, (rank + 1) * n // world)
 
         fake = _fake_partials_factory(np, torch)
@@ -171,6 +222,27 @@ def _worker_full_branch(rank, world, port, fail_q):
                 np.testing.assert_allclose(
                     got.numpy() if hasattr(got, "numpy") else got, want,
                     equal_nan=True, rtol=1e-10, atol=1e-10)
+            # order-dependent funcs need the global row offset of this shard;
+            # the arg-reductions cover the p2 present/count combine (a rank
+            # holding no rows of a group must still return the global index)
+            for func in ["min", "max", "nanmin", "nanmax",
+                         "argmax", "argmin", "nanargmax", "nanargmin",
+                         "first", "last", "nanfirst", "nanlast"]:
+                got, *_ = core.groupby_reduce(
+                    vals[sl], labels[sl], func=func,
+                    expected_groups=np.arange(ng), shard_row_offset=sl.start)
+                want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+                np.testing.assert_allclose(
+                    got.numpy() if hasattr(got, "numpy") else got, want,
+                    equal_nan=True, rtol=1e-12, atol=0, err_msg=func)
+            # rank-local group discovery is rejected under the distributed
+            # combine (per-rank code spaces would misalign the bins)
+            try:
+                core.groupby_reduce(vals[sl], labels[sl], func="sum")
+            except NotImplementedError:
+                pass
+            else:
+                raise AssertionError("expected NotImplementedError for dist + expected_groups=None")
         finally:
             torch.cuda.is_available = orig_cuda
         dist.destroy_process_group()
@@ -196,6 +268,10 @@ def _worker_scan(rank, world, port, fail_q):
         fvals[rng.random(n) < 0.3] = np.nan  # big NaN runs so carries matter
         ivals = rng.integers(-50, 50, n).astype(np.int64)
         labels = rng.integers(0, ng, n)
+        # ~4% of rows fall outside expected_groups: the reference's
+        # NaN-sentinel group (factorize_:201-210) must carry across ranks
+        # too (via its dedicated slot ngroups, mirroring scan.py's call site)
+        labels[rng.random(n) < 0.04] = ng + 76
         eg = np.arange(ng)
         sl = slice(rank * n // world, (rank + 1) * n // world)
 
@@ -205,11 +281,13 @@ def _worker_scan(rank, world, port, fail_q):
         ]:
             for func in funcs:
                 local = oracle_scan(vals[sl], labels[sl], func=func, expected_groups=eg)
+                codes = labels[sl].copy()
+                codes[(codes < 0) | (codes >= ng)] = ng
                 got = fdist.scan_carry_exchange(
                     torch.from_numpy(np.ascontiguousarray(local)),
                     torch.from_numpy(np.ascontiguousarray(vals[sl])),
-                    torch.from_numpy(np.ascontiguousarray(labels[sl])),
-                    ng,
+                    torch.from_numpy(np.ascontiguousarray(codes)),
+                    ng + 1,
                     func,
                 )
                 want = oracle_scan(vals, labels, func=func, expected_groups=eg)[sl]

@@ -181,6 +181,27 @@ def test_any_all_bool():
         np.testing.assert_array_equal(got, want)
 
 
+def test_any_all_seam_float_nan_truthy():
+    """Through the generic_aggregate seam any/all accept float input; NaN is
+    truthy (np.any of a NaN-containing group is True) — a raw float->int
+    cast would turn NaN into 0 on the GPU."""
+    from flox_amd import generic_aggregate
+
+    rng = np.random.default_rng(41)
+    n, ng = 20_000, 64
+    g = rng.integers(0, ng, n)
+    a = (rng.random(n) < 0.05).astype(np.float64)  # mostly 0.0
+    a[rng.random(n) < 0.03] = np.nan
+    for func in ("any", "all"):
+        got = generic_aggregate(g, a, engine="hip", func=func, size=ng).cpu().numpy()
+        red = np.logical_or if func == "any" else np.logical_and
+        want = np.array([
+            red.reduce((a[g == i] != 0) | np.isnan(a[g == i])) if (g == i).any() else False
+            for i in range(ng)
+        ])
+        np.testing.assert_array_equal(got, want, err_msg=func)
+
+
 def test_partition_overflow_falls_back_exact():
     """Extreme label skew overflows the optimistic capacity regions; the
     exact counted path must kick in and produce correct results."""
