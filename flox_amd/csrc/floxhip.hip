@@ -34,6 +34,7 @@
 #include <cstdint>
 #include <type_traits>
 #include <cstdio>
+#include <cstdlib>
 
 #include "../../include/floxhip.h"
 
@@ -1075,6 +1076,135 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
   }
 }
 
+/* one-level DIRECT scatter: every fine bucket (up to 4096) is a pass-A
+ * target, so the second scatter pass disappears entirely — streamed bytes
+ * drop from ~44 B/row (12r+8w, 8r+8w, 8r) to ~28 B/row (12r+8w, 8r).
+ * There is no LDS staging: a register-blocked tile (RPT rows per thread)
+ * is histogrammed in LDS, each nonempty bucket reserves its span with ONE
+ * returning global atomic, and rows store their pair directly to the
+ * reserved slot. Stores are short per-bucket runs (~T/B rows ≈ 100 B at
+ * 1e7 groups): partial 128-B lines merge in the XCD L2 first (one
+ * workgroup's run lands in one L2) and the remainder merges in the
+ * 256 MiB memory-side Infinity Cache, whose capacity easily holds every
+ * bucket's write frontier (B × ~2 lines ≈ 1 MB) — so the true HBM write
+ * traffic stays ≈8 B/row (verify via WRITE_SIZE ratios).
+ * Without the 73 KB staging, LDS is just the 16 KB histogram and the
+ * kernel runs 3 workgroups/CU instead of 2 — more latency cover for the
+ * scatter's pointer-chasing stores. */
+constexpr int PBD = 512;
+
+template <typename V, typename L>
+__launch_bounds__(PBD, 6) __global__ void k_part_scatter_direct(
+    const V* __restrict__ values, const L* __restrict__ labels,
+    const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
+    int64_t g1, int shift, int B, uint32_t* __restrict__ cursors,
+    uint32_t cap /* 0 = exact bases preloaded in cursors */,
+    uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs) {
+  constexpr int RPT = sizeof(V) == 4 ? 24 : 12;
+  constexpr int T = PBD * RPT; /* 12288 rows (4-B V) / 6144 (8-B V) */
+  extern __shared__ __attribute__((aligned(16))) char smem_pd[];
+  uint32_t* s_hist = (uint32_t*)smem_pd; /* [B] counts -> running cursors */
+  const bool twolab = labels2 != nullptr;
+  const int tid = threadIdx.x;
+  const uint32_t lmask = (1u << shift) - 1u;
+  constexpr uint32_t INV = 0xFFFFFFFFu;
+
+  for (int64_t tile = (int64_t)blockIdx.x * T; tile < n;
+       tile += (int64_t)gridDim.x * T) {
+    const int nt = (int)((n - tile < T) ? (n - tile) : T);
+    for (int b = tid; b < B; b += PBD) s_hist[b] = 0u;
+    __syncthreads();
+
+    V rv[RPT];
+    uint32_t rcode[RPT]; /* full code (bucket<<shift | lc), < 2^25; INV = drop */
+    const int base = tid * RPT;
+    if (base + RPT <= nt) {
+      /* per-thread CONTIGUOUS rows -> 16-B vector loads */
+      constexpr int VW = 16 / (int)sizeof(V);
+#pragma unroll
+      for (int k = 0; k < RPT; k += VW) {
+        const Vec<V, VW> vv =
+            *reinterpret_cast<const Vec<V, VW>*>(values + tile + base + k);
+#pragma unroll
+        for (int j = 0; j < VW; ++j) rv[k + j] = vv.v[j];
+      }
+      if (sizeof(L) == 8) {
+#pragma unroll
+        for (int k = 0; k < RPT; k += 2) {
+          const Vec<L, 2> lv =
+              *reinterpret_cast<const Vec<L, 2>*>(labels + tile + base + k);
+          const int64_t i = tile + base + k;
+          const int64_t c0 = code_of((int64_t)lv.v[0],
+                                     twolab ? (int64_t)labels2[i] : 0, twolab,
+                                     g0, g1, ngroups);
+          const int64_t c1 = code_of((int64_t)lv.v[1],
+                                     twolab ? (int64_t)labels2[i + 1] : 0,
+                                     twolab, g0, g1, ngroups);
+          rcode[k] = c0 >= 0 ? (uint32_t)c0 : INV;
+          rcode[k + 1] = c1 >= 0 ? (uint32_t)c1 : INV;
+          if (c0 >= 0) atomicAdd(&s_hist[rcode[k] >> shift], 1u);
+          if (c1 >= 0) atomicAdd(&s_hist[rcode[k + 1] >> shift], 1u);
+        }
+      } else {
+#pragma unroll
+        for (int k = 0; k < RPT; ++k) {
+          const int64_t i = tile + base + k;
+          const int64_t c0 = code_of((int64_t)labels[i],
+                                     twolab ? (int64_t)labels2[i] : 0, twolab,
+                                     g0, g1, ngroups);
+          rcode[k] = c0 >= 0 ? (uint32_t)c0 : INV;
+          if (c0 >= 0) atomicAdd(&s_hist[rcode[k] >> shift], 1u);
+        }
+      }
+    } else {
+#pragma unroll
+      for (int k = 0; k < RPT; ++k) {
+        const int idx = base + k;
+        rcode[k] = INV;
+        if (idx < nt) {
+          const int64_t i = tile + idx;
+          const int64_t c0 = code_of((int64_t)labels[i],
+                                     twolab ? (int64_t)labels2[i] : 0, twolab,
+                                     g0, g1, ngroups);
+          if (c0 >= 0) {
+            rv[k] = values[i];
+            rcode[k] = (uint32_t)c0;
+            atomicAdd(&s_hist[rcode[k] >> shift], 1u);
+          }
+        }
+      }
+    }
+    __syncthreads();
+    /* one returning global atomic per nonempty bucket reserves the tile's
+     * span; s_hist[b] becomes the running intra-tile cursor */
+    for (int b = tid; b < B; b += PBD) {
+      const uint32_t cnt = s_hist[b];
+      if (cnt) {
+        uint32_t gb = atomicAdd(&cursors[b], cnt);
+        if (cap && gb + cnt > (uint32_t)(b + 1) * cap) {
+          /* optimistic region overflow: flag it and keep writes in-bounds
+           * (results are discarded and recomputed by the exact path) */
+          *overflow = 1u;
+          gb = (uint32_t)b * cap;
+        }
+        s_hist[b] = gb;
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int k = 0; k < RPT; ++k) {
+      if (rcode[k] != INV) {
+        const uint32_t pos = atomicAdd(&s_hist[rcode[k] >> shift], 1u);
+        PairT<V> pr{};
+        pr.v = rv[k];
+        pr.lc = rcode[k] & lmask;
+        pairs[pos] = pr;
+      }
+    }
+    __syncthreads();
+  }
+}
+
 /* sorted-labels direct path: bucket b's rows are the contiguous range
  * [base[b], base[b+1]) of the ORIGINAL arrays (no scatter passes at all —
  * 12 B/row instead of the partition's ~44 B/row). base comes from
@@ -1392,6 +1522,32 @@ struct PartPlan {
   bool feasible;
 };
 
+/* FH_PART_MODE: 0 auto (one-level direct scatter once two levels would be
+ * needed), 1 force the two-level staged path, 2 force direct everywhere.
+ * FH_PART_BINKB: pass-C LDS bin budget in KiB (default 150 of the CU's
+ * 160; larger bins -> fewer, bigger buckets -> fewer reservation atomics). */
+inline int fh_part_mode() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("FH_PART_MODE");
+    v = e ? atoi(e) : 0;
+    if (v < 0 || v > 2) v = 0;
+  }
+  return v;
+}
+
+inline int64_t fh_part_bin_bytes() {
+  static int64_t v = -1;
+  if (v < 0) {
+    const char* e = getenv("FH_PART_BINKB");
+    int kb = e ? atoi(e) : 150;
+    if (kb < 16) kb = 16;
+    v = (int64_t)kb * 1024;
+    if (v > LDS_MAX) v = LDS_MAX;
+  }
+  return v;
+}
+
 template <typename V>
 PartPlan part_plan(const fh_call* c) {
   PartPlan p{};
@@ -1400,10 +1556,10 @@ PartPlan part_plan(const fh_call* c) {
   const int bits = set_bits(c->op_set);
   if (bits & (B_IDXMIN | B_IDXMAX)) return p; /* pairs carry no row index */
   int shift = 13;
-  while (shift > 8 && bin_layout<V>(bits, (int64_t)1 << shift, 4).bytes > 120 * 1024)
+  while (shift > 8 && bin_layout<V>(bits, (int64_t)1 << shift, 4).bytes > fh_part_bin_bytes())
     shift--;
   BinLayout lay = bin_layout<V>(bits, (int64_t)1 << shift, 4);
-  if (lay.bytes > 120 * 1024) return p;
+  if (lay.bytes > fh_part_bin_bytes()) return p;
   int64_t B64 = (c->ngroups + ((int64_t)1 << shift) - 1) >> shift;
   if (B64 > 4096) return p;
   p.B = (int)B64;
@@ -1576,11 +1732,84 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
   return 0;
 }
 
+/* one-level direct partition (optimistic): scatter straight into the
+ * fine-bucket capacity regions with k_part_scatter_direct — no second
+ * scatter pass and no LDS staging — then the per-bucket LDS-binned
+ * reduce. ~28 B/row streamed vs the two-level's ~44. Overflow falls back
+ * to the exact two-level path (the scratch keeps its layout). */
+template <typename V, typename L, int OPS>
+int launch_partition_direct(fh_call* c, const PartPlan& pp) {
+  hipStream_t stream = (hipStream_t)c->stream;
+  const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
+  char* scr = (char*)c->scratch;
+  uint32_t* cursors = (uint32_t*)(scr + pp.cursors_off);
+  uint32_t* overflow = (uint32_t*)(scr + pp.overflow_off);
+  PairT<V>* pairs2 = (PairT<V>*)(scr + pp.pairs2_off);
+
+  FH_CHECK(hipMemsetAsync(overflow, 0, 4, stream));
+  hipLaunchKernelGGL(k_init_cursors, dim3((pp.B + 255) / 256), dim3(256), 0,
+                     stream, cursors, pp.B, pp.cap2);
+  FH_CHECK(hipGetLastError());
+  {
+    auto kern = k_part_scatter_direct<V, L>;
+    const int64_t hist_lds = ((int64_t)pp.B + 63) / 64 * 64 * 4;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)hist_lds));
+    constexpr int T = PBD * (sizeof(V) == 4 ? 24 : 12);
+    int64_t wb = (c->n + T - 1) / T;
+    int nb = (int)(wb < 1536 ? (wb > 0 ? wb : 1) : 1536);
+    hipLaunchKernelGGL(kern, dim3(nb), dim3(PBD), hist_lds, stream,
+                       (const V*)c->values, (const L*)c->labels,
+                       (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
+                       pp.shift, pp.B, cursors, pp.cap2, overflow, pairs2);
+    FH_CHECK(hipGetLastError());
+  }
+  {
+    int rc = init_outs<V, OPS>(c, c->ngroups, stream);
+    if (rc) return rc;
+  }
+  {
+    auto kern = k_reduce_bucket<V, OPS>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)pp.lay.bytes));
+    const int64_t chunk = 1 << 19;
+    int maxchunks = (int)(((int64_t)pp.cap2 + chunk - 1) / chunk);
+    if (maxchunks < 1) maxchunks = 1;
+    hipLaunchKernelGGL(kern, dim3(maxchunks, pp.B), dim3(BLOCK_LDS),
+                       pp.lay.bytes, stream, pairs2, cursors, pp.cap2, chunk,
+                       pp.gpb, pp.shift, c->ngroups, c->means, skipnan, pp.lay,
+                       c->out_sum, c->out_count, c->out_present, c->out_min,
+                       c->out_max, c->out_nanflag);
+    FH_CHECK(hipGetLastError());
+  }
+  if (OPS & (B_MIN | B_MAX)) {
+    int db = (int)((c->ngroups + 255) / 256);
+    hipLaunchKernelGGL((k_decode<V, OPS>), dim3(db), dim3(256), 0, stream,
+                       c->ngroups, c->out_min, c->out_max, c->out_count,
+                       c->out_present);
+    FH_CHECK(hipGetLastError());
+  }
+  uint32_t h_ov = 0;
+  FH_CHECK(hipMemcpyAsync(&h_ov, overflow, 4, hipMemcpyDeviceToHost, stream));
+  FH_CHECK(hipStreamSynchronize(stream));
+  if (h_ov) return launch_partition_exact<V, L, OPS>(c, pp);
+  c->path_used = 6;
+  return 0;
+}
+
 /* optimistic partition: capacity regions replace the counting pre-pass and
  * every host round trip except the final 4-byte overflow check */
 template <typename V, typename L, int OPS>
 int launch_partition(fh_call* c, const PartPlan& pp) {
   if (!pp.optimistic) return launch_partition_exact<V, L, OPS>(c, pp);
+  /* The one-level direct variant is kept for evidence (FH_PART_MODE=2) but
+   * is NOT the default: measured on MI355X (profiles/r02_direct_pmc.md) its
+   * partial-line scatter stores amplify 3.5x (WRITE_SIZE 28.1 GB for 8 GB
+   * of pairs) and the ~1e8 reservation atomics add fabric RMW, so its
+   * effective traffic exceeds the staged two-level's 44 GB/row-stream. */
+  if (fh_part_mode() == 2) return launch_partition_direct<V, L, OPS>(c, pp);
   hipStream_t stream = (hipStream_t)c->stream;
   const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
   char* scr = (char*)c->scratch;
