@@ -882,3 +882,31 @@ _LEAD_KINDS = {
 for _n, _k in _LEAD_KINDS.items():
     globals()[_n] = _lead_wrap(_k)(globals()[_n])
 del _n, _k
+
+
+# ---------------------------------------------------------------------------
+# numpy-in -> numpy-out at the seam: the reference's orchestrator
+# (chunk_reduce / _finalize_results) operates on numpy arrays, so when the
+# caller hands numpy in, hand numpy back (callee-allocates convention,
+# reference aggregate_flox.py:174-180). torch callers keep torch tensors
+# on-device. Outermost wrapper (after the lead-dim fold).
+# ---------------------------------------------------------------------------
+
+
+def _numpy_io_wrap(f):
+    import functools
+
+    @functools.wraps(f)
+    def wrapper(group_idx, array, **kw):
+        np_in = not isinstance(array, torch.Tensor)
+        out = f(group_idx, array, **kw)
+        if np_in and isinstance(out, torch.Tensor):
+            return out.cpu().numpy()
+        return out
+
+    return wrapper
+
+
+for _n in list(_LEAD_KINDS) + ["cumsum", "nancumsum", "ffill", "bfill"]:
+    globals()[_n] = _numpy_io_wrap(globals()[_n])
+del _n
