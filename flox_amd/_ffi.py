@@ -31,6 +31,8 @@ SET_PROD = 8
 SET_IDXMIN = 9
 SET_IDXMAX = 10
 SET_WELFORD = 11
+SET_ARGMIN_PAIR = 12
+SET_ARGMAX_PAIR = 13
 FLAG_SKIPNAN = 1
 FLAG_FORCE_LDS = 2
 FLAG_FORCE_ATOMIC = 4

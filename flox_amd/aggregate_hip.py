@@ -65,6 +65,10 @@ _SET_MEMBERS = {
     SET_IDXMIN: ("idx", "count", "present"),
     SET_IDXMAX: ("idx", "count", "present"),
     SET_WELFORD: ("wssd", "wsum", "count"),  # cols path only
+    # pair-payload arg-reductions (8-B dtypes, partition path only):
+    # idx = int64 row (INT64_MAX empty), min/max = the decoded extremum
+    _ffi.SET_ARGMIN_PAIR: ("idx", "count", "present", "min", "nanflag"),
+    _ffi.SET_ARGMAX_PAIR: ("idx", "count", "present", "max", "nanflag"),
 }
 
 IDX_SENTINEL_MIN = (1 << 63) - 1  # untouched IDXMIN bin

@@ -654,70 +654,106 @@ def groupby_reduce(
             if arr.dtype.is_floating_point
             else counts_for_mask == 0
         )
-    elif func in ("argmax", "argmin", "nanargmax", "nanargmin") and (
-        ngroups > PACKED_ARG_THRESHOLD
-        and vals.dtype in (torch.float32, torch.int32)
-        and not dist_on
-        and 0 < vals.numel()
-        and shard_row_offset + vals.numel() < (1 << 32)
-    ):
-        # huge group counts: pack (order-preserving 32-bit value encoding,
-        # row index) into one int64 key and take a single grouped MIN — the
-        # partition path then handles what the 20 B/group IDX bins cannot
-        # (LDS holds ~8e3 of them). Ties break to the smaller row, which is
-        # exactly np.argmin/argmax's first-occurrence rule.
-        skip = agg.skipnan
-        ismax = "max" in func
-        lib = _ffi.load_library()
-        key = torch.empty(vals.numel(), dtype=torch.int64, device=device)
-        vc = vals.contiguous()
-        _ffi.check(lib.fh_pack_argkeys(
-            vc.data_ptr(),
-            _ffi.F32 if vals.dtype == torch.float32 else _ffi.I32,
-            vc.numel(), shard_row_offset, int(ismax), int(skip),
-            key.data_ptr(),
-            torch.cuda.current_stream(device).cuda_stream,
-        ))
-        vc.record_stream(torch.cuda.current_stream(device))
-        p = grouped_partials(
-            _ffi.SET_MIN_COUNT, key, labels, ngroups,
-            skipnan=False, labels2=labels2, grp_shape=grp_pair,
-        )
-        kmin = p["min"]
-        empty_mask = kmin == ((1 << 63) - 1)
-        result = (kmin ^ (-(1 << 63))) & 0xFFFFFFFF
-        counts_for_mask = p["count"]
     elif func in ("argmax", "argmin", "nanargmax", "nanargmin"):
-        # pass 1: the per-group extremum; pass 2: the smallest row index whose
-        # value matches it (ties -> first occurrence, like np.argmax; a NaN
-        # target matches NaN rows, so non-skip arg* land on the first NaN)
         skip = agg.skipnan
         ismax = "max" in func
-        if skip:
-            p1 = run_set(_ffi.SET_MAX_COUNT if ismax else _ffi.SET_MIN_COUNT, True)
+        base_ok = (
+            ngroups > PACKED_ARG_THRESHOLD
+            and shard_row_offset + vals.numel() < (1 << 32)
+        )
+        # huge group counts, 4-byte dtypes: pack (order-preserving 32-bit
+        # value encoding, row index) into one int64 key and take a single
+        # grouped MIN — the partition path then handles what the 20 B/group
+        # IDX bins cannot (LDS holds ~8e3 of them). Ties break to the
+        # smaller row, which is exactly np.argmin/argmax's first-occurrence
+        # rule. Distributed: the packed key is lexicographic (value, row),
+        # so ONE min all-reduce of the key bins is the exact global answer.
+        packed_ok = (
+            base_ok
+            and vals.dtype in (torch.float32, torch.int32)
+            and (0 < vals.numel() or dist_on)
+        )
+        # 8-byte dtypes: no room to pack — the partition pairs carry the
+        # row in their spare pad word and a second bucket pass takes the
+        # min row among rows matching the group extremum (FH_SET_ARG*_PAIR)
+        pair_ok = (
+            base_ok
+            and not dist_on
+            and vals.dtype in (torch.float64, torch.int64)
+            and 0 < vals.numel() < (1 << 31)
+            and ngroups <= (1 << 24)
+        )
+        if dist_on and distributed.is_active():
+            # the branch choice must agree across ranks (shard sizes differ)
+            okt = torch.tensor([1 if packed_ok else 0], dtype=torch.int32,
+                               device=device)
+            distributed.all_reduce_(okt, "min")
+            packed_ok = bool(okt.item())
+        if packed_ok:
+            lib = _ffi.load_library()
+            key = torch.empty(vals.numel(), dtype=torch.int64, device=device)
+            vc = vals.contiguous()
+            _ffi.check(lib.fh_pack_argkeys(
+                vc.data_ptr(),
+                _ffi.F32 if vals.dtype == torch.float32 else _ffi.I32,
+                vc.numel(), shard_row_offset, int(ismax), int(skip),
+                key.data_ptr(),
+                torch.cuda.current_stream(device).cuda_stream,
+            ))
+            vc.record_stream(torch.cuda.current_stream(device))
+            p = grouped_partials(
+                _ffi.SET_MIN_COUNT, key, labels, ngroups,
+                skipnan=False, labels2=labels2, grp_shape=grp_pair,
+            )
+            if dist_on:
+                distributed.all_reduce_(p["min"], "min")
+                distributed.all_reduce_(p["count"], "sum")
+            kmin = p["min"]
+            empty_mask = kmin == ((1 << 63) - 1)
+            result = (kmin ^ (-(1 << 63))) & 0xFFFFFFFF
+            counts_for_mask = p["count"]
+        elif pair_ok:
+            p = grouped_partials(
+                _ffi.SET_ARGMAX_PAIR if ismax else _ffi.SET_ARGMIN_PAIR,
+                vals, labels, ngroups, skipnan=skip,
+                labels2=labels2, grp_shape=grp_pair,
+                row_offset=shard_row_offset,
+            )
+            idx = p["idx"]
+            empty_mask = idx == ((1 << 63) - 1)
+            result = idx
+            counts_for_mask = p["count"]
         else:
-            p1 = run_set(_ffi.SET_MAX_FULL if ismax else _ffi.SET_MIN_FULL, False)
-        if dist_on:
-            distributed.all_reduce_(p1["max" if ismax else "min"], "max" if ismax else "min")
-            distributed.all_reduce_(p1["count"], "sum")
-            if "nanflag" in p1:
-                distributed.all_reduce_(p1["nanflag"], "max")
-        target = p1["max" if ismax else "min"]
-        if "nanflag" in p1 and arr.dtype.is_floating_point:
-            target = torch.where(p1["nanflag"] != 0, torch.full_like(target, float("nan")), target)
-        p2 = run_set(_ffi.SET_IDXMIN, skip, target=target)
-        if dist_on:
-            # present/count are rank-local (the kernel marks presence only
-            # for rows this rank holds): combine them too, or a rank with no
-            # rows of a group would fill -1 while others return the index
-            distributed.all_reduce_(p2["idx"], "min")
-            distributed.all_reduce_(p2["present"], "max")
-            distributed.all_reduce_(p2["count"], "sum")
-        idx = p2["idx"]
-        sentinel = (1 << 63) - 1
-        result = idx
-        empty_mask = (p2["present"] == 0) | (idx == sentinel)
-        counts_for_mask = p2["count"]
+            # pass 1: the per-group extremum; pass 2: the smallest row index
+            # whose value matches it (ties -> first occurrence, like
+            # np.argmax; a NaN target matches NaN rows, so non-skip arg*
+            # land on the first NaN)
+            if skip:
+                p1 = run_set(_ffi.SET_MAX_COUNT if ismax else _ffi.SET_MIN_COUNT, True)
+            else:
+                p1 = run_set(_ffi.SET_MAX_FULL if ismax else _ffi.SET_MIN_FULL, False)
+            if dist_on:
+                distributed.all_reduce_(p1["max" if ismax else "min"], "max" if ismax else "min")
+                distributed.all_reduce_(p1["count"], "sum")
+                if "nanflag" in p1:
+                    distributed.all_reduce_(p1["nanflag"], "max")
+            target = p1["max" if ismax else "min"]
+            if "nanflag" in p1 and arr.dtype.is_floating_point:
+                target = torch.where(p1["nanflag"] != 0, torch.full_like(target, float("nan")), target)
+            p2 = run_set(_ffi.SET_IDXMIN, skip, target=target)
+            if dist_on:
+                # present/count are rank-local (the kernel marks presence
+                # only for rows this rank holds): combine them too, or a
+                # rank with no rows of a group would fill -1 while others
+                # return the index
+                distributed.all_reduce_(p2["idx"], "min")
+                distributed.all_reduce_(p2["present"], "max")
+                distributed.all_reduce_(p2["count"], "sum")
+            idx = p2["idx"]
+            sentinel = (1 << 63) - 1
+            result = idx
+            empty_mask = (p2["present"] == 0) | (idx == sentinel)
+            counts_for_mask = p2["count"]
     elif func in ("first", "last", "nanfirst", "nanlast") and (
         ngroups > PACKED_ARG_THRESHOLD
         and not dist_on

@@ -64,6 +64,8 @@ enum {
   B_IDXMIN = 256,
   B_IDXMAX = 512,
   B_WELFORD = 1024,
+  B_ARGROW = 2048, /* pair-payload arg-reductions: second bucket pass takes
+                      the min row among rows matching the group extremum */
 };
 
 __host__ __device__ constexpr int set_bits(int op_set) {
@@ -80,6 +82,8 @@ __host__ __device__ constexpr int set_bits(int op_set) {
     case FH_SET_IDXMIN: return B_IDXMIN | B_CNT | B_PRESENT;
     case FH_SET_IDXMAX: return B_IDXMAX | B_CNT | B_PRESENT;
     case FH_SET_WELFORD: return B_WELFORD | B_CNT;
+    case FH_SET_ARGMIN_PAIR: return B_MIN | B_CNT | B_PRESENT | B_NANFLAG | B_ARGROW;
+    case FH_SET_ARGMAX_PAIR: return B_MAX | B_CNT | B_PRESENT | B_NANFLAG | B_ARGROW;
     default: return 0;
   }
 }
@@ -831,13 +835,14 @@ __device__ __forceinline__ uint32_t wave_incl_scan(uint32_t v) {
   return v;
 }
 
-template <typename V, typename L>
+template <typename V, typename L, bool CROW = false>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     const V* __restrict__ values, const L* __restrict__ labels,
     const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
     int64_t g1, int shift, int B, int Bpad /* unused, <= 64 buckets */,
     uint32_t* __restrict__ cursors, uint32_t cap /* 0 = exact bases */,
-    uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs) {
+    uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs,
+    int64_t row_base = 0 /* CROW: global row = row_base + i in .pad */) {
   constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
   constexpr int RPT = T / PART_BLOCK;
   constexpr int NB = 64;
@@ -953,6 +958,8 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
         const uint32_t pos = atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
         s_stage[pos].v = rv[k];
         s_stage[pos].lc = rlc[k];
+        if constexpr (CROW && sizeof(PairT<V>) == 16)
+          s_stage[pos].pad = (uint32_t)(tile + base + k + row_base);
         s_dest[pos] = s_gbase[rbk[k]] + pos;
       }
     }
@@ -967,7 +974,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
  * as k_part_scatter, with a fixed 64-entry histogram. */
 constexpr int PART_SUB = 64;
 
-template <typename V>
+template <typename V, bool CROW = false>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     const PairT<V>* __restrict__ in, const uint32_t* __restrict__ baseA,
     uint32_t capA /* 0: baseA[sb]..baseA[sb+1]; else sb*capA..baseA[sb] */,
@@ -1004,6 +1011,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     __syncthreads();
     V rv[RPT];
     uint32_t rlc[RPT];
+    uint32_t rpd[CROW ? RPT : 1];
     int rbk[RPT];
     const int base = tid * RPT;
 #pragma unroll
@@ -1013,6 +1021,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
       if (idx < nt) {
         const PairT<V> pr = in[tile + idx];
         rv[k] = pr.v;
+        if constexpr (CROW && sizeof(PairT<V>) == 16) rpd[k] = pr.pad;
         /* mask: stale pairs after a pass-A overflow can carry garbage lc */
         rbk[k] = (int)((pr.lc >> shift) & (PART_SUB - 1));
         rlc[k] = pr.lc & lmask;
@@ -1067,6 +1076,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
         const uint32_t pos = atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
         s_stage[pos].v = rv[k];
         s_stage[pos].lc = rlc[k];
+        if constexpr (CROW && sizeof(PairT<V>) == 16) s_stage[pos].pad = rpd[k];
         s_dest[pos] = s_gbase[rbk[k]] + pos;
       }
     }
@@ -1093,13 +1103,14 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
  * scatter's pointer-chasing stores. */
 constexpr int PBD = 512;
 
-template <typename V, typename L>
+template <typename V, typename L, bool CROW = false>
 __launch_bounds__(PBD, 6) __global__ void k_part_scatter_direct(
     const V* __restrict__ values, const L* __restrict__ labels,
     const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
     int64_t g1, int shift, int B, uint32_t* __restrict__ cursors,
     uint32_t cap /* 0 = exact bases preloaded in cursors */,
-    uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs) {
+    uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs,
+    int64_t row_base = 0) {
   constexpr int RPT = sizeof(V) == 4 ? 24 : 12;
   constexpr int T = PBD * RPT; /* 12288 rows (4-B V) / 6144 (8-B V) */
   extern __shared__ __attribute__((aligned(16))) char smem_pd[];
@@ -1198,6 +1209,8 @@ __launch_bounds__(PBD, 6) __global__ void k_part_scatter_direct(
         PairT<V> pr{};
         pr.v = rv[k];
         pr.lc = rcode[k] & lmask;
+        if constexpr (CROW && sizeof(PairT<V>) == 16)
+          pr.pad = (uint32_t)(tile + base + k + row_base);
         pairs[pos] = pr;
       }
     }
@@ -1477,6 +1490,70 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket(
 }
 
 
+/* second bucket pass for the pair-payload arg-reductions (FH_SET_ARG*_PAIR):
+ * re-reads the scattered pairs and keeps, per group, the SMALLEST row index
+ * among rows matching the group's extremum (first occurrence = np.argmin /
+ * np.argmax tie rule; numeric == so +-0.0 match each other exactly as the
+ * reference's target-match pass does). Targets and nanflags are staged into
+ * LDS once per bucket — after the partition they are bucket-local, unlike
+ * the two-pass atomic form's random gathers over the full 80 MB bins. */
+template <typename V>
+__launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket_argrow(
+    const PairT<V>* __restrict__ pairs, const uint32_t* __restrict__ base,
+    uint32_t cap2, int64_t chunk, int gpb, int shift, int64_t ngroups,
+    int skipnan, const V* __restrict__ target,
+    const uint32_t* __restrict__ nanflag, int64_t* __restrict__ out_idx) {
+  using TR = Traits<V>;
+  const int b = blockIdx.y;
+  int64_t bkt_begin, bkt_end;
+  if (cap2) {
+    bkt_begin = (int64_t)(uint32_t)b * cap2;
+    bkt_end = base[b];
+    const int64_t lim = bkt_begin + cap2;
+    if (bkt_end > lim) bkt_end = lim;
+  } else {
+    bkt_begin = base[b];
+    bkt_end = base[b + 1];
+  }
+  const int64_t start = bkt_begin + (int64_t)blockIdx.x * chunk;
+  if (start >= bkt_end) return;
+  const int64_t end = (start + chunk < bkt_end) ? start + chunk : bkt_end;
+  const int64_t gbase = (int64_t)b << shift;
+  const int ng_here = (int)(((gbase + gpb) <= ngroups) ? gpb : (ngroups - gbase));
+
+  extern __shared__ __attribute__((aligned(16))) char smem_ar[];
+  V* s_target = (V*)smem_ar;
+  uint32_t* s_row = (uint32_t*)(smem_ar + (int64_t)gpb * sizeof(V));
+  uint32_t* s_nan = nanflag ? s_row + gpb : nullptr;
+  const int tid = threadIdx.x;
+  for (int g = tid; g < ng_here; g += blockDim.x) {
+    s_target[g] = target[gbase + g];
+    s_row[g] = 0xFFFFFFFFu;
+    if (s_nan) s_nan[g] = nanflag[gbase + g];
+  }
+  __syncthreads();
+  for (int64_t i = start + tid; i < end; i += blockDim.x) {
+    const PairT<V> p = pairs[i];
+    const uint32_t lc = p.lc;
+    const V v = p.v;
+    const bool vnan = TR::isnan_(v);
+    bool match;
+    if (skipnan)
+      match = !vnan && v == s_target[lc];
+    else if (s_nan && s_nan[lc])
+      match = vnan; /* non-skip arg* land on the first NaN, as numpy does */
+    else
+      match = !vnan && v == s_target[lc];
+    if (match) atomicMin(&s_row[lc], p.pad);
+  }
+  __syncthreads();
+  for (int g = tid; g < ng_here; g += blockDim.x) {
+    const uint32_t r = s_row[g];
+    if (r != 0xFFFFFFFFu)
+      idx_min(&out_idx[gbase + g], (int64_t)r);
+  }
+}
+
 /* ---- partition-path host plumbing ---------------------------------------- */
 template <typename V, int OPS>
 int init_outs(fh_call* c, int64_t nbins, hipStream_t stream) {
@@ -1487,6 +1564,14 @@ int init_outs(fh_call* c, int64_t nbins, hipStream_t stream) {
     hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
                        (int64_t*)c->out_sum, nbins,
                        (OPS & B_IDXMIN) ? INT64_MAX : (int64_t)-1);
+    FH_CHECK(hipGetLastError());
+  }
+  if (OPS & B_ARGROW) {
+    /* row-index bins (out_sum as int64): min-row on both argmin and argmax
+     * (first occurrence), sentinel INT64_MAX for empty */
+    int fb = (int)((nbins + 255) / 256);
+    hipLaunchKernelGGL(k_fill_i64, dim3(fb), dim3(256), 0, stream,
+                       (int64_t*)c->out_sum, nbins, INT64_MAX);
     FH_CHECK(hipGetLastError());
   }
   if (OPS & B_PROD) {
@@ -1506,6 +1591,31 @@ int init_outs(fh_call* c, int64_t nbins, hipStream_t stream) {
   if (OPS & B_MAX)
     FH_CHECK(hipMemsetAsync(c->out_max, 0x00, nbins * sizeof(typename TR::Enc), stream));
   if (OPS & B_NANFLAG) FH_CHECK(hipMemsetAsync(c->out_nanflag, 0, nbins * 4, stream));
+  return 0;
+}
+
+/* argrow phase launcher shared by the three partition variants */
+template <typename V, int OPS, typename PP>
+int launch_argrow_phase(fh_call* c, const PP& pp, const PairT<V>* pairs,
+                        const uint32_t* ends, uint32_t capF, int64_t maxspan,
+                        hipStream_t stream) {
+  if constexpr ((OPS & B_ARGROW) != 0 && sizeof(PairT<V>) == 16) {
+    const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
+    const V* target = (const V*)((OPS & B_MIN) ? c->out_min : c->out_max);
+    const uint32_t* nan = skipnan ? nullptr : (const uint32_t*)c->out_nanflag;
+    const int64_t chunk = 1 << 19;
+    int maxchunks = (int)((maxspan + chunk - 1) / chunk);
+    if (maxchunks < 1) maxchunks = 1;
+    const int64_t lds = ((int64_t)pp.gpb * (sizeof(V) + 8) + 255) / 256 * 256;
+    auto kern = k_reduce_bucket_argrow<V>;
+    FH_CHECK(hipFuncSetAttribute((const void*)kern,
+                                 hipFuncAttributeMaxDynamicSharedMemorySize,
+                                 (int)lds));
+    hipLaunchKernelGGL(kern, dim3(maxchunks, pp.B), dim3(BLOCK_LDS), lds,
+                       stream, pairs, ends, capF, chunk, pp.gpb, pp.shift,
+                       c->ngroups, skipnan, target, nan, (int64_t*)c->out_sum);
+    FH_CHECK(hipGetLastError());
+  }
   return 0;
 }
 
@@ -1558,6 +1668,9 @@ PartPlan part_plan(const fh_call* c) {
   if (c->n >= ((int64_t)1 << 31) || c->ngroups <= 0) return p;
   const int bits = set_bits(c->op_set);
   if (bits & (B_IDXMIN | B_IDXMAX)) return p; /* pairs carry no row index */
+  if ((bits & B_ARGROW) &&
+      (sizeof(PairT<V>) != 16 || c->n + c->row_offset >= ((int64_t)1 << 32)))
+    return p; /* pad word (8-byte dtypes) carries the 32-bit row */
   int shift = 13;
   while (shift > 8 && bin_layout<V>(bits, (int64_t)1 << shift, 4).bytes > fh_part_bin_bytes())
     shift--;
@@ -1616,6 +1729,7 @@ template <typename V, typename L, int OPS>
 int launch_partition_exact(fh_call* c, const PartPlan& pp) {
   hipStream_t stream = (hipStream_t)c->stream;
   const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
+  constexpr bool CROW = (OPS & B_ARGROW) != 0 && sizeof(PairT<V>) == 16;
   char* scr = (char*)c->scratch;
   uint32_t* overflow = (uint32_t*)(scr + pp.overflow_off);
   uint32_t* counts = (uint32_t*)(scr + pp.counts_off);
@@ -1673,7 +1787,7 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
   }
 
   {
-    auto kern = k_part_scatter<V, L>;
+    auto kern = k_part_scatter<V, L, CROW>;
     FH_CHECK(hipFuncSetAttribute((const void*)kern,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter_lds));
@@ -1685,14 +1799,15 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
     hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
                        stream, (const V*)c->values, (const L*)c->labels,
                        (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       shA, bA, pp.Bpad, cursors, 0u, overflow, pairs);
+                       shA, bA, pp.Bpad, cursors, 0u, overflow, pairs,
+                       c->row_offset);
     FH_CHECK(hipGetLastError());
   }
   if (pp.two_level) {
     /* fine-bucket cursors, then the in-super-bucket scatter */
     FH_CHECK(hipMemcpyAsync(cursors, h_base, (int64_t)pp.B * 4,
                             hipMemcpyHostToDevice, stream));
-    auto kern2 = k_part_scatter2<V>;
+    auto kern2 = k_part_scatter2<V, CROW>;
     FH_CHECK(hipFuncSetAttribute((const void*)kern2,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter2_lds));
@@ -1731,6 +1846,11 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
                        c->out_present);
     FH_CHECK(hipGetLastError());
   }
+  {
+    int rc = launch_argrow_phase<V, OPS>(c, pp, pp.two_level ? pairs2 : pairs,
+                                         based, 0u, (int64_t)maxc, stream);
+    if (rc) return rc;
+  }
   c->path_used = 4;
   return 0;
 }
@@ -1744,6 +1864,7 @@ template <typename V, typename L, int OPS>
 int launch_partition_direct(fh_call* c, const PartPlan& pp) {
   hipStream_t stream = (hipStream_t)c->stream;
   const int skipnan = (c->flags & FH_SKIPNAN) ? 1 : 0;
+  constexpr bool CROW = (OPS & B_ARGROW) != 0 && sizeof(PairT<V>) == 16;
   char* scr = (char*)c->scratch;
   uint32_t* cursors = (uint32_t*)(scr + pp.cursors_off);
   uint32_t* overflow = (uint32_t*)(scr + pp.overflow_off);
@@ -1754,7 +1875,7 @@ int launch_partition_direct(fh_call* c, const PartPlan& pp) {
                      stream, cursors, pp.B, pp.cap2);
   FH_CHECK(hipGetLastError());
   {
-    auto kern = k_part_scatter_direct<V, L>;
+    auto kern = k_part_scatter_direct<V, L, CROW>;
     const int64_t hist_lds = ((int64_t)pp.B + 63) / 64 * 64 * 4;
     FH_CHECK(hipFuncSetAttribute((const void*)kern,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
@@ -1765,7 +1886,8 @@ int launch_partition_direct(fh_call* c, const PartPlan& pp) {
     hipLaunchKernelGGL(kern, dim3(nb), dim3(PBD), hist_lds, stream,
                        (const V*)c->values, (const L*)c->labels,
                        (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       pp.shift, pp.B, cursors, pp.cap2, overflow, pairs2);
+                       pp.shift, pp.B, cursors, pp.cap2, overflow, pairs2,
+                       c->row_offset);
     FH_CHECK(hipGetLastError());
   }
   {
@@ -1794,6 +1916,11 @@ int launch_partition_direct(fh_call* c, const PartPlan& pp) {
                        c->out_present);
     FH_CHECK(hipGetLastError());
   }
+  {
+    int rc = launch_argrow_phase<V, OPS>(c, pp, pairs2, cursors, pp.cap2,
+                                         (int64_t)pp.cap2, stream);
+    if (rc) return rc;
+  }
   uint32_t h_ov = 0;
   FH_CHECK(hipMemcpyAsync(&h_ov, overflow, 4, hipMemcpyDeviceToHost, stream));
   FH_CHECK(hipStreamSynchronize(stream));
@@ -1807,6 +1934,7 @@ int launch_partition_direct(fh_call* c, const PartPlan& pp) {
 template <typename V, typename L, int OPS>
 int launch_partition(fh_call* c, const PartPlan& pp) {
   if (!pp.optimistic) return launch_partition_exact<V, L, OPS>(c, pp);
+  constexpr bool CROW = (OPS & B_ARGROW) != 0 && sizeof(PairT<V>) == 16;
   /* The one-level direct variant is kept for evidence (FH_PART_MODE=2) but
    * is NOT the default: measured on MI355X (profiles/r02_direct_pmc.md) its
    * partial-line scatter stores amplify 3.5x (WRITE_SIZE 28.1 GB for 8 GB
@@ -1829,7 +1957,7 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
                      cursorsA, bA, capA);
   FH_CHECK(hipGetLastError());
   {
-    auto kern = k_part_scatter<V, L>;
+    auto kern = k_part_scatter<V, L, CROW>;
     FH_CHECK(hipFuncSetAttribute((const void*)kern,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter_lds));
@@ -1840,14 +1968,15 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
     hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
                        stream, (const V*)c->values, (const L*)c->labels,
                        (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       shA, bA, pp.Bpad, cursorsA, capA, overflow, pairs);
+                       shA, bA, pp.Bpad, cursorsA, capA, overflow, pairs,
+                       c->row_offset);
     FH_CHECK(hipGetLastError());
   }
   if (pp.two_level) {
     hipLaunchKernelGGL(k_init_cursors, dim3((pp.B + 255) / 256), dim3(256), 0,
                        stream, cursors, pp.B, pp.cap2);
     FH_CHECK(hipGetLastError());
-    auto kern2 = k_part_scatter2<V>;
+    auto kern2 = k_part_scatter2<V, CROW>;
     FH_CHECK(hipFuncSetAttribute((const void*)kern2,
                                  hipFuncAttributeMaxDynamicSharedMemorySize,
                                  (int)pp.scatter2_lds));
@@ -1889,6 +2018,12 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
                        c->out_present);
     FH_CHECK(hipGetLastError());
   }
+  {
+    int rc = launch_argrow_phase<V, OPS>(
+        c, pp, pp.two_level ? pairs2 : pairs,
+        pp.two_level ? cursors : cursorsA, pp.cap2, (int64_t)pp.cap2, stream);
+    if (rc) return rc;
+  }
   uint32_t h_ov = 0;
   FH_CHECK(hipMemcpyAsync(&h_ov, overflow, 4, hipMemcpyDeviceToHost, stream));
   FH_CHECK(hipStreamSynchronize(stream));
@@ -1905,6 +2040,14 @@ int launch_typed(fh_call* c) {
   hipStream_t stream = (hipStream_t)c->stream;
   const bool skipnan = (c->flags & FH_SKIPNAN) != 0;
   BinLayout lay = bin_layout<V>(OPS, c->ngroups, 4);
+
+  if (OPS & B_ARGROW) {
+    /* pair-payload arg-reductions exist only on the partition path */
+    if (c->flags & (FH_FORCE_LDS | FH_FORCE_ATOMIC)) return 11;
+    PartPlan pp = part_plan<V>(c);
+    if (!pp.feasible || c->scratch_bytes < pp.bytes) return 11;
+    return launch_partition<V, L, OPS>(c, pp);
+  }
 
   bool use_lds = lay.bytes <= LDS_MAX;
   if (c->flags & FH_FORCE_ATOMIC) use_lds = false;
@@ -2073,6 +2216,16 @@ int dispatch_ops(fh_call* c) {
       return launch_typed<V, L, B_IDXMIN | B_CNT | B_PRESENT>(c);
     case B_IDXMAX | B_CNT | B_PRESENT:
       return launch_typed<V, L, B_IDXMAX | B_CNT | B_PRESENT>(c);
+    case B_MIN | B_CNT | B_PRESENT | B_NANFLAG | B_ARGROW:
+      if constexpr (sizeof(V) == 8)
+        return launch_typed<V, L, B_MIN | B_CNT | B_PRESENT | B_NANFLAG | B_ARGROW>(c);
+      else
+        return 11; /* 4-byte dtypes pack (enc32,row32) keys instead */
+    case B_MAX | B_CNT | B_PRESENT | B_NANFLAG | B_ARGROW:
+      if constexpr (sizeof(V) == 8)
+        return launch_typed<V, L, B_MAX | B_CNT | B_PRESENT | B_NANFLAG | B_ARGROW>(c);
+      else
+        return 11;
     default: return 4;
   }
 }
@@ -2398,6 +2551,7 @@ const char* fh_error_string(int code) {
     case 7: return "labels2 given but g0*g1 != ngroups";
     case 8: return "FH_SET_SSD requires means";
     case 9: return "unknown value dtype";
+    case 11: return "arg-pair reduction needs the bucket-partition path (8-byte dtype, n < 2^31, ngroups <= 2^24)";
     default: return code >= 1000 ? hipGetErrorString((hipError_t)(code - 1000)) : "unknown error";
   }
 }

@@ -64,6 +64,20 @@ enum fh_opset {
    * reference (aggregations.py:348-389): ssd -> out_sum (f64),
    * sum -> out_min (f64, pointer reused), count -> out_count. */
   FH_SET_WELFORD = 11,
+  /* single-call arg-reductions for 8-byte value dtypes (f64/i64) at group
+   * counts beyond the LDS IDX bins, where no 32-bit value encoding can
+   * pack into one int64 key: the partition pairs carry the row index in
+   * their spare pad word, phase 1 is the ordinary bucketed MIN/MAX (+count
+   * +present +nanflag), and a second bucket pass (k_reduce_bucket_argrow)
+   * re-reads the SCATTERED pairs and takes the smallest row whose value
+   * equals the group's extremum — target lookups are LDS-local after the
+   * partition, unlike the two-pass form's random 80 MB gathers. Requires
+   * the partition path (returns error 11 otherwise); rows must satisfy
+   * row_offset + n < 2^32. Outputs: out_sum = int64 row index (INT64_MAX
+   * for empty; first occurrence on ties, np.argmin semantics),
+   * out_min/out_max = the decoded extremum, plus count/present/nanflag. */
+  FH_SET_ARGMIN_PAIR = 12,
+  FH_SET_ARGMAX_PAIR = 13,
 };
 
 /* flags */

@@ -242,6 +242,52 @@ def test_partition_path_many_groups(func):
     np.testing.assert_allclose(got, want, equal_nan=True, **_tol(func, want.dtype))
 
 
+@pytest.mark.parametrize("func", ["argmin", "argmax", "nanargmin", "nanargmax"])
+@pytest.mark.parametrize("dtype", ["float64", "int64"])
+def test_pair_arg_reductions_many_groups(func, dtype):
+    """8-byte-dtype arg-reductions at huge group counts: the pair-payload
+    partition path (FH_SET_ARG*_PAIR — row rides the pad word, second
+    bucket pass matches the extremum). Bit-exact incl. first-occurrence
+    ties (int values in a tiny range force heavy ties)."""
+    if "nan" in func and dtype == "int64":
+        pytest.skip("nanarg on ints == arg on ints")
+    rng = np.random.default_rng(zlib.crc32(f"pair-{func}-{dtype}".encode()))
+    n, ng = 1_000_000, 2_000_000
+    labels = rng.integers(0, ng, n)
+    if dtype == "float64":
+        vals = np.round(rng.standard_normal(n), 1)  # quantized -> ties
+        vals[rng.random(n) < 0.02] = np.nan
+    else:
+        vals = rng.integers(-5, 5, n).astype(np.int64)  # massive ties
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    assert got.dtype == want.dtype, (got.dtype, want.dtype)
+    np.testing.assert_allclose(got, want, equal_nan=True, rtol=0, atol=0)
+
+
+def test_pair_arg_matches_two_pass_form():
+    """The pair-payload path must agree bit-for-bit with the LDS two-pass
+    form on the same inputs (threshold lowered to force both)."""
+    from flox_amd import core as fa_core
+
+    rng = np.random.default_rng(55)
+    n, ng = 300_000, 9_000  # just above the packed threshold
+    labels = rng.integers(0, ng, n)
+    vals = rng.standard_normal(n)
+    vals[rng.random(n) < 0.05] = np.nan
+    for func in ["argmin", "nanargmax"]:
+        got_pair, *_ = flox_amd.groupby_reduce(
+            vals, labels, func=func, expected_groups=np.arange(ng))
+        old_thr = fa_core.PACKED_ARG_THRESHOLD
+        fa_core.PACKED_ARG_THRESHOLD = 10**9  # force the two-pass form
+        try:
+            got_two, *_ = flox_amd.groupby_reduce(
+                vals, labels, func=func, expected_groups=np.arange(ng))
+        finally:
+            fa_core.PACKED_ARG_THRESHOLD = old_thr
+        np.testing.assert_array_equal(got_pair, got_two, err_msg=func)
+
+
 def test_engine_seam_callables():
     """The reference-shaped seam: generic_aggregate(engine='hip', func=...)
     (reference flox/aggregations.py:60-133 signature)."""

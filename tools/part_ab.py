@@ -66,6 +66,10 @@ CASES = {
     "sum1e5": lambda: run("sum f32 1e5g", _ffi.SET_SUM_COUNT_PRESENT, 100_000),
     "sumf64": lambda: run("sum f64 1e7g", _ffi.SET_SUM_COUNT_PRESENT,
                           10_000_000, dtype=torch.float64),
+    "argf64": lambda: run("argmin-pair f64 1e7g", _ffi.SET_ARGMIN_PAIR,
+                          10_000_000, dtype=torch.float64),
+    "argi64": lambda: run("argmin-pair i64 1e7g", _ffi.SET_ARGMIN_PAIR,
+                          10_000_000, dtype=torch.int64),
 }
 
 if __name__ == "__main__":
