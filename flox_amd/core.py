@@ -383,20 +383,23 @@ def groupby_reduce(
     ngroups = math.prod(grp_shape)
     labels, labels2, grp_pair = _combined_codes(facs)
 
+    nat_skip_after_fold = False
     if dt_dtype is not None and func in ("count", "nanfirst", "nanlast"):
         # NaT rows are missing for count (verified reference behavior) and
         # for nanfirst/nanlast (xrutils.nanfirst's isnull, xrutils.py:389-397):
         # drop them by invalidating their codes; with leading dims NaT-ness
         # is per (lead, row), so count recounts on a float view instead
-        # (NaT -> NaN; exact, count ignores magnitudes)
+        # (NaT -> NaN; exact, count ignores magnitudes) and nanfirst/nanlast
+        # invalidate the per-(lead, row) composite codes after the lead fold
         if lead_M != 1:
-            if func != "count":
-                raise NotImplementedError("datetime NaT skipping with leading dims: next row")
-            arr = torch.where(
-                arr == torch.iinfo(torch.int64).min,
-                torch.tensor(float("nan"), dtype=torch.float64, device=device),
-                arr.to(torch.float64),
-            )
+            if func == "count":
+                arr = torch.where(
+                    arr == torch.iinfo(torch.int64).min,
+                    torch.tensor(float("nan"), dtype=torch.float64, device=device),
+                    arr.to(torch.float64),
+                )
+            else:
+                nat_skip_after_fold = True
         else:
             natm = arr.reshape(-1) == torch.iinfo(torch.int64).min
             labels = torch.where(natm, torch.full_like(labels, -1), labels)
@@ -528,6 +531,12 @@ def groupby_reduce(
             base_ngroups = ngroups
             ngroups = lead_M * ngroups
             lead_folded = True
+            if nat_skip_after_fold:
+                # exact NaT skipping per (lead, row): the folded composite
+                # codes are per flattened element, so invalidating them IS
+                # the int64-sentinel skip (no lossy float view)
+                natm = vals == torch.iinfo(torch.int64).min
+                labels = torch.where(natm, torch.full_like(labels, -1), labels)
 
             def run_set(op_set, skipnan, means=None, target=None):
                 return grouped_partials(

@@ -294,7 +294,9 @@ def groupby_reduce(
             # globally-found groups
             assert set(ax) <= set(trailing)
             if dt_dtype is not None and func in ("nanfirst", "nanlast"):
-                raise NotImplementedError("datetime NaT skipping with an axis subset")
+                # recurse on the datetime view so each slice re-detects NaT
+                # and applies exact int64-sentinel skipping
+                array = array.view(dt_dtype)
             if dt_dtype is not None and func == "count":
                 # count must skip NaT: recurse on a float view with NaT -> NaN
                 # (count ignores magnitudes, so the f64 cast is exact for it)
@@ -382,6 +384,7 @@ def groupby_reduce(
     ngroups = math.prod(grp_shape)
     codes = _ravel_codes(codes_list, grp_shape) if nby > 1 else codes_list[0]
 
+    nat_null_lead = False
     if dt_dtype is not None and func in ("count", "nanfirst", "nanlast"):
         # NaT rows are missing for these (see datetime note above)
         natm = array.reshape(-1) == np.iinfo(np.int64).min
@@ -393,7 +396,9 @@ def groupby_reduce(
                     array == np.iinfo(np.int64).min, np.nan, array.astype(np.float64)
                 )
             else:
-                raise NotImplementedError("datetime NaT skipping with leading dims")
+                # nanfirst/nanlast: NaT-ness is per (lead, row) — treat the
+                # int64 sentinel as null in the skip mask (exact)
+                nat_null_lead = True
         else:
             codes = np.where(natm, -1, codes)
 
@@ -423,6 +428,8 @@ def groupby_reduce(
     vals2d = vals.reshape(M, -1)
 
     nanmask_v = _isnull(vals2d)
+    if nat_null_lead:
+        nanmask_v = nanmask_v | (vals2d == np.iinfo(np.int64).min)
     valid_code = codes >= 0
 
     acc_dtype = np.float64 if array.dtype.kind in "fc" else np.int64
@@ -609,6 +616,13 @@ def groupby_reduce(
         else:
             out = np.zeros((M, ngroups), dtype=array.dtype)
         seen = np.zeros((M, ngroups), dtype=bool)
+
+        def _nullm(vg):
+            m = _isnull(vg)
+            if nat_null_lead:
+                m = m | (vg == np.iinfo(np.int64).min)
+            return m
+
         for r in range(M):
             row = vals2d[r]
             for g in range(ngroups):
@@ -621,7 +635,7 @@ def groupby_reduce(
                 elif func == "argmin":
                     out[r, g] = rows[np.argmin(vg)]
                 elif func in ("nanargmax", "nanargmin"):
-                    ok = ~_isnull(vg)
+                    ok = ~_isnull(vg)  # NaT participates (int64 view), as the reference
                     if not ok.any():
                         continue  # stays -1
                     sub = vg[ok]
@@ -632,7 +646,7 @@ def groupby_reduce(
                 elif func == "last":
                     out[r, g] = vg[-1]
                 elif func in ("nanfirst", "nanlast"):
-                    ok = ~_isnull(vg)
+                    ok = ~_nullm(vg)
                     if not ok.any():
                         if array.dtype.kind in "fc":
                             out[r, g] = np.nan

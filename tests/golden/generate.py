@@ -225,6 +225,22 @@ def gen_cases():
         yield f"{func}_datetime_nat", dict(
             array=tnat, by=tby, func=func, expected_groups=np.arange(6)
         )
+    # nanfirst/nanlast on datetime with leading dims (NaT skipping is per
+    # (lead, row)) and with an axis subset of by's dims
+    tv2 = (np.datetime64("2022-03-01") + rng.integers(0, 9000, (3, 50)).astype("timedelta64[m]"))
+    tv2[rng.random((3, 50)) < 0.3] = np.datetime64("NaT")
+    tb2 = rng.integers(0, 6, 50)
+    for func in ["nanfirst", "nanlast", "first", "last"]:
+        yield f"{func}_datetime_lead", dict(
+            array=tv2, by=tb2, func=func, expected_groups=np.arange(6)
+        )
+    tv3 = (np.datetime64("2022-03-01") + rng.integers(0, 9000, (4, 5, 30)).astype("timedelta64[m]"))
+    tv3[rng.random((4, 5, 30)) < 0.25] = np.datetime64("NaT")
+    tb3 = rng.integers(0, 6, (4, 5, 30))
+    for func in ["nanfirst", "nanlast"]:
+        yield f"{func}_datetime_axis_subset", dict(
+            array=tv3, by=tb3, func=func, axis=(2,), expected_groups=np.arange(6)
+        )
     # pd.IntervalIndex expected_groups (binning without isbin=True)
     import pandas as pd
     iv_vals = rng.standard_normal(300)
