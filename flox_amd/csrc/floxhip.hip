@@ -535,7 +535,7 @@ constexpr int COLS_TTILE = 2048;
  * its section of the scratch slab, identical in layout to the 1-D path's
  * per-block slab, and k_combine folds the chunks; with one chunk the final
  * bins are written directly (cnt as i64). */
-template <typename V, int OPS, int VC, bool SLAB>
+template <typename V, int OPS, int VC, bool SLAB, bool SKIP>
 __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
     const V* __restrict__ values, const int* __restrict__ codes_sorted,
     const int* __restrict__ perm, int64_t n_t, int64_t m, int64_t ldm,
@@ -639,8 +639,12 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
 
   auto consume = [&](V v, int k, bool lanes_ok) {
     if (!lanes_ok) return;
+    /* SKIP is compile-time (skipnan-templated variants): the skip branch
+     * and the count select drop out of the per-element stream */
     const bool vnan = TR::isnan_(v);
-    if (vnan && skipnan) return;
+    if constexpr (SKIP) {
+      if (vnan) return;
+    }
     if (IS_WEL) {
       if (!w_have[k]) {
         w_x0[k] = (double)v; /* non-skip: a NaN first value poisons the group */
@@ -649,7 +653,7 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
       const double d = (double)v - w_x0[k];
       w_s1[k] += d;
       acc[k] += (SumT)(d * d);
-      cnt[k] += vnan ? 0u : 1u;
+      if constexpr (SKIP) cnt[k] += 1u; else cnt[k] += vnan ? 0u : 1u;
       return;
     }
     if (OPS & B_SUM) acc[k] += (SumT)v;
@@ -658,7 +662,9 @@ __launch_bounds__(COLS_BLOCK) __global__ void k_reduce_cols(
       const double d = (double)v - mean_g[k];
       acc[k] += d * d;
     }
-    if (OPS & B_CNT) cnt[k] += vnan ? 0u : 1u;
+    if (OPS & B_CNT) {
+      if constexpr (SKIP) cnt[k] += 1u; else cnt[k] += vnan ? 0u : 1u;
+    }
     if (OPS & (B_MIN | B_MAX)) {
       if (vnan) {
         if (OPS & B_NANFLAG) nanflag[k] = 1u;
@@ -2355,20 +2361,30 @@ int launch_cols(fh_call* c) {
     return (int)hipGetLastError();
   };
   int rc;
+  const bool sk = skipnan != 0;
   if constexpr (sizeof(V) == 4) {
     if (plan.vc == 8) {
-      rc = slab_mode ? launch(k_reduce_cols<V, OPS, 8, true>)
-                     : launch(k_reduce_cols<V, OPS, 8, false>);
+      rc = slab_mode
+               ? launch(sk ? k_reduce_cols<V, OPS, 8, true, true>
+                           : k_reduce_cols<V, OPS, 8, true, false>)
+               : launch(sk ? k_reduce_cols<V, OPS, 8, false, true>
+                           : k_reduce_cols<V, OPS, 8, false, false>);
       if (rc != 0) return rc + 1000;
       goto cols_launched;
     }
   }
   if (slab_mode)
-    rc = plan.vc > 1 ? launch(k_reduce_cols<V, OPS, Traits<V>::VEC, true>)
-                     : launch(k_reduce_cols<V, OPS, 1, true>);
+    rc = plan.vc > 1
+             ? launch(sk ? k_reduce_cols<V, OPS, Traits<V>::VEC, true, true>
+                         : k_reduce_cols<V, OPS, Traits<V>::VEC, true, false>)
+             : launch(sk ? k_reduce_cols<V, OPS, 1, true, true>
+                         : k_reduce_cols<V, OPS, 1, true, false>);
   else
-    rc = plan.vc > 1 ? launch(k_reduce_cols<V, OPS, Traits<V>::VEC, false>)
-                     : launch(k_reduce_cols<V, OPS, 1, false>);
+    rc = plan.vc > 1
+             ? launch(sk ? k_reduce_cols<V, OPS, Traits<V>::VEC, false, true>
+                         : k_reduce_cols<V, OPS, Traits<V>::VEC, false, false>)
+             : launch(sk ? k_reduce_cols<V, OPS, 1, false, true>
+                         : k_reduce_cols<V, OPS, 1, false, false>);
   if (rc != 0) return rc + 1000;
 cols_launched:;
 
