@@ -33,6 +33,80 @@ def is_active() -> bool:
     return dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1
 
 
+# group-count threshold for the shard-aware sparse combine (the cohorts
+# label-locality idea of reference cohorts.py:109-301 restated for ranks:
+# when each rank's shard touches only a subset of the groups, exchanging
+# compressed (index, value) bins beats all-reducing the full 1e7-group
+# buffers). Below it the dense all-reduce always wins (bins are a few MB).
+SPARSE_NGROUPS = 1 << 20
+# engage only when the gathered compressed traffic is under this fraction
+# of the dense all-reduce's
+SPARSE_FRACTION = 0.25
+
+
+def _sparse_combine(partials: dict[str, torch.Tensor], combine: dict[str, str]) -> bool:
+    """Shard-aware combine: each rank gathers only its TOUCHED bins.
+
+    A bin is touched when any partial differs from its combine identity
+    (count/present/nanflag nonzero, or a NaN-poisoned fp sum — the all-NaN
+    non-skip group whose count is 0). One scalar all-reduce decides whether
+    the locality is real; uniform shards fall back to the dense path
+    (returns False). Exact for every combine op (sum/prod/min/max are
+    applied only at touched indices; untouched bins already hold the
+    identity on every rank)."""
+    names = [n for n in combine if n in partials]
+    if not names:
+        return True
+    t0 = partials[names[0]]
+    mask = torch.zeros(t0.shape, dtype=torch.bool, device=t0.device)
+    for name in names:
+        t = partials[name]
+        if name in ("count", "present", "nanflag"):
+            mask |= t != 0
+        elif t.is_floating_point():
+            mask |= torch.isnan(t)
+    if "count" not in partials and "present" not in partials:
+        return False  # no touch signal: stay dense
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+    idx = mask.nonzero(as_tuple=False).flatten()
+    nnz = torch.tensor([idx.numel()], dtype=torch.int64, device=t0.device)
+    tot = nnz.clone()
+    dist.all_reduce(tot, op=dist.ReduceOp.SUM)
+    mx = nnz.clone()
+    dist.all_reduce(mx, op=dist.ReduceOp.MAX)
+    ngroups = t0.numel()
+    if float(tot.item()) > SPARSE_FRACTION * world * ngroups:
+        return False
+    m = int(mx.item())
+    if m == 0:
+        return True
+    pad_idx = torch.full((m,), -1, dtype=torch.int64, device=t0.device)
+    pad_idx[: idx.numel()] = idx
+    g_idx = [torch.empty_like(pad_idx) for _ in range(world)]
+    dist.all_gather(g_idx, pad_idx)
+    for name in names:
+        t = partials[name]
+        vals = t.flatten()[idx] if idx.numel() else t.new_empty(0)
+        pad_v = t.new_zeros(m)
+        pad_v[: idx.numel()] = vals
+        g_v = [torch.empty_like(pad_v) for _ in range(world)]
+        dist.all_gather(g_v, pad_v)
+        op = combine[name]
+        red = {"sum": "sum", "prod": "prod", "min": "amin", "max": "amax"}[op]
+        flat = t.flatten()
+        for r in range(world):
+            if r == rank:
+                continue
+            gi = g_idx[r]
+            ok = gi >= 0
+            if not bool(ok.any().item()):
+                continue
+            flat.scatter_reduce_(0, gi[ok], g_v[r][ok], reduce=red)
+        partials[name] = flat.view(t.shape)
+    return True
+
+
 def combine_partials(partials: dict[str, torch.Tensor], combine: dict[str, str]) -> dict:
     """All-reduce each partial bin across ranks with its combine op, in place.
 
@@ -41,10 +115,15 @@ def combine_partials(partials: dict[str, torch.Tensor], combine: dict[str, str])
     collective needs no validity mask. Partials sharing a combine op ride
     ONE coalesced all-reduce (mean's sum/count/present = a single
     collective per step instead of three — collective launch latency is
-    the scaling tax at small bin sizes).
+    the scaling tax at small bin sizes). At huge group counts the
+    shard-aware sparse combine is tried first (see _sparse_combine).
     """
     if not is_active():
         return partials
+    first = next((partials[n] for n in combine if n in partials), None)
+    if first is not None and first.numel() >= SPARSE_NGROUPS:
+        if _sparse_combine(partials, combine):
+            return partials
     by_op: dict[str, list[torch.Tensor]] = {}
     for name, op in combine.items():
         t = partials.get(name)

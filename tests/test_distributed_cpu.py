@@ -454,6 +454,112 @@ def test_gloo_distributed_mode():
     assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
 
 
+def _worker_sparse_combine(rank, world, port, fail_q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        import numpy as np
+        import torch
+        import flox_amd.distributed as fdist
+
+        ng = 50_000
+        old_thr = fdist.SPARSE_NGROUPS
+        fdist.SPARSE_NGROUPS = 1024  # engage the sparse path at test sizes
+        try:
+            rng = np.random.default_rng(101)
+            n = 30_000
+            vals = rng.standard_normal(world * n)
+            vals[rng.random(world * n) < 0.02] = np.nan
+            # strong label locality: rank r's labels live in its own slice of
+            # the group space (the cohorts case), plus a few shared groups
+            labels = np.concatenate([
+                np.where(
+                    rng.random(n) < 0.05,
+                    rng.integers(0, 64, n),  # shared groups
+                    rng.integers(r * ng // world, (r + 1) * ng // world, n),
+                )
+                for r in range(world)
+            ])
+            sv, sl = vals[rank * n:(rank + 1) * n], labels[rank * n:(rank + 1) * n]
+
+            def local_partials(v, l, skipnan, nbins=ng):
+                sums = np.zeros(nbins)
+                counts = np.zeros(nbins, dtype=np.int64)
+                present = np.zeros(nbins, dtype=np.int32)
+                mins = np.full(nbins, np.inf)
+                np.add.at(present, l, 1)
+                m = ~np.isnan(v) if skipnan else np.ones(len(v), bool)
+                np.add.at(sums, l[m], np.where(np.isnan(v[m]), np.nan, v[m]))
+                np.add.at(counts, l[~np.isnan(v)], 1)
+                np.minimum.at(mins, l[~np.isnan(v)], v[~np.isnan(v)])
+                return {
+                    "sum": torch.tensor(sums), "count": torch.tensor(counts),
+                    "present": torch.tensor((present > 0).astype(np.int32)),
+                    "min": torch.tensor(mins),
+                }
+
+            combine = {"sum": "sum", "count": "sum", "present": "sum", "min": "min"}
+            # sparse path (auto-engaged: locality makes tot_nnz << world*ng)
+            p_sparse = local_partials(sv, sl, skipnan=False)
+            fdist.combine_partials(p_sparse, combine)
+            # dense path for the same inputs
+            fdist.SPARSE_NGROUPS = 1 << 60
+            p_dense = local_partials(sv, sl, skipnan=False)
+            fdist.combine_partials(p_dense, combine)
+            fdist.SPARSE_NGROUPS = 1024
+            for k in combine:
+                np.testing.assert_allclose(
+                    p_sparse[k].numpy(), p_dense[k].numpy(),
+                    equal_nan=True, rtol=1e-13, atol=0, err_msg=k)
+            # whole-data ground truth
+            whole = local_partials(vals, labels, skipnan=False)
+            np.testing.assert_array_equal(p_sparse["count"].numpy(), whole["count"].numpy())
+            np.testing.assert_allclose(p_sparse["sum"].numpy(), whole["sum"].numpy(),
+                                       equal_nan=True, rtol=1e-12)
+            np.testing.assert_array_equal(p_sparse["min"].numpy(), whole["min"].numpy())
+            # uniform labels must REJECT the sparse path (decision collective
+            # agrees on every rank) and still combine exactly
+            labels_u = np.concatenate([
+                rng2.integers(0, 2000, n) for rng2 in
+                [np.random.default_rng(7 + r) for r in range(world)]
+            ])
+            slu = labels_u[rank * n:(rank + 1) * n]
+            fdist.SPARSE_NGROUPS = 1024
+            p_u = local_partials(sv, slu, skipnan=False, nbins=2000)
+            fdist.combine_partials(p_u, combine)
+            whole_u = local_partials(vals, labels_u, skipnan=False, nbins=2000)
+            np.testing.assert_array_equal(p_u["count"].numpy(), whole_u["count"].numpy())
+            np.testing.assert_allclose(p_u["sum"].numpy(), whole_u["sum"].numpy(),
+                                       equal_nan=True, rtol=1e-12)
+        finally:
+            fdist.SPARSE_NGROUPS = old_thr
+        dist.destroy_process_group()
+    except Exception as e:  # pragma: no cover
+        import traceback
+        fail_q.put(f"rank {rank}: {type(e).__name__}: {e}\n{traceback.format_exc()}")
+        raise
+
+
+def test_gloo_sparse_combine():
+    """Shard-aware sparse combine (cohorts label-locality restated for
+    ranks): compressed touched-bin exchange must equal both the dense
+    all-reduce and the whole-data partials; uniform labels auto-reject."""
+    ctx = mp.get_context("spawn")
+    fail_q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_sparse_combine, args=(r, 2, 29527, fail_q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=240)
+    errs = []
+    while not fail_q.empty():
+        errs.append(fail_q.get())
+    assert not errs, errs[0]
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]
+
+
 def test_gloo_world3_scan_quantile_mode():
     """world_size=3: middle ranks both receive and forward carries/counts."""
     ctx = mp.get_context("spawn")
