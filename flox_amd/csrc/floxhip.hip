@@ -899,7 +899,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
         if (code >= 0) {
           rbk[k] = (int)(code >> shift);
           rlc[k] = (uint32_t)(code - ((int64_t)rbk[k] << shift));
-          atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
+          atomicAdd(&s_slot[wid * NB + rbk[k]], 1u);
         }
       }
     } else {
@@ -916,7 +916,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
             rv[k] = values[i];
             rbk[k] = (int)(code >> shift);
             rlc[k] = (uint32_t)(code - ((int64_t)rbk[k] << shift));
-            atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
+            atomicAdd(&s_slot[wid * NB + rbk[k]], 1u);
           }
         }
       }
@@ -924,7 +924,16 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     __syncthreads();
     /* exclusive scan over the NE slots: wave shuffle scans + wave totals */
     {
-      const uint32_t mine = s_slot[tid];
+      /* s_slot layout is [wave][bucket] (addr = w*NB + b): the per-row
+       * histogram/cursor atomics touch all 32 LDS banks (the previous
+       * [bucket][wave] stride-NW layout mapped random buckets onto only 4
+       * banks). Measured NEUTRAL on the 1e9/1e7 sum (10.60 vs 10.62 ms) —
+       * the wave parking is memory-latency-bound, not LDS-bound — kept for
+       * the clean banking. The scan is still over the semantic
+       * (bucket, wave) order = tid; only this once-per-tile gather reads
+       * transposed. */
+      const int tb = tid / PART_NW, tw = tid % PART_NW;
+      const uint32_t mine = s_slot[tw * NB + tb];
       const uint32_t incl = wave_incl_scan(mine);
       if ((tid & 63) == 63) s_wtot[wid] = incl;
       __syncthreads();
@@ -938,13 +947,13 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
         s_wtot[PART_NW] = run;
       }
       __syncthreads();
-      s_slot[tid] = incl - mine + s_wtot[wid];
+      s_slot[tw * NB + tb] = incl - mine + s_wtot[wid];
     }
     __syncthreads();
     const uint32_t total = s_wtot[PART_NW];
     for (int b = tid; b < B; b += PART_BLOCK) {
-      const uint32_t excl = s_slot[b * PART_NW];
-      const uint32_t nxt = (b + 1 < NB) ? s_slot[(b + 1) * PART_NW] : total;
+      const uint32_t excl = s_slot[b]; /* (b, w=0) in [w][b] layout */
+      const uint32_t nxt = (b + 1 < NB) ? s_slot[b + 1] : total;
       const uint32_t cnt = nxt - excl;
       if (cnt) {
         uint32_t gb = atomicAdd(&cursors[b], cnt);
@@ -961,7 +970,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
 #pragma unroll
     for (int k = 0; k < RPT; ++k) {
       if (rbk[k] >= 0) {
-        const uint32_t pos = atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
+        const uint32_t pos = atomicAdd(&s_slot[wid * NB + rbk[k]], 1u);
         s_stage[pos].v = rv[k];
         s_stage[pos].lc = rlc[k];
         if constexpr (CROW && sizeof(PairT<V>) == 16)
@@ -1031,12 +1040,21 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
         /* mask: stale pairs after a pass-A overflow can carry garbage lc */
         rbk[k] = (int)((pr.lc >> shift) & (PART_SUB - 1));
         rlc[k] = pr.lc & lmask;
-        atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
+        atomicAdd(&s_slot[wid * NB + rbk[k]], 1u);
       }
     }
     __syncthreads();
     {
-      const uint32_t mine = s_slot[tid];
+      /* s_slot layout is [wave][bucket] (addr = w*NB + b): the per-row
+       * histogram/cursor atomics touch all 32 LDS banks (the previous
+       * [bucket][wave] stride-NW layout mapped random buckets onto only 4
+       * banks). Measured NEUTRAL on the 1e9/1e7 sum (10.60 vs 10.62 ms) —
+       * the wave parking is memory-latency-bound, not LDS-bound — kept for
+       * the clean banking. The scan is still over the semantic
+       * (bucket, wave) order = tid; only this once-per-tile gather reads
+       * transposed. */
+      const int tb = tid / PART_NW, tw = tid % PART_NW;
+      const uint32_t mine = s_slot[tw * NB + tb];
       const uint32_t incl = wave_incl_scan(mine);
       if ((tid & 63) == 63) s_wtot[wid] = incl;
       __syncthreads();
@@ -1050,13 +1068,13 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
         s_wtot[PART_NW] = run;
       }
       __syncthreads();
-      s_slot[tid] = incl - mine + s_wtot[wid];
+      s_slot[tw * NB + tb] = incl - mine + s_wtot[wid];
     }
     __syncthreads();
     const uint32_t total = s_wtot[PART_NW];
     for (int b = tid; b < NB; b += PART_BLOCK) {
-      const uint32_t excl = s_slot[b * PART_NW];
-      const uint32_t nxt = (b + 1 < NB) ? s_slot[(b + 1) * PART_NW] : total;
+      const uint32_t excl = s_slot[b]; /* (b, w=0) in [w][b] layout */
+      const uint32_t nxt = (b + 1 < NB) ? s_slot[b + 1] : total;
       const uint32_t cnt = nxt - excl;
       if (cnt) {
         const uint32_t fb = (uint32_t)sb * PART_SUB + (uint32_t)b;
@@ -1079,7 +1097,7 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
 #pragma unroll
     for (int k = 0; k < RPT; ++k) {
       if (rbk[k] >= 0) {
-        const uint32_t pos = atomicAdd(&s_slot[rbk[k] * PART_NW + wid], 1u);
+        const uint32_t pos = atomicAdd(&s_slot[wid * NB + rbk[k]], 1u);
         s_stage[pos].v = rv[k];
         s_stage[pos].lc = rlc[k];
         if constexpr (CROW && sizeof(PairT<V>) == 16) s_stage[pos].pad = rpd[k];
