@@ -1,0 +1,189 @@
+"""Huge-size randomized GPU fuzz: 2e6..2e7-row cases
+through the partition / packed-arg / pair-arg / sorted-direct / atomic /
+overflow-fallback paths at realistic group counts (8e3..1.2e7).
+
+Checkers:
+  * reductions (sum/nansum/mean/nanmean/count/min/nanmin/max/nanmax/var):
+    vectorized numpy oracle (np.bincount / minimum.at with f64 accumulation,
+    the npg contract) — no per-group Python loops, so 2e7 rows is seconds;
+  * arg-reductions: the product's two INDEPENDENT implementations must agree
+    bit-for-bit (partition pair/packed path vs the forced two-pass
+    LDS/atomic form), plus the gathered value at each index must equal the
+    per-group extremum from the numpy oracle.
+
+Label layouts: uniform, sorted (FH_SORTED_LABELS direct path), heavily
+skewed (optimistic-region overflow -> exact fallback).
+
+Usage: FUZZ_HUGE_CASES=N python tools/fuzz_huge_gpu.py [seed]
+"""
+
+import os
+import sys
+import time
+
+sys.path.insert(0, os.environ.get("GRAFT_REPO_ROOT", "/root/repo"))
+import numpy as np  # noqa: E402
+import torch  # noqa: E402
+
+import flox_amd  # noqa: E402
+from flox_amd import core as fa_core  # noqa: E402
+
+N_CASES = int(os.environ.get("FUZZ_HUGE_CASES", "20"))
+
+RED_FUNCS = ["sum", "nansum", "mean", "nanmean", "count", "min", "nanmin",
+             "max", "nanmax", "var", "nanvar"]
+ARG_FUNCS = ["argmin", "argmax", "nanargmin", "nanargmax"]
+
+
+def np_reduce(func, vals, labels, ng):
+    """Vectorized f64-accumulating reference (npg semantics)."""
+    v = vals.astype(np.float64) if vals.dtype.kind == "f" else vals.astype(np.int64)
+    nan = np.isnan(v) if vals.dtype.kind == "f" else np.zeros(len(v), bool)
+    skip = func.startswith("nan") or func == "count"
+    m = ~nan if skip else np.ones(len(v), bool)
+    counts = np.bincount(labels[~nan], minlength=ng)
+    present = np.bincount(labels, minlength=ng) > 0
+    if func == "count":
+        return counts.astype(np.intp), counts == 0
+    if func in ("sum", "nansum"):
+        if vals.dtype.kind == "f":
+            s = np.bincount(labels[m], weights=v[m], minlength=ng)
+        else:
+            s = np.zeros(ng, np.int64)
+            np.add.at(s, labels[m], v[m])
+        if not skip:
+            nansum_mark = np.bincount(labels[nan], minlength=ng) > 0
+            if vals.dtype.kind == "f":
+                s[nansum_mark] = np.nan
+        out = s.astype(vals.dtype if vals.dtype.kind == "f" else np.int64)
+        return out, ~present
+    if func in ("mean", "nanmean"):
+        s = np.bincount(labels[m], weights=v[m], minlength=ng)
+        if not skip and vals.dtype.kind == "f":
+            s[np.bincount(labels[nan], minlength=ng) > 0] = np.nan
+        with np.errstate(invalid="ignore", divide="ignore"):
+            out = s / counts
+        return out.astype(vals.dtype if vals.dtype.kind == "f" else np.float64), counts == 0
+    if func in ("var", "nanvar"):
+        s = np.bincount(labels[m], weights=v[m], minlength=ng)
+        with np.errstate(invalid="ignore", divide="ignore"):
+            mu = s / counts
+        d = v - mu[labels]
+        ssd = np.bincount(labels[m], weights=(d * d)[m], minlength=ng)
+        if not skip and vals.dtype.kind == "f":
+            ssd[np.bincount(labels[nan], minlength=ng) > 0] = np.nan
+        with np.errstate(invalid="ignore", divide="ignore"):
+            out = ssd / counts
+        out[counts == 0] = np.nan
+        return out.astype(vals.dtype if vals.dtype.kind == "f" else np.float64), counts == 0
+    # min / max families
+    ismin = "min" in func
+    ext = np.full(ng, np.inf if ismin else -np.inf)
+    mm = ~nan
+    if ismin:
+        np.minimum.at(ext, labels[mm], v[mm])
+    else:
+        np.maximum.at(ext, labels[mm], v[mm])
+    if not skip and vals.dtype.kind == "f":
+        hasnan = np.bincount(labels[nan], minlength=ng) > 0
+        ext[hasnan] = np.nan
+    out = ext.astype(vals.dtype) if vals.dtype.kind == "f" else ext
+    if vals.dtype.kind != "f":
+        out = np.where(counts > 0, ext, 0).astype(np.int64)
+    empty = counts == 0 if skip else ~present
+    return out, empty
+
+
+def make_labels(rng, n, ng, layout):
+    if layout == "uniform":
+        return rng.integers(0, ng, n)
+    if layout == "sorted":
+        return np.sort(rng.integers(0, ng, n))
+    # skew: most rows in a handful of groups (overflow -> exact fallback)
+    lab = rng.integers(0, max(ng // 1000, 2), n)
+    lab[rng.random(n) < 0.05] = rng.integers(0, ng, int((rng.random(n) < 0.05).sum()))
+    m = rng.random(n) < 0.05
+    lab[m] = rng.integers(0, ng, m.sum())
+    return lab
+
+
+def one_case(i, rng):
+    n = int(rng.integers(2_000_000, 20_000_001))
+    ng = int(rng.choice([8192, 100_000, 1_000_000, 12_000_000]))
+    dtype = str(rng.choice(["float32", "float64", "int64"]))
+    layout = str(rng.choice(["uniform", "uniform", "sorted", "skew"]))
+    labels = make_labels(rng, n, ng, layout)
+    if dtype in ("float32", "float64"):
+        vals = (rng.standard_normal(n) * 100).astype(dtype)
+        if rng.random() < 0.6:
+            vals[rng.random(n) < 0.05] = np.nan
+    else:
+        vals = rng.integers(-1000, 1000, n).astype(np.int64)
+
+    kind = "arg" if rng.random() < 0.35 else "red"
+    t0 = time.perf_counter()
+    if kind == "red":
+        func = str(rng.choice(RED_FUNCS))
+        want, empty = np_reduce(func, vals, labels, ng)
+        got, *_ = flox_amd.groupby_reduce(vals, labels, func=func,
+                                          expected_groups=range(ng))
+        got = np.asarray(got)
+        ok = ~empty
+        if want.dtype.kind == "f":
+            scale = 1 + np.nanmax(np.abs(want[ok]), initial=0.0)
+            rtol, atol = (3e-5, 1e-4 * scale) if want.dtype.itemsize == 4 else (1e-10, 1e-10 * scale)
+            np.testing.assert_allclose(got[ok].astype(np.float64),
+                                       want[ok].astype(np.float64),
+                                       equal_nan=True, rtol=rtol, atol=atol,
+                                       err_msg=f"[{i}] {func} {dtype} {layout} n={n} ng={ng}")
+        else:
+            np.testing.assert_array_equal(got[ok], want[ok],
+                                          err_msg=f"[{i}] {func} {dtype} {layout} n={n} ng={ng}")
+    else:
+        func = str(rng.choice(ARG_FUNCS))
+        if "nan" in func and dtype == "int64":
+            func = func.replace("nan", "")
+        fast, *_ = flox_amd.groupby_reduce(vals, labels, func=func,
+                                           expected_groups=range(ng))
+        old = fa_core.PACKED_ARG_THRESHOLD
+        fa_core.PACKED_ARG_THRESHOLD = 10**18  # force the two-pass form
+        try:
+            slow, *_ = flox_amd.groupby_reduce(vals, labels, func=func,
+                                               expected_groups=range(ng))
+        finally:
+            fa_core.PACKED_ARG_THRESHOLD = old
+        np.testing.assert_array_equal(
+            np.asarray(fast), np.asarray(slow),
+            err_msg=f"[{i}] {func} {dtype} {layout} n={n} ng={ng} (cross-path)")
+        # value-at-index must equal the per-group extremum
+        base = func.replace("arg", "")
+        want_ext, empty = np_reduce(base, vals, labels, ng)
+        idx = np.asarray(fast)
+        ok = ~empty & (idx >= 0) & (idx < n)
+        picked = vals[idx[ok]]
+        w = want_ext[ok]
+        if vals.dtype.kind == "f":
+            np.testing.assert_array_equal(
+                np.isnan(picked), np.isnan(w.astype(np.float64)),
+                err_msg=f"[{i}] {func} nan-pos")
+            fin = ~np.isnan(picked)
+            np.testing.assert_array_equal(picked[fin], w[fin].astype(vals.dtype),
+                                          err_msg=f"[{i}] {func} value@idx")
+        else:
+            np.testing.assert_array_equal(picked, w, err_msg=f"[{i}] {func} value@idx")
+    dt = time.perf_counter() - t0
+    print(f"[{i}] OK {kind}:{func} {dtype} {layout} n={n:.1e} ng={ng:.0e} ({dt:.1f}s)",
+          flush=True)
+
+
+def main():
+    seed = int(sys.argv[1]) if len(sys.argv) > 1 else 20260915
+    rng = np.random.default_rng(seed)
+    for i in range(N_CASES):
+        one_case(i, rng)
+        torch.cuda.empty_cache()
+    print(f"huge fuzz: {N_CASES} cases clean (seed {seed})")
+
+
+if __name__ == "__main__":
+    main()
