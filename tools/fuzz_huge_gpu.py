@@ -101,9 +101,8 @@ def make_labels(rng, n, ng, layout):
         return np.sort(rng.integers(0, ng, n))
     # skew: most rows in a handful of groups (overflow -> exact fallback)
     lab = rng.integers(0, max(ng // 1000, 2), n)
-    lab[rng.random(n) < 0.05] = rng.integers(0, ng, int((rng.random(n) < 0.05).sum()))
     m = rng.random(n) < 0.05
-    lab[m] = rng.integers(0, ng, m.sum())
+    lab[m] = rng.integers(0, ng, int(m.sum()))
     return lab
 
 
