@@ -511,9 +511,16 @@ def groupby_reduce(
             # order-dependent reductions with leading dims: fold the lead
             # index into the group codes (lead*ngroups + code) and run the
             # 1-D machinery over the C-order flattened stream — the
-            # offset-labels trick of reference factorize.py:24-39
-            if subset_keep_shape is not None:
-                raise NotImplementedError(f"{func} with an axis subset: next row")
+            # offset-labels trick of reference factorize.py:24-39. With an
+            # axis subset the two folds COMPOSE: labels already carry the
+            # kept-by-dims offset (ngroups = keep*base), and the extra
+            # leading array dims fold on top of that
+            if subset_keep_shape is not None and func in (
+                "argmax", "argmin", "nanargmax", "nanargmin"
+            ):
+                raise NotImplementedError(
+                    f"{func} with an axis subset and extra leading array dims: next row"
+                )
             if dist_on:
                 raise NotImplementedError(f"distributed {func} with leading dims: next row")
             if N * lead_M >= (1 << 31) or ngroups * lead_M >= (1 << 31):

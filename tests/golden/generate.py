@@ -270,6 +270,18 @@ def gen_cases():
             array=a4, by=b4, func=func, axis=axs,
             expected_groups=np.arange(6), fill_value=-99.0,
         )
+    # order funcs with subset + extra lead dims (the composed folds)
+    a4n = a4.copy()
+    a4n[rng.random(a4.shape) < 0.15] = np.nan
+    yield "median_lead_axis_subset", dict(
+        array=a4, by=b4, func="median", axis=(3,),
+        expected_groups=np.arange(6), fill_value=-99.0,
+    )
+    yield "nanquantile_lead_axis_subset", dict(
+        array=a4n, by=b4, func="nanquantile", axis=(3,),
+        expected_groups=np.arange(6), fill_value=-99.0,
+        finalize_kwargs={"q": 0.75},
+    )
     # sort=False: groups in first-appearance order (factorize.py:96)
     ub = np.array([30, 5, 30, 17, 5, 2, 17, 30, 2, 9])
     uv = rng.standard_normal(10)

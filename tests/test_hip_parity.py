@@ -270,6 +270,31 @@ def test_nanfirst_nanlast_datetime_lead_and_subset(func):
     np.testing.assert_array_equal(got, want)
 
 
+@pytest.mark.parametrize("func", ["mode", "first", "nanlast", "nanmedian",
+                                  "quantile"])
+def test_order_funcs_subset_plus_extra_lead(func):
+    """Order-dependent funcs with an axis subset AND extra leading array
+    dims: the kept-by-dims fold and the lead fold compose (r2; the
+    reference supports this shape — median/quantile golden-pinned, the
+    rest oracle-anchored)."""
+    rng = np.random.default_rng(abs(zlib.crc32(func.encode())) % 2**31)
+    a4 = rng.standard_normal((2, 4, 5, 30))
+    if func != "mode":
+        a4[rng.random(a4.shape) < 0.15] = np.nan
+    else:
+        a4 = np.round(a4)  # repeats so modes exist
+    b4 = rng.integers(0, 6, (4, 5, 30))
+    kw = dict(axis=(3,), expected_groups=np.arange(6), fill_value=-99.0)
+    if func == "quantile":
+        kw["finalize_kwargs"] = {"q": [0.25, 0.75]}
+    want, *_ = oracle_reduce(a4, b4, func=func, **kw)
+    got, *_ = flox_amd.groupby_reduce(a4, b4, func=func, **kw)
+    assert np.asarray(got).shape == want.shape
+    assert np.asarray(got).dtype == want.dtype
+    np.testing.assert_allclose(np.asarray(got), want, equal_nan=True,
+                               rtol=1e-12, atol=1e-12, err_msg=func)
+
+
 @pytest.mark.parametrize("func", ["argmin", "argmax", "nanargmin", "nanargmax"])
 @pytest.mark.parametrize("dtype", ["float64", "int64"])
 def test_pair_arg_reductions_many_groups(func, dtype):
