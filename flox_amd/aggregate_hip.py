@@ -550,6 +550,11 @@ def grouped_mode(
     frequent value, ties -> smallest; NaN propagates unless skipnan)."""
     lib = _ffi.load_library()
     _require_gpu_tensor(values, "values")
+    if values.is_floating_point():
+        # scipy counts by NUMERIC equality: -0.0 and +0.0 are one value, but
+        # their order-preserving encodings differ — canonicalize to +0.0 so
+        # the sorted runs merge
+        values = torch.where(values == 0, torch.zeros_like(values), values)
     values = values.contiguous()
     labels = labels.contiguous()
     if labels.dtype not in _TORCH_LDTYPE:

@@ -250,6 +250,10 @@ def distributed_grouped_mode(vals, codes, ngroups, skipnan):
     device = vals.device
     codes = codes.to(torch.int64)
     ok = (codes >= 0) & (codes < ngroups)
+    if vals.is_floating_point():
+        # mode counts by NUMERIC equality: canonicalize -0.0 -> +0.0 so the
+        # encoded runs merge (scipy.stats.mode treats them as one value)
+        vals = torch.where(vals == 0, torch.zeros_like(vals), vals)
     if skipnan and vals.is_floating_point():
         ok &= ~torch.isnan(vals)
     keys, dec = _enc_full(vals[ok])

@@ -270,6 +270,21 @@ def test_nanfirst_nanlast_datetime_lead_and_subset(func):
     np.testing.assert_array_equal(got, want)
 
 
+def test_mode_signed_zero():
+    """scipy.stats.mode counts -0.0 and +0.0 as ONE value; the encoded-key
+    runs must merge them (regression: the subset+lead test caught split
+    counts flipping multimodal winners)."""
+    rng = np.random.default_rng(77)
+    n, ng = 30_000, 41
+    vals = np.round(rng.standard_normal(n))  # plenty of -0.0 and +0.0
+    labels = rng.integers(0, ng, n)
+    for func in ("mode", "nanmode"):
+        want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+        got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+        np.testing.assert_allclose(got, want, equal_nan=True, rtol=0, atol=0,
+                                   err_msg=func)
+
+
 @pytest.mark.parametrize("func", ["mode", "first", "nanlast", "nanmedian",
                                   "quantile"])
 def test_order_funcs_subset_plus_extra_lead(func):
