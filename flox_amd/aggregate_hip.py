@@ -75,6 +75,14 @@ IDX_SENTINEL_MIN = (1 << 63) - 1  # untouched IDXMIN bin
 IDX_SENTINEL_MAX = -1             # untouched IDXMAX bin
 
 
+
+def _record(t, stream):
+    """record_stream guard: illegal during hipGraph capture (graph-pool
+    allocations have static lifetime there, so it is also unnecessary)."""
+    if isinstance(t, torch.Tensor) and not torch.cuda.is_current_stream_capturing():
+        t.record_stream(stream)
+
+
 def _acc_dtype(value_dtype: torch.dtype) -> torch.dtype:
     """sum/ssd accumulator dtype: f64 for floats (npg contract), i64 for ints."""
     return torch.float64 if value_dtype.is_floating_point else torch.int64
@@ -160,6 +168,7 @@ def grouped_partials(
         and ngroups >= 8192
         and values.numel() >= 1_000_000
         and labels.numel() > 1
+        and not torch.cuda.is_current_stream_capturing()
     ):
         idx = torch.randint(0, labels.numel() - 1, (4096,), device=labels.device)
         if bool((labels[idx + 1] >= labels[idx]).all().item()):
@@ -230,8 +239,7 @@ def grouped_partials(
     # ties lifetime to the stream via recorded events only for torch ops; we
     # record explicitly)
     for t in (values, labels, labels2, means, target, scratch):
-        if isinstance(t, torch.Tensor):
-            t.record_stream(torch.cuda.current_stream(dev))
+        _record(t, torch.cuda.current_stream(dev))
     return out
 
 
@@ -479,8 +487,7 @@ def grouped_partials_cols(
     _ffi.check(lib.fh_grouped_reduce_cols(ctypes.byref(c)))
     out["_path"] = c.path_used  # type: ignore[assignment]
     for t in (values2d, codes_sorted, perm, means, scratch, chunk_t):
-        if isinstance(t, torch.Tensor):
-            t.record_stream(torch.cuda.current_stream(dev))
+        _record(t, torch.cuda.current_stream(dev))
     return out
 
 
@@ -532,8 +539,7 @@ def grouped_quantile(
     c.stream = torch.cuda.current_stream(dev).cuda_stream
     _ffi.check(lib.fh_grouped_quantile(ctypes.byref(c), nq))
     for t in (values, labels, labels2, q_t, scratch, out):
-        if isinstance(t, torch.Tensor):
-            t.record_stream(torch.cuda.current_stream(dev))
+        _record(t, torch.cuda.current_stream(dev))
     return out
 
 
@@ -583,8 +589,7 @@ def grouped_mode(
     c.stream = torch.cuda.current_stream(dev).cuda_stream
     _ffi.check(lib.fh_grouped_quantile(ctypes.byref(c), 0))
     for t in (values, labels, labels2, scratch, out):
-        if isinstance(t, torch.Tensor):
-            t.record_stream(torch.cuda.current_stream(dev))
+        _record(t, torch.cuda.current_stream(dev))
     return out
 
 

@@ -65,11 +65,23 @@ def _torch_dtype(d: np.dtype) -> torch.dtype:
     raise NotImplementedError(f"unsupported dtype {d}")
 
 
+# host staging buffers whose H2D copies get captured into a hipGraph must
+# outlive every replay (the graph re-executes the copy from the same host
+# address) — pin them here while capturing
+_CAPTURE_KEEPALIVE: list = []
+
+
+def _keepalive_if_capturing(t: torch.Tensor) -> torch.Tensor:
+    if torch.cuda.is_current_stream_capturing():
+        _CAPTURE_KEEPALIVE.append(t)
+    return t
+
+
 def _as_device_tensor(x, device) -> torch.Tensor:
     if isinstance(x, torch.Tensor):
         t = x
     else:
-        t = torch.from_numpy(np.ascontiguousarray(x))
+        t = _keepalive_if_capturing(torch.from_numpy(np.ascontiguousarray(x)))
     if t.device != device:
         t = t.to(device, non_blocking=True)
     return t
@@ -100,7 +112,9 @@ def _factorize_bins(flat: torch.Tensor, edges: np.ndarray) -> _FactorizedBy:
     assert edges.ndim == 1 and len(edges) >= 2, "bin edges need >= 2 values"
     nbins = len(edges) - 1
     fl = flat
-    edges_t = torch.from_numpy(np.ascontiguousarray(edges)).to(flat.device)
+    edges_t = _keepalive_if_capturing(
+        torch.from_numpy(np.ascontiguousarray(edges))
+    ).to(flat.device)
     if edges_t.dtype != fl.dtype:
         common = torch.promote_types(edges_t.dtype, fl.dtype)
         edges_t = edges_t.to(common)
@@ -153,7 +167,9 @@ def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
     sorted_expect = np.sort(expect_np) if sort else expect_np
     if sort is False and not np.all(np.diff(expect_np) >= 0):
         raise NotImplementedError("unsorted expected_groups with sort=False")
-    exp_t = torch.from_numpy(np.ascontiguousarray(sorted_expect)).to(flat.device)
+    exp_t = _keepalive_if_capturing(
+        torch.from_numpy(np.ascontiguousarray(sorted_expect))
+    ).to(flat.device)
     if exp_t.dtype != flat.dtype:
         common = torch.promote_types(exp_t.dtype, flat.dtype)
         exp_t = exp_t.to(common)

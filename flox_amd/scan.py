@@ -144,9 +144,10 @@ def groupby_scan(
         if in_range and bool((labels[1:] >= labels[:-1]).all().item()):
             c.flags |= _ffi.FLAG_SORTED_LABELS
     _ffi.check(lib.fh_grouped_scan(ctypes.byref(c), SCAN_OPS[func]))
+    from .aggregate_hip import _record
+
     for t in (vals, labels, labels2, scratch, out):
-        if isinstance(t, torch.Tensor):
-            t.record_stream(torch.cuda.current_stream(device))
+        _record(t, torch.cuda.current_stream(device))
 
     if distributed_combine is None:
         distributed_combine = distributed.is_active()
