@@ -106,7 +106,50 @@ def make_labels(rng, n, ng, layout):
     return lab
 
 
+def cols_case(i, rng):
+    """Column-path (leading-dims) case at realistic shapes: exercises the
+    VC=8/VC=vec/vc=1 kernels and the multi-chunk slab mode (row range split
+    + k_combine fold), checked against the oracle (vectorized per lead
+    row)."""
+    from oracle import groupby_reduce as oracle_reduce
+
+    n_t = int(rng.integers(2_000, 30_001))
+    m = int(rng.integers(64, 1_501))
+    ng = int(rng.choice([8, 128, 1000]))
+    dtype = str(rng.choice(["float32", "float64"]))
+    func = str(rng.choice(["mean", "nanmean", "sum", "var", "nanvar",
+                           "min", "nanmax", "count"]))
+    vals = (rng.standard_normal((m, n_t)) * 10).astype(dtype)
+    if rng.random() < 0.6:
+        vals[rng.random((m, n_t)) < 0.05] = np.nan
+    labels = rng.integers(0, ng, n_t)
+    if rng.random() < 0.3:
+        labels = np.sort(labels)
+    t0 = time.perf_counter()
+    want, *_ = oracle_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=np.arange(ng))
+    got = np.asarray(got)
+    assert got.shape == want.shape and got.dtype == want.dtype
+    if want.dtype.kind == "f":
+        fin = want[np.isfinite(want)]
+        scale = 1 + float(np.max(np.abs(fin), initial=0.0))
+        tol = (dict(rtol=3e-5, atol=1e-4 * scale) if want.dtype.itemsize == 4
+               else dict(rtol=1e-10, atol=1e-10 * scale))
+        np.testing.assert_allclose(got, want, equal_nan=True,
+                                   err_msg=f"[{i}] cols {func} {dtype} ({m}x{n_t}, ng={ng})",
+                                   **tol)
+    else:
+        np.testing.assert_array_equal(got, want,
+                                      err_msg=f"[{i}] cols {func} {dtype}")
+    dt = time.perf_counter() - t0
+    print(f"[{i}] OK cols:{func} {dtype} m={m} n_t={n_t} ng={ng} ({dt:.1f}s)",
+          flush=True)
+
+
 def one_case(i, rng):
+    if rng.random() < 0.25:
+        cols_case(i, rng)
+        return
     n = int(rng.integers(2_000_000, 20_000_001))
     ng = int(rng.choice([8192, 100_000, 1_000_000, 12_000_000]))
     dtype = str(rng.choice(["float32", "float64", "int64"]))
