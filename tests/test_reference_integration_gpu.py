@@ -94,6 +94,36 @@ def _cases():
             array=iv, by=(rng.integers(0, 9, 300),), func=func,
             expected_groups=np.arange(9),
         )
+    # datetime/timedelta through the reference's int64-view machinery
+    tv = (np.datetime64("2021-01-01")
+          + rng.integers(0, 10**6, 500).astype("timedelta64[s]"))
+    tvn = tv.copy()
+    tvn[rng.random(500) < 0.2] = np.datetime64("NaT")
+    byd = rng.integers(0, 9, 500)
+    yield "min_datetime", dict(array=tv, by=(byd,), func="min",
+                               expected_groups=np.arange(9))
+    yield "max_datetime_nat", dict(array=tvn, by=(byd,), func="max",
+                                   expected_groups=np.arange(9))
+    yield "count_datetime_nat", dict(array=tvn, by=(byd,), func="count",
+                                     expected_groups=np.arange(9))
+    yield "median_datetime", dict(array=tv, by=(byd,), func="median",
+                                  expected_groups=np.arange(9))
+    yield "sum_timedelta", dict(
+        array=rng.integers(0, 3600, 500).astype("timedelta64[s]"),
+        by=(byd,), func="sum", expected_groups=np.arange(9))
+    # bin-edge grouping and first-appearance ordering
+    yield "mean_isbin", dict(
+        array=rng.standard_normal(500), by=(rng.standard_normal(500) * 2,),
+        func="mean", expected_groups=np.array([-3.0, -1.0, 0.0, 1.0, 3.0]),
+        isbin=True)
+    yield "sum_nosort", dict(
+        array=rng.standard_normal(500), by=(rng.choice([30, 5, 17, 2, 9], 500),),
+        func="sum", sort=False)
+    # NaN labels drop rows (the reference's sentinel-group machinery)
+    yield "nanvar_nanby", dict(
+        array=rng.standard_normal(500),
+        by=(np.where(rng.random(500) < 0.1, np.nan, byd.astype(float)),),
+        func="nanvar")
 
 
 CASES = list(_cases())
@@ -112,11 +142,14 @@ def test_reference_drives_hip_engine(name, kw):
     want = np.asarray(want)
     got = np.asarray(got)
     assert got.dtype == want.dtype, (got.dtype, want.dtype)
-    np.testing.assert_allclose(
-        got.astype(np.float64, copy=False) if want.dtype.kind in "fc" else got,
-        want.astype(np.float64, copy=False) if want.dtype.kind in "fc" else want,
-        equal_nan=True, err_msg=name, **_tol(kw.get("func", ""), want.dtype),
-    )
+    if want.dtype.kind in "Mm":
+        np.testing.assert_array_equal(got, want, err_msg=name)
+    else:
+        np.testing.assert_allclose(
+            got.astype(np.float64, copy=False) if want.dtype.kind in "fc" else got,
+            want.astype(np.float64, copy=False) if want.dtype.kind in "fc" else want,
+            equal_nan=True, err_msg=name, **_tol(kw.get("func", ""), want.dtype),
+        )
 
 
 ORACLE_BASELINED = [
