@@ -725,14 +725,15 @@ def groupby_reduce(
             lib = _ffi.load_library()
             key = torch.empty(vals.numel(), dtype=torch.int64, device=device)
             vc = vals.contiguous()
-            _ffi.check(lib.fh_pack_argkeys(
-                vc.data_ptr(),
-                _ffi.F32 if vals.dtype == torch.float32 else _ffi.I32,
-                vc.numel(), shard_row_offset, int(ismax), int(skip),
-                key.data_ptr(),
-                torch.cuda.current_stream(device).cuda_stream,
-            ))
-            vc.record_stream(torch.cuda.current_stream(device))
+            if vc.numel():  # a zero-row dist rank still joins the collectives
+                _ffi.check(lib.fh_pack_argkeys(
+                    vc.data_ptr(),
+                    _ffi.F32 if vals.dtype == torch.float32 else _ffi.I32,
+                    vc.numel(), shard_row_offset, int(ismax), int(skip),
+                    key.data_ptr(),
+                    torch.cuda.current_stream(device).cuda_stream,
+                ))
+                vc.record_stream(torch.cuda.current_stream(device))
             p = grouped_partials(
                 _ffi.SET_MIN_COUNT, key, labels, ngroups,
                 skipnan=False, labels2=labels2, grp_shape=grp_pair,
