@@ -333,6 +333,23 @@ def test_pair_arg_reductions_many_groups(func, dtype):
     np.testing.assert_allclose(got, want, equal_nan=True, rtol=0, atol=0)
 
 
+def test_pair_arg_multi_by_fused():
+    """Pair-payload f64 arg-reductions through the fused 2-D groupby path
+    (labels2 codes ravel in-kernel) at group counts past the threshold."""
+    rng = np.random.default_rng(91)
+    n = 400_000
+    by_a = rng.integers(0, 40, n)
+    by_b = rng.integers(0, 300, n)  # 12000 groups > PACKED_ARG_THRESHOLD
+    vals = np.round(rng.standard_normal(n), 1)
+    vals[rng.random(n) < 0.03] = np.nan
+    eg = (np.arange(40), np.arange(300))
+    for func in ["argmin", "nanargmax"]:
+        want, *_ = oracle_reduce(vals, by_a, by_b, func=func, expected_groups=eg)
+        got, *_ = flox_amd.groupby_reduce(vals, by_a, by_b, func=func, expected_groups=eg)
+        assert np.asarray(got).shape == want.shape
+        np.testing.assert_array_equal(np.asarray(got), want, err_msg=func)
+
+
 def test_pair_arg_matches_two_pass_form():
     """The pair-payload path must agree bit-for-bit with the LDS two-pass
     form on the same inputs (threshold lowered to force both)."""
