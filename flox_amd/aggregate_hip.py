@@ -249,11 +249,31 @@ def grouped_partials(
 
 
 def _prep(group_idx, array):
-    """Coerce engine-seam inputs (torch CUDA tensors; numpy moves to GPU)."""
+    """Coerce engine-seam inputs (torch CUDA tensors; numpy moves to GPU).
+
+    Small/unsigned/half dtypes promote (the reference itself casts arrays to
+    the aggregation's intermediate dtype before calling the engine, so what
+    arrives here for e.g. a uint8 sum is already uint64): uint64 computes on
+    its int64 VIEW — sums are wrap-exact mod 2^64; min/max are correct for
+    values < 2^63 (documented) — and u8..u32/i8/i16 promote exactly."""
     if not isinstance(array, torch.Tensor):
-        array = torch.as_tensor(np.ascontiguousarray(array))
+        a_np = np.ascontiguousarray(array)
+        if a_np.dtype == np.uint64:
+            a_np = a_np.view(np.int64)
+        elif a_np.dtype.kind in "iu" and a_np.dtype.itemsize < 8 and a_np.dtype not in (
+            np.dtype(np.int32), np.dtype(np.int64)
+        ):
+            a_np = a_np.astype(np.int64)
+        elif a_np.dtype == np.float16:
+            a_np = a_np.astype(np.float32)
+        array = torch.as_tensor(a_np)
     if not isinstance(group_idx, torch.Tensor):
-        group_idx = torch.as_tensor(np.ascontiguousarray(group_idx))
+        g_np = np.ascontiguousarray(group_idx)
+        if g_np.dtype.kind in "iub" and g_np.dtype not in (
+            np.dtype(np.int32), np.dtype(np.int64)
+        ):
+            g_np = g_np.astype(np.int64)
+        group_idx = torch.as_tensor(g_np)
     if not array.is_cuda:
         if not torch.cuda.is_available():
             raise RuntimeError("engine='hip' requires a GPU; none is available")

@@ -65,6 +65,66 @@ def _torch_dtype(d: np.dtype) -> torch.dtype:
     raise NotImplementedError(f"unsupported dtype {d}")
 
 
+# small / unsigned / half value dtypes compute on a promoted dtype and cast
+# back per the reference's own promotion rules (it casts BEFORE reducing:
+# _initialize_aggregation + xrdtypes - u* sums land in uint64, i8/i16 in
+# int64, min/max/first/last/mode preserve the input dtype, mean/var/std and
+# the quantile family preserve a floating input dtype)
+_SMALL_PROMOTE = {
+    np.dtype("uint8"): np.dtype("int32"),
+    np.dtype("uint16"): np.dtype("int32"),
+    np.dtype("int8"): np.dtype("int32"),
+    np.dtype("int16"): np.dtype("int32"),
+    np.dtype("uint32"): np.dtype("int64"),
+    np.dtype("float16"): np.dtype("float32"),
+}
+_PRESERVE_SMALL = (
+    "min", "nanmin", "max", "nanmax", "first", "last", "nanfirst", "nanlast",
+    "mode", "nanmode",
+)
+
+
+def _cast_back_small(out_np: np.ndarray, func: str, small) -> np.ndarray:
+    # final cast from the promoted compute dtype back to the reference's
+    # output dtype for a small input dtype (see _SMALL_PROMOTE)
+    if small is None:
+        return out_np
+    small = np.dtype(small)
+    if func in _PRESERVE_SMALL:
+        if small.kind in "iu" and out_np.dtype.kind == "i":
+            # the promoted NA fill (iinfo(promoted).min) maps to the small
+            # dtype's NA (iinfo(small).min); real data never hits it
+            out_np = np.where(
+                out_np == np.iinfo(out_np.dtype).min, np.iinfo(small).min, out_np
+            )
+        return out_np.astype(small)
+    if func in ("sum", "nansum", "prod", "nanprod", "cumsum", "nancumsum"):
+        if small.kind == "u":
+            return out_np.astype(np.uint64)
+        if small.kind == "i":
+            return out_np.astype(np.int64)
+        return out_np.astype(small)  # float16 sums stay float16
+    if small.kind == "f" and func in (
+        "mean", "nanmean", "var", "nanvar", "std", "nanstd",
+        "median", "nanmedian", "quantile", "nanquantile", "ffill", "bfill",
+    ):
+        return out_np.astype(small)
+    return out_np
+
+
+def _coerce_by(b):
+    # by / expected arrays in dtypes torch cannot ingest (u16/u32/i8/f16...)
+    # carry the same information as int64/f32 - convert before the device
+    if isinstance(b, torch.Tensor):
+        return b
+    b = np.asarray(b)
+    if b.dtype.kind in "iub" and b.dtype not in (np.dtype(np.int32), np.dtype(np.int64)):
+        return b.astype(np.int64)
+    if b.dtype == np.float16:
+        return b.astype(np.float32)
+    return b
+
+
 # host staging buffers whose H2D copies get captured into a hipGraph must
 # outlive every replay (the graph re-executes the copy from the same host
 # address) — pin them here while capturing
@@ -167,8 +227,15 @@ def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
     sorted_expect = np.sort(expect_np) if sort else expect_np
     if sort is False and not np.all(np.diff(expect_np) >= 0):
         raise NotImplementedError("unsorted expected_groups with sort=False")
+    exp_np_t = sorted_expect
+    if exp_np_t.dtype.kind in "iub" and exp_np_t.dtype not in (
+        np.dtype(np.int32), np.dtype(np.int64)
+    ):
+        exp_np_t = exp_np_t.astype(np.int64)
+    elif exp_np_t.dtype == np.float16:
+        exp_np_t = exp_np_t.astype(np.float32)
     exp_t = _keepalive_if_capturing(
-        torch.from_numpy(np.ascontiguousarray(sorted_expect))
+        torch.from_numpy(np.ascontiguousarray(exp_np_t))
     ).to(flat.device)
     if exp_t.dtype != flat.dtype:
         common = torch.promote_types(exp_t.dtype, flat.dtype)
@@ -297,8 +364,13 @@ def groupby_reduce(
     # datetime64/timedelta64 values compute on their int64 view and
     # dtype-preserving results view back (reference core.py:985-1001)
     dt_dtype = None
+    small_dtype = None
     if return_numpy:
         arr_np = np.asarray(array)
+        if arr_np.dtype in _SMALL_PROMOTE:
+            small_dtype = arr_np.dtype
+            array = arr_np.astype(_SMALL_PROMOTE[arr_np.dtype])
+            arr_np = np.asarray(array)
         if arr_np.dtype.kind in "Mm":
             dt_dtype = arr_np.dtype
             array = arr_np.view("i8")
@@ -309,11 +381,18 @@ def groupby_reduce(
             if func in xrdtypes.PRESERVES_DTYPE and fill_value is None:
                 fill_value = np.iinfo(np.int64).min
     arr = _as_device_tensor(array, device)
+    _torch_small = {
+        torch.uint8: np.dtype("uint8"), torch.int8: np.dtype("int8"),
+        torch.int16: np.dtype("int16"), torch.float16: np.dtype("float16"),
+    }
+    if arr.dtype in _torch_small:
+        small_dtype = _torch_small[arr.dtype]
+        arr = arr.to(_torch_dtype(_SMALL_PROMOTE[small_dtype]))
     was_bool = arr.dtype == torch.bool
     if was_bool:
         arr = arr.to(torch.int64)  # reference core.py:916-917
     in_np_dtype = _np_dtype(arr.dtype)
-    bys = tuple(_as_device_tensor(b, device) for b in by)
+    bys = tuple(_as_device_tensor(_coerce_by(b), device) for b in by)
     nby = len(bys)
     if nby == 0:
         raise ValueError("need at least one by array")
@@ -641,6 +720,7 @@ def groupby_reduce(
             out_np = result.cpu().numpy()
             if dt_dtype is not None:
                 out_np = out_np.astype(dt_dtype)  # reference core.py:1209-1211
+            out_np = _cast_back_small(out_np, func, small_dtype)
             return (out_np, *groups)
         return (result, *groups)
 
@@ -1006,5 +1086,14 @@ def groupby_reduce(
             # dtype (core.py:1209-1211): int64 counts reinterpret, float
             # results truncate, NaN -> NaT
             out_np = out_np.astype(dt_dtype)
+        out_np = _cast_back_small(out_np, func, small_dtype)
         return (out_np, *groups)
+    if small_dtype is not None and func in _PRESERVE_SMALL:
+        # torch outputs for torch small-dtype inputs: preserve-funcs cast
+        # back (torch has these dtypes); sums stay in the promoted
+        # accumulator dtype (torch has no uint64) - documented
+        _back = {np.dtype("uint8"): torch.uint8, np.dtype("int8"): torch.int8,
+                 np.dtype("int16"): torch.int16, np.dtype("float16"): torch.float16}
+        if np.dtype(small_dtype) in _back:
+            result = result.to(_back[np.dtype(small_dtype)])
     return (result, *groups)

@@ -43,13 +43,22 @@ def groupby_scan(
     # passing through as a plain value — the reference's behavior (its ffill
     # isnull sees no missing values on the int64 view)
     dt_dtype = None
+    small_dtype = None
     if return_numpy:
         arr_np0 = np.asarray(array)
+        from .core import _SMALL_PROMOTE, _cast_back_small, _coerce_by
+
+        if arr_np0.dtype in _SMALL_PROMOTE:
+            small_dtype = arr_np0.dtype
+            array = arr_np0.astype(_SMALL_PROMOTE[arr_np0.dtype])
+            arr_np0 = np.asarray(array)
         if arr_np0.dtype.kind in "Mm":
             dt_dtype = arr_np0.dtype
             array = arr_np0.view("i8")
+    else:
+        from .core import _cast_back_small, _coerce_by
     arr = _as_device_tensor(array, device)
-    bys = tuple(_as_device_tensor(b, device) for b in by)
+    bys = tuple(_as_device_tensor(_coerce_by(b), device) for b in by)
     if len(bys) == 0:
         raise ValueError("need at least one by array")
     by_shape = bys[0].shape
@@ -177,5 +186,7 @@ def groupby_scan(
         out_np = out.cpu().numpy()
         if dt_dtype is not None:
             out_np = out_np.astype(dt_dtype)  # int64 counts reinterpret
+        if small_dtype is not None and dtype is None:
+            out_np = _cast_back_small(out_np, func, small_dtype)
         return out_np
     return out

@@ -766,7 +766,9 @@ def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
         ngs = ngroups + 1
         codes = ((np.arange(lead_M) * ngs)[:, None] + codes[None, :]).reshape(-1)
     if func in ("cumsum", "nancumsum") and flat.dtype.kind in "iub" and flat.dtype.itemsize < 8:
-        flat = flat.astype(np.int64)
+        # np.cumsum promotes sub-platform ints to the platform int, keeping
+        # unsignedness (uint8 -> uint64), like the reference
+        flat = flat.astype(np.uint64 if flat.dtype.kind == "u" else np.int64)
     perm = np.argsort(codes, kind="stable")
     sv = flat[perm].astype(flat.dtype)
     sc = codes[perm]

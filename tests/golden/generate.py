@@ -133,6 +133,22 @@ def gen_cases():
                 dict(array=vals_f8, by=labels_basic, func=func, finalize_kwargs={"ddof": 1}),
             )
 
+    # small integer / unsigned / half dtypes (the reference promotes before
+    # reducing: u* sums -> uint64, i8/i16 -> int64, min/max preserve dtype,
+    # mean/var -> float64 for ints and float16 for float16)
+    sd_by = rng.integers(0, 5, 64)
+    for dt in ["uint8", "uint16", "uint32", "int8", "int16", "float16"]:
+        if np.dtype(dt).kind == "f":
+            sdv = (rng.standard_normal(64) * 10).astype(dt)
+        else:
+            info = np.iinfo(dt)
+            sdv = rng.integers(max(info.min, -120), min(info.max, 250), 64).astype(dt)
+        for func in ["sum", "min", "max", "mean", "count", "var", "prod",
+                     "first", "last", "median"]:
+            yield f"{func}_{dt}", dict(
+                array=sdv.copy(), by=sd_by, func=func, expected_groups=np.arange(5)
+            )
+
     # min_count masking and explicit output dtype
     vmc = rng.standard_normal(60)
     vmc[rng.random(60) < 0.5] = np.nan
@@ -351,6 +367,14 @@ def gen_scan_cases():
         yield f"scan_{func}_f32", dict(array=vals.astype(np.float32), by=by, func=func)
     ints = rng.integers(-50, 50, 200).astype(np.int32)
     yield "scan_cumsum_i32", dict(array=ints, by=by, func="cumsum")
+    yield "scan_cumsum_u8", dict(
+        array=rng.integers(0, 250, 200).astype(np.uint8), by=by, func="cumsum")
+    yield "scan_cumsum_i16", dict(
+        array=rng.integers(-120, 120, 200).astype(np.int16), by=by, func="cumsum")
+    f16 = rng.standard_normal(200).astype(np.float16)
+    f16[rng.random(200) < 0.2] = np.nan
+    yield "scan_ffill_f16", dict(array=f16, by=by, func="ffill")
+    yield "scan_cumsum_f16", dict(array=f16, by=by, func="cumsum")
     # NaN labels -> sentinel group scans together
     nby = by.astype(float)
     nby[rng.random(200) < 0.1] = np.nan

@@ -86,6 +86,11 @@ def tolerance_for(name, result_dtype):
     rtol=1e-15/atol=1e-18 for f64, var/std rtol 1e-13 test_core.py:259)."""
     if result_dtype.kind in "iub":
         return dict(rtol=0, atol=0)
+    if result_dtype.itemsize == 2:
+        # float16: the reference's intermediate arithmetic is f16 itself
+        # (e.g. its 3-pass var rounds each pass); we compute in f32/f64 and
+        # cast back — agree to f16 ulp scale
+        return dict(rtol=2e-3, atol=1e-3)
     if result_dtype.itemsize == 4:
         if name.startswith("scan_"):
             # the reference computes segment scans as (global cumsum) minus
@@ -93,6 +98,11 @@ def tolerance_for(name, result_dtype):
             # direct per-segment sum by ~1 ulp of the global running sum
             return dict(rtol=2e-6, atol=1e-5)
         return dict(rtol=2e-6, atol=1e-7)
+    if name.startswith("scan_"):
+        # f64 scans: the reference forms segment scans as global-cumsum
+        # minus prior-group offsets — the cancellation differs from a direct
+        # per-segment sum by ~1 ulp of the running sum
+        return dict(rtol=1e-11, atol=1e-12)
     if "var" in name or "std" in name:
         return dict(rtol=1e-12, atol=1e-14)
     if "quantile" in name or "median" in name:
