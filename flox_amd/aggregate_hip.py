@@ -339,7 +339,8 @@ def _mean_like(group_idx, array, *, skipnan, size=None, fill_value=None, dtype=N
     group_idx, array = _prep(group_idx, array)
     ng = _size_of(group_idx, size)
     p = grouped_partials(SET_SUM_COUNT, array, group_idx, ng, skipnan=skipnan)
-    res = p["sum"] / p["count"]
+    # f64 division explicitly: torch resolves int64/int64 to float32
+    res = p["sum"].to(torch.float64) / p["count"]
     if array.dtype.is_floating_point and dtype is None:
         res = res.to(array.dtype)
     res = _fill(res, p["count"] == 0, fill_value)
