@@ -188,9 +188,36 @@ def _factorize_bins(flat: torch.Tensor, edges: np.ndarray) -> _FactorizedBy:
     return _FactorizedBy(codes, pd.IntervalIndex.from_breaks(edges), direct=False)
 
 
-def _factorize_device(flat: torch.Tensor, expect, sort: bool) -> _FactorizedBy:
+def _factorize_device(flat: torch.Tensor, expect, sort: bool, dt_by=None) -> _FactorizedBy:
     """labels -> codes in [0, ngroups), invalid/NaN -> out-of-range
-    (restates reference _factorize_single, factorize.py:42-99)."""
+    (restates reference _factorize_single, factorize.py:42-99).
+
+    dt_by: original numpy datetime64/timedelta64 dtype when the by array was
+    viewed as int64 — NaT (int64 min) rows are missing (pd.factorize drops
+    NaT) and found/expected groups convert back to dt_by."""
+    if dt_by is not None:
+        nat = torch.iinfo(torch.int64).min
+        if expect is None:
+            fl = flat[flat != nat]
+            uniq = torch.unique(fl)  # sorted; NaT excluded
+            codes = torch.searchsorted(uniq, flat)
+            codes = torch.where(flat == nat, torch.full_like(codes, -1), codes)
+            codes = torch.clamp(codes, max=max(uniq.numel() - 1, 0))
+            return _FactorizedBy(codes, uniq.cpu().numpy().view(dt_by), direct=False)
+        expect_np = np.asarray(expect)
+        assert expect_np.dtype.kind in "Mm", "datetime by needs datetime expected_groups"
+        # align to the by array's unit (the reference returns groups in the
+        # by's unit, e.g. expected [D] against a [s] by -> groups [s])
+        sorted_expect = (np.sort(expect_np) if sort else expect_np).astype(dt_by)
+        exp_t = _keepalive_if_capturing(
+            torch.from_numpy(sorted_expect.view("i8").copy())
+        ).to(flat.device)
+        n = exp_t.numel()
+        codes = torch.searchsorted(exp_t, flat)
+        clipped = torch.clamp(codes, max=n - 1)
+        bad = (codes == n) | (exp_t[clipped] != flat) | (flat == nat)
+        codes = torch.where(bad, torch.full_like(codes, -1), clipped)
+        return _FactorizedBy(codes, sorted_expect, direct=False)
     if expect is None:
         fl = flat
         if fl.dtype.is_floating_point:
@@ -393,7 +420,19 @@ def groupby_reduce(
     if was_bool:
         arr = arr.to(torch.int64)  # reference core.py:916-917
     in_np_dtype = _np_dtype(arr.dtype)
-    bys = tuple(_as_device_tensor(_coerce_by(b), device) for b in by)
+    by_dts = []
+    _coerced = []
+    for b in by:
+        if not isinstance(b, torch.Tensor):
+            bn = np.asarray(b)
+            if bn.dtype.kind in "Mm":
+                # group BY datetimes on the int64 view; NaT rows are missing
+                by_dts.append(bn.dtype)
+                _coerced.append(bn.view("i8"))
+                continue
+        by_dts.append(None)
+        _coerced.append(_coerce_by(b))
+    bys = tuple(_as_device_tensor(b, device) for b in _coerced)
     nby = len(bys)
     if nby == 0:
         raise ValueError("need at least one by array")
@@ -472,8 +511,10 @@ def groupby_reduce(
 
     isbins = isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby
     facs = [
-        _factorize_bins(b.reshape(-1), e) if ib else _factorize_device(b.reshape(-1), e, sort)
-        for b, e, ib in zip(bys, expected_groups, isbins)
+        _factorize_bins(b.reshape(-1), e)
+        if ib
+        else _factorize_device(b.reshape(-1), e, sort, dt_by=dtb)
+        for b, e, ib, dtb in zip(bys, expected_groups, isbins, by_dts)
     ]
     grp_shape = tuple(f.ngroups for f in facs)
     ngroups = math.prod(grp_shape)

@@ -58,7 +58,18 @@ def groupby_scan(
     else:
         from .core import _cast_back_small, _coerce_by
     arr = _as_device_tensor(array, device)
-    bys = tuple(_as_device_tensor(_coerce_by(b), device) for b in by)
+    by_dts = []
+    _coerced = []
+    for b in by:
+        if not isinstance(b, torch.Tensor):
+            bn = np.asarray(b)
+            if bn.dtype.kind in "Mm":
+                by_dts.append(bn.dtype)
+                _coerced.append(bn.view("i8"))
+                continue
+        by_dts.append(None)
+        _coerced.append(_coerce_by(b))
+    bys = tuple(_as_device_tensor(b, device) for b in _coerced)
     if len(bys) == 0:
         raise ValueError("need at least one by array")
     by_shape = bys[0].shape
@@ -91,7 +102,10 @@ def groupby_scan(
         expected_groups = (expected_groups,)
     if expected_groups is None:
         expected_groups = (None,) * len(bys)
-    facs = [_factorize_device(b.reshape(-1), e, True) for b, e in zip(bys, expected_groups)]
+    facs = [
+        _factorize_device(b.reshape(-1), e, True, dt_by=dtb)
+        for b, e, dtb in zip(bys, expected_groups, by_dts)
+    ]
     ngroups = 1
     for f in facs:
         ngroups *= f.ngroups

@@ -76,9 +76,11 @@ _PRESERVES_DTYPE = {
 
 
 def _isnull(a: np.ndarray) -> np.ndarray:
-    # reference xrutils.isnull, restricted to numeric dtypes
+    # reference xrutils.isnull: NaN for floats, NaT for datetimes
     if a.dtype.kind in "fc":
         return np.isnan(a)
+    if a.dtype.kind in "Mm":
+        return np.isnat(a)
     return np.zeros(a.shape, dtype=bool)
 
 

@@ -257,6 +257,20 @@ def gen_cases():
         yield f"{func}_datetime_axis_subset", dict(
             array=tv3, by=tb3, func=func, axis=(2,), expected_groups=np.arange(6)
         )
+    # grouping BY datetime labels (int64-view factorize; NaT drops rows;
+    # expected datetime groups align to the by's unit)
+    dby = (np.datetime64("2022-01-01")
+           + rng.integers(0, 6, 300).astype("timedelta64[D]")).astype("datetime64[s]")
+    dbyn = dby.copy()
+    dbyn[rng.random(300) < 0.15] = np.datetime64("NaT")
+    dvals = rng.standard_normal(300)
+    for func in ["mean", "sum", "count", "nanmax", "var"]:
+        yield f"{func}_dtby", dict(array=dvals, by=dbyn, func=func)
+    yield "mean_dtby_expected", dict(
+        array=dvals, by=dby, func="mean",
+        expected_groups=np.datetime64("2022-01-01") + np.arange(6).astype("timedelta64[D]"),
+    )
+    yield "median_dtby", dict(array=dvals, by=dbyn, func="median")
     # pd.IntervalIndex expected_groups (binning without isbin=True)
     import pandas as pd
     iv_vals = rng.standard_normal(300)
