@@ -31,7 +31,8 @@ def make_case(rng, big=False):
     packed-arg and atomic kernel paths (used by the GPU product fuzz; the
     CPU oracle fuzz keeps the small default so old seeds reproduce)."""
     n = int(rng.integers(10_000, 300_000)) if big else int(rng.integers(1, 400))
-    dt = rng.choice(["float64", "float32", "int64", "int32", "datetime"])
+    dt = rng.choice(["float64", "float32", "int64", "int32", "datetime",
+                     "uint8", "int16", "uint32", "float16"])
     if dt == "datetime":
         vals = np.datetime64("2020-01-01") + rng.integers(0, 10**6, n).astype("timedelta64[s]")
         if rng.random() < 0.5:
@@ -40,12 +41,30 @@ def make_case(rng, big=False):
         vals = (rng.standard_normal(n) * 10 ** rng.integers(0, 4)).astype(dt)
         if rng.random() < 0.6:
             vals[rng.random(n) < rng.choice([0.05, 0.5, 0.95])] = np.nan
+    elif dt == "float16":
+        vals = rng.standard_normal(n).astype(np.float16)
+        if rng.random() < 0.5:
+            vals[rng.random(n) < 0.1] = np.nan
+    elif dt in ("uint8", "int16", "uint32"):
+        info = np.iinfo(dt)
+        vals = rng.integers(max(info.min, -500), min(info.max, 500), n).astype(dt)
     else:
         vals = rng.integers(-1000, 1000, n).astype(dt)
     ng = int(rng.integers(2, 100_000)) if big else int(rng.integers(1, 25))
     shape_kind = rng.choice(["1d", "lead", "multiby", "subset"])
     kw = {}
     if shape_kind == "1d":
+        if rng.random() < 0.15:
+            # group BY datetime labels (NaT rows drop; unit-aligned expected)
+            base = np.datetime64("2021-06-01")
+            by = (base + rng.integers(0, ng, n).astype("timedelta64[h]")).astype("datetime64[s]")
+            if rng.random() < 0.4:
+                by = by.copy()
+                by[rng.random(n) < 0.1] = np.datetime64("NaT")
+            arr = vals
+            if rng.random() < 0.5:
+                kw["expected_groups"] = base + np.arange(ng).astype("timedelta64[h]")
+            return arr, by, kw
         by = rng.integers(0, ng, n)
         arr = vals
     elif shape_kind == "lead":
@@ -170,8 +189,25 @@ def main():
             elif want.dtype.kind in "iubMm":
                 np.testing.assert_array_equal(got, want)
             else:
-                rtol = 2e-5 if want.dtype.itemsize == 4 else 1e-11
-                atol = (1e-4 if want.dtype.itemsize == 4 else 1e-9) * (
+                # tolerance follows the NARROWER of output and INPUT dtype:
+                # the reference computes in the input precision (f16
+                # quantiles lerp in f16 even though the output is f64)
+                in_dt = np.asarray(arr).dtype
+                eff = min(want.dtype.itemsize,
+                          in_dt.itemsize if in_dt.kind == "f" else 8)
+                if eff == 2:
+                    if is_scan:
+                        # f16 cumsum: the reference accumulates natively in
+                        # f16 (per-step rounding); we promote to f32 — agree
+                        # only to the f16 error-accumulation scale
+                        rtol, base_atol = 5e-2, 5e-2
+                    else:
+                        rtol, base_atol = 2e-3, 2e-3
+                elif eff == 4:
+                    rtol, base_atol = 2e-5, 1e-4
+                else:
+                    rtol, base_atol = 1e-11, 1e-9
+                atol = base_atol * (
                     1 + float(np.nanmax(np.abs(want[np.isfinite(want)]), initial=0)))
                 np.testing.assert_allclose(got, want, equal_nan=True, rtol=rtol, atol=atol)
             n_ok += 1
