@@ -85,12 +85,24 @@ _PRESERVE_SMALL = (
 )
 
 
-def _cast_back_small(out_np: np.ndarray, func: str, small) -> np.ndarray:
+def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> np.ndarray:
     # final cast from the promoted compute dtype back to the reference's
     # output dtype for a small input dtype (see _SMALL_PROMOTE)
     if small is None:
         return out_np
     small = np.dtype(small)
+    if fill_value is not None:
+        # a user fill promotes the output dtype unconditionally (reference
+        # xrdtypes.py:170-171) — core already computed in the promoted
+        # result dtype, so no cast-back applies
+        try:
+            tgt = small if func in _PRESERVE_SMALL else (
+                np.dtype(np.uint64) if small.kind == "u"
+                else np.dtype(np.int64) if small.kind == "i" else small)
+            if np.result_type(tgt, fill_value) != tgt:
+                return out_np
+        except TypeError:
+            return out_np
     if func in _PRESERVE_SMALL:
         if small.kind in "iu" and out_np.dtype.kind == "i":
             # the promoted NA fill (iinfo(promoted).min) maps to the small
@@ -389,6 +401,8 @@ def groupby_reduce(
     device = torch.device("cuda", torch.cuda.current_device())
 
     return_numpy = not isinstance(array, torch.Tensor)
+    user_fill_value = fill_value  # internal fill defaults must not trigger
+    # the user-fill dtype-promotion rule in _cast_back_small
     # datetime64/timedelta64 values compute on their int64 view and
     # dtype-preserving results view back (reference core.py:985-1001)
     dt_dtype = None
@@ -762,7 +776,7 @@ def groupby_reduce(
             out_np = result.cpu().numpy()
             if dt_dtype is not None:
                 out_np = out_np.astype(dt_dtype)  # reference core.py:1209-1211
-            out_np = _cast_back_small(out_np, func, small_dtype)
+            out_np = _cast_back_small(out_np, func, small_dtype, user_fill_value)
             return (out_np, *groups)
         return (result, *groups)
 
@@ -1128,7 +1142,7 @@ def groupby_reduce(
             # dtype (core.py:1209-1211): int64 counts reinterpret, float
             # results truncate, NaN -> NaT
             out_np = out_np.astype(dt_dtype)
-        out_np = _cast_back_small(out_np, func, small_dtype)
+        out_np = _cast_back_small(out_np, func, small_dtype, user_fill_value)
         return (out_np, *groups)
     if small_dtype is not None and func in _PRESERVE_SMALL:
         # torch outputs for torch small-dtype inputs: preserve-funcs cast
