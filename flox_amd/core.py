@@ -119,8 +119,10 @@ def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> n
         return out_np.astype(small)  # float16 sums stay float16
     if small.kind == "f" and func in (
         "mean", "nanmean", "var", "nanvar", "std", "nanstd",
-        "median", "nanmedian", "quantile", "nanquantile", "ffill", "bfill",
+        "median", "nanmedian", "ffill", "bfill",
     ):
+        # NOT quantile/nanquantile: their final dtype is ALWAYS float64
+        # (reference aggregations.py:695-710); median preserves floating
         return out_np.astype(small)
     return out_np
 
