@@ -111,6 +111,9 @@ def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> n
                 out_np == np.iinfo(out_np.dtype).min, np.iinfo(small).min, out_np
             )
         return out_np.astype(small)
+    if func in ("ffill", "bfill"):
+        # identity-carry preserves the input dtype for every kind
+        return out_np.astype(small)
     if func in ("sum", "nansum", "prod", "nanprod", "cumsum", "nancumsum"):
         if small.kind == "u":
             return out_np.astype(np.uint64)
