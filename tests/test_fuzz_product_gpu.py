@@ -49,7 +49,9 @@ def _run_case(i, rng):
         return None  # datetime wrap arithmetic: out of parity scope
     if func in ("prod", "nanprod"):
         if a.dtype.kind in "iuMm":
-            arr = np.clip(arr, -1, 1)
+            # unsigned: a -1 bound wraps to the dtype max and products then
+            # overflow int64 order-dependently (documented wrap class)
+            arr = np.clip(arr, 0 if a.dtype.kind == "u" else -1, 1)
         elif np.asarray(arr).size > 1000:
             # big groups: fp products overflow/underflow at order-dependent
             # points; sign-only values keep them exact

@@ -145,7 +145,8 @@ def main():
             if np.asarray(arr).dtype.kind in "Mm":
                 n_skip += 1
                 continue
-            arr = np.clip(arr, -1, 1)
+            a0 = np.asarray(arr)
+            arr = np.clip(a0, 0 if a0.dtype.kind == "u" else -1, 1)
         elif func in ("prod", "nanprod") and np.asarray(arr).size > 1000:
             # big groups: fp products overflow/underflow at order-dependent
             # points; sign-only values keep them exact
