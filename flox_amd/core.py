@@ -116,6 +116,12 @@ def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> n
         return out_np.astype(small)
     if func in ("sum", "nansum", "prod", "nanprod", "cumsum", "nancumsum"):
         if small.kind == "u":
+            if out_np.dtype.kind == "i":
+                # empty-group NA fill: iinfo(int64).min on the promoted
+                # dtype maps to iinfo(uint64).min == 0 on the true output
+                out_np = np.where(
+                    out_np == np.iinfo(np.int64).min, 0, out_np
+                )
             return out_np.astype(np.uint64)
         if small.kind == "i":
             return out_np.astype(np.int64)
