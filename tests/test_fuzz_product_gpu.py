@@ -91,11 +91,12 @@ def _run_case(i, rng):
     else:
         fin = want[np.isfinite(want)]
         scale = 1 + float(np.max(np.abs(fin), initial=0.0))
-        if want.dtype.itemsize == 2:
-            # f16 outputs: product (promoted f32) and oracle (f64) can round
-            # to adjacent f16 values at half-ulp boundaries
+        if want.dtype.itemsize == 2 or np.asarray(arr).dtype == np.float16:
+            # f16 precision class: the oracle computes quantile/lerp
+            # arithmetic in the INPUT precision (np.quantile preserves f16),
+            # the product lerps the promoted f32 values in f64
             tol = dict(rtol=2e-3, atol=1e-3 * scale)
-        elif want.dtype.itemsize == 4 or np.asarray(arr).dtype.itemsize in (2, 4):
+        elif want.dtype.itemsize == 4 or np.asarray(arr).dtype.itemsize == 4:
             tol = dict(rtol=3e-5, atol=1e-4 * scale)
         else:
             tol = dict(rtol=1e-10, atol=1e-10 * scale)
