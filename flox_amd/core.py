@@ -91,6 +91,18 @@ def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> n
     if small is None:
         return out_np
     small = np.dtype(small)
+    if small == np.dtype(np.uint64):
+        if func in _PRESERVE_SMALL:
+            # values < 2**63 (guarded at entry); NA fill int64-min maps to
+            # uint64 NA = 0
+            return np.where(
+                out_np == np.iinfo(np.int64).min, 0, out_np
+            ).astype(np.uint64)
+        if func in ("sum", "nansum", "prod", "nanprod", "cumsum", "nancumsum"):
+            out_np = np.where(out_np == np.iinfo(np.int64).min, 0, out_np) \
+                if func in ("sum", "nansum", "prod", "nanprod") else out_np
+            return out_np.astype(np.int64).view(np.uint64)  # wrap-exact
+        return out_np
     if fill_value is not None:
         # a user fill promotes the output dtype unconditionally (reference
         # xrdtypes.py:170-171) — core already computed in the promoted
@@ -424,6 +436,19 @@ def groupby_reduce(
             small_dtype = arr_np.dtype
             array = arr_np.astype(_SMALL_PROMOTE[arr_np.dtype])
             arr_np = np.asarray(array)
+        elif arr_np.dtype == np.uint64:
+            # uint64 computes on its int64 VIEW: sums/prods/counts are
+            # wrap-exact mod 2^64; order-dependent funcs are only correct
+            # below 2^63, so values with the high bit set are rejected there
+            small_dtype = arr_np.dtype
+            array = arr_np.view(np.int64)
+            arr_np = np.asarray(array)
+            if func not in (
+                "sum", "nansum", "count", "prod", "nanprod", "any", "all",
+            ) and bool((arr_np < 0).any()):
+                raise NotImplementedError(
+                    f"{func} on uint64 values >= 2**63 (int64-view order breaks)"
+                )
         if arr_np.dtype.kind in "Mm":
             dt_dtype = arr_np.dtype
             array = arr_np.view("i8")

@@ -137,7 +137,7 @@ def gen_cases():
     # reducing: u* sums -> uint64, i8/i16 -> int64, min/max preserve dtype,
     # mean/var -> float64 for ints and float16 for float16)
     sd_by = rng.integers(0, 5, 64)
-    for dt in ["uint8", "uint16", "uint32", "int8", "int16", "float16"]:
+    for dt in ["uint8", "uint16", "uint32", "uint64", "int8", "int16", "float16"]:
         if np.dtype(dt).kind == "f":
             sdv = (rng.standard_normal(64) * 10).astype(dt)
         else:
