@@ -32,7 +32,7 @@ def make_case(rng, big=False):
     CPU oracle fuzz keeps the small default so old seeds reproduce)."""
     n = int(rng.integers(10_000, 300_000)) if big else int(rng.integers(1, 400))
     dt = rng.choice(["float64", "float32", "int64", "int32", "datetime",
-                     "uint8", "int16", "uint32", "float16"])
+                     "uint8", "int16", "uint32", "uint64", "float16"])
     if dt == "datetime":
         vals = np.datetime64("2020-01-01") + rng.integers(0, 10**6, n).astype("timedelta64[s]")
         if rng.random() < 0.5:
@@ -45,7 +45,7 @@ def make_case(rng, big=False):
         vals = rng.standard_normal(n).astype(np.float16)
         if rng.random() < 0.5:
             vals[rng.random(n) < 0.1] = np.nan
-    elif dt in ("uint8", "int16", "uint32"):
+    elif dt in ("uint8", "int16", "uint32", "uint64"):
         info = np.iinfo(dt)
         vals = rng.integers(max(info.min, -500), min(info.max, 500), n).astype(dt)
     else:
