@@ -91,18 +91,6 @@ def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> n
     if small is None:
         return out_np
     small = np.dtype(small)
-    if small == np.dtype(np.uint64):
-        if func in _PRESERVE_SMALL:
-            # values < 2**63 (guarded at entry); NA fill int64-min maps to
-            # uint64 NA = 0
-            return np.where(
-                out_np == np.iinfo(np.int64).min, 0, out_np
-            ).astype(np.uint64)
-        if func in ("sum", "nansum", "prod", "nanprod", "cumsum", "nancumsum"):
-            out_np = np.where(out_np == np.iinfo(np.int64).min, 0, out_np) \
-                if func in ("sum", "nansum", "prod", "nanprod") else out_np
-            return out_np.astype(np.int64).view(np.uint64)  # wrap-exact
-        return out_np
     if fill_value is not None:
         # a user fill promotes the output dtype unconditionally (reference
         # xrdtypes.py:170-171) — core already computed in the promoted
@@ -115,6 +103,18 @@ def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> n
                 return out_np
         except TypeError:
             return out_np
+    if small == np.dtype(np.uint64):
+        if func in _PRESERVE_SMALL:
+            # values < 2**63 (guarded at entry); NA fill int64-min maps to
+            # uint64 NA = 0
+            return np.where(
+                out_np == np.iinfo(np.int64).min, 0, out_np
+            ).astype(np.uint64)
+        if func in ("sum", "nansum", "prod", "nanprod", "cumsum", "nancumsum"):
+            out_np = np.where(out_np == np.iinfo(np.int64).min, 0, out_np) \
+                if func in ("sum", "nansum", "prod", "nanprod") else out_np
+            return out_np.astype(np.int64).view(np.uint64)  # wrap-exact
+        return out_np
     if func in _PRESERVE_SMALL:
         if small.kind in "iu" and out_np.dtype.kind == "i":
             # the promoted NA fill (iinfo(promoted).min) maps to the small
