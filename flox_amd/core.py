@@ -114,6 +114,8 @@ def _cast_back_small(out_np: np.ndarray, func: str, small, fill_value=None) -> n
             out_np = np.where(out_np == np.iinfo(np.int64).min, 0, out_np) \
                 if func in ("sum", "nansum", "prod", "nanprod") else out_np
             return out_np.astype(np.int64).view(np.uint64)  # wrap-exact
+        if func in ("ffill", "bfill"):
+            return out_np.view(np.uint64)  # identity carry
         return out_np
     if func in _PRESERVE_SMALL:
         if small.kind in "iu" and out_np.dtype.kind == "i":

@@ -52,6 +52,11 @@ def groupby_scan(
             small_dtype = arr_np0.dtype
             array = arr_np0.astype(_SMALL_PROMOTE[arr_np0.dtype])
             arr_np0 = np.asarray(array)
+        elif arr_np0.dtype == np.uint64:
+            # int64 view: cumsum wrap-exact mod 2^64; ffill/bfill identity
+            small_dtype = arr_np0.dtype
+            array = arr_np0.view(np.int64)
+            arr_np0 = np.asarray(array)
         if arr_np0.dtype.kind in "Mm":
             dt_dtype = arr_np0.dtype
             array = arr_np0.view("i8")
