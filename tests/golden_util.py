@@ -108,4 +108,6 @@ def tolerance_for(name, result_dtype):
     if "quantile" in name or "median" in name:
         # the lerp t*(b-a)+a vs (1-t)*a+t*b forms differ by ~1 ulp
         return dict(rtol=1e-12, atol=1e-14)
-    return dict(rtol=1e-13, atol=1e-16)
+    # atol absorbs atomic-order cancellation on near-zero f64 sums
+    # (group sums of O(10) addends cancelling to ~1e-3)
+    return dict(rtol=1e-13, atol=1e-11)
