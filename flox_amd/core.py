@@ -199,15 +199,20 @@ class _FactorizedBy:
         self.direct = direct
 
 
-def _factorize_bins(flat: torch.Tensor, edges: np.ndarray) -> _FactorizedBy:
+def _factorize_bins(flat: torch.Tensor, edges: np.ndarray, dt_by=None) -> _FactorizedBy:
     """Bin-edge grouping (isbin=True): right-closed intervals (edges[i-1],
     edges[i]], values outside every interval -> invalid (restates reference
     factorize.py:55-82: np.digitize(right=True) - 1). Groups are the
-    pandas IntervalIndex the reference returns."""
+    pandas IntervalIndex the reference returns. Datetime edges bin the
+    int64-viewed by on unit-aligned views (dt_by = the by dtype)."""
     import pandas as pd
 
     edges = np.asarray(edges)
     assert edges.ndim == 1 and len(edges) >= 2, "bin edges need >= 2 values"
+    iv_groups = pd.IntervalIndex.from_breaks(edges)
+    if edges.dtype.kind in "Mm":
+        assert dt_by is not None, "datetime bin edges need a datetime by"
+        edges = edges.astype(dt_by).view("i8")
     nbins = len(edges) - 1
     fl = flat
     edges_t = _keepalive_if_capturing(
@@ -222,7 +227,7 @@ def _factorize_bins(flat: torch.Tensor, edges: np.ndarray) -> _FactorizedBy:
     if fl.dtype.is_floating_point:
         bad |= torch.isnan(fl)
     codes = torch.where(bad, torch.full_like(codes, -1), codes)
-    return _FactorizedBy(codes, pd.IntervalIndex.from_breaks(edges), direct=False)
+    return _FactorizedBy(codes, iv_groups, direct=False)
 
 
 def _factorize_device(flat: torch.Tensor, expect, sort: bool, dt_by=None) -> _FactorizedBy:
@@ -563,7 +568,7 @@ def groupby_reduce(
 
     isbins = isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby
     facs = [
-        _factorize_bins(b.reshape(-1), e)
+        _factorize_bins(b.reshape(-1), e, dt_by=dtb)
         if ib
         else _factorize_device(b.reshape(-1), e, sort, dt_by=dtb)
         for b, e, ib, dtb in zip(bys, expected_groups, isbins, by_dts)

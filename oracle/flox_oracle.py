@@ -185,16 +185,25 @@ def _fill_default(func: str, out_dtype: np.dtype):
 
 
 def _factorize_bins_np(flat, edges):
-    """right-closed binning, reference factorize.py:55-82."""
+    """right-closed binning, reference factorize.py:55-82 (datetime edges
+    bin on unit-aligned int64 views; groups stay the datetime intervals)."""
     import pandas as pd
 
     edges = np.asarray(edges)
+    flat = np.asarray(flat)
+    iv = pd.IntervalIndex.from_breaks(edges)
+    if flat.dtype.kind in "Mm":
+        nanm = np.isnat(flat)
+        edges = edges.astype(flat.dtype).view("i8")
+        flat = flat.view("i8")
+    else:
+        nanm = _isnull(np.asarray(flat, dtype=float)) if flat.dtype.kind in "fc" else np.zeros(len(flat), bool)
     nbins = len(edges) - 1
     codes = np.digitize(flat, bins=edges, right=True) - 1
     within = flat <= edges.max()
     codes[(codes < 0) | (codes >= nbins) | ~within] = -1
-    codes[_isnull(np.asarray(flat, dtype=float)) if flat.dtype.kind in "fc" else np.zeros(len(flat), bool)] = -1
-    return codes.astype(np.int64), pd.IntervalIndex.from_breaks(edges)
+    codes[nanm] = -1
+    return codes.astype(np.int64), iv
 
 
 def groupby_reduce(

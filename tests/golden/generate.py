@@ -271,6 +271,12 @@ def gen_cases():
         expected_groups=np.datetime64("2022-01-01") + np.arange(6).astype("timedelta64[D]"),
     )
     yield "median_dtby", dict(array=dvals, by=dbyn, func="median")
+    # datetime bin edges (isbin on datetime labels)
+    dt_edges = np.datetime64("2022-01-01") + (np.arange(5) * 2).astype("timedelta64[D]")
+    for func in ["mean", "count", "sum"]:
+        yield f"{func}_dt_isbin", dict(
+            array=dvals, by=dbyn, func=func, expected_groups=dt_edges, isbin=True
+        )
     # pd.IntervalIndex expected_groups (binning without isbin=True)
     import pandas as pd
     iv_vals = rng.standard_normal(300)
