@@ -450,7 +450,9 @@ def grouped_partials_cols(
     chunk_t = None
     col_blocks = builtins.max(1, m // 1024)
     desired = int(builtins.min(64, builtins.max(1, -(-2048 // col_blocks))))
-    if desired > 1 and n_t > 1:
+    if desired > 1 and n_t > 1 and not torch.cuda.is_current_stream_capturing():
+        # host-side group-aligned chunk planning needs a D2H sync — illegal
+        # during hipGraph capture; captured column ops use the slab path
         sc = codes_sorted.cpu().numpy()
         seg = np.flatnonzero(np.diff(sc)) + 1
         target = builtins.max(1, -(-n_t // desired))
