@@ -237,8 +237,26 @@ def main():
                 .all().item()
             )
             if ok:
-                graph = g
-                log("hipGraph step replay enabled (replay == eager verified)")
+                # keep the graph only if replay is actually faster (a
+                # captured column op falls to the slab path, which can lose
+                # to the eager group-aligned chunks)
+                t0 = time.perf_counter()
+                for _ in range(3):
+                    g.replay()
+                torch.cuda.synchronize()
+                t_replay = time.perf_counter() - t0
+                t0 = time.perf_counter()
+                for _ in range(3):
+                    step()
+                torch.cuda.synchronize()
+                t_eager = time.perf_counter() - t0
+                if t_replay < t_eager:
+                    graph = g
+                    log(f"hipGraph step replay enabled (verified; "
+                        f"{t_replay / 3 * 1e3:.2f} vs eager {t_eager / 3 * 1e3:.2f} ms)")
+                else:
+                    log(f"hipGraph replay slower than eager "
+                        f"({t_replay / 3 * 1e3:.2f} vs {t_eager / 3 * 1e3:.2f} ms); eager steps")
             else:
                 log("hipGraph replay mismatch vs eager; falling back to eager")
         except Exception as e:  # pragma: no cover - path depends on config
