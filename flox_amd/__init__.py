@@ -13,6 +13,7 @@ from .aggregations import (  # noqa: F401
     generic_aggregate,
 )
 from .core import groupby_reduce  # noqa: F401
+from .options import OPTIONS, set_options  # noqa: F401
 from .scan import groupby_scan  # noqa: F401
 
 __version__ = "0.1.0"

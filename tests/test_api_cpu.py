@@ -81,3 +81,26 @@ def test_unknown_func_rejected():
         pytest.skip("GPU present")
     with pytest.raises(NotImplementedError):
         flox_amd.groupby_reduce(np.ones(4), np.zeros(4, dtype=np.int64), func="cumsum")
+
+
+def test_set_options_roundtrip():
+    """flox_amd.set_options mirrors the reference's options context
+    (reference options.py:9-64), re-keyed to this engine's knobs."""
+    import flox_amd
+    from flox_amd import core, distributed
+
+    old_thr = core.PACKED_ARG_THRESHOLD
+    old_ng = distributed.SPARSE_NGROUPS
+    with flox_amd.set_options(packed_arg_threshold=123,
+                              sparse_combine_ngroups=456,
+                              sparse_combine_fraction=0.5):
+        assert core.PACKED_ARG_THRESHOLD == 123
+        assert distributed.SPARSE_NGROUPS == 456
+        assert distributed.SPARSE_FRACTION == 0.5
+    assert core.PACKED_ARG_THRESHOLD == old_thr
+    assert distributed.SPARSE_NGROUPS == old_ng
+    import pytest
+
+    with pytest.raises(ValueError):
+        with flox_amd.set_options(nonsense=1):
+            pass
