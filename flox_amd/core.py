@@ -924,6 +924,17 @@ def groupby_reduce(
             empty_mask = kmin == ((1 << 63) - 1)
             result = (kmin ^ (-(1 << 63))) & 0xFFFFFFFF
             counts_for_mask = p["count"]
+            if min_count_ > 0 and skip:
+                # the packed keys are always-valid ints, so p["count"]
+                # counts NaN rows too; nanarg min_count masks on the
+                # NaN-aware count
+                pc = grouped_partials(
+                    _ffi.SET_COUNT, vals, labels, ngroups, skipnan=True,
+                    labels2=labels2, grp_shape=grp_pair,
+                )
+                if dist_on:
+                    distributed.all_reduce_(pc["count"], "sum")
+                counts_for_mask = pc["count"]
         elif pair_ok:
             p = grouped_partials(
                 _ffi.SET_ARGMAX_PAIR if ismax else _ffi.SET_ARGMIN_PAIR,

@@ -350,6 +350,21 @@ def test_pair_arg_multi_by_fused():
         np.testing.assert_array_equal(np.asarray(got), want, err_msg=func)
 
 
+def test_packed_nanarg_min_count():
+    """nanarg min_count masks on the NaN-aware count — the packed keys are
+    always-valid ints, so their count pass alone would include NaN rows."""
+    rng = np.random.default_rng(17)
+    n, ng = 120_000, 8_000  # past the packed threshold
+    labels = rng.integers(0, ng, n)
+    vals = rng.standard_normal(n).astype(np.float32)
+    vals[rng.random(n) < 0.6] = np.nan  # many groups fall below min_count
+    kw = dict(expected_groups=np.arange(ng), min_count=8, fill_value=-1.0)
+    want, *_ = oracle_reduce(vals, labels, func="nanargmin", **kw)
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func="nanargmin", **kw)
+    assert got.dtype == want.dtype
+    np.testing.assert_array_equal(got, want)
+
+
 def test_pair_arg_matches_two_pass_form():
     """The pair-payload path must agree bit-for-bit with the LDS two-pass
     form on the same inputs (threshold lowered to force both)."""
