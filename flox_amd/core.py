@@ -56,7 +56,13 @@ _TORCH_TO_NP = {
 
 
 def _np_dtype(t: torch.dtype) -> np.dtype:
-    return _TORCH_TO_NP[t]
+    try:
+        return _TORCH_TO_NP[t]
+    except KeyError:
+        raise NotImplementedError(
+            f"engine='hip' does not support torch dtype {t} (numpy inputs "
+            "of small/unsigned dtypes promote automatically)"
+        ) from None
 
 
 def _torch_dtype(d: np.dtype) -> torch.dtype:
