@@ -12,8 +12,11 @@ leading array dims, axis subsets of ``by``'s dims, 1+ ``by`` arrays
 first-appearance ordering, expected_groups/fill_value/min_count/dtype
 rules, custom Aggregation(chunk/combine/finalize) instances; dtypes
 f32/f64/i32/i64/bool/datetime64/timedelta64 (NaT as int64-min, like the
-reference) plus u8/u16/u32/i8/i16/f16 via promote-compute-cast following
-the reference's own promotion rules (see _SMALL_PROMOTE). Multi-GPU: every op combines across ranks (partial-bin
+reference) plus u8/u16/u32/u64/i8/i16/f16 via promote-compute-cast
+following the reference's own promotion rules (see _SMALL_PROMOTE; uint64
+computes on the int64 view). Grouping BY datetime labels works on the
+int64 view (NaT rows drop; expected groups unit-align), including datetime
+bin edges (isbin). Multi-GPU: every op combines across ranks (partial-bin
 all-reduce / scan carries / quantile radix selection / mode run merge).
 """
 
