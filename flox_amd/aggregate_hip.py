@@ -947,3 +947,46 @@ def _numpy_io_wrap(f):
 for _n in list(_LEAD_KINDS) + ["cumsum", "nancumsum", "ffill", "bfill"]:
     globals()[_n] = _numpy_io_wrap(globals()[_n])
 del _n
+
+
+# ---------------------------------------------------------------------------
+# complex input at the seam: the linear funcs compute componentwise on the
+# re/im planes (the reference casts arrays to the aggregation's intermediate
+# dtype before the engine call, so complex sums arrive as complex128).
+# Whole-value nulls (either component NaN — the reference's isnull) premask
+# BOTH components for the skipna funcs. Outermost wrapper.
+# ---------------------------------------------------------------------------
+
+_COMPLEX_SKIPNA = {"nansum", "nanmean", "nanfirst", "nanlast", "nancumsum",
+                   "nanlen", "count"}
+
+
+def _complex_io_wrap(f, name):
+    import functools
+
+    @functools.wraps(f)
+    def wrapper(group_idx, array, **kw):
+        if isinstance(array, torch.Tensor) or np.asarray(array).dtype.kind != "c":
+            return f(group_idx, array, **kw)
+        a = np.asarray(array)
+        comp_dt = a.dtype
+        fw = np.float64 if comp_dt == np.dtype(np.complex128) else np.float32
+        if name in ("nanlen", "count"):
+            real = a.real.astype(np.float64)
+            real[np.isnan(a)] = np.nan  # whole-value null
+            return f(group_idx, real, **kw)
+        fv = a.view(fw).reshape(a.shape + (2,)).copy()
+        if name in _COMPLEX_SKIPNA:
+            fv[np.isnan(a)] = np.nan
+        kw.pop("dtype", None)
+        rr = f(group_idx, np.ascontiguousarray(fv[..., 0]), **kw)
+        ri = f(group_idx, np.ascontiguousarray(fv[..., 1]), **kw)
+        return (np.asarray(rr) + 1j * np.asarray(ri)).astype(comp_dt)
+
+    return wrapper
+
+
+for _n in ["sum", "nansum", "mean", "nanmean", "nanlen", "count",
+           "first", "last", "nanfirst", "nanlast", "cumsum", "nancumsum"]:
+    globals()[_n] = _complex_io_wrap(globals()[_n], _n)
+del _n

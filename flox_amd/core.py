@@ -452,6 +452,63 @@ def groupby_reduce(
             small_dtype = arr_np.dtype
             array = arr_np.astype(_SMALL_PROMOTE[arr_np.dtype])
             arr_np = np.asarray(array)
+        elif arr_np.dtype.kind == "c":
+            # complex: the linear set computes on a (2, ...) re/im component
+            # view (components are independent for sums/means/first/last and
+            # scans); whole-value nulls (either component NaN — the
+            # reference's isnull) premask BOTH components for skipna funcs.
+            # Order/product/var families have no componentwise form: raise.
+            _C_OK = ("sum", "nansum", "mean", "nanmean", "count",
+                     "first", "last", "nanfirst", "nanlast")
+            if not isinstance(func, str) or func not in _C_OK:
+                raise NotImplementedError(
+                    f"complex input for {func!r} (supported: {_C_OK})"
+                )
+            if fill_value is not None and not (
+                isinstance(fill_value, (float, complex)) and np.isnan(
+                    np.asarray(fill_value).real
+                )
+            ):
+                raise NotImplementedError("complex input with a non-NaN fill_value")
+            comp_dt = arr_np.dtype
+            if func == "count":
+                # count ignores magnitudes: a real view with whole-value
+                # nulls mapped to NaN is exact
+                real = arr_np.real.astype(np.float64)
+                real[np.isnan(arr_np)] = np.nan
+                r, *grps = groupby_reduce(
+                    real, *by, func=func, expected_groups=expected_groups,
+                    sort=sort, isbin=isbin, axis=axis, fill_value=fill_value,
+                    dtype=dtype, min_count=min_count, method=method,
+                    engine=engine, reindex=reindex,
+                    finalize_kwargs=finalize_kwargs,
+                    distributed_combine=distributed_combine,
+                    shard_row_offset=shard_row_offset,
+                )
+                return (r, *grps)
+            fwidth = np.float64 if comp_dt == np.dtype(np.complex128) else np.float32
+            fv2 = arr_np.view(fwidth).reshape(arr_np.shape + (2,)).copy()
+            if REDUCTIONS[func].skipnan:
+                fv2[np.isnan(arr_np)] = np.nan  # both components
+            comps = np.moveaxis(fv2, -1, 0)  # (2, ...original dims)
+            ax2 = axis
+            if axis is not None:
+                axl = axis if isinstance(axis, (tuple, list)) else (axis,)
+                ax2 = tuple((a % arr_np.ndim) + 1 for a in axl)
+            r, *grps = groupby_reduce(
+                np.ascontiguousarray(comps), *by, func=func,
+                expected_groups=expected_groups, sort=sort, isbin=isbin,
+                axis=ax2, fill_value=fill_value, dtype=None,
+                min_count=min_count, method=method, engine=engine,
+                reindex=reindex, finalize_kwargs=finalize_kwargs,
+                distributed_combine=distributed_combine,
+                shard_row_offset=shard_row_offset,
+            )
+            out = np.asarray(r)
+            res = (out[0] + 1j * out[1]).astype(
+                comp_dt if dtype is None else np.dtype(dtype)
+            )
+            return (res, *grps)
         elif arr_np.dtype == np.uint64:
             # uint64 computes on its int64 VIEW: sums/prods/counts are
             # wrap-exact mod 2^64; order-dependent funcs are only correct

@@ -446,18 +446,28 @@ def groupby_reduce(
     acc_dtype = np.float64 if array.dtype.kind in "fc" else np.int64
     if array.dtype == np.complex64 or array.dtype == np.complex128:
         acc_dtype = np.complex128
+        if func not in ("sum", "nansum", "mean", "nanmean", "count",
+                        "first", "last", "nanfirst", "nanlast"):
+            raise NotImplementedError(f"complex input for {func!r}")
 
     def bincount_rows(weights2d, mask2d):
         """per-row np.bincount with f64 accumulation (npg semantics)."""
-        out = np.zeros((M, ngroups), dtype=acc_dtype)
+        out = np.zeros((M, ngroups),
+                       dtype=np.int64 if weights2d is None else acc_dtype)
         for r in range(M):
             m = mask2d[r] if mask2d is not None else valid_code
             if weights2d is None:
-                out[r] = np.bincount(codes[m], minlength=ngroups).astype(acc_dtype)
+                out[r] = np.bincount(codes[m], minlength=ngroups)
             elif acc_dtype == np.int64:
                 # integer sums accumulate (and wrap) in int64, like the
                 # reference's reduceat — np.bincount would force float64
                 np.add.at(out[r], codes[m], weights2d[r][m].astype(np.int64))
+            elif acc_dtype == np.complex128:
+                w = weights2d[r][m].astype(np.complex128)
+                out[r] = (
+                    np.bincount(codes[m], weights=w.real, minlength=ngroups)
+                    + 1j * np.bincount(codes[m], weights=w.imag, minlength=ngroups)
+                )
             else:
                 out[r] = np.bincount(codes[m], weights=weights2d[r][m].astype(acc_dtype), minlength=ngroups)
         return out
@@ -790,15 +800,24 @@ def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
     for s0, s1 in zip(starts, ends):
         seg = sv[s0:s1]
         if func == "cumsum":
-            acc = np.nancumsum(seg)
-            if seg.dtype.kind in "fc":
-                nanpos = np.cumsum(_isnull(seg)) > 0
-                acc = acc.astype(float)
-                acc[nanpos] = np.nan
-            out_sorted[s0:s1] = acc
+            if seg.dtype.kind == "c":
+                # complex: per-component NaN propagation (np.cumsum)
+                out_sorted[s0:s1] = np.cumsum(seg)
+            else:
+                acc = np.nancumsum(seg)
+                if seg.dtype.kind == "f":
+                    nanpos = np.cumsum(_isnull(seg)) > 0
+                    acc = acc.astype(float)
+                    acc[nanpos] = np.nan
+                out_sorted[s0:s1] = acc
         elif func == "nancumsum":
             out_sorted[s0:s1] = np.nancumsum(seg)
         elif func in ("ffill", "bfill"):
+            if seg.dtype.kind == "c":
+                # the reference's dtype gate (scan.py:199: kind != "f" ->
+                # identity scan) leaves complex ffill/bfill as identity
+                out_sorted[s0:s1] = seg
+                continue
             seg2 = seg[::-1] if func == "bfill" else seg.copy()
             mask = _isnull(seg2)
             idx = np.where(mask, 0, np.arange(len(seg2)))

@@ -149,6 +149,16 @@ def gen_cases():
                 array=sdv.copy(), by=sd_by, func=func, expected_groups=np.arange(5)
             )
 
+    # complex values: the linear set (componentwise re/im)
+    cv = (rng.standard_normal(80) + 1j * rng.standard_normal(80)).astype(np.complex128)
+    cv[rng.random(80) < 0.15] = np.nan
+    cv.real[rng.random(80) < 0.05] = np.nan  # partial-NaN values (null per isnull)
+    c_by = rng.integers(0, 6, 80)
+    for func in ["sum", "nansum", "mean", "nanmean", "count"]:
+        yield f"{func}_c128", dict(array=cv.copy(), by=c_by, func=func,
+                                   expected_groups=np.arange(6))
+        yield f"{func}_c64", dict(array=cv.astype(np.complex64), by=c_by, func=func,
+                                  expected_groups=np.arange(6))
     # min_count masking and explicit output dtype
     vmc = rng.standard_normal(60)
     vmc[rng.random(60) < 0.5] = np.nan
@@ -395,6 +405,10 @@ def gen_scan_cases():
     f16[rng.random(200) < 0.2] = np.nan
     yield "scan_ffill_f16", dict(array=f16, by=by, func="ffill")
     yield "scan_cumsum_f16", dict(array=f16, by=by, func="cumsum")
+    cvs = (rng.standard_normal(200) + 1j * rng.standard_normal(200)).astype(np.complex128)
+    cvs[rng.random(200) < 0.2] = np.nan
+    for func in ["cumsum", "nancumsum", "ffill", "bfill"]:
+        yield f"scan_{func}_c128", dict(array=cvs.copy(), by=by, func=func)
     # NaN labels -> sentinel group scans together
     nby = by.astype(float)
     nby[rng.random(200) < 0.1] = np.nan

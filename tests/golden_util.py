@@ -86,6 +86,11 @@ def tolerance_for(name, result_dtype):
     rtol=1e-15/atol=1e-18 for f64, var/std rtol 1e-13 test_core.py:259)."""
     if result_dtype.kind in "iub":
         return dict(rtol=0, atol=0)
+    if result_dtype.kind == "c":
+        # complex: per-component precision class (c64 components are f32)
+        if result_dtype.itemsize == 8:
+            return dict(rtol=2e-6, atol=1e-5)
+        return dict(rtol=1e-12, atol=1e-11)
     if result_dtype.itemsize == 2:
         # float16: the reference's intermediate arithmetic is f16 itself
         # (e.g. its 3-pass var rounds each pass); we compute in f32/f64 and

@@ -29,6 +29,10 @@ core = ref_loader.load_reference()
 def _tol(func, dtype):
     if np.dtype(dtype).kind in "iuMmb":
         return dict(rtol=0, atol=0)
+    if np.dtype(dtype).kind == "c":
+        # per-component precision (c64 components are f32)
+        return (dict(rtol=3e-6, atol=1e-5) if np.dtype(dtype).itemsize == 8
+                else dict(rtol=1e-12, atol=1e-10))
     if np.dtype(dtype).itemsize == 2:
         # engine="flox" computes f16 in f16; hip promotes to f32/f64
         return dict(rtol=2e-3, atol=1e-3)
@@ -113,6 +117,16 @@ def _cases():
                                expected_groups=np.arange(5))
     yield "min_float16", dict(array=f16v, by=(by_s,), func="min",
                               expected_groups=np.arange(5))
+    # complex values: componentwise seam compute (reference casts to the
+    # complex128 intermediate before the engine call)
+    cvals = (rng.standard_normal(300) + 1j * rng.standard_normal(300)).astype(np.complex128)
+    cvals[rng.random(300) < 0.1] = np.nan
+    cby = rng.integers(0, 7, 300)
+    for func in ["sum", "nansum", "mean", "nanmean", "count"]:
+        yield f"{func}_c128", dict(array=cvals.copy(), by=(cby,), func=func,
+                                   expected_groups=np.arange(7))
+    yield "mean_c64", dict(array=cvals.astype(np.complex64), by=(cby,),
+                           func="mean", expected_groups=np.arange(7))
     # datetime/timedelta through the reference's int64-view machinery
     tv = (np.datetime64("2021-01-01")
           + rng.integers(0, 10**6, 500).astype("timedelta64[s]"))
@@ -172,9 +186,10 @@ def test_reference_drives_hip_engine(name, kw):
     if want.dtype.kind in "Mm":
         np.testing.assert_array_equal(got, want, err_msg=name)
     else:
+        cast = np.complex128 if want.dtype.kind == "c" else np.float64
         np.testing.assert_allclose(
-            got.astype(np.float64, copy=False) if want.dtype.kind in "fc" else got,
-            want.astype(np.float64, copy=False) if want.dtype.kind in "fc" else want,
+            got.astype(cast, copy=False) if want.dtype.kind in "fc" else got,
+            want.astype(cast, copy=False) if want.dtype.kind in "fc" else want,
             equal_nan=True, err_msg=name, **_tol(kw.get("func", ""), want.dtype),
         )
 
