@@ -329,6 +329,27 @@ def _factorize_device(flat: torch.Tensor, expect, sort: bool, dt_by=None) -> _Fa
     return _FactorizedBy(codes, sorted_expect, direct=False)
 
 
+def _factorize_strings_host(flat_np: np.ndarray, expect, sort: bool, device) -> _FactorizedBy:
+    """String/object labels: the reference's pd.factorize hash path runs on
+    the HOST (strings have no device representation); the dense int64 codes
+    ship to the GPU (reference factorize.py:96)."""
+    if expect is None:
+        uniq, inv = np.unique(flat_np, return_inverse=True)  # sorted
+        codes = inv.astype(np.int64)
+        groups = uniq
+    else:
+        expect_np = np.asarray(expect)
+        sorted_expect = np.sort(expect_np) if sort else expect_np
+        idx = np.searchsorted(sorted_expect, flat_np).astype(np.int64)
+        n = len(sorted_expect)
+        clip = np.clip(idx, 0, max(n - 1, 0))
+        bad = (idx == n) | (sorted_expect[clip] != flat_np)
+        codes = np.where(bad, -1, clip)
+        groups = sorted_expect
+    t = torch.from_numpy(np.ascontiguousarray(codes)).to(device)
+    return _FactorizedBy(t, groups, direct=False)
+
+
 def _combined_codes(facs: list[_FactorizedBy]):
     """Combine multiple factorized by-arrays into one code stream.
 
@@ -553,9 +574,19 @@ def groupby_reduce(
                 by_dts.append(bn.dtype)
                 _coerced.append(bn.view("i8"))
                 continue
+            if bn.dtype.kind in "US" or bn.dtype == object:
+                # string/object labels factorize on the host at the facs
+                # stage; the array stays numpy until then
+                by_dts.append(None)
+                _coerced.append(bn)
+                continue
         by_dts.append(None)
         _coerced.append(_coerce_by(b))
-    bys = tuple(_as_device_tensor(b, device) for b in _coerced)
+    bys = tuple(
+        b if isinstance(b, np.ndarray) and (b.dtype.kind in "US" or b.dtype == object)
+        else _as_device_tensor(b, device)
+        for b in _coerced
+    )
     nby = len(bys)
     if nby == 0:
         raise ValueError("need at least one by array")
@@ -571,7 +602,12 @@ def groupby_reduce(
         if trail != tuple(by_shape) and all(
             bs in (1, ts) for bs, ts in zip(by_shape, trail)
         ):
-            bys = tuple(b.broadcast_to(trail).contiguous() for b in bys)
+            bys = tuple(
+                np.ascontiguousarray(np.broadcast_to(b, trail))
+                if isinstance(b, np.ndarray)
+                else b.broadcast_to(trail).contiguous()
+                for b in bys
+            )
             by_shape = bys[0].shape
     if tuple(arr.shape[arr.ndim - len(by_shape) :]) != tuple(by_shape):
         raise ValueError(f"by {tuple(by_shape)} must align with trailing dims of array {tuple(arr.shape)}")
@@ -591,7 +627,12 @@ def groupby_reduce(
             perm = keep + sorted(ax)
             arr = arr.permute(perm).contiguous()
             by_perm = [d - nlead_s for d in perm if d >= nlead_s]
-            bys = tuple(b.permute(by_perm).contiguous() for b in bys)
+            bys = tuple(
+                np.ascontiguousarray(np.transpose(b, by_perm))
+                if isinstance(b, np.ndarray)
+                else b.permute(by_perm).contiguous()
+                for b in bys
+            )
             by_shape = bys[0].shape  # (kept by dims..., reduced dims...)
             subset_keep_shape = tuple(arr.shape[nlead_s : len(keep)])
             # for sort=False: first-appearance order is defined on the
@@ -634,7 +675,9 @@ def groupby_reduce(
 
     isbins = isbin if isinstance(isbin, (tuple, list)) else (isbin,) * nby
     facs = [
-        _factorize_bins(b.reshape(-1), e, dt_by=dtb)
+        _factorize_strings_host(b.reshape(-1), e, sort, device)
+        if isinstance(b, np.ndarray)
+        else _factorize_bins(b.reshape(-1), e, dt_by=dtb)
         if ib
         else _factorize_device(b.reshape(-1), e, sort, dt_by=dtb)
         for b, e, ib, dtb in zip(bys, expected_groups, isbins, by_dts)

@@ -96,9 +96,17 @@ def groupby_scan(
                 by_dts.append(bn.dtype)
                 _coerced.append(bn.view("i8"))
                 continue
+            if bn.dtype.kind in "US" or bn.dtype == object:
+                by_dts.append(None)
+                _coerced.append(bn)  # host-factorized at the facs stage
+                continue
         by_dts.append(None)
         _coerced.append(_coerce_by(b))
-    bys = tuple(_as_device_tensor(b, device) for b in _coerced)
+    bys = tuple(
+        b if isinstance(b, np.ndarray) and (b.dtype.kind in "US" or b.dtype == object)
+        else _as_device_tensor(b, device)
+        for b in _coerced
+    )
     if len(bys) == 0:
         raise ValueError("need at least one by array")
     by_shape = bys[0].shape
@@ -131,8 +139,12 @@ def groupby_scan(
         expected_groups = (expected_groups,)
     if expected_groups is None:
         expected_groups = (None,) * len(bys)
+    from .core import _factorize_strings_host
+
     facs = [
-        _factorize_device(b.reshape(-1), e, True, dt_by=dtb)
+        _factorize_strings_host(b.reshape(-1), e, True, device)
+        if isinstance(b, np.ndarray)
+        else _factorize_device(b.reshape(-1), e, True, dt_by=dtb)
         for b, e, dtb in zip(bys, expected_groups, by_dts)
     ]
     ngroups = 1

@@ -287,6 +287,16 @@ def gen_cases():
         yield f"{func}_dt_isbin", dict(
             array=dvals, by=dbyn, func=func, expected_groups=dt_edges, isbin=True
         )
+    # grouping BY string labels (the pd.factorize hash path)
+    sb = rng.choice(np.array(["north", "south", "east", "west", "up"]), 250)
+    sv2 = rng.standard_normal(250)
+    for func in ["mean", "sum", "count", "nanmax", "var"]:
+        yield f"{func}_strby", dict(array=sv2, by=sb.copy(), func=func)
+    yield "mean_strby_expected", dict(
+        array=sv2, by=sb.copy(), func="mean",
+        expected_groups=np.array(["east", "north", "zz"]), fill_value=-5.0,
+    )
+    yield "sum_strby_nosort", dict(array=sv2, by=sb.copy(), func="sum", sort=False)
     # pd.IntervalIndex expected_groups (binning without isbin=True)
     import pandas as pd
     iv_vals = rng.standard_normal(300)
