@@ -297,6 +297,12 @@ def gen_cases():
         expected_groups=np.array(["east", "north", "zz"]), fill_value=-5.0,
     )
     yield "sum_strby_nosort", dict(array=sv2, by=sb.copy(), func="sum", sort=False)
+    # mixed multi-by: string x int
+    yield "nanmean_str_int_multiby", dict(
+        array=sv2, by=(sb.copy(), rng.integers(0, 4, 250)), func="nanmean",
+        expected_groups=(np.array(["east", "north", "south", "up", "west"]),
+                         np.arange(4)),
+    )
     # pd.IntervalIndex expected_groups (binning without isbin=True)
     import pandas as pd
     iv_vals = rng.standard_normal(300)
