@@ -61,6 +61,13 @@ def make_case(rng, big=False):
     shape_kind = rng.choice(["1d", "lead", "multiby", "subset"])
     kw = {}
     if shape_kind == "1d":
+        if rng.random() < 0.1:
+            # group BY string labels (host pd.factorize hash path)
+            cats = np.array([f"s{j:03d}" for j in range(int(min(ng, 50)))])
+            by = rng.choice(cats, n)
+            if rng.random() < 0.5:
+                kw["expected_groups"] = np.sort(cats)
+            return vals, by, kw
         if rng.random() < 0.15:
             # group BY datetime labels (NaT rows drop; unit-aligned expected)
             base = np.datetime64("2021-06-01")
