@@ -91,12 +91,19 @@ def _run_case(i, rng):
     else:
         fin = want[np.isfinite(want)]
         scale = 1 + float(np.max(np.abs(fin), initial=0.0))
-        if want.dtype.itemsize == 2 or np.asarray(arr).dtype == np.float16:
+
+        def _prec(dtp):
+            if dtp.kind == "c":
+                return dtp.itemsize // 2  # per-component precision
+            return dtp.itemsize if dtp.kind == "f" else 8
+
+        eff = min(_prec(want.dtype), _prec(np.asarray(arr).dtype))
+        if eff == 2:
             # f16 precision class: the oracle computes quantile/lerp
             # arithmetic in the INPUT precision (np.quantile preserves f16),
             # the product lerps the promoted f32 values in f64
             tol = dict(rtol=2e-3, atol=1e-3 * scale)
-        elif want.dtype.itemsize == 4 or np.asarray(arr).dtype.itemsize == 4:
+        elif eff == 4:
             tol = dict(rtol=3e-5, atol=1e-4 * scale)
         else:
             tol = dict(rtol=1e-10, atol=1e-10 * scale)
