@@ -340,6 +340,10 @@ def _factorize_strings_host(flat_np: np.ndarray, expect, sort: bool, device) -> 
     else:
         expect_np = np.asarray(expect)
         sorted_expect = np.sort(expect_np) if sort else expect_np
+        if not sort and len(sorted_expect) > 1 and not np.all(
+            sorted_expect[:-1] <= sorted_expect[1:]
+        ):
+            raise NotImplementedError("unsorted expected_groups with sort=False")
         idx = np.searchsorted(sorted_expect, flat_np).astype(np.int64)
         n = len(sorted_expect)
         clip = np.clip(idx, 0, max(n - 1, 0))
