@@ -212,6 +212,31 @@ def test_count_datetime_nat_numpy_side():
     np.testing.assert_array_equal(np.asarray(got), want)
 
 
+def test_reference_custom_aggregation_hip():
+    """The reference's own Aggregation blueprint (chunk/combine/finalize)
+    driven with engine="hip" must match engine="flox" (fresh instances per
+    engine: the reference mutates agg.dtype during initialization)."""
+    import importlib
+
+    refaggs = importlib.import_module("floxref.aggregations")
+
+    def mk():
+        return refaggs.Aggregation(
+            name="custommean", numpy="mean", chunk=("sum", "nanlen"),
+            combine=("sum", "sum"), finalize=lambda s, c: s / c,
+            fill_value=0, final_fill_value=np.nan)
+
+    rng = np.random.default_rng(12)
+    vals = rng.standard_normal(2_000)
+    by = rng.integers(0, 11, 2_000)
+    want, *_ = core.groupby_reduce(vals, by, func=mk(), engine="flox",
+                                   expected_groups=np.arange(11))
+    got, *_ = core.groupby_reduce(vals, by, func=mk(), engine="hip",
+                                  expected_groups=np.arange(11))
+    np.testing.assert_allclose(np.asarray(got), np.asarray(want),
+                               equal_nan=True, rtol=1e-12, atol=1e-12)
+
+
 ORACLE_BASELINED = [
     "first", "last", "nanfirst", "nanlast",
     "argmax", "argmin", "nanargmax", "nanargmin", "any", "all",
