@@ -1371,8 +1371,11 @@ __launch_bounds__(BLOCK_LDS) __global__ void k_reduce_bucket_direct(
       const uint32_t nlc = __shfl_down(lc, 1, 64);
       const bool boundary = (lane == 63) || (nlc != lc);
       if (boundary && lc != LC_INVALID) {
-        if (OPS & (B_SUM | B_SSD)) acc_add(&s_sum[lc], (Acc)s);
-        if (IS_PROD) acc_mul(&s_sum[lc], (Acc)s);
+        /* s is already SumT: never narrow through Acc here — for SSD with
+         * integer V, Acc is int64 and the cast would truncate the run's
+         * double d^2 partial (found by the huge fuzz, seed 246810 case 21) */
+        if (OPS & (B_SUM | B_SSD)) acc_add(&s_sum[lc], s);
+        if (IS_PROD) acc_mul(&s_sum[lc], s);
         if ((OPS & B_CNT) && cn) atomicAdd(&s_cnt[lc], cn);
         if (OPS & B_PRESENT) s_present[lc] = 1u;
         /* gate on cn (count of non-NaN rows in the run): a real extreme can

@@ -826,6 +826,32 @@ def test_sorted_labels_direct_reduce(func):
     np.testing.assert_allclose(np.asarray(got), want, equal_nan=True, **tol)
 
 
+@pytest.mark.parametrize("func", ["var", "nanvar", "std"])
+def test_sorted_direct_int_var_regression(func):
+    """Huge-fuzz find (seed 246810 case 21): the sorted-direct bucket
+    kernel's run flush narrowed the double SSD partial through Acc=int64
+    for integer inputs, truncating d^2 fractions (~2.6e-7 relative error).
+    Check against exact integer algebra var = (c*sum(v^2) - s^2) / c^2."""
+    rng = np.random.default_rng(zlib.crc32(f"sdi-{func}".encode()))
+    n, ng = 3_000_000, 200_000
+    labels = np.sort(rng.integers(0, ng, n))
+    vals = rng.integers(-1000, 1000, n).astype(np.int64)
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func, expected_groups=range(ng))
+    c = np.bincount(labels, minlength=ng)
+    s = np.zeros(ng, dtype=np.int64)
+    np.add.at(s, labels, vals)
+    sq = np.zeros(ng, dtype=np.int64)
+    np.add.at(sq, labels, vals * vals)
+    exact = np.full(ng, np.nan)
+    ok = c > 0
+    exact[ok] = (c[ok] * sq[ok] - s[ok] * s[ok]).astype(np.float64) / (
+        c[ok].astype(np.float64) ** 2)
+    if func == "std":
+        exact = np.sqrt(exact)
+    np.testing.assert_allclose(np.asarray(got), exact, equal_nan=True,
+                               rtol=1e-12, atol=1e-12)
+
+
 @pytest.mark.parametrize("func", [
     "sum", "nansum", "mean", "min", "nanmin", "max", "nanmax", "var",
     "argmin", "argmax", "first", "median", "cumsum", "ffill",
