@@ -841,7 +841,8 @@ __device__ __forceinline__ uint32_t wave_incl_scan(uint32_t v) {
   return v;
 }
 
-template <typename V, typename L, bool CROW = false>
+template <typename V, typename L, bool CROW = false,
+          int T = (sizeof(V) == 4 ? 6144 : 3072)>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     const V* __restrict__ values, const L* __restrict__ labels,
     const L* __restrict__ labels2, int64_t n, int64_t ngroups, int64_t g0,
@@ -849,7 +850,6 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
     uint32_t* __restrict__ cursors, uint32_t cap /* 0 = exact bases */,
     uint32_t* __restrict__ overflow, PairT<V>* __restrict__ pairs,
     int64_t row_base = 0 /* CROW: global row = row_base + i in .pad */) {
-  constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
   constexpr int RPT = T / PART_BLOCK;
   constexpr int NB = 64;
   constexpr int NE = NB * PART_NW; /* = PART_BLOCK slots */
@@ -989,7 +989,8 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter(
  * as k_part_scatter, with a fixed 64-entry histogram. */
 constexpr int PART_SUB = 64;
 
-template <typename V, bool CROW = false>
+template <typename V, bool CROW = false,
+          int T = (sizeof(V) == 4 ? 6144 : 3072)>
 __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     const PairT<V>* __restrict__ in, const uint32_t* __restrict__ baseA,
     uint32_t capA /* 0: baseA[sb]..baseA[sb+1]; else sb*capA..baseA[sb] */,
@@ -997,7 +998,6 @@ __launch_bounds__(PART_BLOCK) __global__ void k_part_scatter2(
     uint32_t* __restrict__ cursors,
     uint32_t cap2 /* 0 = exact fine bases */, uint32_t* __restrict__ overflow,
     PairT<V>* __restrict__ out) {
-  constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
   constexpr int RPT = T / PART_BLOCK;
   constexpr int NB = PART_SUB;
   constexpr int NE = NB * PART_NW;
@@ -1673,6 +1673,23 @@ inline int fh_part_mode() {
   return v;
 }
 
+/* FH_PART_TILE: 0 (default) = full scatter staging tile (f32 6144 / f64
+ * 3072 pairs, ~76 KB LDS, 2 blocks/CU); 1 = small tile (f32 4096 / f64
+ * 2048, ~51/43 KB, 3 blocks/CU) — occupancy-vs-coalescing A/B knob. */
+inline int fh_part_tile() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("FH_PART_TILE");
+    v = (e && atoi(e) != 0) ? 1 : 0;
+  }
+  return v;
+}
+
+constexpr int64_t part_tile_lds(int T, int pairsz) {
+  return (int64_t)(PART_SUB * PART_NW + PART_NW + 1 + PART_SUB + 3) * 4 +
+         (int64_t)T * 4 + (int64_t)T * pairsz;
+}
+
 inline int64_t fh_part_bin_bytes() {
   static int64_t v = -1;
   if (v < 0) {
@@ -1814,38 +1831,58 @@ int launch_partition_exact(fh_call* c, const PartPlan& pp) {
   }
 
   {
-    auto kern = k_part_scatter<V, L, CROW>;
-    FH_CHECK(hipFuncSetAttribute((const void*)kern,
-                                 hipFuncAttributeMaxDynamicSharedMemorySize,
-                                 (int)pp.scatter_lds));
-    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
-    int64_t wb = (c->n + T - 1) / T;
-    int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
     const int shA = pp.two_level ? pp.shift + 6 : pp.shift;
     const int bA = pp.two_level ? pp.B1 : pp.B;
-    hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
-                       stream, (const V*)c->values, (const L*)c->labels,
-                       (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       shA, bA, pp.Bpad, cursors, 0u, overflow, pairs,
-                       c->row_offset);
-    FH_CHECK(hipGetLastError());
+    auto launch1 = [&](auto tc) {
+      constexpr int T = decltype(tc)::value;
+      auto kern = k_part_scatter<V, L, CROW, T>;
+      const int64_t lds = part_tile_lds(T, (int)sizeof(PairT<V>));
+      hipError_t e = hipFuncSetAttribute(
+          (const void*)kern, hipFuncAttributeMaxDynamicSharedMemorySize,
+          (int)lds);
+      if (e != hipSuccess) return e;
+      int64_t wb = (c->n + T - 1) / T;
+      int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
+      hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), lds, stream,
+                         (const V*)c->values, (const L*)c->labels,
+                         (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
+                         shA, bA, pp.Bpad, cursors, 0u, overflow, pairs,
+                         c->row_offset);
+      return hipGetLastError();
+    };
+    hipError_t e1;
+    if (fh_part_tile())
+      e1 = launch1(std::integral_constant<int, (sizeof(V) == 4 ? 4096 : 2048)>{});
+    else
+      e1 = launch1(std::integral_constant<int, (sizeof(V) == 4 ? 6144 : 3072)>{});
+    FH_CHECK(e1);
   }
   if (pp.two_level) {
     /* fine-bucket cursors, then the in-super-bucket scatter */
     FH_CHECK(hipMemcpyAsync(cursors, h_base, (int64_t)pp.B * 4,
                             hipMemcpyHostToDevice, stream));
-    auto kern2 = k_part_scatter2<V, CROW>;
-    FH_CHECK(hipFuncSetAttribute((const void*)kern2,
-                                 hipFuncAttributeMaxDynamicSharedMemorySize,
-                                 (int)pp.scatter2_lds));
-    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
-    int tiles_x = (int)((maxA + T - 1) / T);
-    if (tiles_x > 64) tiles_x = 64;
-    if (tiles_x < 1) tiles_x = 1;
-    hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK),
-                       pp.scatter2_lds, stream, pairs, baseAd, 0u, pp.shift,
-                       pp.B, cursors, 0u, overflow, pairs2);
-    FH_CHECK(hipGetLastError());
+    auto launch2 = [&](auto tc) {
+      constexpr int T = decltype(tc)::value;
+      auto kern2 = k_part_scatter2<V, CROW, T>;
+      const int64_t lds = part_tile_lds(T, (int)sizeof(PairT<V>));
+      hipError_t e = hipFuncSetAttribute(
+          (const void*)kern2, hipFuncAttributeMaxDynamicSharedMemorySize,
+          (int)lds);
+      if (e != hipSuccess) return e;
+      int tiles_x = (int)((maxA + T - 1) / T);
+      if (tiles_x > 64) tiles_x = 64;
+      if (tiles_x < 1) tiles_x = 1;
+      hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK), lds,
+                         stream, pairs, baseAd, 0u, pp.shift,
+                         pp.B, cursors, 0u, overflow, pairs2);
+      return hipGetLastError();
+    };
+    hipError_t e2;
+    if (fh_part_tile())
+      e2 = launch2(std::integral_constant<int, (sizeof(V) == 4 ? 4096 : 2048)>{});
+    else
+      e2 = launch2(std::integral_constant<int, (sizeof(V) == 4 ? 6144 : 3072)>{});
+    FH_CHECK(e2);
   }
   {
     int rc = init_outs<V, OPS>(c, c->ngroups, stream);
@@ -1984,37 +2021,57 @@ int launch_partition(fh_call* c, const PartPlan& pp) {
                      cursorsA, bA, capA);
   FH_CHECK(hipGetLastError());
   {
-    auto kern = k_part_scatter<V, L, CROW>;
-    FH_CHECK(hipFuncSetAttribute((const void*)kern,
-                                 hipFuncAttributeMaxDynamicSharedMemorySize,
-                                 (int)pp.scatter_lds));
-    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
-    int64_t wb = (c->n + T - 1) / T;
-    int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
     const int shA = pp.two_level ? pp.shift + 6 : pp.shift;
-    hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), pp.scatter_lds,
-                       stream, (const V*)c->values, (const L*)c->labels,
-                       (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
-                       shA, bA, pp.Bpad, cursorsA, capA, overflow, pairs,
-                       c->row_offset);
-    FH_CHECK(hipGetLastError());
+    auto launch1 = [&](auto tc) {
+      constexpr int T = decltype(tc)::value;
+      auto kern = k_part_scatter<V, L, CROW, T>;
+      const int64_t lds = part_tile_lds(T, (int)sizeof(PairT<V>));
+      hipError_t e = hipFuncSetAttribute(
+          (const void*)kern, hipFuncAttributeMaxDynamicSharedMemorySize,
+          (int)lds);
+      if (e != hipSuccess) return e;
+      int64_t wb = (c->n + T - 1) / T;
+      int nb1 = (int)(wb < 1024 ? (wb > 0 ? wb : 1) : 1024);
+      hipLaunchKernelGGL(kern, dim3(nb1), dim3(PART_BLOCK), lds, stream,
+                         (const V*)c->values, (const L*)c->labels,
+                         (const L*)c->labels2, c->n, c->ngroups, c->g0, c->g1,
+                         shA, bA, pp.Bpad, cursorsA, capA, overflow, pairs,
+                         c->row_offset);
+      return hipGetLastError();
+    };
+    hipError_t e1;
+    if (fh_part_tile())
+      e1 = launch1(std::integral_constant<int, (sizeof(V) == 4 ? 4096 : 2048)>{});
+    else
+      e1 = launch1(std::integral_constant<int, (sizeof(V) == 4 ? 6144 : 3072)>{});
+    FH_CHECK(e1);
   }
   if (pp.two_level) {
     hipLaunchKernelGGL(k_init_cursors, dim3((pp.B + 255) / 256), dim3(256), 0,
                        stream, cursors, pp.B, pp.cap2);
     FH_CHECK(hipGetLastError());
-    auto kern2 = k_part_scatter2<V, CROW>;
-    FH_CHECK(hipFuncSetAttribute((const void*)kern2,
-                                 hipFuncAttributeMaxDynamicSharedMemorySize,
-                                 (int)pp.scatter2_lds));
-    constexpr int T = sizeof(V) == 4 ? 6144 : 3072;
-    int tiles_x = (int)(((int64_t)pp.cap1 + T - 1) / T);
-    if (tiles_x > 64) tiles_x = 64;
-    if (tiles_x < 1) tiles_x = 1;
-    hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK),
-                       pp.scatter2_lds, stream, pairs, cursorsA, pp.cap1,
-                       pp.shift, pp.B, cursors, pp.cap2, overflow, pairs2);
-    FH_CHECK(hipGetLastError());
+    auto launch2 = [&](auto tc) {
+      constexpr int T = decltype(tc)::value;
+      auto kern2 = k_part_scatter2<V, CROW, T>;
+      const int64_t lds = part_tile_lds(T, (int)sizeof(PairT<V>));
+      hipError_t e = hipFuncSetAttribute(
+          (const void*)kern2, hipFuncAttributeMaxDynamicSharedMemorySize,
+          (int)lds);
+      if (e != hipSuccess) return e;
+      int tiles_x = (int)(((int64_t)pp.cap1 + T - 1) / T);
+      if (tiles_x > 64) tiles_x = 64;
+      if (tiles_x < 1) tiles_x = 1;
+      hipLaunchKernelGGL(kern2, dim3(tiles_x, pp.B1), dim3(PART_BLOCK), lds,
+                         stream, pairs, cursorsA, pp.cap1,
+                         pp.shift, pp.B, cursors, pp.cap2, overflow, pairs2);
+      return hipGetLastError();
+    };
+    hipError_t e2;
+    if (fh_part_tile())
+      e2 = launch2(std::integral_constant<int, (sizeof(V) == 4 ? 4096 : 2048)>{});
+    else
+      e2 = launch2(std::integral_constant<int, (sizeof(V) == 4 ? 6144 : 3072)>{});
+    FH_CHECK(e2);
   }
   {
     int rc = init_outs<V, OPS>(c, c->ngroups, stream);
