@@ -240,6 +240,7 @@ def test_reference_custom_aggregation_hip():
 ORACLE_BASELINED = [
     "first", "last", "nanfirst", "nanlast",
     "argmax", "argmin", "nanargmax", "nanargmin", "any", "all",
+    "mode", "nanmode",
 ]
 
 
