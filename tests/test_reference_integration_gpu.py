@@ -62,6 +62,18 @@ def _cases():
             array=bnan, by=(big_by,), func=func,
             expected_groups=np.arange(10_000),
         )
+    # partition-path sizes under the reference's orchestrator: huge group
+    # counts (two-level scatter path) and sorted labels (sorted-direct path)
+    huge_by = rng.integers(0, 2_000_000, 1_000_000)
+    for func in ["mean", "var"]:
+        yield f"{func}_big_f32_2e6g", dict(
+            array=bnan, by=(huge_by,), func=func,
+            expected_groups=np.arange(2_000_000),
+        )
+    yield "count_big_sorted_2e6g", dict(
+        array=bnan, by=(np.sort(huge_by),), func="count",
+        expected_groups=np.arange(2_000_000),
+    )
     # expected superset + fill, min_count, dtype
     yield "sum_expected_fill", dict(
         array=vals, by=(labels_basic,), func="sum",
