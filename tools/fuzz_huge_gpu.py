@@ -146,9 +146,52 @@ def cols_case(i, rng):
           flush=True)
 
 
+def scan_case(i, rng):
+    """Huge-size grouped scans (block-carry chains over thousands of blocks,
+    sentinel carry slot, sorted fast path) against the pinned oracle."""
+    from oracle import groupby_scan as oracle_scan
+
+    n = int(rng.integers(2_000_000, 8_000_001))
+    ng = int(rng.choice([8192, 100_000, 1_000_000]))
+    dtype = str(rng.choice(["float32", "float64", "int64"]))
+    layout = str(rng.choice(["uniform", "sorted"]))
+    labels = rng.integers(0, ng, n)
+    if layout == "sorted":
+        labels = np.sort(labels)
+    if dtype == "int64":
+        vals = rng.integers(-1000, 1000, n).astype(np.int64)
+        func = str(rng.choice(["cumsum", "ffill", "bfill"]))
+    else:
+        vals = (rng.standard_normal(n) * 100).astype(dtype)
+        if rng.random() < 0.7:
+            vals[rng.random(n) < 0.05] = np.nan
+        func = str(rng.choice(["cumsum", "nancumsum", "ffill", "bfill"]))
+    t0 = time.perf_counter()
+    want = oracle_scan(vals, labels, func=func, expected_groups=np.arange(ng))
+    got = np.asarray(flox_amd.groupby_scan(vals, labels, func=func,
+                                           expected_groups=np.arange(ng)))
+    ctx = f"[{i}] scan:{func} {dtype} {layout} n={n} ng={ng}"
+    assert got.shape == want.shape and got.dtype == want.dtype, ctx
+    if want.dtype.kind == "f":
+        fin = np.isfinite(want)
+        scale = 1 + float(np.max(np.abs(want[fin]), initial=0.0))
+        tol = (dict(rtol=3e-5, atol=1e-4 * scale) if want.dtype.itemsize == 4
+               else dict(rtol=1e-10, atol=1e-10 * scale))
+        np.testing.assert_allclose(got, want, equal_nan=True, err_msg=ctx, **tol)
+    else:
+        np.testing.assert_array_equal(got, want, err_msg=ctx)
+    dt = time.perf_counter() - t0
+    print(f"[{i}] OK scan:{func} {dtype} {layout} n={n:.1e} ng={ng:.0e} ({dt:.1f}s)",
+          flush=True)
+
+
 def one_case(i, rng):
-    if rng.random() < 0.25:
+    r = rng.random()
+    if r < 0.25:
         cols_case(i, rng)
+        return
+    if r < 0.40:
+        scan_case(i, rng)
         return
     n = int(rng.integers(2_000_000, 20_000_001))
     ng = int(rng.choice([8192, 100_000, 1_000_000, 12_000_000]))
