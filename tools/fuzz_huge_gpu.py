@@ -185,6 +185,43 @@ def scan_case(i, rng):
           flush=True)
 
 
+def order_case(i, rng):
+    """Order statistics (rocprim-sorted quantile path) at 1e6+ rows against
+    the pinned oracle. Sizes are capped by the oracle's per-group sort cost
+    (~5-15 s/case on host cores); mode stays in the 300k product fuzz."""
+    from oracle import groupby_reduce as oracle_reduce
+
+    n = int(rng.integers(1_000_000, 3_000_001))
+    ng = int(rng.choice([8192, 100_000]))
+    dtype = str(rng.choice(["float32", "float64"]))
+    vals = (rng.standard_normal(n) * 100).astype(dtype)
+    if rng.random() < 0.7:
+        vals[rng.random(n) < 0.05] = np.nan
+    labels = rng.integers(0, ng, n)
+    func = str(rng.choice(["quantile", "nanquantile", "median", "nanmedian"]))
+    kw = {}
+    if "quantile" in func:
+        q = [0.25, 0.5, 0.75] if rng.random() < 0.3 else float(rng.uniform(0.02, 0.98))
+        kw["finalize_kwargs"] = {"q": q}
+    t0 = time.perf_counter()
+    want, *_ = oracle_reduce(vals, labels, func=func,
+                             expected_groups=np.arange(ng), **kw)
+    got, *_ = flox_amd.groupby_reduce(vals, labels, func=func,
+                                      expected_groups=np.arange(ng), **kw)
+    got = np.asarray(got)
+    ctx = f"[{i}] order:{func} {dtype} n={n} ng={ng}"
+    assert got.shape == want.shape, ctx
+    fin = np.isfinite(want)
+    scale = 1 + float(np.max(np.abs(want[fin]), initial=0.0))
+    tol = (dict(rtol=3e-5, atol=1e-4 * scale) if want.dtype.itemsize == 4
+           else dict(rtol=1e-10, atol=1e-10 * scale))
+    np.testing.assert_allclose(got.astype(np.float64), want.astype(np.float64),
+                               equal_nan=True, err_msg=ctx, **tol)
+    dt = time.perf_counter() - t0
+    print(f"[{i}] OK order:{func} {dtype} n={n:.1e} ng={ng:.0e} ({dt:.1f}s)",
+          flush=True)
+
+
 def one_case(i, rng):
     r = rng.random()
     if r < 0.25:
@@ -192,6 +229,9 @@ def one_case(i, rng):
         return
     if r < 0.40:
         scan_case(i, rng)
+        return
+    if r < 0.48:
+        order_case(i, rng)
         return
     n = int(rng.integers(2_000_000, 20_000_001))
     ng = int(rng.choice([8192, 100_000, 1_000_000, 12_000_000]))
