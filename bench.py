@@ -222,10 +222,35 @@ def main():
     graph = None
     graph_res = None
     if not args.no_graph and world == 1:
+        # eager probe BEFORE any capture attempt: graphs only pay where
+        # launch overhead is a visible fraction of the step, and a capture
+        # attempt that dies mid-step (partition path's host sync, allocator
+        # growth — both capture-forbidden on HIP) leaves torch allocator /
+        # RNG state subtly damaged even when caught. Long-step configs never
+        # attempt capture at all.
+        step()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(3):
+            step()
+        torch.cuda.synchronize()
+        t_probe = (time.perf_counter() - t0) / 3
+        if t_probe > 3e-3:
+            log(f"eager step {t_probe * 1e3:.2f} ms: above the hipGraph "
+                "payoff range; skipping capture")
+            args.no_graph = True
+    if not args.no_graph and world == 1:
         try:
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                graph_res = step()
+            # own outer stream context: if step() raises mid-capture, the
+            # graph ctx's __exit__ raises from capture_end BEFORE restoring
+            # the current stream, which would strand later eager steps on
+            # the invalidated capture stream (hipError 901). A python
+            # `with` unwinds this one even when the inner ctx throws.
+            cap_stream = torch.cuda.Stream()
+            with torch.cuda.stream(cap_stream):
+                with torch.cuda.graph(g):
+                    graph_res = step()
             g.replay()
             torch.cuda.synchronize()
             eager_res = step()
@@ -260,8 +285,12 @@ def main():
             else:
                 log("hipGraph replay mismatch vs eager; falling back to eager")
         except Exception as e:  # pragma: no cover - path depends on config
-            log(f"hipGraph capture unavailable ({type(e).__name__}); eager steps")
+            log(f"hipGraph capture unavailable ({type(e).__name__}: {e}); eager steps")
             graph = None
+            try:  # drain any half-ended capture state before eager steps
+                torch.cuda.synchronize()
+            except Exception:
+                pass
 
     # roofline instrumentation: HIP events around every fused-kernel launch
     # on the launch stream (kernel + its O(ngroups) slab-combine tail)

@@ -37,6 +37,7 @@ FLAG_SKIPNAN = 1
 FLAG_FORCE_LDS = 2
 FLAG_FORCE_ATOMIC = 4
 FLAG_SORTED_LABELS = 8
+FLAG_NO_HOST_SYNC = 16
 
 VDTYPE_OF = {
     np.dtype("float32"): F32,

@@ -158,6 +158,11 @@ def grouped_partials(
     c.flags = (FLAG_SKIPNAN if skipnan else 0) | (
         _ffi.FLAG_FORCE_LDS if force_path == 1 else _ffi.FLAG_FORCE_ATOMIC if force_path == 2 else 0
     )
+    if torch.cuda.is_current_stream_capturing():
+        # hipGraph capture: the partition paths read back to the host
+        # (overflow check / counting pre-pass), which would invalidate the
+        # capture — restrict the C dispatcher to sync-free paths
+        c.flags |= _ffi.FLAG_NO_HOST_SYNC
     # sorted-labels direct path at huge group counts: skips every scatter
     # pass (~44 -> 12 B/row). A 4K-pair sample rejects random labels for
     # ~20 us; the full O(n) verification only runs when the sample passes

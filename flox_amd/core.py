@@ -1006,6 +1006,9 @@ def groupby_reduce(
             and vals.dtype in (torch.float64, torch.int64)
             and 0 < vals.numel() < (1 << 31)
             and ngroups <= (1 << 24)
+            # pair-arg rides the partition path, whose overflow check
+            # host-syncs: under hipGraph capture use the two-pass form
+            and not (vals.is_cuda and torch.cuda.is_current_stream_capturing())
         )
         if dist_on and distributed.is_active():
             # the branch choice must agree across ranks (shard sizes differ)

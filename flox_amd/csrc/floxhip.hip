@@ -2126,8 +2126,11 @@ int launch_typed(fh_call* c) {
   BinLayout lay = bin_layout<V>(OPS, c->ngroups, 4);
 
   if (OPS & B_ARGROW) {
-    /* pair-payload arg-reductions exist only on the partition path */
-    if (c->flags & (FH_FORCE_LDS | FH_FORCE_ATOMIC)) return 11;
+    /* pair-payload arg-reductions exist only on the partition path (which
+     * host-syncs, so also refused during capture — callers use the
+     * two-pass form there) */
+    if (c->flags & (FH_FORCE_LDS | FH_FORCE_ATOMIC | FH_NO_HOST_SYNC))
+      return 11;
     PartPlan pp = part_plan<V>(c);
     if (!pp.feasible || c->scratch_bytes < pp.bytes) return 11;
     return launch_partition<V, L, OPS>(c, pp);
@@ -2224,8 +2227,11 @@ int launch_typed(fh_call* c) {
     }
   }
 
-  /* huge group counts: bucket-partition path when scratch allows */
-  if (!(c->flags & FH_FORCE_ATOMIC)) {
+  /* huge group counts: bucket-partition path when scratch allows. Not
+   * during hipGraph capture (FH_NO_HOST_SYNC): the optimistic overflow
+   * check and the exact path's counting pre-pass read back to the host
+   * and synchronize the stream, which invalidates a capture. */
+  if (!(c->flags & (FH_FORCE_ATOMIC | FH_NO_HOST_SYNC))) {
     PartPlan pp = part_plan<V>(c);
     if (pp.feasible && c->scratch_bytes >= pp.bytes)
       return launch_partition<V, L, OPS>(c, pp);

@@ -89,6 +89,11 @@ enum fh_flags {
                               * all in [0, ngroups): fh_grouped_scan skips the
                               * radix sort (the reference's issorted fast path,
                               * aggregate_flox.py:9-23) */
+  FH_NO_HOST_SYNC = 1 << 4, /* stream is being captured into a hipGraph: only
+                             * paths with no D2H readback / stream sync may
+                             * run (LDS, sorted-direct, atomic — NOT the
+                             * bucket-partition paths, whose overflow check
+                             * and counting pre-pass synchronize) */
 };
 
 typedef struct fh_call {
