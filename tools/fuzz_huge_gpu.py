@@ -213,7 +213,10 @@ def order_case(i, rng):
     assert got.shape == want.shape, ctx
     fin = np.isfinite(want)
     scale = 1 + float(np.max(np.abs(want[fin]), initial=0.0))
-    tol = (dict(rtol=3e-5, atol=1e-4 * scale) if want.dtype.itemsize == 4
+    # tier by INPUT precision: quantile outputs are always f64, but the
+    # interpolation endpoints are the f32 inputs, so product-vs-oracle
+    # arithmetic-order differences sit at f32 eps of the operands
+    tol = (dict(rtol=3e-5, atol=1e-4 * scale) if dtype == "float32"
            else dict(rtol=1e-10, atol=1e-10 * scale))
     np.testing.assert_allclose(got.astype(np.float64), want.astype(np.float64),
                                equal_nan=True, err_msg=ctx, **tol)
