@@ -775,6 +775,25 @@ def groupby_scan(array, *by, func, expected_groups=None, axis=None, dtype=None):
         found.append(f)
     grp_shape = tuple(len(f) for f in found)
     ngroups = math.prod(grp_shape)
+    # reference scan.py:286-291 ("avoid some roundoff error when we can"):
+    # a length-1 trailing axis, or a 1-D by where every observed row is its
+    # own group, returns the INPUT unchanged (cast to the scan's output
+    # dtype) — including nancumsum of a NaN row, which stays NaN instead of
+    # the semantic identity 0. Mirrored only under the reference's own
+    # precondition (scans there reject expected_groups). Found by the
+    # oracle-vs-reference fuzz, seed 606162 case 17401.
+    if all(e is None for e in (expected_groups or ())) and (
+        bys[0].shape[-1] == 1 or (len(bys) == 1 and bys[0].shape == (ngroups,))
+    ):
+        out = array.reshape(-1)
+        if func in ("cumsum", "nancumsum") and out.dtype.kind in "iub" and out.dtype.itemsize < 8:
+            out = out.astype(np.uint64 if out.dtype.kind == "u" else np.int64)
+        out = out.copy()
+        if dt_dtype is not None:
+            out = out.astype(dt_dtype)
+        if dtype is not None:
+            out = out.astype(dtype)
+        return out.reshape(array.shape)
     codes = _ravel_codes(codes_list, grp_shape) if len(bys) > 1 else codes_list[0]
     codes = codes.copy()
     codes[codes < 0] = ngroups  # sentinel group scans together, like factorize_:201-210

@@ -405,6 +405,17 @@ def gen_cases():
 
 def gen_scan_cases():
     rng = np.random.default_rng(77)
+    # reference scan.py:286-291 identity shortcut: length-1 trailing axis /
+    # all-distinct 1-D by return the input unchanged — nancumsum of a NaN
+    # row stays NaN instead of the identity 0 (fuzz seed 606162)
+    yield "scan_nancumsum_quirklen1", dict(
+        array=np.array([np.nan]), by=np.array([0.0]), func="nancumsum")
+    yield "scan_nancumsum_quirkdistinct", dict(
+        array=np.array([np.nan, 2.0, np.nan, 7.5]),
+        by=np.array([4, 1, 9, 2]), func="nancumsum")
+    yield "scan_nancumsum_quirku8", dict(
+        array=np.arange(5).astype(np.uint8),
+        by=np.array([3, 1, 4, 0, 2]), func="nancumsum")
     vals = rng.standard_normal(200)
     vals[rng.random(200) < 0.15] = np.nan
     by = rng.integers(0, 7, 200)
