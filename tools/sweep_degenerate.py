@@ -93,7 +93,7 @@ def main():
         try:
             assert got.shape == want.shape, (got.shape, want.shape)
             assert got.dtype == want.dtype, (got.dtype, want.dtype)
-            if want.dtype.kind in "iub":
+            if want.dtype.kind in "iubMm":
                 np.testing.assert_array_equal(got, want)
             else:
                 np.testing.assert_allclose(got, want, equal_nan=True,
@@ -122,6 +122,65 @@ def main():
                         run_case(vals, labels, func, True, {})
             print(f"[{dt.__name__} n={n}] cumulative: {n_ok} ok, {n_skip} skipped, "
                   f"{n_artifact} inf-artifact, {n_bad} mismatches", flush=True)
+
+    # kwarg corners at n <= 2: expected supersets + fill_value, min_count
+    # bigger than any group, ddof on 1-2 element groups
+    for n in (1, 2):
+        for vt in itertools.product(fpool, repeat=n):
+            vals = np.array(vt)
+            for lp in label_patterns(n):
+                labels = np.array(lp)
+                for exp in (np.array([0.0, 1.0]), np.array([0.0, 1.0, 5.0])):
+                    for func in FUNCS:
+                        kw = {"expected_groups": exp}
+                        if func in ("quantile", "nanquantile"):
+                            kw["finalize_kwargs"] = {"q": 0.4}
+                        run_case(vals, labels, func, False, kw)
+                    for func, fv in [("sum", -7.5), ("min", -7.5), ("last", -7.5)]:
+                        run_case(vals, labels, func, False,
+                                 {"expected_groups": exp, "fill_value": fv})
+                    for mc in (1, 2, 3):
+                        run_case(vals, labels, "nansum", False,
+                                 {"expected_groups": exp, "min_count": mc,
+                                  "fill_value": np.nan})
+                for func in ("var", "nanvar", "std", "nanstd"):
+                    run_case(vals, labels, func, False,
+                             {"finalize_kwargs": {"ddof": 1}})
+    print(f"[kwargs n<=2] cumulative: {n_ok} ok, {n_skip} skipped, "
+          f"{n_artifact} inf-artifact, {n_bad} mismatches", flush=True)
+
+    # datetime values with NaT (min/max/count/first/last/median + the
+    # non-float ffill/bfill early identity, scan.py:199-201)
+    dpool = [np.datetime64("NaT"), np.datetime64("2020-01-01"),
+             np.datetime64("1970-01-01")]
+    for n in (1, 2):
+        for vt in itertools.product(dpool, repeat=n):
+            vals = np.array(vt, dtype="datetime64[s]")
+            for lp in label_patterns(n):
+                labels = np.array(lp)
+                for func in ("min", "nanmin", "max", "nanmax", "count",
+                             "first", "last", "nanfirst", "nanlast",
+                             "median", "nanmedian"):
+                    run_case(vals, labels, func, False, {})
+                for func in ("ffill", "bfill"):
+                    run_case(vals, labels, func, True, {})
+    print(f"[datetime n<=2] cumulative: {n_ok} ok, {n_skip} skipped, "
+          f"{n_artifact} inf-artifact, {n_bad} mismatches", flush=True)
+
+    # complex values, linear set (inf excluded: same artifact regions)
+    cpool = [complex("nan"), 1.5 + 2.5j, complex(0, float("nan")), -0.0 + 0j]
+    for n in (1, 2):
+        for vt in itertools.product(cpool, repeat=n):
+            vals = np.array(vt, dtype=np.complex128)
+            for lp in label_patterns(n):
+                labels = np.array(lp)
+                for func in ("sum", "nansum", "mean", "nanmean", "count",
+                             "first", "last", "nanfirst", "nanlast"):
+                    run_case(vals, labels, func, False, {})
+                for func in SCANS:
+                    run_case(vals, labels, func, True, {})
+    print(f"[complex n<=2] cumulative: {n_ok} ok, {n_skip} skipped, "
+          f"{n_artifact} inf-artifact, {n_bad} mismatches", flush=True)
     print(f"degenerate sweep done: {n_ok} ok, {n_skip} skipped, "
           f"{n_artifact} inf-artifact, {n_bad} mismatches")
     sys.exit(1 if n_bad else 0)
