@@ -167,6 +167,23 @@ def main():
     print(f"[datetime n<=2] cumulative: {n_ok} ok, {n_skip} skipped, "
           f"{n_artifact} inf-artifact, {n_bad} mismatches", flush=True)
 
+    # 2-D lead dims at degenerate sizes: exhaustive value grids over
+    # (m, n) in {1,2}^2 (trimmed pool at 2x2), default axis (the random
+    # fuzz covers axis subsets at larger shapes)
+    for m, ncol in ((1, 1), (1, 2), (2, 1), (2, 2)):
+        pool2 = fpool if m * ncol <= 2 else [float("nan"), 1.5, float("inf"), -0.0]
+        for vt in itertools.product(pool2, repeat=m * ncol):
+            vals = np.array(vt).reshape(m, ncol)
+            for lp in label_patterns(ncol):
+                labels = np.array(lp)
+                for func in FUNCS:
+                    kw = {}
+                    if func in ("quantile", "nanquantile"):
+                        kw["finalize_kwargs"] = {"q": 0.4}
+                    run_case(vals, labels, func, False, kw)
+    print(f"[2d lead m,n<=2] cumulative: {n_ok} ok, {n_skip} skipped, "
+          f"{n_artifact} inf-artifact, {n_bad} mismatches", flush=True)
+
     # complex values, linear set (inf excluded: same artifact regions)
     cpool = [complex("nan"), 1.5 + 2.5j, complex(0, float("nan")), -0.0 + 0j]
     for n in (1, 2):
