@@ -20,6 +20,7 @@ pytestmark = pytest.mark.skipif(
 )
 
 
+@pytest.mark.filterwarnings("ignore::RuntimeWarning")
 def test_degenerate_sweep_n2():
     import itertools
 
