@@ -152,6 +152,14 @@ def main():
                     continue
             if func in ("var", "nanvar", "std", "nanstd") and rng.random() < 0.3:
                 kw["finalize_kwargs"] = {"ddof": 1}
+            if (np.asarray(arr).dtype == np.float16 and np.asarray(arr).size > 20_000
+                    and func in ("sum", "nansum", "mean", "nanmean",
+                                 "var", "nanvar", "std", "nanstd")):
+                # same unbounded native-f16 accumulation drift as the scan
+                # case above: reference np.add.reduceat accumulates in f16,
+                # we accumulate in f64 — parity only at bounded group sizes
+                n_skip += 1
+                continue
         if np.asarray(arr).dtype.kind in "Mm" and func not in (
             "min", "nanmin", "max", "nanmax", "count",
             "median", "nanmedian", "quantile", "nanquantile",
